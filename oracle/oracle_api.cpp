@@ -1,0 +1,133 @@
+// TEST INFRASTRUCTURE — C ABI over the CPU oracle (see oracle_filter.h).
+// Only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg may
+// load this library.
+#include <atomic>
+#include <cstring>
+#include <thread>
+
+#include "oracle_filter.h"
+#include "../victorialogs_amd/csrc/core/gen.h"
+
+using namespace vl;
+using namespace vl::oracle;
+
+namespace {
+thread_local std::string g_err;
+
+struct OrcPart {
+  PartReader pr;
+  std::vector<BlockHeader> bhs;
+  explicit OrcPart(const std::string& dir) : pr(dir) {
+    bhs = pr.read_all_block_headers();
+  }
+};
+
+int set_err(const std::exception& e) {
+  g_err = e.what();
+  return -1;
+}
+}  // namespace
+
+extern "C" {
+
+const char* orc_errstr() { return g_err.c_str(); }
+
+void* orc_open_part(const char* dir) {
+  try {
+    return new OrcPart(dir);
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return nullptr;
+  }
+}
+
+void orc_close_part(void* p) { delete (OrcPart*)p; }
+
+long orc_block_count(void* p) { return long(((OrcPart*)p)->bhs.size()); }
+long orc_block_rows(void* p, long i) {
+  return long(((OrcPart*)p)->bhs[size_t(i)].rows_count);
+}
+long orc_part_rows(void* p) { return long(((OrcPart*)p)->pr.header().rows_count); }
+
+void* orc_compile_filter(const char* json) {
+  try {
+    return new FilterNode(compile_filter(json));
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return nullptr;
+  }
+}
+
+void orc_free_filter(void* f) { delete (FilterNode*)f; }
+
+// Scans blocks [block_lo, block_hi) with `threads` worker threads.  Bitmap
+// words for block i are written at out_words[word_off[i]] where word_off is
+// the running sum of (rows+63)/64 over scanned blocks, in block order.
+// Returns total matched rows, or -1 on error.
+long long orc_scan_blocks(void* part, void* filter, long block_lo, long block_hi,
+                          unsigned long long* out_words, long long out_words_cap,
+                          int threads) {
+  try {
+    OrcPart* p = (OrcPart*)part;
+    FilterNode* f = (FilterNode*)filter;
+    if (block_hi < 0 || size_t(block_hi) > p->bhs.size()) block_hi = long(p->bhs.size());
+    if (block_lo < 0) block_lo = 0;
+
+    std::vector<uint64_t> word_off(size_t(block_hi - block_lo) + 1, 0);
+    for (long i = block_lo; i < block_hi; i++) {
+      word_off[size_t(i - block_lo) + 1] =
+          word_off[size_t(i - block_lo)] + (p->bhs[size_t(i)].rows_count + 63) / 64;
+    }
+    if (out_words && int64_t(word_off.back()) > out_words_cap) {
+      g_err = "orc_scan_blocks: output buffer too small";
+      return -1;
+    }
+
+    std::atomic<long> next{block_lo};
+    std::atomic<long long> hits{0};
+    if (threads < 1) threads = 1;
+    std::vector<std::thread> pool;
+    auto worker = [&]() {
+      long long local = 0;
+      for (;;) {
+        long i = next.fetch_add(1);
+        if (i >= block_hi) break;
+        Bitmap bm;
+        search_block(*f, p->pr, p->bhs[size_t(i)], bm);
+        local += (long long)bm.ones_count();
+        if (out_words) {
+          memcpy(out_words + word_off[size_t(i - block_lo)], bm.a.data(),
+                 bm.a.size() * 8);
+        }
+      }
+      hits += local;
+    };
+    for (int t = 1; t < threads; t++) pool.emplace_back(worker);
+    worker();
+    for (auto& t : pool) t.join();
+    return hits.load();
+  } catch (const std::exception& e) {
+    return set_err(e);
+  }
+}
+
+// Generates a synthetic part (vlogsgenerator shapes; gen.h).  Returns total
+// _msg bytes or -1.
+long long orc_generate_part(const char* dir, unsigned long long rows,
+                            unsigned long long streams,
+                            unsigned long long rows_per_block,
+                            unsigned long long msg_len, unsigned long long seed) {
+  try {
+    GenConfig cfg;
+    cfg.rows = rows;
+    cfg.streams = streams;
+    cfg.rows_per_block = rows_per_block;
+    cfg.msg_len = size_t(msg_len);
+    cfg.seed = seed;
+    return (long long)generate_part(dir, cfg);
+  } catch (const std::exception& e) {
+    return set_err(e);
+  }
+}
+
+}  // extern "C"

@@ -1,0 +1,723 @@
+// TEST INFRASTRUCTURE — see oracle_filter.h.  Each function cites the
+// reference function it restates.
+#include "oracle_filter.h"
+
+#include <cmath>
+#include <cstring>
+
+#include "../victorialogs_amd/csrc/core/bloom.h"
+#include "../victorialogs_amd/csrc/core/match.h"
+#include "../victorialogs_amd/csrc/core/values.h"
+
+namespace vl {
+namespace oracle {
+
+const PartReader::BlockColumns& BlockCtx::columns() {
+  if (!bc_loaded) {
+    pr->read_block_columns(*bh, bc);
+    bc_loaded = true;
+  }
+  return bc;
+}
+
+std::string BlockCtx::const_value(const std::string& name) {
+  std::string v;
+  if (!pr->get_const_column(columns(), name, &v)) return "";
+  return v;
+}
+
+bool BlockCtx::column_header(const std::string& name, ColumnHeader* ch) {
+  return pr->get_column_header(columns(), name, ch);
+}
+
+const StringsBlockDec& BlockCtx::values(const ColumnHeader& ch) {
+  auto it = values_cache.find(ch.name);
+  if (it == values_cache.end()) {
+    StringsBlockDec dec;
+    pr->read_values(ch, bh->rows_count, dec);
+    it = values_cache.emplace(ch.name, std::move(dec)).first;
+  }
+  return it->second;
+}
+
+const std::vector<uint64_t>& BlockCtx::bloom(const ColumnHeader& ch) {
+  auto it = bloom_cache.find(ch.name);
+  if (it == bloom_cache.end()) {
+    std::vector<uint64_t> words;
+    pr->read_bloom(ch, words);
+    it = bloom_cache.emplace(ch.name, std::move(words)).first;
+  }
+  return it->second;
+}
+
+const std::vector<int64_t>& BlockCtx::get_timestamps() {
+  if (!ts_loaded) {
+    pr->read_timestamps(*bh, timestamps);
+    ts_loaded = true;
+  }
+  return timestamps;
+}
+
+// matchBloomFilterAllTokens (filter_phrase.go:302-308)
+static bool match_bloom_all(BlockCtx& ctx, const ColumnHeader& ch,
+                            const std::vector<uint64_t>& hashes) {
+  if (hashes.empty()) return true;
+  const auto& words = ctx.bloom(ch);
+  return bloom_contains_all(words.data(), words.size(), hashes.data(), hashes.size());
+}
+
+// visitValues (filter_phrase.go:291-300)
+template <typename F>
+static void visit_values(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm, F&& f) {
+  if (bm.is_zero()) return;
+  const StringsBlockDec& vals = ctx.values(ch);
+  bm.for_each_set_bit([&](uint64_t idx) { return f(vals.row(idx)); });
+}
+
+// matchBinaryValue (filter_exact.go:356-364)
+static void match_binary_value(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                               const bytes& bin,
+                               const std::vector<uint64_t>& hashes) {
+  if (!match_bloom_all(ctx, ch, hashes)) {
+    bm.reset_bits();
+    return;
+  }
+  strview b((const char*)bin.data(), bin.size());
+  visit_values(ctx, ch, bm, [&](strview v) { return v == b; });
+}
+
+// matchEncodedValuesDict (filter_phrase.go:272-289)
+static void match_encoded_dict(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                               const std::vector<uint8_t>& encoded) {
+  bool any = false;
+  for (uint8_t c : encoded) any |= (c == 1);
+  if (!any) {
+    bm.reset_bits();
+    return;
+  }
+  visit_values(ctx, ch, bm, [&](strview v) {
+    if (v.n != 1) fail("unexpected dict value length");
+    uint8_t idx = uint8_t(v.p[0]);
+    if (idx >= encoded.size()) fail("too big dict index");
+    return encoded[idx] == 1;
+  });
+}
+
+// ---- exact-value matchers (filter_exact.go:237-364) ----
+
+static void match_uint_by_exact(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                                strview phrase, const std::vector<uint64_t>& hashes,
+                                int width) {
+  uint64_t n;
+  if (!try_parse_uint64(phrase, &n) || n < ch.min_value || n > ch.max_value) {
+    bm.reset_bits();
+    return;
+  }
+  bytes bin;
+  switch (width) {
+    case 1: bin.push_back(uint8_t(n)); break;
+    case 2: put_u16be(bin, uint16_t(n)); break;
+    case 4: put_u32be(bin, uint32_t(n)); break;
+    default: put_u64be(bin, n); break;
+  }
+  match_binary_value(ctx, ch, bm, bin, hashes);
+}
+
+static void match_int64_by_exact(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                                 strview phrase, const std::vector<uint64_t>& hashes) {
+  int64_t n;
+  if (!try_parse_int64(phrase, &n) || n < int64_t(ch.min_value) ||
+      n > int64_t(ch.max_value)) {
+    bm.reset_bits();
+    return;
+  }
+  bytes bin;
+  put_i64be_zigzag(bin, n);
+  match_binary_value(ctx, ch, bm, bin, hashes);
+}
+
+static void match_float64_by_exact(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                                   strview value, const std::vector<uint64_t>& hashes) {
+  double f, mn, mx;
+  uint64_t mnu = ch.min_value, mxu = ch.max_value;
+  memcpy(&mn, &mnu, 8);
+  memcpy(&mx, &mxu, 8);
+  if (!try_parse_float64_exact(value, &f) || f < mn || f > mx) {
+    bm.reset_bits();
+    return;
+  }
+  bytes bin;
+  uint64_t u;
+  memcpy(&u, &f, 8);
+  put_u64be(bin, u);
+  match_binary_value(ctx, ch, bm, bin, hashes);
+}
+
+static void match_ipv4_by_exact(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                                strview value, const std::vector<uint64_t>& hashes) {
+  uint32_t n;
+  if (!try_parse_ipv4(value, &n) || uint64_t(n) < ch.min_value ||
+      uint64_t(n) > ch.max_value) {
+    bm.reset_bits();
+    return;
+  }
+  bytes bin;
+  put_u32be(bin, n);
+  match_binary_value(ctx, ch, bm, bin, hashes);
+}
+
+static void match_iso8601_by_exact(BlockCtx& ctx, const ColumnHeader& ch, Bitmap& bm,
+                                   strview value,
+                                   const std::vector<uint64_t>& hashes) {
+  int64_t n;
+  if (!try_parse_timestamp_iso8601(value, &n) || n < int64_t(ch.min_value) ||
+      n > int64_t(ch.max_value)) {
+    bm.reset_bits();
+    return;
+  }
+  bytes bin;
+  put_u64be(bin, uint64_t(n));
+  match_binary_value(ctx, ch, bm, bin, hashes);
+}
+
+// ---- string conversions (filter_phrase.go:321-346) ----
+
+static std::string to_uint_string(strview v, int width) {
+  uint64_t n = 0;
+  switch (width) {
+    case 1: n = uint8_t(v.p[0]); break;
+    case 2: n = get_u16be((const uint8_t*)v.p); break;
+    case 4: n = get_u32be((const uint8_t*)v.p); break;
+    default: n = get_u64be((const uint8_t*)v.p); break;
+  }
+  std::string s;
+  format_uint64(s, n);
+  return s;
+}
+
+// ---- filterPhrase (filter_phrase.go:61-209) ----
+
+static void apply_phrase(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  strview phrase(f.phrase);
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_phrase(strview(cv), phrase)) bm.reset_bits();
+    return;
+  }
+
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (phrase.n > 0) bm.reset_bits();  // filter_phrase.go:76-83
+    return;
+  }
+
+  switch (ch.type) {
+    case ValueType::String: {
+      // matchStringByPhrase (filter_phrase.go:201-209)
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) { return match_phrase(v, phrase); });
+      return;
+    }
+    case ValueType::Dict: {
+      // matchValuesDictByPhrase (filter_phrase.go:188-199)
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) {
+        enc.push_back(match_phrase(strview(dv), phrase) ? 1 : 0);
+      }
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8:
+      match_uint_by_exact(ctx, ch, bm, phrase, f.token_hashes, 1);
+      return;
+    case ValueType::Uint16:
+      match_uint_by_exact(ctx, ch, bm, phrase, f.token_hashes, 2);
+      return;
+    case ValueType::Uint32:
+      match_uint_by_exact(ctx, ch, bm, phrase, f.token_hashes, 4);
+      return;
+    case ValueType::Uint64:
+      match_uint_by_exact(ctx, ch, bm, phrase, f.token_hashes, 8);
+      return;
+    case ValueType::Int64:
+      match_int64_by_exact(ctx, ch, bm, phrase, f.token_hashes);
+      return;
+    case ValueType::Float64: {
+      // matchFloat64ByPhrase (filter_phrase.go:159-186)
+      double ff;
+      bool ok = try_parse_float64_exact(phrase, &ff);
+      bool special = phrase == std::string(".") || phrase == std::string("+") ||
+                     phrase == std::string("-");
+      if (!ok && !special) {
+        bm.reset_bits();
+        return;
+      }
+      const char* dot = phrase.n ? (const char*)memchr(phrase.p, '.', phrase.n) : nullptr;
+      long ndot = dot ? dot - phrase.p : -1;
+      if (ndot > 0 && size_t(ndot) < phrase.n - 1) {
+        match_float64_by_exact(ctx, ch, bm, phrase, f.token_hashes);
+        return;
+      }
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        if (v.n != 8) fail("unexpected float64 binary length");
+        uint64_t u = get_u64be((const uint8_t*)v.p);
+        double d;
+        memcpy(&d, &u, 8);
+        std::string s;
+        format_float64(s, d);
+        return match_phrase(strview(s), phrase);
+      });
+      return;
+    }
+    case ValueType::IPv4: {
+      // matchIPv4ByPhrase (filter_phrase.go:135-157)
+      uint32_t ip;
+      if (try_parse_ipv4(phrase, &ip)) {
+        match_ipv4_by_exact(ctx, ch, bm, phrase, f.token_hashes);
+        return;
+      }
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        if (v.n != 4) fail("unexpected ipv4 binary length");
+        std::string s;
+        format_ipv4(s, get_u32be((const uint8_t*)v.p));
+        return match_phrase(strview(s), phrase);
+      });
+      return;
+    }
+    case ValueType::TimestampISO8601: {
+      // matchTimestampISO8601ByPhrase (filter_phrase.go:113-133)
+      int64_t ts;
+      if (try_parse_timestamp_iso8601(phrase, &ts)) {
+        match_iso8601_by_exact(ctx, ch, bm, phrase, f.token_hashes);
+        return;
+      }
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        if (v.n != 8) fail("unexpected iso8601 binary length");
+        std::string s;
+        format_timestamp_iso8601(s, int64_t(get_u64be((const uint8_t*)v.p)));
+        return match_phrase(strview(s), phrase);
+      });
+      return;
+    }
+    default:
+      fail("unknown valueType in phrase filter");
+  }
+}
+
+// ---- filterExact (filter_exact.go:178-235) ----
+
+static void apply_exact(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  strview value(f.phrase);
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!(strview(cv) == value)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (value.n > 0) bm.reset_bits();
+    return;
+  }
+  switch (ch.type) {
+    case ValueType::String: {
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) { return v == value; });
+      return;
+    }
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) enc.push_back(strview(dv) == value ? 1 : 0);
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8: match_uint_by_exact(ctx, ch, bm, value, f.token_hashes, 1); return;
+    case ValueType::Uint16: match_uint_by_exact(ctx, ch, bm, value, f.token_hashes, 2); return;
+    case ValueType::Uint32: match_uint_by_exact(ctx, ch, bm, value, f.token_hashes, 4); return;
+    case ValueType::Uint64: match_uint_by_exact(ctx, ch, bm, value, f.token_hashes, 8); return;
+    case ValueType::Int64: match_int64_by_exact(ctx, ch, bm, value, f.token_hashes); return;
+    case ValueType::Float64: match_float64_by_exact(ctx, ch, bm, value, f.token_hashes); return;
+    case ValueType::IPv4: match_ipv4_by_exact(ctx, ch, bm, value, f.token_hashes); return;
+    case ValueType::TimestampISO8601: match_iso8601_by_exact(ctx, ch, bm, value, f.token_hashes); return;
+    default: fail("unknown valueType in exact filter");
+  }
+}
+
+// ---- filterRegexp (filter_regexp.go:78-254) ----
+
+static void apply_regexp(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!regex_match(f.re, strview(cv))) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!regex_match(f.re, strview("", 0))) bm.reset_bits();
+    return;
+  }
+
+  if (ch.type == ValueType::Dict) {
+    std::vector<uint8_t> enc;
+    for (const auto& dv : ch.dict) enc.push_back(regex_match(f.re, strview(dv)) ? 1 : 0);
+    match_encoded_dict(ctx, ch, bm, enc);
+    return;
+  }
+
+  if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+    bm.reset_bits();
+    return;
+  }
+  switch (ch.type) {
+    case ValueType::String:
+      visit_values(ctx, ch, bm, [&](strview v) { return regex_match(f.re, v); });
+      return;
+    case ValueType::Uint8:
+    case ValueType::Uint16:
+    case ValueType::Uint32:
+    case ValueType::Uint64: {
+      int width = ch.type == ValueType::Uint8 ? 1
+                  : ch.type == ValueType::Uint16 ? 2
+                  : ch.type == ValueType::Uint32 ? 4 : 8;
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string s = to_uint_string(v, width);
+        return regex_match(f.re, strview(s));
+      });
+      return;
+    }
+    case ValueType::Int64:
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string s;
+        format_int64(s, get_i64be_zigzag((const uint8_t*)v.p));
+        return regex_match(f.re, strview(s));
+      });
+      return;
+    case ValueType::Float64:
+      visit_values(ctx, ch, bm, [&](strview v) {
+        uint64_t u = get_u64be((const uint8_t*)v.p);
+        double d;
+        memcpy(&d, &u, 8);
+        std::string s;
+        format_float64(s, d);
+        return regex_match(f.re, strview(s));
+      });
+      return;
+    case ValueType::IPv4:
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string s;
+        format_ipv4(s, get_u32be((const uint8_t*)v.p));
+        return regex_match(f.re, strview(s));
+      });
+      return;
+    case ValueType::TimestampISO8601:
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string s;
+        format_timestamp_iso8601(s, int64_t(get_u64be((const uint8_t*)v.p)));
+        return regex_match(f.re, strview(s));
+      });
+      return;
+    default:
+      fail("unknown valueType in regexp filter");
+  }
+}
+
+// ---- filterTime (filter_time.go:114-137) ----
+
+static void apply_time(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  if (f.min_ts > f.max_ts) {
+    bm.reset_bits();
+    return;
+  }
+  const TimestampsHeader& th = ctx.bh->timestamps_header;
+  if (f.min_ts > th.max_timestamp || f.max_ts < th.min_timestamp) {
+    bm.reset_bits();
+    return;
+  }
+  if (f.min_ts <= th.min_timestamp && f.max_ts >= th.max_timestamp) return;
+  const auto& ts = ctx.get_timestamps();
+  bm.for_each_set_bit(
+      [&](uint64_t idx) { return ts[idx] >= f.min_ts && ts[idx] <= f.max_ts; });
+}
+
+// ---- filterRange (filter_range.go:180-372) ----
+
+static uint64_t to_u64_clamp(double f) {
+  if (f < 0) return 0;
+  if (f > double(UINT64_MAX)) return UINT64_MAX;
+  return uint64_t(f);
+}
+static int64_t to_i64_clamp(double f) {
+  if (f < double(INT64_MIN)) return INT64_MIN;
+  if (f >= double(INT64_MAX)) return INT64_MAX;
+  return int64_t(f);
+}
+static uint32_t to_u32_clamp(double f) {
+  if (f < 0) return 0;
+  if (f > double(UINT32_MAX)) return UINT32_MAX;
+  return uint32_t(f);
+}
+
+static void apply_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  double min_v = f.min_f, max_v = f.max_f;
+  if (min_v > max_v) {
+    bm.reset_bits();
+    return;
+  }
+  std::string name = canonical_field(f.field);
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    double x = parse_math_number(strview(cv));
+    if (!(x >= min_v && x <= max_v)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    bm.reset_bits();  // filter_range.go:199-204
+    return;
+  }
+
+  switch (ch.type) {
+    case ValueType::String:
+      // matchStringByRange (filter_range.go:261-265)
+      visit_values(ctx, ch, bm, [&](strview v) {
+        double x = parse_math_number(v);
+        return x >= min_v && x <= max_v;
+      });
+      return;
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) {
+        double x = parse_math_number(strview(dv));
+        enc.push_back(x >= min_v && x <= max_v ? 1 : 0);
+      }
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8:
+    case ValueType::Uint16:
+    case ValueType::Uint32:
+    case ValueType::Uint64: {
+      // matchUintNByRange (filter_range.go:267-333): bounds via
+      // ceil/floor + clamp (toUint64Range, :374-388)
+      uint64_t mn = to_u64_clamp(std::ceil(min_v));
+      uint64_t mx = to_u64_clamp(std::floor(max_v));
+      if (max_v < 0 || mn > ch.max_value || mx < ch.min_value) {
+        bm.reset_bits();
+        return;
+      }
+      int width = ch.type == ValueType::Uint8 ? 1
+                  : ch.type == ValueType::Uint16 ? 2
+                  : ch.type == ValueType::Uint32 ? 4 : 8;
+      visit_values(ctx, ch, bm, [&](strview v) {
+        uint64_t n;
+        switch (width) {
+          case 1: n = uint8_t(v.p[0]); break;
+          case 2: n = get_u16be((const uint8_t*)v.p); break;
+          case 4: n = get_u32be((const uint8_t*)v.p); break;
+          default: n = get_u64be((const uint8_t*)v.p); break;
+        }
+        return n >= mn && n <= mx;
+      });
+      return;
+    }
+    case ValueType::Int64: {
+      // matchInt64ByRange (filter_range.go:335-350)
+      int64_t mn = to_i64_clamp(std::ceil(min_v));
+      int64_t mx = to_i64_clamp(std::floor(max_v));
+      if (mn > int64_t(ch.max_value) || mx < int64_t(ch.min_value)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        int64_t n = get_i64be_zigzag((const uint8_t*)v.p);
+        return n >= mn && n <= mx;
+      });
+      return;
+    }
+    case ValueType::Float64: {
+      // matchFloat64ByRange (filter_range.go:233-246)
+      double cmn, cmx;
+      uint64_t mnu = ch.min_value, mxu = ch.max_value;
+      memcpy(&cmn, &mnu, 8);
+      memcpy(&cmx, &mxu, 8);
+      if (min_v > cmx || max_v < cmn) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        if (v.n != 8) fail("unexpected float64 binary length");
+        uint64_t u = get_u64be((const uint8_t*)v.p);
+        double d;
+        memcpy(&d, &u, 8);
+        return d >= min_v && d <= max_v;
+      });
+      return;
+    }
+    case ValueType::IPv4: {
+      // filter_range.go:223-225 + matchIPv4ByRange
+      uint32_t mn = to_u32_clamp(std::ceil(min_v));
+      uint32_t mx = to_u32_clamp(std::floor(max_v));
+      if (max_v < 0 || uint64_t(mn) > ch.max_value || uint64_t(mx) < ch.min_value) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        uint32_t n = get_u32be((const uint8_t*)v.p);
+        return n >= mn && n <= mx;
+      });
+      return;
+    }
+    case ValueType::TimestampISO8601: {
+      // matchTimestampISO8601ByRange (filter_range.go:352-367)
+      int64_t mn = to_i64_clamp(std::ceil(min_v));
+      int64_t mx = to_i64_clamp(std::floor(max_v));
+      if (max_v < 0 || mn > int64_t(ch.max_value) || mx < int64_t(ch.min_value)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        int64_t n = int64_t(get_u64be((const uint8_t*)v.p));
+        return n >= mn && n <= mx;
+      });
+      return;
+    }
+    default:
+      fail("unknown valueType in range filter");
+  }
+}
+
+// ---- AND/OR bloom prefilters (filter_and.go:76-111, filter_or.go:80-115) ----
+
+static bool and_match_bloom(const FilterNode& f, BlockCtx& ctx) {
+  for (const auto& ft : f.by_field_tokens) {
+    std::string name = canonical_field(ft.field);
+    std::string cv = ctx.const_value(name);
+    if (!cv.empty()) {
+      if (match_string_by_all_tokens(strview(cv), ft.tokens)) continue;
+      return false;
+    }
+    ColumnHeader ch;
+    if (!ctx.column_header(name, &ch)) return false;
+    if (ch.type == ValueType::Dict) {
+      if (match_dict_values_by_all_tokens(ch.dict, ft.tokens)) continue;
+      return false;
+    }
+    if (!match_bloom_all(ctx, ch, ft.hashes)) return false;
+  }
+  return true;
+}
+
+static bool or_match_bloom(const FilterNode& f, BlockCtx& ctx) {
+  if (f.by_field_tokens.empty()) return true;
+  for (const auto& ft : f.by_field_tokens) {
+    std::string name = canonical_field(ft.field);
+    std::string cv = ctx.const_value(name);
+    if (!cv.empty()) {
+      if (match_string_by_all_tokens(strview(cv), ft.tokens)) return true;
+      continue;
+    }
+    ColumnHeader ch;
+    if (!ctx.column_header(name, &ch)) continue;
+    if (ch.type == ValueType::Dict) {
+      if (match_dict_values_by_all_tokens(ch.dict, ft.tokens)) return true;
+      continue;
+    }
+    if (match_bloom_all(ctx, ch, ft.hashes)) return true;
+  }
+  return false;
+}
+
+void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  switch (f.type) {
+    case FilterNode::Phrase:
+      apply_phrase(f, ctx, bm);
+      return;
+    case FilterNode::Exact:
+      apply_exact(f, ctx, bm);
+      return;
+    case FilterNode::Regexp:
+      apply_regexp(f, ctx, bm);
+      return;
+    case FilterNode::Time:
+      apply_time(f, ctx, bm);
+      return;
+    case FilterNode::Range:
+      apply_range(f, ctx, bm);
+      return;
+    case FilterNode::Noop:
+      return;
+    case FilterNode::And: {
+      // filter_and.go:58-74
+      if (!and_match_bloom(f, ctx)) {
+        bm.reset_bits();
+        return;
+      }
+      for (const auto& c : f.children) {
+        apply_filter(c, ctx, bm);
+        if (bm.is_zero()) return;
+      }
+      return;
+    }
+    case FilterNode::Or: {
+      // filter_or.go:55-78: andNot double-buffer
+      if (!or_match_bloom(f, ctx)) {
+        bm.reset_bits();
+        return;
+      }
+      Bitmap bm_result, bm_tmp;
+      bm_result.bits_len = bm.bits_len;
+      bm_result.a = bm.a;
+      for (const auto& c : f.children) {
+        bm_tmp.bits_len = bm_result.bits_len;
+        bm_tmp.a = bm_result.a;
+        apply_filter(c, ctx, bm_tmp);
+        bm_result.and_not(bm_tmp);
+        if (bm_result.is_zero()) return;
+      }
+      bm.and_not(bm_result);
+      return;
+    }
+    case FilterNode::Not: {
+      // filter_not.go:38-46
+      Bitmap bm_tmp;
+      bm_tmp.bits_len = bm.bits_len;
+      bm_tmp.a = bm.a;
+      apply_filter(f.children[0], ctx, bm_tmp);
+      bm.and_not(bm_tmp);
+      return;
+    }
+  }
+}
+
+void search_block(const FilterNode& f, const PartReader& pr, const BlockHeader& bh,
+                  Bitmap& bm) {
+  BlockCtx ctx(&pr, &bh);
+  bm.init_ones(bh.rows_count);
+  apply_filter(f, ctx, bm);
+}
+
+}  // namespace oracle
+}  // namespace vl

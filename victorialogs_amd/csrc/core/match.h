@@ -1,0 +1,30 @@
+// Host-side match semantics restating lib/logstorage/filter_phrase.go:211-300
+// and the token helpers used by the AND/OR bloom prefilters
+// (filter_and.go:189-208).  The HIP kernels implement the same semantics
+// independently on-device; the CPU oracle checks them row-for-row.
+#pragma once
+
+#include <string>
+#include <vector>
+
+#include "vl_base.h"
+
+namespace vl {
+
+// matchPhrase (filter_phrase.go:211-218): empty phrase matches only "".
+bool match_phrase(strview s, strview phrase);
+// getPhrasePos (filter_phrase.go:220-270): first occurrence with non-token
+// boundary runes; -1 if none.
+long get_phrase_pos(strview s, strview phrase);
+
+// skipFirstLastToken (filter_regexp.go:53-69)
+std::string skip_first_last_token(const std::string& s);
+
+// matchStringByAllTokens (filter_and.go:189-196)
+bool match_string_by_all_tokens(strview v, const std::vector<std::string>& tokens);
+// matchDictValuesByAllTokens (filter_and.go:198-208): match against
+// comma-joined dict values.
+bool match_dict_values_by_all_tokens(const std::vector<std::string>& dict_values,
+                                     const std::vector<std::string>& tokens);
+
+}  // namespace vl
