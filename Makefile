@@ -1,0 +1,49 @@
+# Build for the MI355X-native VictoriaLogs block-scan engine.
+#
+#   make            -> oracle/liboracle.so (CPU) + victorialogs_amd/libvlogsql.so (HIP, gfx950)
+#   make oracle     -> CPU-only pieces (no hipcc needed)
+#
+# The HIP library cross-compiles for gfx950 without a GPU; .so files are
+# built in-tree so they travel with the gpurun snapshot.
+
+CXX      ?= g++
+HIPCC    ?= hipcc
+ARCH     ?= gfx950
+CXXFLAGS ?= -O3 -std=c++17 -fPIC -Wall
+HIPFLAGS ?= -O3 -std=c++17 -fPIC --offload-arch=$(ARCH)
+
+CORE_SRCS := $(wildcard victorialogs_amd/csrc/core/*.cpp)
+CORE_OBJS := $(patsubst victorialogs_amd/csrc/core/%.cpp,build/core/%.o,$(CORE_SRCS))
+
+all: oracle hip
+
+oracle: oracle/liboracle.so
+
+hip: victorialogs_amd/libvlogsql.so
+
+build/core/%.o: victorialogs_amd/csrc/core/%.cpp $(wildcard victorialogs_amd/csrc/core/*.h) victorialogs_amd/csrc/core/unicode_ranges.inc
+	@mkdir -p build/core
+	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+build/oracle/%.o: oracle/%.cpp oracle/oracle_filter.h $(wildcard victorialogs_amd/csrc/core/*.h)
+	@mkdir -p build/oracle
+	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+oracle/liboracle.so: $(CORE_OBJS) build/oracle/oracle_filter.o build/oracle/oracle_api.o
+	$(CXX) -shared $^ -o $@ -ldl -lpthread
+
+build/hip/scan_kernels.o: victorialogs_amd/csrc/hip/scan_kernels.hip victorialogs_amd/csrc/hip/scan_types.h victorialogs_amd/csrc/core/unicode_ranges.inc
+	@mkdir -p build/hip
+	$(HIPCC) $(HIPFLAGS) -x hip -c $< -o $@
+
+build/hip/vql_api.o: victorialogs_amd/csrc/vql_api.cpp victorialogs_amd/csrc/hip/scan_types.h $(wildcard victorialogs_amd/csrc/core/*.h)
+	@mkdir -p build/hip
+	$(HIPCC) $(HIPFLAGS) -Ivictorialogs_amd/csrc -c $< -o $@
+
+victorialogs_amd/libvlogsql.so: $(CORE_OBJS) build/hip/scan_kernels.o build/hip/vql_api.o
+	$(HIPCC) $(HIPFLAGS) -shared $^ -o $@ -ldl -lpthread
+
+clean:
+	rm -rf build oracle/liboracle.so victorialogs_amd/libvlogsql.so
+
+.PHONY: all oracle hip clean

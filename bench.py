@@ -1,0 +1,277 @@
+#!/usr/bin/env python3
+"""Benchmark for the MI355X-native VictoriaLogs block-scan engine.
+
+Workload (BASELINE.json configs[1], the largest single-GPU config the metric
+is quoted on): 100M vlogsgenerator-shaped rows with 256-byte _msg, phrase
+filter on _msg — the per-block filter evaluation of lib/logstorage
+(blockSearch.search, block_search.go:207-226) rebuilt as HIP kernels.
+
+A "step" is one pass of the hot path over the staged dataset: the filter
+program is evaluated over every block's rows (one kernel launch per staged
+part), with inputs already resident in HBM.  value = whole-job rows
+scanned/sec across all ranks (weak scaling: each rank owns its own shard).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--rows R]
+For N>1 the driver launches this under torch.distributed.run, one rank per
+GPU; the only collective is an all_reduce(SUM) of matched-row counters
+(RCCL over xGMI).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+from concurrent.futures import ThreadPoolExecutor
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK_BYTES_PER_S = 8.0e12  # MI355X spec peak (MI355X_MICROARCH.md)
+
+PHRASE_FILTER = '{"type":"phrase","field":"_msg","phrase":"message for the stream"}'
+AND_REGEX_FILTER = (
+    '{"type":"and","filters":['
+    '{"type":"phrase","field":"_msg","phrase":"message for the stream"},'
+    '{"type":"regexp","field":"var_0","re":"some value"}]}'
+)
+
+WORKLOADS = {
+    "phrase": {
+        "name": "100M rows, phrase filter on _msg column, 1xMI355X "
+                "(bloom + substring kernel)",
+        "filter": PHRASE_FILTER,
+    },
+    "phrase_regex": {
+        "name": "100M rows, AND(phrase, regex) on two string columns, 1xMI355X",
+        "filter": AND_REGEX_FILTER,
+    },
+    "dict_time": {
+        "name": "dict column + timestamp range",
+        "filter": '{"type":"and","filters":['
+                  '{"type":"phrase","field":"dict_0","phrase":"error"},'
+                  '{"type":"time","min":1700000000000000000,'
+                  '"max":1700000049999000000}]}',
+    },
+}
+
+
+def log(msg):
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def prepare_parts(data_dir, total_rows, nparts, msg_len, seed_base):
+    """Generates nparts reference-format parts in parallel (cached on disk)."""
+    from victorialogs_amd import generate_part
+
+    os.makedirs(data_dir, exist_ok=True)
+    dirs = []
+    jobs = []
+    rows_per = total_rows // nparts
+    for i in range(nparts):
+        rows = rows_per + (total_rows % nparts if i == nparts - 1 else 0)
+        d = os.path.join(data_dir, f"part_{rows}_{msg_len}_{seed_base + i}")
+        dirs.append(d)
+        if not os.path.exists(os.path.join(d, "metadata.json")):
+            jobs.append((d, rows, seed_base + i))
+    if jobs:
+        t0 = time.time()
+        log(f"generating {len(jobs)} parts ({total_rows} rows total)...")
+        with ThreadPoolExecutor(max_workers=min(len(jobs), os.cpu_count() or 8)) as ex:
+            futs = [
+                ex.submit(generate_part, d, rows, 1, 8192, msg_len, seed)
+                for d, rows, seed in jobs
+            ]
+            for f in futs:
+                f.result()
+        log(f"generation took {time.time() - t0:.1f}s")
+    return dirs
+
+
+def cpu_baseline(part_dirs, filter_json, budget_s=20.0):
+    """Times the CPU oracle (restatement of the reference's Go scan path —
+    no Go toolchain exists on this image) on a bounded sample
+    of the same workload, all host cores."""
+    from victorialogs_amd import OracleScanner
+
+    threads = os.cpu_count() or 8
+    orc = OracleScanner(part_dirs[0])
+    try:
+        nblocks = orc.blocks
+        # calibrate on a few blocks
+        t0 = time.perf_counter()
+        probe_blocks = min(8, nblocks)
+        orc.scan(filter_json, lo=0, hi=probe_blocks, threads=threads)
+        dt = max(time.perf_counter() - t0, 1e-6)
+        per_block = dt / probe_blocks
+        sample_blocks = min(nblocks, max(probe_blocks, int(budget_s / per_block)))
+        rows = sum(orc.block_rows(i) for i in range(sample_blocks))
+        t0 = time.perf_counter()
+        orc.scan(filter_json, lo=0, hi=sample_blocks, threads=threads)
+        dt = time.perf_counter() - t0
+        return {
+            "value": rows / dt,
+            "unit": "rows/s",
+            "cores": threads,
+            "kind": "port",
+            "sample": f"{sample_blocks} blocks / {rows} rows of the same part, "
+                      f"{dt:.1f}s on {threads} host threads (C++ oracle "
+                      f"restatement; reference Go binary not buildable here)",
+        }
+    finally:
+        orc.close()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--rows", type=int,
+                    default=int(os.environ.get("VQL_BENCH_ROWS", 100_000_000)))
+    ap.add_argument("--msg-len", type=int, default=256)
+    ap.add_argument("--workload", choices=sorted(WORKLOADS), default="phrase")
+    ap.add_argument("--data-dir", default=os.environ.get(
+        "VQL_DATA_DIR", "/tmp/vql_bench_data"))
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+
+    from victorialogs_amd import Filter, Part, Stage
+
+    wl = WORKLOADS[args.workload]
+    filter_json = wl["filter"]
+
+    # Each rank generates and stages its own shard (weak scaling; blocks are
+    # independent units, storage_search.go:1035-1067).
+    nparts = max(1, min(8, (os.cpu_count() or 8) // max(1, world)))
+    part_dirs = prepare_parts(
+        os.path.join(args.data_dir, f"rank{rank}"), args.rows, nparts,
+        args.msg_len, seed_base=1000 * rank + 1)
+
+    log(f"rank {rank}: staging {len(part_dirs)} parts onto device {local_rank}")
+    t0 = time.time()
+    parts = [Part(d) for d in part_dirs]
+    filt = Filter(filter_json)
+    stages = [Stage(p, filt, device=local_rank) for p in parts]
+    staged_bytes = sum(s.staged_bytes for s in stages)
+    algo_bytes = sum(s.algo_bytes for s in stages)
+    rows = sum(s.rows for s in stages)
+    log(f"rank {rank}: staged {staged_bytes / 1e9:.2f} GB "
+        f"({rows} rows) in {time.time() - t0:.1f}s")
+
+    def one_step():
+        hits = 0
+        kms = 0.0
+        for s in stages:
+            hits += s.scan()
+            kms += s.last_kernel_ms
+        return hits, kms
+
+    # warmup
+    for _ in range(args.warmup):
+        hits, _ = one_step()
+
+    # timed region: barrier + sync on both sides, MAX over ranks
+    if distributed:
+        dist.barrier()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    kernel_ms_total = 0.0
+    for _ in range(args.steps):
+        hits, kms = one_step()
+        kernel_ms_total += kms
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # whole-job aggregation: MAX(elapsed) over ranks; SUM(rows, hits)
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        c = torch.tensor([float(rows), float(hits), float(algo_bytes)],
+                         dtype=torch.float64, device="cuda")
+        dist.all_reduce(c, op=dist.ReduceOp.SUM)
+        total_rows, total_hits, total_algo = (int(c[0].item()), int(c[1].item()),
+                                              int(c[2].item()))
+    else:
+        total_rows, total_hits, total_algo = rows, hits, algo_bytes
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    rows_per_s = total_rows * args.steps / elapsed
+    gb_per_s = total_algo * args.steps / elapsed / 1e9
+
+    # roofline for the dominant (only) kernel, HIP-event timed per launch on
+    # its own stream; algorithmic bytes per launch / avg launch duration.
+    avg_kernel_s = (kernel_ms_total / 1000.0) / (args.steps * max(1, len(stages)))
+    algo_per_launch = algo_bytes / max(1, len(stages))
+    achieved = algo_per_launch / avg_kernel_s if avg_kernel_s > 0 else 0.0
+    roofline = {
+        "bound": "hbm",
+        "achieved": achieved,
+        "peak": HBM_PEAK_BYTES_PER_S,
+        "unit": "B/s",
+        "frac": achieved / HBM_PEAK_BYTES_PER_S,
+        "traffic": None,  # PMC counters come from the committed rocprofv3 runs
+        "kernel": "scan_program_kernel",
+        "avg_kernel_ms": avg_kernel_s * 1000.0,
+    }
+
+    result = None
+    if rank == 0:
+        cb = None
+        if world == 1 and not args.skip_cpu_baseline:
+            log("timing CPU baseline (oracle restatement)...")
+            cb = cpu_baseline(part_dirs, filter_json)
+        result = {
+            "metric": "matched-rows/sec",
+            "value": rows_per_s,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": wl["name"],
+                "rows_per_gpu": rows,
+                "msg_len": args.msg_len,
+                "filter": filter_json,
+                "matched_rows_per_pass": total_hits,
+                "parallelism": f"dp{world}",
+            },
+            "gb_scanned_per_sec": gb_per_s,
+            "hbm_resident_bytes": staged_bytes * world,
+            "roofline": roofline,
+            "cpu_baseline": cb,
+        }
+        print(json.dumps(result), flush=True)
+
+    for s in stages:
+        s.close()
+    filt.close()
+    for p in parts:
+        p.close()
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,75 @@
+/* vlogsql.h — C ABI of the MI355X-native VictoriaLogs block-scan engine.
+ *
+ * This is the drop-in boundary: the entry points a cgo shim
+ * inside VictoriaLogs would bind to replace the per-block filter evaluation
+ * invoked at lib/logstorage/block_search.go:215
+ * (filter.applyToBlockSearch(bs, bm), interface lib/logstorage/filter.go:8-20)
+ * from the search worker loop (lib/logstorage/storage_search.go:1040-1066).
+ * See INTEGRATION.md for the Go-side binding a maintainer would add.
+ *
+ * Bitmap layout contract: bit i of a block's result = row i of the block,
+ * LSB-first within little-endian u64 words (lib/logstorage/bitmap.go:113-125).
+ * Result bitmaps for a scanned range are concatenated per block in block
+ * order, each block padded to a word boundary.
+ *
+ * Threading: one stage/scan context per caller thread; parts and compiled
+ * filters are immutable after creation and may be shared
+ * (mirrors blockSearch pooling, block_search.go:79-96).
+ *
+ * Errors: functions returning pointers return NULL on failure; functions
+ * returning counts return -1.  vql_errstr() returns the thread-local message
+ * (the reference panics on corruption, block_search.go:264,318 — we surface
+ * hard errors instead).
+ */
+#ifndef VLOGSQL_H
+#define VLOGSQL_H
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+const char* vql_errstr(void);
+
+/* Opens a reference-format part directory (FormatVersion 1..3; the file set
+ * of lib/logstorage/filenames.go:3-24, read as part.go:105-173 does). */
+void* vql_open_part(const char* dir);
+void vql_close_part(void* part);
+long vql_part_blocks(void* part);
+long long vql_part_rows(void* part);
+long vql_block_rows(void* part, long block);
+
+/* Compiles a JSON filter tree (same shapes filter_test.go:34-59 builds
+ * programmatically).  Node types: phrase, exact, regexp, and, or, not, time,
+ * range, noop. */
+void* vql_compile_filter(const char* json);
+void vql_free_filter(void* filter);
+
+/* Steady-state path: decode blocks [block_lo, block_hi) of the part on the
+ * host once and stage the needed columns + blooms into HBM of `device`.
+ * block_hi < 0 means "all blocks". */
+void* vql_stage(void* part, void* filter, int device, long block_lo,
+                long block_hi);
+void vql_stage_free(void* stage);
+long long vql_stage_bytes(void* stage);      /* bytes resident in HBM */
+long long vql_stage_algo_bytes(void* stage); /* algorithmic bytes per pass */
+long long vql_stage_rows(void* stage);
+
+/* One scan pass over the staged blocks (one kernel launch).  Returns the
+ * number of matched rows; bitmaps stay device-resident. */
+long long vql_scan_staged(void* stage);
+/* Kernel time of the last vql_scan_staged, measured with HIP events on the
+ * stream the kernel was launched on. */
+double vql_last_kernel_ms(void* stage);
+/* Copies result bitmaps to host (concatenated per-block u64 words). */
+int vql_fetch_bitmaps(void* stage, unsigned long long* out_words,
+                      long long cap_words);
+
+/* Cold path (§8b vql_scan_batch): stage + scan + fetch + free in one call. */
+long long vql_scan_batch(void* part, void* filter, long block_lo, long block_hi,
+                         unsigned long long* out_words, long long cap_words);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* VLOGSQL_H */

@@ -7,6 +7,13 @@
 
 #include "oracle_filter.h"
 #include "../victorialogs_amd/csrc/core/gen.h"
+#include "../victorialogs_amd/csrc/core/bloom.h"
+#include "../victorialogs_amd/csrc/core/json.h"
+#include "../victorialogs_amd/csrc/core/match.h"
+#include "../victorialogs_amd/csrc/core/part_writer.h"
+#include "../victorialogs_amd/csrc/core/tokenizer.h"
+#include "../victorialogs_amd/csrc/core/values.h"
+#include "../victorialogs_amd/csrc/core/xxhash64.h"
 
 using namespace vl;
 using namespace vl::oracle;
@@ -127,6 +134,107 @@ long long orc_generate_part(const char* dir, unsigned long long rows,
     return (long long)generate_part(dir, cfg);
   } catch (const std::exception& e) {
     return set_err(e);
+  }
+}
+
+
+// ---- test-support exports (golden-vector checks from pytest) ----
+
+unsigned long long orc_xxhash64(const uint8_t* p, long n) { return xxhash64(p, size_t(n)); }
+
+// bloomFilterMarshalTokens over newline-separated tokens; returns marshaled
+// size, writes up to cap bytes.
+long orc_bloom_marshal_tokens(const char* tokens_nl, uint8_t* out, long cap) {
+  std::vector<std::string> tokens;
+  std::string s(tokens_nl);
+  size_t pos = 0;
+  while (pos <= s.size() && !s.empty()) {
+    size_t nl = s.find('\n', pos);
+    if (nl == std::string::npos) {
+      tokens.push_back(s.substr(pos));
+      break;
+    }
+    tokens.push_back(s.substr(pos, nl - pos));
+    pos = nl + 1;
+  }
+  bytes b = bloom_marshal_tokens(tokens);
+  long n = long(b.size()) < cap ? long(b.size()) : cap;
+  memcpy(out, b.data(), size_t(n));
+  return long(b.size());
+}
+
+int orc_match_phrase(const char* s, long sn, const char* ph, long pn) {
+  return match_phrase(strview(s, size_t(sn)), strview(ph, size_t(pn))) ? 1 : 0;
+}
+
+// tokenizeStrings of one string; returns newline-joined tokens.
+long orc_tokenize(const char* s, long sn, char* out, long cap) {
+  auto tokens = tokenize_strings({std::string(s, size_t(sn))});
+  std::string joined;
+  for (size_t i = 0; i < tokens.size(); i++) {
+    if (i) joined += '\n';
+    joined += tokens[i];
+  }
+  long n = long(joined.size()) < cap ? long(joined.size()) : cap;
+  memcpy(out, joined.data(), size_t(n));
+  return long(joined.size());
+}
+
+long orc_format_float64(double f, char* out, long cap) {
+  std::string s;
+  format_float64(s, f);
+  long n = long(s.size()) < cap ? long(s.size()) : cap;
+  memcpy(out, s.data(), size_t(n));
+  return long(s.size());
+}
+
+long orc_format_iso8601(long long nsecs, char* out, long cap) {
+  std::string s;
+  format_timestamp_iso8601(s, nsecs);
+  long n = long(s.size()) < cap ? long(s.size()) : cap;
+  memcpy(out, s.data(), size_t(n));
+  return long(s.size());
+}
+
+int orc_parse_iso8601(const char* s, long sn, long long* out) {
+  int64_t v;
+  if (!try_parse_timestamp_iso8601(strview(s, size_t(sn)), &v)) return 0;
+  *out = v;
+  return 1;
+}
+
+// Writes a custom part from a JSON fixture (filter_test.go fixture style):
+// {"blocks":[{"stream":0,"timestamps":[...],
+//             "columns":[{"name":"x","values":["a","b",...]}]}]}
+// Returns 0 on success, -1 on error.
+int orc_write_custom_part(const char* dir, const char* spec_json) {
+  try {
+    JValue v = json_parse(spec_json);
+    PartWriter w(dir, 1);
+    for (const auto& bj : jget(v, "blocks").arr) {
+      StreamID sid;
+      auto it = bj.obj.find("stream");
+      uint64_t snum = it != bj.obj.end() ? uint64_t(it->second.num) : 0;
+      sid.id_hi = snum;
+      sid.id_lo = 1;
+      std::vector<int64_t> ts;
+      for (const auto& t : jget(bj, "timestamps").arr) ts.push_back(int64_t(t.num));
+      std::vector<InputColumn> cols;
+      for (const auto& cj : jget(bj, "columns").arr) {
+        InputColumn c;
+        c.name = jget(cj, "name").str;
+        if (c.name == "_msg") c.name = "";  // canonical (log_rows.go:508-513)
+        for (const auto& vv : jget(cj, "values").arr) c.values.push_back(vv.str);
+        if (c.values.size() != ts.size()) fail("column length != timestamps length");
+        cols.push_back(std::move(c));
+      }
+      w.add_block(sid, ts, cols);
+    }
+    w.finish();
+    return 0;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
   }
 }
 

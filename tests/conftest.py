@@ -1,0 +1,159 @@
+import os
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, ROOT)
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires a real MI355X (run via gpurun)"
+    )
+
+
+@pytest.fixture(scope="session")
+def gen_part(tmp_path_factory):
+    """A small vlogsgenerator-shaped part shared across tests."""
+    from victorialogs_amd import generate_part
+
+    d = str(tmp_path_factory.mktemp("parts") / "gen")
+    generate_part(d, rows=30000, streams=2, rows_per_block=4096, msg_len=256,
+                  seed=42)
+    return d
+
+
+@pytest.fixture(scope="session")
+def typed_part(tmp_path_factory):
+    """A custom part exercising every valueType, with edge-case values
+    (filter_test.go fixture style)."""
+    from victorialogs_amd import write_custom_part
+
+    rows = 300
+    ts = [1700000000000000000 + i * 1000000 for i in range(rows)]
+    dict_vals = ["debug", "info", "warn", "error"]
+    spec = {
+        "blocks": [
+            {
+                "stream": 0,
+                "timestamps": ts,
+                "columns": [
+                    {"name": "_msg", "values": [
+                        f"log line {i} level={dict_vals[i % 4]} took {i % 50}ms"
+                        for i in range(rows)]},
+                    {"name": "lvl", "values": [dict_vals[i % 4] for i in range(rows)]},
+                    {"name": "u8", "values": [str(i % 250) for i in range(rows)]},
+                    {"name": "u16", "values": [str(300 + i * 7) for i in range(rows)]},
+                    {"name": "u32", "values": [str(70000 + i * 1009) for i in range(rows)]},
+                    {"name": "u64", "values": [str(5000000000 + i * 999999) for i in range(rows)]},
+                    {"name": "i64", "values": [str((i - 150) * 37) for i in range(rows)]},
+                    {"name": "f64", "values": [f"{(i - 150) / 8}" for i in range(rows)]},
+                    {"name": "ip", "values": [f"10.{i % 256}.{(i * 3) % 256}.{(i * 7) % 256}"
+                                              for i in range(rows)]},
+                    {"name": "iso", "values": [
+                        "2024-01-%02dT%02d:%02d:%02d.%03dZ"
+                        % (1 + i % 28, i % 24, i % 60, (i * 3) % 60, i % 1000)
+                        for i in range(rows)]},
+                    {"name": "constcol", "values": ["fixed value 42"] * rows},
+                    {"name": "uni", "values": [
+                        ("раз два три" if i % 3 == 0 else "foo bar") for i in range(rows)]},
+                ],
+            },
+            {
+                # second block, different dict content and some empty values
+                "stream": 1,
+                "timestamps": [t + 10**12 for t in ts],
+                "columns": [
+                    {"name": "_msg", "values": [
+                        f"other stream row {i}; ip=192.168.1.{i % 256}"
+                        for i in range(rows)]},
+                    {"name": "lvl", "values": [["fatal", "ERROR"][i % 2] for i in range(rows)]},
+                    {"name": "u8", "values": [str(i % 7) for i in range(rows)]},
+                ],
+            },
+        ]
+    }
+    d = str(tmp_path_factory.mktemp("parts") / "typed")
+    write_custom_part(d, spec)
+    return d
+
+
+# A shared battery of filters used by both the CPU consistency tests and the
+# GPU parity tests.  Every entry must compile on both paths.
+FILTERS = [
+    '{"type":"phrase","field":"_msg","phrase":"message"}',
+    '{"type":"phrase","field":"_msg","phrase":"the stream 1 and"}',
+    '{"type":"phrase","field":"_msg","phrase":"absent_token_zzz"}',
+    '{"type":"phrase","field":"_msg","phrase":""}',
+    '{"type":"phrase","field":"dict_0","phrase":"error"}',
+    '{"type":"phrase","field":"missing_col","phrase":"x"}',
+    '{"type":"phrase","field":"host","phrase":"host_0"}',
+    '{"type":"exact","field":"host","value":"host_1"}',
+    '{"type":"phrase","field":"u8_0","phrase":"17"}',
+    '{"type":"phrase","field":"u16_0","phrase":"300"}',
+    '{"type":"phrase","field":"u64_0","phrase":"123"}',
+    '{"type":"phrase","field":"i64_0","phrase":"-55"}',
+    '{"type":"regexp","field":"_msg","re":"stream (0|1) and"}',
+    '{"type":"regexp","field":"_msg","re":"uuid=[0-9a-f]"}',
+    '{"type":"regexp","field":"dict_0","re":"err(or)|info"}',
+    '{"type":"regexp","field":"_msg","re":"ip=1.*uuid"}',
+    '{"type":"time","min":1700000000000000000,"max":1700000004999000000}',
+    '{"type":"time","min":0,"max":1}',
+    '{"type":"range","field":"u8_0","min":10,"max":99.5}',
+    '{"type":"range","field":"u64_0","min":0,"max":1e18}',
+    '{"type":"range","field":"i64_0","min":-1e18,"max":0}',
+    '{"type":"range","field":"float_0","min":1.25,"max":7.5}',
+    '{"type":"and","filters":['
+    '{"type":"phrase","field":"_msg","phrase":"message"},'
+    '{"type":"phrase","field":"dict_0","phrase":"info"}]}',
+    '{"type":"or","filters":['
+    '{"type":"phrase","field":"dict_0","phrase":"error"},'
+    '{"type":"phrase","field":"dict_0","phrase":"fatal"},'
+    '{"type":"phrase","field":"dict_0","phrase":"INFO"}]}',
+    '{"type":"not","filter":{"type":"phrase","field":"dict_0","phrase":"error"}}',
+    '{"type":"and","filters":['
+    '{"type":"time","min":1700000000000000000,"max":1700000019999000000},'
+    '{"type":"not","filter":{"type":"regexp","field":"_msg","re":"stream 1"}},'
+    '{"type":"or","filters":['
+    '{"type":"phrase","field":"dict_0","phrase":"warn"},'
+    '{"type":"phrase","field":"dict_1","phrase":"debug"}]}]}',
+]
+
+TYPED_FILTERS = [
+    '{"type":"phrase","field":"_msg","phrase":"level=error"}',
+    '{"type":"phrase","field":"lvl","phrase":"warn"}',
+    '{"type":"exact","field":"lvl","value":"ERROR"}',
+    '{"type":"phrase","field":"u8","phrase":"13"}',
+    '{"type":"phrase","field":"u16","phrase":"1000"}',
+    '{"type":"phrase","field":"u32","phrase":"171512"}',
+    '{"type":"phrase","field":"u64","phrase":"5099999901"}',
+    '{"type":"phrase","field":"i64","phrase":"-37"}',
+    '{"type":"phrase","field":"f64","phrase":"-18.625"}',
+    '{"type":"phrase","field":"ip","phrase":"10.5.15.35"}',
+    '{"type":"phrase","field":"ip","phrase":"10.5"}',
+    '{"type":"phrase","field":"iso","phrase":"2024-01-03T02:02:06.002Z"}',
+    '{"type":"phrase","field":"constcol","phrase":"fixed value 42"}',
+    '{"type":"phrase","field":"constcol","phrase":"value"}',
+    '{"type":"phrase","field":"uni","phrase":"два"}',
+    '{"type":"exact","field":"u8","value":"3"}',
+    '{"type":"exact","field":"f64","value":"0.125"}',
+    '{"type":"exact","field":"ip","value":"10.9.27.63"}',
+    '{"type":"range","field":"u8","min":100,"max":200}',
+    '{"type":"range","field":"i64","min":-100.5,"max":100.5}',
+    '{"type":"range","field":"f64","min":-1,"max":1}',
+    '{"type":"range","field":"lvl","min":0,"max":1}',
+    '{"type":"range","field":"ip","min":0,"max":4e9}',
+    '{"type":"regexp","field":"lvl","re":"warn|error"}',
+    '{"type":"regexp","field":"u8","re":"^?(1|2)"}'.replace("^?", ""),
+    '{"type":"regexp","field":"ip","re":"192\\\\.168"}',
+    '{"type":"regexp","field":"iso","re":"2024-01-0"}',
+    '{"type":"regexp","field":"i64","re":"-3"}',
+    '{"type":"and","filters":['
+    '{"type":"phrase","field":"lvl","phrase":"error"},'
+    '{"type":"range","field":"u8","min":0,"max":50},'
+    '{"type":"not","filter":{"type":"phrase","field":"_msg","phrase":"took 13ms"}}]}',
+    '{"type":"or","filters":['
+    '{"type":"phrase","field":"_msg","phrase":"level=warn"},'
+    '{"type":"phrase","field":"_msg","phrase":"other stream row 7"}]}',
+]

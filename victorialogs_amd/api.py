@@ -1,0 +1,285 @@
+"""ctypes bindings for the product (HIP) and oracle (CPU) C ABIs.
+
+The product library is the hot path (include/vlogsql.h); the oracle library
+is TEST INFRASTRUCTURE only — used by tests/, __graft_entry__.smoke() and
+bench.py's cpu_baseline leg (see oracle/oracle_filter.h).
+"""
+
+import ctypes
+import os
+
+_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def lib_path() -> str:
+    return os.path.join(_ROOT, "victorialogs_amd", "libvlogsql.so")
+
+
+def oracle_lib_path() -> str:
+    return os.path.join(_ROOT, "oracle", "liboracle.so")
+
+
+_product = None
+_oracle = None
+
+
+def load_product() -> ctypes.CDLL:
+    """Loads the HIP scan library.  Raises if missing — no CPU fallback."""
+    global _product
+    if _product is not None:
+        return _product
+    path = lib_path()
+    if not os.path.exists(path):
+        raise RuntimeError(
+            f"HIP extension missing: {path}. Run `make hip` (hipcc, gfx950); "
+            "the product path has no CPU fallback."
+        )
+    lib = ctypes.CDLL(path)
+    lib.vql_errstr.restype = ctypes.c_char_p
+    lib.vql_open_part.restype = ctypes.c_void_p
+    lib.vql_open_part.argtypes = [ctypes.c_char_p]
+    lib.vql_close_part.argtypes = [ctypes.c_void_p]
+    lib.vql_part_blocks.restype = ctypes.c_long
+    lib.vql_part_blocks.argtypes = [ctypes.c_void_p]
+    lib.vql_part_rows.restype = ctypes.c_longlong
+    lib.vql_part_rows.argtypes = [ctypes.c_void_p]
+    lib.vql_block_rows.restype = ctypes.c_long
+    lib.vql_block_rows.argtypes = [ctypes.c_void_p, ctypes.c_long]
+    lib.vql_compile_filter.restype = ctypes.c_void_p
+    lib.vql_compile_filter.argtypes = [ctypes.c_char_p]
+    lib.vql_free_filter.argtypes = [ctypes.c_void_p]
+    lib.vql_stage.restype = ctypes.c_void_p
+    lib.vql_stage.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+                              ctypes.c_long, ctypes.c_long]
+    lib.vql_stage_free.argtypes = [ctypes.c_void_p]
+    lib.vql_stage_bytes.restype = ctypes.c_longlong
+    lib.vql_stage_bytes.argtypes = [ctypes.c_void_p]
+    lib.vql_stage_algo_bytes.restype = ctypes.c_longlong
+    lib.vql_stage_algo_bytes.argtypes = [ctypes.c_void_p]
+    lib.vql_stage_rows.restype = ctypes.c_longlong
+    lib.vql_stage_rows.argtypes = [ctypes.c_void_p]
+    lib.vql_scan_staged.restype = ctypes.c_longlong
+    lib.vql_scan_staged.argtypes = [ctypes.c_void_p]
+    lib.vql_last_kernel_ms.restype = ctypes.c_double
+    lib.vql_last_kernel_ms.argtypes = [ctypes.c_void_p]
+    lib.vql_fetch_bitmaps.restype = ctypes.c_int
+    lib.vql_fetch_bitmaps.argtypes = [ctypes.c_void_p,
+                                      ctypes.POINTER(ctypes.c_ulonglong),
+                                      ctypes.c_longlong]
+    lib.vql_scan_batch.restype = ctypes.c_longlong
+    lib.vql_scan_batch.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
+                                   ctypes.c_long,
+                                   ctypes.POINTER(ctypes.c_ulonglong),
+                                   ctypes.c_longlong]
+    _product = lib
+    return lib
+
+
+def _load_oracle() -> ctypes.CDLL:
+    global _oracle
+    if _oracle is not None:
+        return _oracle
+    path = oracle_lib_path()
+    if not os.path.exists(path):
+        raise RuntimeError(f"oracle library missing: {path}. Run `make oracle`.")
+    lib = ctypes.CDLL(path)
+    lib.orc_errstr.restype = ctypes.c_char_p
+    lib.orc_open_part.restype = ctypes.c_void_p
+    lib.orc_open_part.argtypes = [ctypes.c_char_p]
+    lib.orc_close_part.argtypes = [ctypes.c_void_p]
+    lib.orc_block_count.restype = ctypes.c_long
+    lib.orc_block_count.argtypes = [ctypes.c_void_p]
+    lib.orc_block_rows.restype = ctypes.c_long
+    lib.orc_block_rows.argtypes = [ctypes.c_void_p, ctypes.c_long]
+    lib.orc_part_rows.restype = ctypes.c_long
+    lib.orc_part_rows.argtypes = [ctypes.c_void_p]
+    lib.orc_compile_filter.restype = ctypes.c_void_p
+    lib.orc_compile_filter.argtypes = [ctypes.c_char_p]
+    lib.orc_free_filter.argtypes = [ctypes.c_void_p]
+    lib.orc_scan_blocks.restype = ctypes.c_longlong
+    lib.orc_scan_blocks.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
+                                    ctypes.c_long,
+                                    ctypes.POINTER(ctypes.c_ulonglong),
+                                    ctypes.c_longlong, ctypes.c_int]
+    lib.orc_generate_part.restype = ctypes.c_longlong
+    lib.orc_generate_part.argtypes = [ctypes.c_char_p, ctypes.c_ulonglong,
+                                      ctypes.c_ulonglong, ctypes.c_ulonglong,
+                                      ctypes.c_ulonglong, ctypes.c_ulonglong]
+    lib.orc_write_custom_part.restype = ctypes.c_int
+    lib.orc_write_custom_part.argtypes = [ctypes.c_char_p, ctypes.c_char_p]
+    lib.orc_xxhash64.restype = ctypes.c_ulonglong
+    lib.orc_xxhash64.argtypes = [ctypes.c_char_p, ctypes.c_long]
+    lib.orc_bloom_marshal_tokens.restype = ctypes.c_long
+    lib.orc_bloom_marshal_tokens.argtypes = [ctypes.c_char_p, ctypes.c_char_p,
+                                             ctypes.c_long]
+    lib.orc_match_phrase.restype = ctypes.c_int
+    lib.orc_match_phrase.argtypes = [ctypes.c_char_p, ctypes.c_long,
+                                     ctypes.c_char_p, ctypes.c_long]
+    lib.orc_tokenize.restype = ctypes.c_long
+    lib.orc_tokenize.argtypes = [ctypes.c_char_p, ctypes.c_long, ctypes.c_char_p,
+                                 ctypes.c_long]
+    lib.orc_format_float64.restype = ctypes.c_long
+    lib.orc_format_float64.argtypes = [ctypes.c_double, ctypes.c_char_p,
+                                       ctypes.c_long]
+    lib.orc_format_iso8601.restype = ctypes.c_long
+    lib.orc_format_iso8601.argtypes = [ctypes.c_longlong, ctypes.c_char_p,
+                                       ctypes.c_long]
+    lib.orc_parse_iso8601.restype = ctypes.c_int
+    lib.orc_parse_iso8601.argtypes = [ctypes.c_char_p, ctypes.c_long,
+                                      ctypes.POINTER(ctypes.c_longlong)]
+    _oracle = lib
+    return lib
+
+
+def generate_part(dir_path, rows, streams=1, rows_per_block=8192, msg_len=256,
+                  seed=1):
+    """Generates a reference-format part with vlogsgenerator-shaped rows
+    (app/vlogsgenerator/main.go:234-297 shapes).  Returns total _msg bytes."""
+    lib = _load_oracle()
+    r = lib.orc_generate_part(dir_path.encode(), rows, streams, rows_per_block,
+                              msg_len, seed)
+    if r < 0:
+        raise RuntimeError(lib.orc_errstr().decode())
+    return r
+
+
+def write_custom_part(dir_path, spec):
+    """Writes a part from a JSON fixture spec (test infrastructure; mirrors
+    the filter_test.go:158-277 fixture pattern)."""
+    import json as _json
+    lib = _load_oracle()
+    if lib.orc_write_custom_part(dir_path.encode(), _json.dumps(spec).encode()) != 0:
+        raise RuntimeError(lib.orc_errstr().decode())
+
+
+def oracle_helpers():
+    """Raw oracle lib for golden-vector tests."""
+    return _load_oracle()
+
+
+class OracleScanner:
+    """TEST INFRASTRUCTURE — CPU oracle scans for parity checks and the
+    cpu_baseline bench leg."""
+
+    def __init__(self, part_dir):
+        self.lib = _load_oracle()
+        self.part = self.lib.orc_open_part(part_dir.encode())
+        if not self.part:
+            raise RuntimeError(self.lib.orc_errstr().decode())
+
+    def close(self):
+        if self.part:
+            self.lib.orc_close_part(self.part)
+            self.part = None
+
+    @property
+    def blocks(self):
+        return self.lib.orc_block_count(self.part)
+
+    def block_rows(self, i):
+        return self.lib.orc_block_rows(self.part, i)
+
+    def words_for(self, lo=0, hi=-1):
+        if hi < 0:
+            hi = self.blocks
+        return sum((self.block_rows(i) + 63) // 64 for i in range(lo, hi))
+
+    def scan(self, filter_json, lo=0, hi=-1, with_bitmaps=False, threads=1):
+        f = self.lib.orc_compile_filter(filter_json.encode())
+        if not f:
+            raise RuntimeError(self.lib.orc_errstr().decode())
+        try:
+            if with_bitmaps:
+                nwords = self.words_for(lo, hi if hi >= 0 else self.blocks)
+                buf = (ctypes.c_ulonglong * max(nwords, 1))()
+                hits = self.lib.orc_scan_blocks(self.part, f, lo, hi, buf, nwords,
+                                                threads)
+                if hits < 0:
+                    raise RuntimeError(self.lib.orc_errstr().decode())
+                return hits, bytes(buf)[: nwords * 8]
+            hits = self.lib.orc_scan_blocks(self.part, f, lo, hi, None, 0, threads)
+            if hits < 0:
+                raise RuntimeError(self.lib.orc_errstr().decode())
+            return hits, None
+        finally:
+            self.lib.orc_free_filter(f)
+
+
+class Part:
+    def __init__(self, part_dir):
+        self.lib = load_product()
+        self.h = self.lib.vql_open_part(part_dir.encode())
+        if not self.h:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+
+    def close(self):
+        if self.h:
+            self.lib.vql_close_part(self.h)
+            self.h = None
+
+    @property
+    def blocks(self):
+        return self.lib.vql_part_blocks(self.h)
+
+    @property
+    def rows(self):
+        return self.lib.vql_part_rows(self.h)
+
+    def block_rows(self, i):
+        return self.lib.vql_block_rows(self.h, i)
+
+
+class Filter:
+    def __init__(self, filter_json):
+        self.lib = load_product()
+        self.h = self.lib.vql_compile_filter(filter_json.encode())
+        if not self.h:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+
+    def close(self):
+        if self.h:
+            self.lib.vql_free_filter(self.h)
+            self.h = None
+
+
+class Stage:
+    """Staged (HBM-resident) scan context: decode-once, scan-many."""
+
+    def __init__(self, part: Part, filt: Filter, device=0, lo=0, hi=-1):
+        self.lib = load_product()
+        self.h = self.lib.vql_stage(part.h, filt.h, device, lo, hi)
+        if not self.h:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+
+    def close(self):
+        if self.h:
+            self.lib.vql_stage_free(self.h)
+            self.h = None
+
+    @property
+    def staged_bytes(self):
+        return self.lib.vql_stage_bytes(self.h)
+
+    @property
+    def algo_bytes(self):
+        return self.lib.vql_stage_algo_bytes(self.h)
+
+    @property
+    def rows(self):
+        return self.lib.vql_stage_rows(self.h)
+
+    def scan(self):
+        hits = self.lib.vql_scan_staged(self.h)
+        if hits < 0:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+        return hits
+
+    @property
+    def last_kernel_ms(self):
+        return self.lib.vql_last_kernel_ms(self.h)
+
+    def fetch_bitmaps(self, nwords):
+        buf = (ctypes.c_ulonglong * max(nwords, 1))()
+        if self.lib.vql_fetch_bitmaps(self.h, buf, nwords) != 0:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+        return bytes(buf)[: nwords * 8]

@@ -1,0 +1,24 @@
+// Minimal JSON parser used for programmatic filter trees and test fixtures.
+#pragma once
+
+#include <map>
+#include <string>
+#include <vector>
+
+#include "vl_base.h"
+
+namespace vl {
+
+struct JValue {
+  enum Kind { Obj, Arr, Str, Num, Bool, Null } kind = Null;
+  std::map<std::string, JValue> obj;
+  std::vector<JValue> arr;
+  std::string str;
+  double num = 0;
+  bool b = false;
+};
+
+JValue json_parse(const std::string& s);
+const JValue& jget(const JValue& o, const char* key);
+
+}  // namespace vl

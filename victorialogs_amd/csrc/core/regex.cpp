@@ -383,6 +383,15 @@ RegexProg regex_compile(const std::string& expr) {
       re.substr_dot_plus = suffix.subs[1].lit;
     }
   }
+  // prefix + ".*lit" (e.g. "abc.*def"): Go's suffixRe here is "^(?:.*lit)",
+  // whose unanchored-tail semantics in matchStringWithPrefix reduce to
+  // strings.Contains(tail, lit) — identical to the substrDotStar branch
+  // (regex.go:177-180), so classify it there.
+  if (!re.prefix.empty() && suffix.kind == RNode::Concat &&
+      suffix.subs.size() == 2 && is_dot_star(suffix.subs[0]) &&
+      suffix.subs[1].kind == RNode::Lit) {
+    re.substr_dot_star = suffix.subs[1].lit;
+  }
   if (or_ok && !ors.empty()) {
     re.or_values = std::move(ors);
     re.has_or_values = true;
