@@ -8,6 +8,14 @@
 // drop-in boundary requires).  Bloom gates (bloomfilter.go:173-191) run
 // cooperatively per chunk before any row is touched.
 //
+// String columns are scanned through an LDS tile: 256-row groups are copied
+// from HBM with coalesced 16-byte loads (streaming reads == algorithmic
+// bytes; the naive per-lane global scan fetched ~3x the data, measured with
+// rocprofv3 FETCH_SIZE), then each lane scans its own row from LDS.  The
+// tile is bank-swizzled at 16-byte-slot granularity so that the ~256-byte
+// row stride does not put all 64 lanes of a wave on the same LDS bank pair
+// (cdna_hip_programming.md §2, Guideline 4).
+//
 // This path is HBM-bandwidth-bound integer/byte work — no MFMA by design
 // (BASELINE.json north star).
 #include <hip/hip_runtime.h>
@@ -18,6 +26,9 @@ namespace vl {
 
 // Non-ASCII token-rune ranges (tokenizer.go:142-148); see unicode_ranges.inc.
 #include "../core/unicode_ranges.inc"
+
+constexpr uint32_t kTileBytes = 66 * 1024;  // 256 rows x <=~256 B + slack
+constexpr uint32_t kGroupRows = 256;
 
 __device__ __forceinline__ bool d_is_token_char(uint8_t c) {
   // tokenizer.go:132-140: [a-zA-Z0-9_]
@@ -41,11 +52,39 @@ __device__ bool d_is_token_rune(uint32_t r) {
   return false;
 }
 
-// Go utf8.DecodeRuneInString semantics (0xFFFD,1 on invalid).
-__device__ uint32_t d_utf8_decode(const char* p, long n, int* size) {
+// ---- memory accessors ----
+// All string matching routines are templated over an accessor that serves
+// 8-byte-aligned u64 loads and byte loads at tile/global byte offsets.
+
+struct GlobalAcc {
+  const uint8_t* base;  // 16-byte aligned (arena allocations)
+  __device__ __forceinline__ uint64_t u64a(long off) const {  // off % 8 == 0
+    return *(const uint64_t*)(base + off);
+  }
+  __device__ __forceinline__ uint8_t u8(long off) const { return base[off]; }
+};
+
+// LDS tile with 16-byte-slot swizzle: slot u -> u ^ ((u>>4) & 15), bijective
+// within every 4 KiB window; an aligned u64 never crosses its 16-byte slot.
+struct TileAcc {
+  const uint8_t* tile;
+  __device__ __forceinline__ long swz(long off) const {
+    long u = off >> 4;
+    return ((u ^ ((u >> 4) & 15)) << 4) | (off & 15);
+  }
+  __device__ __forceinline__ uint64_t u64a(long off) const {
+    return *(const uint64_t*)(tile + swz(off));
+  }
+  __device__ __forceinline__ uint8_t u8(long off) const { return tile[swz(off)]; }
+};
+
+// Go utf8.DecodeRuneInString semantics (0xFFFD,1 on invalid), reading bytes
+// [s, s+n) of the accessor at base offset `off`.
+template <typename A>
+__device__ uint32_t d_utf8_decode(const A& a, long off, long n, int* size) {
   *size = 1;
   if (n <= 0) return 0xFFFD;
-  uint8_t c0 = uint8_t(p[0]);
+  uint8_t c0 = a.u8(off);
   if (c0 < 0x80) return c0;
   int len;
   uint32_t r, lo;
@@ -60,7 +99,7 @@ __device__ uint32_t d_utf8_decode(const char* p, long n, int* size) {
   }
   if (len > n) return 0xFFFD;
   for (int i = 1; i < len; i++) {
-    uint8_t c = uint8_t(p[i]);
+    uint8_t c = a.u8(off + i);
     if ((c & 0xC0) != 0x80) return 0xFFFD;
     r = (r << 6) | (c & 0x3F);
   }
@@ -69,80 +108,73 @@ __device__ uint32_t d_utf8_decode(const char* p, long n, int* size) {
   return r;
 }
 
-__device__ uint32_t d_utf8_decode_last(const char* p, long n, int* size) {
+template <typename A>
+__device__ uint32_t d_utf8_decode_last(const A& a, long off, long n, int* size) {
   *size = 1;
   if (n <= 0) return 0xFFFD;
   long start = n - 1;
-  if (uint8_t(p[start]) < 0x80) return uint8_t(p[start]);
+  if (a.u8(off + start) < 0x80) return a.u8(off + start);
   long lim = n >= 4 ? n - 4 : 0;
-  while (start > lim && (uint8_t(p[start]) & 0xC0) == 0x80) start--;
+  while (start > lim && (a.u8(off + start) & 0xC0) == 0x80) start--;
   int sz;
-  uint32_t r = d_utf8_decode(p + start, n - start, &sz);
+  uint32_t r = d_utf8_decode(a, off + start, n - start, &sz);
   if (start + sz != n) return 0xFFFD;
   *size = sz;
   return r;
 }
 
-// strings.Index: first occurrence of sub in s, -1 if absent.  SWAR first-byte
-// candidate scan over 8-byte windows, then verify.
-__device__ long d_index(const char* s, long sn, const char* sub, long subn) {
+// strings.Index over accessor bytes [s0, s0+sn): first occurrence of the
+// operand (global memory, byte-addressable) or -1.  SWAR first-byte scan over
+// ALIGNED u64 windows; head/tail bytes are masked out.
+template <typename A>
+__device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
+                           long subn) {
   if (subn == 0) return 0;
   if (subn > sn) return -1;
-  const uint8_t c0 = uint8_t(sub[0]);
+  const uint8_t c0 = sub[0];
   const uint64_t pat = 0x0101010101010101ULL * c0;
-  long last = sn - subn;
-  long i = 0;
-  while (i <= last) {
-    // SWAR zero-byte trick over the next 8 bytes (bounded by last+1)
-    long win = last + 1 - i;
-    if (win >= 8) {
-      uint64_t x;
-      __builtin_memcpy(&x, s + i, 8);
-      uint64_t t = x ^ pat;
-      uint64_t hit = (t - 0x0101010101010101ULL) & ~t & 0x8080808080808080ULL;
-      if (hit == 0) {
-        i += 8;
-        continue;
-      }
-      i += long(__builtin_ctzll(hit) >> 3);
-      if (i > last) return -1;
-    } else {
-      if (uint8_t(s[i]) != c0) {
-        i++;
-        continue;
-      }
-    }
-    if (uint8_t(s[i]) == c0) {
+  const long last = s0 + sn - subn;  // last valid start (absolute)
+  long w = s0 & ~7L;
+  for (; w <= last; w += 8) {
+    uint64_t x = a.u64a(w);
+    uint64_t t = x ^ pat;
+    uint64_t hit = (t - 0x0101010101010101ULL) & ~t & 0x8080808080808080ULL;
+    // mask hits before s0 (head window) -- hit bit for byte k is bit 8k+7
+    if (w < s0) hit &= ~((uint64_t(1) << ((s0 - w) * 8)) - 1);
+    while (hit) {
+      long k = long(__builtin_ctzll(hit) >> 3);
+      hit &= hit - 1;
+      long pos = w + k;
+      if (pos > last) return -1;
       bool eq = true;
-      for (long k = 1; k < subn; k++) {
-        if (s[i + k] != sub[k]) {
+      for (long i = 1; i < subn; i++) {
+        if (a.u8(pos + i) != sub[i]) {
           eq = false;
           break;
         }
       }
-      if (eq) return i;
+      if (eq) return pos - s0;
     }
-    i++;
   }
   return -1;
 }
 
-// getPhrasePos (filter_phrase.go:220-270); starts/ends-with-token flags are
-// precomputed on the host from the phrase.
-__device__ bool d_match_phrase(const char* s, long sn, const char* ph, long phn,
-                               uint8_t flags) {
+// getPhrasePos (filter_phrase.go:220-270) over accessor bytes [s0, s0+sn).
+template <typename A>
+__device__ bool d_match_phrase_at(const A& a, long s0, long sn, const uint8_t* ph,
+                                  long phn, uint8_t flags) {
   if (phn == 0) return sn == 0;  // filter_phrase.go:212-215
   if (phn > sn) return false;
   long pos = 0;
   for (;;) {
-    long n = d_index(s + pos, sn - pos, ph, phn);
+    long n = d_index_at(a, s0 + pos, sn - pos, ph, phn);
     if (n < 0) return false;
     pos += n;
     if ((flags & kPhraseStartsToken) && pos > 0) {
-      uint32_t rb = uint8_t(s[pos - 1]);
+      uint32_t rb = a.u8(s0 + pos - 1);
       if (rb >= 0x80) {
         int sz;
-        rb = d_utf8_decode_last(s, pos, &sz);
+        rb = d_utf8_decode_last(a, s0, pos, &sz);
       }
       if (rb == 0xFFFD || d_is_token_rune(rb)) {
         pos++;
@@ -150,10 +182,10 @@ __device__ bool d_match_phrase(const char* s, long sn, const char* ph, long phn,
       }
     }
     if ((flags & kPhraseEndsToken) && pos + phn < sn) {
-      uint32_t ra = uint8_t(s[pos + phn]);
+      uint32_t ra = a.u8(s0 + pos + phn);
       if (ra >= 0x80) {
         int sz;
-        ra = d_utf8_decode(s + pos + phn, sn - pos - phn, &sz);
+        ra = d_utf8_decode(a, s0 + pos + phn, sn - pos - phn, &sz);
       }
       if (ra == 0xFFFD || d_is_token_rune(ra)) {
         pos++;
@@ -180,8 +212,7 @@ __device__ int d_format_u64(char* buf, uint64_t v) {
 __device__ int d_format_i64(char* buf, int64_t v) {
   if (v < 0) {
     buf[0] = '-';
-    // careful with INT64_MIN
-    uint64_t u = ~uint64_t(v) + 1;
+    uint64_t u = ~uint64_t(v) + 1;  // handles INT64_MIN
     return 1 + d_format_u64(buf + 1, u);
   }
   return d_format_u64(buf, uint64_t(v));
@@ -217,7 +248,6 @@ __device__ int d_format_iso8601(char* buf, int64_t nsecs) {
     sod += 86400;
     days--;
   }
-  // civil_from_days
   int64_t z = days + 719468;
   int64_t era = (z >= 0 ? z : z - 146096) / 146097;
   int64_t doe = z - era * 146097;
@@ -229,16 +259,14 @@ __device__ int d_format_iso8601(char* buf, int64_t nsecs) {
   int m = int(mp + (mp < 10 ? 3 : -9));
   int64_t y = yy + (m <= 2);
   int msec = int(rem / 1000000);
-  // yyyy-mm-ddThh:mm:ss.mmmZ (year assumed 0..9999 for %04d)
   int n = 0;
   if (y >= 1000) {
     n = d_format_u64(buf, uint64_t(y));
   } else {
-    buf[0] = '0'; buf[1] = '0'; buf[2] = '0'; buf[3] = char('0' + y % 10);
-    if (y >= 10) d_pad2(buf + 2, int(y % 100));
-    if (y >= 100) {
-      buf[1] = char('0' + (y / 100) % 10);
-    }
+    buf[0] = '0';
+    buf[1] = char('0' + (y / 100) % 10);
+    buf[2] = char('0' + (y / 10) % 10);
+    buf[3] = char('0' + y % 10);
     n = 4;
   }
   buf[n++] = '-';
@@ -259,13 +287,24 @@ __device__ int d_format_iso8601(char* buf, int64_t nsecs) {
   return n;
 }
 
+// Accessor over a tiny per-thread scratch buffer (formatted numbers).
+struct BufAcc {
+  const uint8_t* base;
+  __device__ __forceinline__ uint64_t u64a(long off) const {
+    uint64_t v;
+    __builtin_memcpy(&v, base + off, 8);
+    return v;
+  }
+  __device__ __forceinline__ uint8_t u8(long off) const { return base[off]; }
+};
+
 // ---- regex fast paths on serialized blob (regex.go:86-212) ----
 
 struct DRegex {
   uint8_t flags;
   uint16_t prefix_len, substr_len, n_or;
-  const char* prefix;
-  const char* substr;
+  const uint8_t* prefix;
+  const uint8_t* substr;
   const uint8_t* ors;  // sequence of {u16 len, bytes}
 };
 
@@ -275,24 +314,26 @@ __device__ DRegex d_regex_load(const uint8_t* blob) {
   re.prefix_len = uint16_t(blob[1]) | uint16_t(blob[2]) << 8;
   re.substr_len = uint16_t(blob[3]) | uint16_t(blob[4]) << 8;
   re.n_or = uint16_t(blob[5]) | uint16_t(blob[6]) << 8;
-  re.prefix = (const char*)blob + 7;
+  re.prefix = blob + 7;
   re.substr = re.prefix + re.prefix_len;
-  re.ors = (const uint8_t*)(re.substr + re.substr_len);
+  re.ors = re.substr + re.substr_len;
   return re;
 }
 
-__device__ bool d_regex_or_contains(const DRegex& re, const char* s, long sn) {
+template <typename A>
+__device__ bool d_regex_or_contains(const DRegex& re, const A& a, long s0, long sn) {
   const uint8_t* p = re.ors;
   for (int i = 0; i < re.n_or; i++) {
     uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
     p += 2;
-    if (d_index(s, sn, (const char*)p, len) >= 0) return true;
+    if (d_index_at(a, s0, sn, p, len) >= 0) return true;
     p += len;
   }
   return false;
 }
 
-__device__ bool d_regex_or_hasprefix(const DRegex& re, const char* s, long sn) {
+template <typename A>
+__device__ bool d_regex_or_hasprefix(const DRegex& re, const A& a, long s0, long sn) {
   const uint8_t* p = re.ors;
   for (int i = 0; i < re.n_or; i++) {
     uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
@@ -300,7 +341,7 @@ __device__ bool d_regex_or_hasprefix(const DRegex& re, const char* s, long sn) {
     if (long(len) <= sn) {
       bool eq = true;
       for (int k = 0; k < len; k++) {
-        if (s[k] != ((const char*)p)[k]) {
+        if (a.u8(s0 + k) != p[k]) {
           eq = false;
           break;
         }
@@ -312,52 +353,56 @@ __device__ bool d_regex_or_hasprefix(const DRegex& re, const char* s, long sn) {
   return false;
 }
 
-__device__ bool d_regex_match(const uint8_t* blob, const char* s, long sn) {
+// Regex.MatchString (regex.go:86-212) over accessor bytes [s0, s0+sn).
+template <typename A>
+__device__ bool d_regex_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
   DRegex re = d_regex_load(blob);
   if (re.flags & kReOnlyPrefix) {
     if (re.prefix_len == 0) return true;
-    return d_index(s, sn, re.prefix, re.prefix_len) >= 0;
+    return d_index_at(a, s0, sn, re.prefix, re.prefix_len) >= 0;
   }
   if (re.prefix_len == 0) {
     // matchStringNoPrefix (regex.go:131-160)
     if (re.flags & kReDotStar) return true;
     if (re.flags & kReDotPlus) return sn > 0;
-    if (re.flags & kReSubstrStar) return d_index(s, sn, re.substr, re.substr_len) >= 0;
+    if (re.flags & kReSubstrStar) {
+      return d_index_at(a, s0, sn, re.substr, re.substr_len) >= 0;
+    }
     if (re.flags & kReSubstrPlus) {
-      long n = d_index(s, sn, re.substr, re.substr_len);
+      long n = d_index_at(a, s0, sn, re.substr, re.substr_len);
       return n > 0 && n + re.substr_len < sn;
     }
-    return d_regex_or_contains(re, s, sn);
+    return d_regex_or_contains(re, a, s0, sn);
   }
   // matchStringWithPrefix (regex.go:162-212)
-  long n = d_index(s, sn, re.prefix, re.prefix_len);
+  long n = d_index_at(a, s0, sn, re.prefix, re.prefix_len);
   if (n < 0) return false;
-  const char* snext = s + n + 1;
-  long snext_n = sn - n - 1;
-  const char* t = s + n + re.prefix_len;
-  long tn = sn - n - re.prefix_len;
+  long next0 = s0 + n + 1, next_n = sn - n - 1;
+  long t0 = s0 + n + re.prefix_len, tn = sn - n - re.prefix_len;
 
   if (re.flags & kReDotStar) return true;
   if (re.flags & kReDotPlus) return tn > 0;
-  if (re.flags & kReSubstrStar) return d_index(t, tn, re.substr, re.substr_len) >= 0;
+  if (re.flags & kReSubstrStar) {
+    return d_index_at(a, t0, tn, re.substr, re.substr_len) >= 0;
+  }
   if (re.flags & kReSubstrPlus) {
-    long k = d_index(t, tn, re.substr, re.substr_len);
+    long k = d_index_at(a, t0, tn, re.substr, re.substr_len);
     return k > 0 && k + re.substr_len < tn;
   }
   for (;;) {
-    if (d_regex_or_hasprefix(re, t, tn)) return true;
-    s = snext;
-    sn = snext_n;
-    n = d_index(s, sn, re.prefix, re.prefix_len);
+    if (d_regex_or_hasprefix(re, a, t0, tn)) return true;
+    s0 = next0;
+    sn = next_n;
+    n = d_index_at(a, s0, sn, re.prefix, re.prefix_len);
     if (n < 0) return false;
-    snext = s + n + 1;
-    snext_n = sn - n - 1;
-    t = s + n + re.prefix_len;
+    next0 = s0 + n + 1;
+    next_n = sn - n - 1;
+    t0 = s0 + n + re.prefix_len;
     tn = sn - n - re.prefix_len;
   }
 }
 
-// ---- per-row leaf predicate ----
+// ---- per-row predicates ----
 
 __device__ __forceinline__ uint64_t d_get_u64be(const uint8_t* p) {
   uint64_t v;
@@ -373,24 +418,28 @@ __device__ __forceinline__ uint16_t d_get_u16be(const uint8_t* p) {
   return uint16_t(p[0]) << 8 | p[1];
 }
 
-__device__ bool d_eval_row(const DevLeafBlock& lb, uint32_t row) {
+// String-kind predicate over an accessor (tile or global).
+template <typename A>
+__device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
+                                  long sn) {
   switch (lb.kind) {
-    case kScanPhraseStr: {
-      uint32_t off = lb.offsets[row];
-      uint32_t len = lb.offsets[row + 1] - off;
-      return d_match_phrase((const char*)lb.data + off, len, (const char*)lb.operand,
-                            lb.operand_len, lb.flags);
-    }
+    case kScanPhraseStr:
+      return d_match_phrase_at(a, s0, sn, lb.operand, lb.operand_len, lb.flags);
     case kScanEqStr: {
-      uint32_t off = lb.offsets[row];
-      uint32_t len = lb.offsets[row + 1] - off;
-      if (len != lb.operand_len) return false;
-      const char* s = (const char*)lb.data + off;
-      for (uint32_t k = 0; k < len; k++) {
-        if (s[k] != ((const char*)lb.operand)[k]) return false;
+      if (sn != long(lb.operand_len)) return false;
+      for (long k = 0; k < sn; k++) {
+        if (a.u8(s0 + k) != lb.operand[k]) return false;
       }
       return true;
     }
+    default:  // kScanRegexStr
+      return d_regex_match_at(lb.operand, a, s0, sn);
+  }
+}
+
+// Fixed-width / dict / timestamp predicate (coalesced global reads).
+__device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
+  switch (lb.kind) {
     case kScanEqBin: {
       const uint8_t* p = lb.data + size_t(row) * lb.width;
       switch (lb.width) {
@@ -441,20 +490,17 @@ __device__ bool d_eval_row(const DevLeafBlock& lb, uint32_t row) {
       double mx = __builtin_bit_cast(double, lb.vmax);
       return v >= mn && v <= mx;
     }
-    case kScanRegexStr: {
-      uint32_t off = lb.offsets[row];
-      uint32_t len = lb.offsets[row + 1] - off;
-      return d_regex_match(lb.operand, (const char*)lb.data + off, len);
-    }
     case kScanPhraseIp: {
       char buf[16];
       int n = d_format_ipv4(buf, d_get_u32be(lb.data + size_t(row) * 4));
-      return d_match_phrase(buf, n, (const char*)lb.operand, lb.operand_len, lb.flags);
+      BufAcc a{(const uint8_t*)buf};
+      return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
     }
     case kScanPhraseIso: {
       char buf[32];
       int n = d_format_iso8601(buf, int64_t(d_get_u64be(lb.data + size_t(row) * 8)));
-      return d_match_phrase(buf, n, (const char*)lb.operand, lb.operand_len, lb.flags);
+      BufAcc a{(const uint8_t*)buf};
+      return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
     }
     case kScanRegexU: {
       const uint8_t* p = lb.data + size_t(row) * lb.width;
@@ -465,30 +511,38 @@ __device__ bool d_eval_row(const DevLeafBlock& lb, uint32_t row) {
         case 4: v = d_get_u32be(p); break;
         default: v = d_get_u64be(p); break;
       }
-      char buf[20];
+      char buf[24];
       int n = d_format_u64(buf, v);
-      return d_regex_match(lb.operand, buf, n);
+      BufAcc a{(const uint8_t*)buf};
+      return d_regex_match_at(lb.operand, a, 0, n);
     }
     case kScanRegexI: {
       uint64_t u = d_get_u64be(lb.data + size_t(row) * 8);
       int64_t v = int64_t(u >> 1) ^ (int64_t(u << 63) >> 63);
-      char buf[21];
+      char buf[24];
       int n = d_format_i64(buf, v);
-      return d_regex_match(lb.operand, buf, n);
+      BufAcc a{(const uint8_t*)buf};
+      return d_regex_match_at(lb.operand, a, 0, n);
     }
     case kScanRegexIp: {
       char buf[16];
       int n = d_format_ipv4(buf, d_get_u32be(lb.data + size_t(row) * 4));
-      return d_regex_match(lb.operand, buf, n);
+      BufAcc a{(const uint8_t*)buf};
+      return d_regex_match_at(lb.operand, a, 0, n);
     }
     case kScanRegexIso: {
       char buf[32];
       int n = d_format_iso8601(buf, int64_t(d_get_u64be(lb.data + size_t(row) * 8)));
-      return d_regex_match(lb.operand, buf, n);
+      BufAcc a{(const uint8_t*)buf};
+      return d_regex_match_at(lb.operand, a, 0, n);
     }
     default:
       return false;
   }
+}
+
+__device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
+  return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr;
 }
 
 // ---- the program kernel ----
@@ -498,6 +552,7 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
     int nleaves, const DevBlock* __restrict__ blocks,
     const DevChunk* __restrict__ chunks, unsigned long long* __restrict__ hits) {
   __shared__ uint64_t stack[kMaxStackDepth][kChunkWords];
+  __shared__ __attribute__((aligned(16))) uint8_t tile[kTileBytes];
   __shared__ int bloom_ok;
   __shared__ unsigned long long wave_sums[4];
 
@@ -528,9 +583,7 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
       if (lb.nhashes) {
         if (tid == 0) bloom_ok = 1;
         __syncthreads();
-        if (lb.bloom_words == 0) {
-          // empty bloom = containsAll true (bloomfilter.go:174-177)
-        } else {
+        if (lb.bloom_words > 0) {
           const uint64_t max_bits = uint64_t(lb.bloom_words) * 64;
           bool miss = false;
           for (uint32_t k = tid; k < lb.nhashes; k += blockDim.x) {
@@ -546,11 +599,52 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
           continue;
         }
       }
-      for (uint32_t w = wave; w < nwords; w += nwaves) {
-        uint32_t row = r0 + w * 64 + lane;
-        bool pred = row < r1 ? d_eval_row(lb, row) : false;
-        uint64_t word = __ballot(pred);
-        if (lane == 0) out[w] = word;
+
+      if (d_is_string_kind(lb.kind)) {
+        // LDS-tiled string scan: 256-row groups, coalesced 16 B copies,
+        // lane-per-row match from the swizzled tile.
+        for (uint32_t g0 = r0; g0 < r1; g0 += kGroupRows) {
+          const uint32_t g1 = min(g0 + kGroupRows, r1);
+          const uint32_t ng = g1 - g0;
+          const uint32_t byte0 = lb.offsets[g0] & ~15u;
+          const uint32_t byte1 = lb.offsets[g1];
+          const uint32_t nbytes = byte1 - byte0;
+          const bool use_tile = nbytes <= kTileBytes;
+          if (use_tile) {
+            const uint4* src = (const uint4*)(lb.data + byte0);
+            const uint32_t n16 = (nbytes + 15) >> 4;
+            for (uint32_t k = tid; k < n16; k += blockDim.x) {
+              uint32_t d = k ^ ((k >> 4) & 15);  // TileAcc swizzle, 16 B slots
+              ((uint4*)tile)[d] = src[k];
+            }
+          }
+          __syncthreads();
+          bool pred = false;
+          const uint32_t row = g0 + tid;
+          if (uint32_t(tid) < ng) {
+            const long s = lb.offsets[row];
+            const long e = lb.offsets[row + 1];
+            if (use_tile) {
+              TileAcc a{tile};
+              pred = d_eval_string_row(lb, a, s - byte0, e - s);
+            } else {
+              GlobalAcc a{lb.data};
+              pred = d_eval_string_row(lb, a, s, e - s);
+            }
+          }
+          const uint64_t word = __ballot(pred);
+          if (lane == 0 && uint32_t(wave * 64) < ng) {
+            out[(g0 - r0) / 64 + wave] = word;
+          }
+          __syncthreads();
+        }
+      } else {
+        for (uint32_t w = wave; w < nwords; w += nwaves) {
+          uint32_t row = r0 + w * 64 + lane;
+          bool pred = row < r1 ? d_eval_fixed_row(lb, row) : false;
+          uint64_t word = __ballot(pred);
+          if (lane == 0) out[w] = word;
+        }
       }
       sp++;
     } else if (op.kind == kOpNot) {
@@ -583,7 +677,6 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
     blk.bitmap_out[r0 / 64 + w] = word;
     local += __popcll(word);
   }
-  // workgroup reduction
   for (int off = 32; off > 0; off >>= 1) local += __shfl_down(local, off, 64);
   if (lane == 0) wave_sums[wave] = local;
   __syncthreads();

@@ -257,6 +257,7 @@ struct BlockStageCtx {
       sc.const_value.assign((const char*)dec.data.data(), dec.data.size());
     } else {
       sc.d_data = st->push(dec.data.data(), dec.data.size());
+      st->reserve(16);  // tiled 16 B copy may read up to 15 B past the data end
       sc.d_offsets = (const uint32_t*)st->push(dec.offsets.data(),
                                                dec.offsets.size() * 4, 4);
       sc.data_bytes = dec.data.size();
