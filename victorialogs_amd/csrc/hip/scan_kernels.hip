@@ -27,8 +27,12 @@ namespace vl {
 // Non-ASCII token-rune ranges (tokenizer.go:142-148); see unicode_ranges.inc.
 #include "../core/unicode_ranges.inc"
 
-constexpr uint32_t kTileBytes = 66 * 1024;  // 256 rows x <=~256 B + slack
-constexpr uint32_t kGroupRows = 256;
+// Each wavefront owns a private LDS tile and scans 64-row groups (one bitmap
+// word per group) with no workgroup barriers: the wave copies its group's
+// bytes with coalesced 16-byte loads (4-deep batches keep ~4 KiB per wave in
+// flight), then matches lane-per-row from the tile.
+constexpr uint32_t kWaveTileBytes = 16896;  // 64 rows x <=264 B avg
+constexpr uint32_t kNumWaves = 4;
 
 __device__ __forceinline__ bool d_is_token_char(uint8_t c) {
   // tokenizer.go:132-140: [a-zA-Z0-9_]
@@ -552,7 +556,7 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
     int nleaves, const DevBlock* __restrict__ blocks,
     const DevChunk* __restrict__ chunks, unsigned long long* __restrict__ hits) {
   __shared__ uint64_t stack[kMaxStackDepth][kChunkWords];
-  __shared__ __attribute__((aligned(16))) uint8_t tile[kTileBytes];
+  __shared__ __attribute__((aligned(16))) uint8_t tile[kNumWaves * kWaveTileBytes];
   __shared__ int bloom_ok;
   __shared__ unsigned long long wave_sums[4];
 
@@ -601,31 +605,46 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
       }
 
       if (d_is_string_kind(lb.kind)) {
-        // LDS-tiled string scan: 256-row groups, coalesced 16 B copies,
-        // lane-per-row match from the swizzled tile.
-        for (uint32_t g0 = r0; g0 < r1; g0 += kGroupRows) {
-          const uint32_t g1 = min(g0 + kGroupRows, r1);
+        // Per-wave LDS-tiled string scan: each wave independently copies the
+        // bytes of a 64-row group into its private tile (coalesced 16 B
+        // loads) and matches lane-per-row; one wavefront ballot = one
+        // bitmap word.  No workgroup barriers on this path.
+        uint8_t* wtile = tile + wave * kWaveTileBytes;
+        uint4* dst = (uint4*)wtile;
+        for (uint32_t wd = wave; wd < nwords; wd += nwaves) {
+          const uint32_t g0 = r0 + wd * 64;
+          const uint32_t g1 = min(g0 + 64, r1);
           const uint32_t ng = g1 - g0;
           const uint32_t byte0 = lb.offsets[g0] & ~15u;
           const uint32_t byte1 = lb.offsets[g1];
           const uint32_t nbytes = byte1 - byte0;
-          const bool use_tile = nbytes <= kTileBytes;
+          const bool use_tile = nbytes <= kWaveTileBytes;
           if (use_tile) {
             const uint4* src = (const uint4*)(lb.data + byte0);
             const uint32_t n16 = (nbytes + 15) >> 4;
-            for (uint32_t k = tid; k < n16; k += blockDim.x) {
-              uint32_t d = k ^ ((k >> 4) & 15);  // TileAcc swizzle, 16 B slots
-              ((uint4*)tile)[d] = src[k];
+            uint32_t k = lane;
+            // 4-deep batches: 4 independent loads in flight per lane
+            for (; k + 192 < n16; k += 256) {
+              uint4 a0 = src[k];
+              uint4 a1 = src[k + 64];
+              uint4 a2 = src[k + 128];
+              uint4 a3 = src[k + 192];
+              dst[k ^ ((k >> 4) & 15)] = a0;
+              dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+              dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+              dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
             }
+            for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
+            // every lane's ds_writes must land before cross-lane reads below
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
           }
-          __syncthreads();
           bool pred = false;
-          const uint32_t row = g0 + tid;
-          if (uint32_t(tid) < ng) {
+          const uint32_t row = g0 + lane;
+          if (uint32_t(lane) < ng) {
             const long s = lb.offsets[row];
             const long e = lb.offsets[row + 1];
             if (use_tile) {
-              TileAcc a{tile};
+              TileAcc a{wtile};
               pred = d_eval_string_row(lb, a, s - byte0, e - s);
             } else {
               GlobalAcc a{lb.data};
@@ -633,10 +652,7 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
             }
           }
           const uint64_t word = __ballot(pred);
-          if (lane == 0 && uint32_t(wave * 64) < ng) {
-            out[(g0 - r0) / 64 + wave] = word;
-          }
-          __syncthreads();
+          if (lane == 0) out[wd] = word;
         }
       } else {
         for (uint32_t w = wave; w < nwords; w += nwaves) {
