@@ -611,48 +611,82 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
         // bitmap word.  No workgroup barriers on this path.
         uint8_t* wtile = tile + wave * kWaveTileBytes;
         uint4* dst = (uint4*)wtile;
-        for (uint32_t wd = wave; wd < nwords; wd += nwaves) {
+        uint32_t wd = wave;
+        // Offsets pipeline: each lane holds offsets[g0+lane] (clamped to g1);
+        // a lane's row end is the next lane's start (shfl), lane ng-1's end is
+        // the clamped value itself.  The next group's offsets are prefetched
+        // while this group's tile copy is in flight.
+        uint32_t o_lane = 0, o_end = 0;
+        if (wd < nwords) {
+          o_lane = lb.offsets[min(r0 + wd * 64 + uint32_t(lane), r1)];
+          if (lane == 0) o_end = lb.offsets[min(r0 + wd * 64 + 64, r1)];
+        }
+        while (wd < nwords) {
           const uint32_t g0 = r0 + wd * 64;
           const uint32_t g1 = min(g0 + 64, r1);
           const uint32_t ng = g1 - g0;
-          const uint32_t byte0 = lb.offsets[g0] & ~15u;
-          const uint32_t byte1 = lb.offsets[g1];
+          const uint32_t byte0 =
+              uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
+          const uint32_t byte1 = uint32_t(__shfl(int(o_end), 0, 64));
           const uint32_t nbytes = byte1 - byte0;
           const bool use_tile = nbytes <= kWaveTileBytes;
           if (use_tile) {
             const uint4* src = (const uint4*)(lb.data + byte0);
             const uint32_t n16 = (nbytes + 15) >> 4;
             uint32_t k = lane;
-            // 4-deep batches: 4 independent loads in flight per lane
+            // 8-deep batches: 8 independent loads in flight per lane
+            for (; k + 448 < n16; k += 512) {
+              uint4 a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
+                    a3 = src[k + 192], a4 = src[k + 256], a5 = src[k + 320],
+                    a6 = src[k + 384], a7 = src[k + 448];
+              dst[k ^ ((k >> 4) & 15)] = a0;
+              dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+              dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+              dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+              dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
+              dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
+              dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
+              dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
+            }
             for (; k + 192 < n16; k += 256) {
-              uint4 a0 = src[k];
-              uint4 a1 = src[k + 64];
-              uint4 a2 = src[k + 128];
-              uint4 a3 = src[k + 192];
+              uint4 a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
+                    a3 = src[k + 192];
               dst[k ^ ((k >> 4) & 15)] = a0;
               dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
               dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
               dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
             }
             for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
+          }
+          // prefetch next group's offsets while the copy is in flight
+          const uint32_t next_wd = wd + nwaves;
+          uint32_t o_next = 0, o_end_next = 0;
+          if (next_wd < nwords) {
+            o_next = lb.offsets[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
+            if (lane == 0) o_end_next = lb.offsets[min(r0 + next_wd * 64 + 64, r1)];
+          }
+          if (use_tile) {
             // every lane's ds_writes must land before cross-lane reads below
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
           }
           bool pred = false;
-          const uint32_t row = g0 + lane;
           if (uint32_t(lane) < ng) {
-            const long s = lb.offsets[row];
-            const long e = lb.offsets[row + 1];
+            const long s = o_lane;
+            const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
+            const long e_fix = uint32_t(lane) == ng - 1 ? long(byte1) : e;
             if (use_tile) {
               TileAcc a{wtile};
-              pred = d_eval_string_row(lb, a, s - byte0, e - s);
+              pred = d_eval_string_row(lb, a, s - byte0, e_fix - s);
             } else {
               GlobalAcc a{lb.data};
-              pred = d_eval_string_row(lb, a, s, e - s);
+              pred = d_eval_string_row(lb, a, s, e_fix - s);
             }
           }
           const uint64_t word = __ballot(pred);
           if (lane == 0) out[wd] = word;
+          o_lane = o_next;
+          o_end = o_end_next;
+          wd = next_wd;
         }
       } else {
         for (uint32_t w = wave; w < nwords; w += nwaves) {
