@@ -50,7 +50,7 @@ WORKLOADS = {
         "filter": '{"type":"and","filters":['
                   '{"type":"phrase","field":"dict_0","phrase":"error"},'
                   '{"type":"time","min":1700000000000000000,'
-                  '"max":1700000049999000000}]}',
+                  '"max":1700006250000000000}]}',
     },
 }
 

@@ -609,8 +609,9 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
         // bytes of a 64-row group into its private tile (coalesced 16 B
         // loads) and matches lane-per-row; one wavefront ballot = one
         // bitmap word.  No workgroup barriers on this path.
+        typedef uint32_t v4u __attribute__((ext_vector_type(4)));
         uint8_t* wtile = tile + wave * kWaveTileBytes;
-        uint4* dst = (uint4*)wtile;
+        v4u* dst = (v4u*)wtile;
         uint32_t wd = wave;
         // Offsets pipeline: each lane holds offsets[g0+lane] (clamped to g1);
         // a lane's row end is the next lane's start (shfl), lane ng-1's end is
@@ -631,14 +632,21 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
           const uint32_t nbytes = byte1 - byte0;
           const bool use_tile = nbytes <= kWaveTileBytes;
           if (use_tile) {
-            const uint4* src = (const uint4*)(lb.data + byte0);
+            const v4u* src = (const v4u*)(lb.data + byte0);
             const uint32_t n16 = (nbytes + 15) >> 4;
             uint32_t k = lane;
             // 8-deep batches: 8 independent loads in flight per lane
             for (; k + 448 < n16; k += 512) {
-              uint4 a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
-                    a3 = src[k + 192], a4 = src[k + 256], a5 = src[k + 320],
-                    a6 = src[k + 384], a7 = src[k + 448];
+              // nt loads: each byte is read once per kernel; keep L2 for
+              // the offsets/bitmap traffic (cdna guide: nt-weights row)
+              v4u a0 = __builtin_nontemporal_load(src + k);
+              v4u a1 = __builtin_nontemporal_load(src + k + 64);
+              v4u a2 = __builtin_nontemporal_load(src + k + 128);
+              v4u a3 = __builtin_nontemporal_load(src + k + 192);
+              v4u a4 = __builtin_nontemporal_load(src + k + 256);
+              v4u a5 = __builtin_nontemporal_load(src + k + 320);
+              v4u a6 = __builtin_nontemporal_load(src + k + 384);
+              v4u a7 = __builtin_nontemporal_load(src + k + 448);
               dst[k ^ ((k >> 4) & 15)] = a0;
               dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
               dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
@@ -649,8 +657,8 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
               dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
             }
             for (; k + 192 < n16; k += 256) {
-              uint4 a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
-                    a3 = src[k + 192];
+              v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
+                  a3 = src[k + 192];
               dst[k ^ ((k >> 4) & 15)] = a0;
               dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
               dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
