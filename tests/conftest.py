@@ -56,6 +56,9 @@ def typed_part(tmp_path_factory):
                         % (1 + i % 28, i % 24, i % 60, (i * 3) % 60, i % 1000)
                         for i in range(rows)]},
                     {"name": "constcol", "values": ["fixed value 42"] * rows},
+                    {"name": "mix", "values": [
+                        ["12345", "1.5", "-7.25", "200ms", "1.5GiB", "abc def",
+                         "3d2h", "100KB", "", "nan?"][i % 10] for i in range(rows)]},
                     {"name": "uni", "values": [
                         ("раз два три" if i % 3 == 0 else "foo bar") for i in range(rows)]},
                 ],
@@ -149,6 +152,14 @@ TYPED_FILTERS = [
     '{"type":"regexp","field":"ip","re":"192\\\\.168"}',
     '{"type":"regexp","field":"iso","re":"2024-01-0"}',
     '{"type":"regexp","field":"i64","re":"-3"}',
+    # float64 slow paths (device Ryu formatting) + range on string column
+    '{"type":"phrase","field":"f64","phrase":"18"}',
+    '{"type":"phrase","field":"f64","phrase":"-18"}',
+    '{"type":"phrase","field":"f64","phrase":"625"}',
+    '{"type":"regexp","field":"f64","re":"18\\.625|0\\.5"}',
+    '{"type":"regexp","field":"f64","re":"-1"}',
+    '{"type":"range","field":"mix","min":0,"max":10000}',
+    '{"type":"range","field":"mix","min":-10,"max":1.6e9}',
     '{"type":"and","filters":['
     '{"type":"phrase","field":"lvl","phrase":"error"},'
     '{"type":"range","field":"u8","min":0,"max":50},'

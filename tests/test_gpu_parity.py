@@ -103,12 +103,8 @@ def test_scan_is_deterministic(gen_part):
         part.close()
 
 
-def test_unsupported_paths_fail_loudly(typed_part):
-    """Round-1 GPU gaps (float formatting, range-on-string) must raise a
-    clear staging error, never fall back to CPU silently."""
-    part = Part(typed_part)
-    filt = Filter('{"type":"phrase","field":"f64","phrase":"18"}')
-    with pytest.raises(RuntimeError, match="round 2"):
-        Stage(part, filt, device=0)
-    filt.close()
-    part.close()
+def test_unsupported_regex_fails_at_compile(typed_part):
+    """Unsupported constructs must raise a clear error at compile time,
+    never fall back to CPU silently."""
+    with pytest.raises(RuntimeError, match="not supported|fast-path"):
+        Filter('{"type":"regexp","field":"_msg","re":"a{2,3}"}')

@@ -20,6 +20,7 @@
 // (BASELINE.json north star).
 #include <hip/hip_runtime.h>
 
+#include "../core/ryu.h"
 #include "scan_types.h"
 
 namespace vl {
@@ -302,6 +303,240 @@ struct BufAcc {
   __device__ __forceinline__ uint8_t u8(long off) const { return base[off]; }
 };
 
+
+// ---- device number parsing (mirrors values.cpp; for range-on-string) ----
+
+// Go math.Pow10 restated (values.cpp go_pow10)
+__device__ static const double kPow10Tab[32] = {
+    1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7, 1e8, 1e9, 1e10, 1e11, 1e12,
+    1e13, 1e14, 1e15, 1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22, 1e23,
+    1e24, 1e25, 1e26, 1e27, 1e28, 1e29, 1e30, 1e31};
+__device__ static const double kPow10PosTab32[10] = {
+    1e0, 1e32, 1e64, 1e96, 1e128, 1e160, 1e192, 1e224, 1e256, 1e288};
+
+__device__ inline double d_go_pow10(int n) {
+  if (n >= 0 && n <= 308) return kPow10PosTab32[unsigned(n) / 32] * kPow10Tab[unsigned(n) % 32];
+  if (n <= 0 && n >= -323) {
+    return 1.0 / (kPow10PosTab32[unsigned(-n) / 32] * kPow10Tab[unsigned(-n) % 32]);
+  }
+  return n > 0 ? __builtin_inf() : 0.0;
+}
+
+// tryParseUint64 (values_encoder.go:553-585)
+template <typename A>
+__device__ bool d_try_parse_uint64(const A& a, long s0, long sn, uint64_t* out) {
+  if (sn == 0 || sn > 26) return false;
+  if (sn > 1 && a.u8(s0) == '0') return false;
+  uint64_t n = 0;
+  for (long i = 0; i < sn; i++) {
+    uint8_t ch = a.u8(s0 + i);
+    if (ch == '_') continue;
+    if (ch < '0' || ch > '9') return false;
+    if (n > 1844674407370955161ULL) return false;  // UINT64_MAX/10
+    n *= 10;
+    uint64_t d = ch - '0';
+    uint64_t n1 = n + d;
+    if (n1 < n) return false;
+    n = n1;
+  }
+  *out = n;
+  return true;
+}
+
+// tryParseFloat64 non-exact (values_encoder.go:788-848, isExact=false)
+template <typename A>
+__device__ bool d_try_parse_float64(const A& a, long s0, long sn, double* out) {
+  if (sn == 0 || sn > 27) return false;
+  bool minus = a.u8(s0) == '-';
+  if (minus) {
+    s0++;
+    sn--;
+  }
+  long ndot = -1;
+  for (long i = 0; i < sn; i++) {
+    if (a.u8(s0 + i) == '.') {
+      ndot = i;
+      break;
+    }
+  }
+  if (ndot < 0) {
+    uint64_t n;
+    if (!d_try_parse_uint64(a, s0, sn, &n)) return false;
+    double f = double(n);
+    *out = minus ? -f : f;
+    return true;
+  }
+  if (ndot == 0 || ndot == sn - 1) return false;
+  uint64_t n_int;
+  if (!d_try_parse_uint64(a, s0, ndot, &n_int)) return false;
+  long f0 = s0 + ndot + 1, fn = sn - ndot - 1;
+  long skip = 0;
+  while (skip < fn - 1 && a.u8(f0 + skip) == '0') skip++;
+  uint64_t n_frac;
+  if (!d_try_parse_uint64(a, f0 + skip, fn - skip, &n_frac)) return false;
+  int underscores = 0;
+  for (long i = 0; i < fn; i++) {
+    if (a.u8(f0 + i) == '_') underscores++;
+  }
+  double p10 = d_go_pow10(underscores - int(fn));
+  double f = fma(double(n_frac), p10, double(n_int));
+  *out = minus ? -f : f;
+  return true;
+}
+
+// tryParseFloat64Prefix (values_encoder.go:762-773); advances *s0/*sn
+template <typename A>
+__device__ bool d_parse_float64_prefix(const A& a, long* s0, long* sn, double* f) {
+  long i = 0;
+  while (i < *sn) {
+    uint8_t c = a.u8(*s0 + i);
+    if ((c >= '0' && c <= '9') || c == '.' || c == '_') {
+      i++;
+    } else {
+      break;
+    }
+  }
+  if (i == 0) return false;
+  if (!d_try_parse_float64(a, *s0, i, f)) return false;
+  *s0 += i;
+  *sn -= i;
+  return true;
+}
+
+__device__ inline long long d_add_i64_no_overflow(long long n, double f) {
+  long long x = (long long)(f);
+  if (n < 0 || x < 0 || x > 0x7FFFFFFFFFFFFFFFLL - n) return 0x7FFFFFFFFFFFFFFFLL;
+  return n + x;
+}
+
+template <typename A>
+__device__ bool d_has_prefix(const A& a, long s0, long sn, const char* p, int np) {
+  if (sn < np) return false;
+  for (int i = 0; i < np; i++) {
+    if (a.u8(s0 + i) != uint8_t(p[i])) return false;
+  }
+  return true;
+}
+
+// tryParseDuration (values_encoder.go:990-1061)
+template <typename A>
+__device__ bool d_try_parse_duration(const A& a, long s0, long sn, long long* out) {
+  if (sn == 0) return false;
+  bool minus = a.u8(s0) == '-';
+  if (minus) {
+    s0++;
+    sn--;
+  }
+  long long nsecs = 0;
+  while (sn > 0) {
+    double f;
+    if (!d_parse_float64_prefix(a, &s0, &sn, &f)) return false;
+    if (sn == 0) return false;
+    if (sn >= 3 && a.u8(s0) == 0xC2 && a.u8(s0 + 1) == 0xB5 && a.u8(s0 + 2) == 's') {
+      nsecs = d_add_i64_no_overflow(nsecs, f * 1000);
+      s0 += 3; sn -= 3;
+      continue;
+    }
+    if (d_has_prefix(a, s0, sn, "ms", 2)) {
+      nsecs = d_add_i64_no_overflow(nsecs, f * 1000000);
+      s0 += 2; sn -= 2;
+      continue;
+    }
+    if (d_has_prefix(a, s0, sn, "ns", 2)) {
+      nsecs = d_add_i64_no_overflow(nsecs, f);
+      s0 += 2; sn -= 2;
+      continue;
+    }
+    uint8_t c = a.u8(s0);
+    double mult;
+    switch (c) {
+      case 'y': mult = 365.0 * 24 * 3600 * 1e9; break;
+      case 'w': mult = 7.0 * 24 * 3600 * 1e9; break;
+      case 'd': mult = 24.0 * 3600 * 1e9; break;
+      case 'h': mult = 3600e9; break;
+      case 'm': mult = 60e9; break;
+      case 's': mult = 1e9; break;
+      default: return false;
+    }
+    nsecs = d_add_i64_no_overflow(nsecs, f * mult);
+    s0 += 1; sn -= 1;
+  }
+  *out = minus ? -nsecs : nsecs;
+  return true;
+}
+
+// tryParseBytes (values_encoder.go:855-966)
+template <typename A>
+__device__ bool d_try_parse_bytes(const A& a, long s0, long sn, long long* out) {
+  if (sn == 0) return false;
+  bool minus = a.u8(s0) == '-';
+  if (minus) {
+    s0++;
+    sn--;
+  }
+  long long n = 0;
+  while (sn > 0) {
+    double f;
+    if (!d_parse_float64_prefix(a, &s0, &sn, &f)) return false;
+    if (sn == 0) {
+      double ip = trunc(f);
+      if (f != ip) return false;  // no suffix: integers only
+      n = d_add_i64_no_overflow(n, f);
+      continue;
+    }
+    bool matched = false;
+    if (sn >= 3) {
+      const char* s3[] = {"KiB", "MiB", "GiB", "TiB"};
+      const double m3[] = {1024.0, 1048576.0, 1073741824.0, 1099511627776.0};
+      for (int i = 0; i < 4 && !matched; i++) {
+        if (d_has_prefix(a, s0, sn, s3[i], 3)) {
+          n = d_add_i64_no_overflow(n, f * m3[i]);
+          s0 += 3; sn -= 3;
+          matched = true;
+        }
+      }
+    }
+    if (!matched && sn >= 2) {
+      const char* s2[] = {"Ki", "Mi", "Gi", "Ti", "KB", "MB", "GB", "TB"};
+      const double m2[] = {1024.0, 1048576.0, 1073741824.0, 1099511627776.0,
+                           1e3, 1e6, 1e9, 1e12};
+      for (int i = 0; i < 8 && !matched; i++) {
+        if (d_has_prefix(a, s0, sn, s2[i], 2)) {
+          n = d_add_i64_no_overflow(n, f * m2[i]);
+          s0 += 2; sn -= 2;
+          matched = true;
+        }
+      }
+    }
+    if (!matched) {
+      const char* s1 = "BKMGT";
+      const double m1[] = {1.0, 1e3, 1e6, 1e9, 1e12};
+      for (int i = 0; i < 5 && !matched; i++) {
+        if (a.u8(s0) == uint8_t(s1[i])) {
+          n = d_add_i64_no_overflow(n, f * m1[i]);
+          s0 += 1; sn -= 1;
+          matched = true;
+        }
+      }
+    }
+    if (!matched) return false;
+  }
+  *out = minus ? -n : n;
+  return true;
+}
+
+// parseMathNumber subset (pipe_math.go:1066-1080; same legs as the host
+// parse_math_number in values.cpp -- float, duration, bytes; others NaN)
+template <typename A>
+__device__ double d_parse_math_number(const A& a, long s0, long sn) {
+  double f;
+  if (sn > 0 && d_try_parse_float64(a, s0, sn, &f)) return f;
+  long long v;
+  if (d_try_parse_duration(a, s0, sn, &v)) return double(v);
+  if (d_try_parse_bytes(a, s0, sn, &v)) return double(v);
+  return __builtin_nan("");
+}
+
 // ---- regex fast paths on serialized blob (regex.go:86-212) ----
 
 struct DRegex {
@@ -436,6 +671,13 @@ __device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
       }
       return true;
     }
+    case kScanRangeStr: {
+      // matchRange (filter_range.go:369-372)
+      double x = d_parse_math_number(a, s0, sn);
+      double mn = __builtin_bit_cast(double, lb.vmin);
+      double mx = __builtin_bit_cast(double, lb.vmax);
+      return x >= mn && x <= mx;
+    }
     default:  // kScanRegexStr
       return d_regex_match_at(lb.operand, a, s0, sn);
   }
@@ -540,13 +782,28 @@ __device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
     }
+    case kScanPhraseF64: {
+      // matchFloat64ByPhrase slow path (filter_phrase.go:175-186): format the
+      // stored float with Ryu (== Go strconv 'f' -1) and substring-match
+      char buf[344];
+      int n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
+      BufAcc a{(const uint8_t*)buf};
+      return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
+    }
+    case kScanRegexF64: {
+      char buf[344];
+      int n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
+      BufAcc a{(const uint8_t*)buf};
+      return d_regex_match_at(lb.operand, a, 0, n);
+    }
     default:
       return false;
   }
 }
 
 __device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
-  return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr;
+  return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr ||
+         kind == kScanRangeStr;
 }
 
 // ---- the program kernel ----

@@ -40,6 +40,9 @@ enum ScanKind : uint8_t {
   kScanRegexI = 12,     // regex over formatted int64 (filter_regexp.go:243-254)
   kScanRegexIp = 13,    // regex over formatted ipv4 (filter_regexp.go:142-153)
   kScanRegexIso = 14,   // regex over formatted iso8601 (filter_regexp.go:129-140)
+  kScanPhraseF64 = 15,  // matchPhrase over Ryu-formatted float64 (filter_phrase.go:159-186)
+  kScanRegexF64 = 16,   // regex over Ryu-formatted float64 (filter_regexp.go:155-166)
+  kScanRangeStr = 17,   // matchRange via parseMathNumber per row (filter_range.go:261-265,369-372)
 };
 
 // phrase flags

@@ -517,9 +517,26 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
             stage_eq_bin(lb, ctx, ch, li, st, bin);
             return;
           }
-          fail("phrase \"" + f.phrase + "\" on float64 column \"" + f.field +
-               "\": partial float phrase match requires per-row float "
-               "formatting, not yet on the GPU path (round 2; DESIGN.md)");
+          // slow path: per-row Ryu formatting + matchPhrase on device
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            std::string fs;
+            uint64_t u = get_u64be((const uint8_t*)sc.const_value.data());
+            double d;
+            memcpy(&d, &u, 8);
+            format_float64(fs, d);
+            lb.mode = match_phrase(strview(fs), phrase) ? kModeAll : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanPhraseF64;
+          lb.width = 8;
+          lb.flags = li.phrase_flags;
+          lb.operand = li.d_operand;
+          lb.operand_len = uint32_t(phrase.n);
+          lb.data = sc.d_data;
+          set_bloom_gate(lb, ctx, ch, li);
+          return;
         }
         case ValueType::IPv4: {
           uint32_t ip;
@@ -644,11 +661,6 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
         stage_dict(lb, ctx, ch, mask);
         return;
       }
-      if (ch.type == ValueType::Float64) {
-        fail("regexp on float64 column \"" + f.field +
-             "\" requires per-row float formatting, not yet on the GPU path "
-             "(round 2; DESIGN.md)");
-      }
       const StagedStrCol& sc = ctx.stage_column(ch);
       if (sc.is_const) {
         // evaluate the single encoded value on the host
@@ -661,6 +673,13 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
           case ValueType::Uint32: format_uint64(s, get_u32be((const uint8_t*)v.p)); break;
           case ValueType::Uint64: format_uint64(s, get_u64be((const uint8_t*)v.p)); break;
           case ValueType::Int64: format_int64(s, get_i64be_zigzag((const uint8_t*)v.p)); break;
+          case ValueType::Float64: {
+            uint64_t u = get_u64be((const uint8_t*)v.p);
+            double d;
+            memcpy(&d, &u, 8);
+            format_float64(s, d);
+            break;
+          }
           case ValueType::IPv4: format_ipv4(s, get_u32be((const uint8_t*)v.p)); break;
           case ValueType::TimestampISO8601:
             format_timestamp_iso8601(s, int64_t(get_u64be((const uint8_t*)v.p)));
@@ -685,6 +704,7 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
         case ValueType::Uint32: lb.kind = kScanRegexU; lb.width = 4; return;
         case ValueType::Uint64: lb.kind = kScanRegexU; lb.width = 8; return;
         case ValueType::Int64: lb.kind = kScanRegexI; lb.width = 8; return;
+        case ValueType::Float64: lb.kind = kScanRegexF64; lb.width = 8; return;
         case ValueType::IPv4: lb.kind = kScanRegexIp; lb.width = 4; return;
         case ValueType::TimestampISO8601: lb.kind = kScanRegexIso; lb.width = 8; return;
         default: fail("unknown valueType while staging regexp filter");
@@ -717,9 +737,24 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
         return;
       }
       if (ch.type == ValueType::String) {
-        fail("range filter on string column \"" + f.field +
-             "\" requires per-row number parsing, not yet on the GPU path "
-             "(round 2; DESIGN.md)");
+        // matchStringByRange (filter_range.go:261-265): parseMathNumber per
+        // row on device; no bloom gate in the reference either
+        const StagedStrCol& sc = ctx.stage_column(ch);
+        if (sc.is_const) {
+          double x = parse_math_number(strview(sc.const_value));
+          lb.mode = (x >= min_v && x <= max_v) ? kModeAll : kModeNone;
+          return;
+        }
+        lb.mode = kModeScan;
+        lb.kind = kScanRangeStr;
+        lb.data = sc.d_data;
+        lb.offsets = sc.d_offsets;
+        uint64_t vmin, vmax;
+        memcpy(&vmin, &min_v, 8);
+        memcpy(&vmax, &max_v, 8);
+        lb.vmin = vmin;
+        lb.vmax = vmax;
+        return;
       }
 
       auto clamp_u64 = [](double x) -> uint64_t {
@@ -951,7 +986,8 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
         switch (lb.kind) {
           case kScanPhraseStr:
           case kScanEqStr:
-          case kScanRegexStr: {
+          case kScanRegexStr:
+          case kScanRangeStr: {
             const StagedStrCol& sc = ctx.cols.at(leaf_infos[size_t(l)].cname);
             st->algo_bytes += sc.data_bytes + (bh.rows_count + 1) * 4;
             break;
