@@ -156,7 +156,7 @@ TYPED_FILTERS = [
     '{"type":"phrase","field":"f64","phrase":"18"}',
     '{"type":"phrase","field":"f64","phrase":"-18"}',
     '{"type":"phrase","field":"f64","phrase":"625"}',
-    '{"type":"regexp","field":"f64","re":"18\\.625|0\\.5"}',
+    '{"type":"regexp","field":"f64","re":"18\\\\.625|0\\\\.5"}',
     '{"type":"regexp","field":"f64","re":"-1"}',
     '{"type":"range","field":"mix","min":0,"max":10000}',
     '{"type":"range","field":"mix","min":-10,"max":1.6e9}',
