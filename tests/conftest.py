@@ -160,6 +160,14 @@ TYPED_FILTERS = [
     '{"type":"regexp","field":"f64","re":"-1"}',
     '{"type":"range","field":"mix","min":0,"max":10000}',
     '{"type":"range","field":"mix","min":-10,"max":1.6e9}',
+    # general regex class (Glushkov NFA on device)
+    '{"type":"regexp","field":"_msg","re":"took \\\\d+ms"}',
+    '{"type":"regexp","field":"_msg","re":"level=[a-z]+ took"}',
+    '{"type":"regexp","field":"lvl","re":"(warn|err)o?r?"}',
+    '{"type":"regexp","field":"mix","re":"\\\\d+KB|GiB"}',
+    '{"type":"regexp","field":"uni","re":"р.з"}',
+    '{"type":"regexp","field":"u8","re":"1\\\\d"}',
+    '{"type":"regexp","field":"ip","re":"10\\\\.\\\\d+\\\\.1?5"}',
     '{"type":"and","filters":['
     '{"type":"phrase","field":"lvl","phrase":"error"},'
     '{"type":"range","field":"u8","min":0,"max":50},'

@@ -41,8 +41,9 @@ def test_filter_compile_errors():
         '{"type":"phrase"}',                       # missing fields
         '{"type":"wat","field":"x","phrase":"y"}', # unknown type
         'not json at all',
-        '{"type":"regexp","field":"x","re":"a{2,3}"}',  # unsupported regex
-        '{"type":"regexp","field":"x","re":"\\\\d+"}',  # escape class
+        '{"type":"regexp","field":"x","re":"a{2,3}"}',  # {m,n} unsupported
+        '{"type":"regexp","field":"x","re":"a\\\\b"}',  # word-boundary assertion
+        '{"type":"regexp","field":"x","re":"^foo"}',    # anchors unsupported
     ]
     for f in bad:
         h = lib.vql_compile_filter(f.encode())

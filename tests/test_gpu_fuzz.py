@@ -21,6 +21,9 @@ GEN_FIELDS_NUM = ["u8_0", "u16_0", "u32_0", "u64_0", "i64_0", "float_0"]
 GEN_REGEXES = [
     "stream (0|1)", "uuid=[0-9a-f]", "ip=1.*uuid", "message.+", "err(or|)x|info",
     "pad", ".*u64.*", "host_.+",
+    # NFA class
+    "stream \\d+ and", "u64=\\d?\\d", "ip=[0-9]+\\.[0-9]+", "pa?d=",
+    "[^;]*uuid", "w\\w+ker",
 ]
 
 TYPED_PHRASES = [
@@ -32,6 +35,9 @@ TYPED_FIELDS_NUM = ["u8", "u16", "u32", "u64", "i64", "f64", "ip", "iso"]
 TYPED_REGEXES = [
     "level=(error|warn)", "took 1.*ms", "row [0-9]", "два|foo", "19(2|3)",
     "-3", "2024-01", "fixed.+",
+    # NFA class
+    "took \\d+ms", "l[a-z]+e\\d", "ро?w", "[0-9]+\\.[0-9]+", "\\dms|GiB",
+    "2024-\\d+-0[1-5]T",
 ]
 
 

@@ -12,9 +12,12 @@ namespace {
 // Minimal RE2-subset AST.  Group nodes are preserved so GetLiterals sees the
 // same literal boundaries as Go's parse tree (regex.go:101-124).
 struct RNode {
-  enum Kind { Lit, Dot, Concat, Alt, Star, Plus, Quest, Group, Empty } kind;
+  enum Kind { Lit, Dot, Concat, Alt, Star, Plus, Quest, Group, Empty, Class } kind;
   std::string lit;
   std::vector<RNode> subs;
+  // Class: 256-bit byte set (ASCII bits 0..127 used) + "all non-ASCII runes"
+  uint8_t cls[32] = {0};
+  bool cls_nonascii = false;
 };
 
 struct Parser {
@@ -114,10 +117,14 @@ struct Parser {
       case '\\':
         return parse_escape();
       default: {
-        pos++;
+        // consume a full UTF-8 rune so postfix quantifiers bind to the rune
+        // (Go binds ? * + to the preceding rune, not byte)
         RNode n;
         n.kind = RNode::Lit;
-        n.lit.push_back(c);
+        uint8_t c0 = uint8_t(c);
+        int len = c0 < 0x80 ? 1 : (c0 & 0xE0) == 0xC0 ? 2
+                  : (c0 & 0xF0) == 0xE0 ? 3 : (c0 & 0xF8) == 0xF0 ? 4 : 1;
+        for (int i = 0; i < len && !eof(); i++) n.lit.push_back(s[pos++]);
         return n;
       }
     }
@@ -133,7 +140,31 @@ struct Parser {
       case 'n': n.lit.push_back('\n'); return n;
       case 't': n.lit.push_back('\t'); return n;
       case 'r': n.lit.push_back('\r'); return n;
-      case 'd': case 'D': case 'w': case 'W': case 's': case 'S':
+      case 'd': case 'D': case 'w': case 'W': case 's': case 'S': {
+        // Go regexp/syntax Perl classes: \d=[0-9], \w=[0-9A-Za-z_],
+        // \s=[\t\n\f\r ]; upper-case = negation (includes non-ASCII runes)
+        RNode cn;
+        cn.kind = RNode::Class;
+        auto set = [&](uint8_t ch) { cn.cls[ch >> 3] |= uint8_t(1) << (ch & 7); };
+        char base = char(c | 0x20);
+        if (base == 'd') {
+          for (uint8_t ch = '0'; ch <= '9'; ch++) set(ch);
+        } else if (base == 'w') {
+          for (uint8_t ch = '0'; ch <= '9'; ch++) set(ch);
+          for (uint8_t ch = 'a'; ch <= 'z'; ch++) set(ch);
+          for (uint8_t ch = 'A'; ch <= 'Z'; ch++) set(ch);
+          set('_');
+        } else {
+          set('\t'); set('\n'); set('\f'); set('\r'); set(' ');
+        }
+        if (c >= 'A' && c <= 'Z') {
+          // negate within ASCII; all non-ASCII runes are included
+          for (int i = 0; i < 16; i++) cn.cls[i] = uint8_t(~cn.cls[i]);
+          for (int i = 16; i < 32; i++) cn.cls[i] = 0;
+          cn.cls_nonascii = true;
+        }
+        return cn;
+      }
       case 'b': case 'B': case 'p': case 'P':
         err(std::string("escape class \\") + c + " is not supported");
       default:
@@ -148,56 +179,95 @@ struct Parser {
 
   RNode parse_char_class() {
     pos++;  // '['
-    if (!eof() && peek() == '^') err("negated char classes are not supported");
-    std::vector<char> chars;
+    bool negated = false;
+    if (!eof() && peek() == '^') {
+      negated = true;
+      pos++;
+    }
+    uint8_t bits[32] = {0};
+    auto set = [&](uint8_t ch) { bits[ch >> 3] |= uint8_t(1) << (ch & 7); };
     bool first = true;
     while (!eof() && (peek() != ']' || first)) {
-      char c = s[pos++];
+      uint8_t c = uint8_t(s[pos++]);
+      if (c >= 0x80) err("non-ASCII characters in [...] are not supported");
       if (c == '\\') {
         if (eof()) err("trailing backslash in class");
-        c = s[pos++];
-        if (c == 'n') c = '\n';
-        else if (c == 't') c = '\t';
-        else if (c == 'r') c = '\r';
-        else if ((c >= 'a' && c <= 'z' && c != 'n' && c != 't' && c != 'r') &&
-                 strchr("dwsb", c)) {
-          err("escape classes inside [...] are not supported");
+        char e = s[pos++];
+        if (e == 'n') c = '\n';
+        else if (e == 't') c = '\t';
+        else if (e == 'r') c = '\r';
+        else if (e == 'f') c = '\f';
+        else if (e == 'd' || e == 'w' || e == 's') {
+          if (e == 'd') {
+            for (uint8_t ch = '0'; ch <= '9'; ch++) set(ch);
+          } else if (e == 'w') {
+            for (uint8_t ch = '0'; ch <= '9'; ch++) set(ch);
+            for (uint8_t ch = 'a'; ch <= 'z'; ch++) set(ch);
+            for (uint8_t ch = 'A'; ch <= 'Z'; ch++) set(ch);
+            set('_');
+          } else {
+            set('\t'); set('\n'); set('\f'); set('\r'); set(' ');
+          }
+          first = false;
+          continue;
+        } else if ((e >= 'a' && e <= 'z') || (e >= 'A' && e <= 'Z') ||
+                   (e >= '0' && e <= '9')) {
+          err(std::string("escape \\") + e + " inside [...] is not supported");
+        } else {
+          c = uint8_t(e);
         }
       }
       if (!eof() && peek() == '-' && pos + 1 < s.size() && s[pos + 1] != ']') {
         pos++;  // '-'
-        char hi = s[pos++];
+        uint8_t hi = uint8_t(s[pos++]);
         if (hi == '\\') {
           if (eof()) err("trailing backslash in class");
-          hi = s[pos++];
+          hi = uint8_t(s[pos++]);
         }
+        if (hi >= 0x80) err("non-ASCII characters in [...] are not supported");
         if (hi < c) err("invalid char class range");
-        for (char x = c;; x++) {
-          chars.push_back(x);
-          if (x == hi) break;
-          if (chars.size() > 100) err("char class too large for or-values");
-        }
+        for (int x = c; x <= int(hi); x++) set(uint8_t(x));
       } else {
-        chars.push_back(c);
+        set(c);
       }
       first = false;
-      if (chars.size() > 100) err("char class too large for or-values");
     }
     if (eof()) err("missing ]");
     pos++;  // ']'
-    // Expand as alternation of single chars (regexutil.go:93-107)
-    RNode alt;
-    alt.kind = RNode::Alt;
-    for (char c : chars) {
-      RNode l;
-      l.kind = RNode::Lit;
-      l.lit.push_back(c);
-      alt.subs.push_back(std::move(l));
+
+    if (negated) {
+      RNode cn;
+      cn.kind = RNode::Class;
+      for (int i = 0; i < 16; i++) cn.cls[i] = uint8_t(~bits[i]);
+      cn.cls_nonascii = true;  // Go: [^...] matches any rune outside the set
+      return cn;
     }
-    if (alt.subs.size() == 1) return std::move(alt.subs[0]);
-    return alt;
+    // expand small positive classes as alternation of single chars so the
+    // or-values fast path classifies them like Go (regexutil.go:93-107)
+    int count = 0;
+    for (int b = 0; b < 128; b++) count += (bits[b >> 3] >> (b & 7)) & 1;
+    if (count == 0) err("empty char class");
+    if (count <= 100) {
+      RNode alt;
+      alt.kind = RNode::Alt;
+      for (int b = 0; b < 128; b++) {
+        if ((bits[b >> 3] >> (b & 7)) & 1) {
+          RNode l;
+          l.kind = RNode::Lit;
+          l.lit.push_back(char(b));
+          alt.subs.push_back(std::move(l));
+        }
+      }
+      if (alt.subs.size() == 1) return std::move(alt.subs[0]);
+      return alt;
+    }
+    RNode cn;
+    cn.kind = RNode::Class;
+    memcpy(cn.cls, bits, 32);
+    return cn;
   }
 };
+
 
 bool is_dot_star(const RNode& n) { return n.kind == RNode::Star && n.subs[0].kind == RNode::Dot; }
 bool is_dot_plus(const RNode& n) { return n.kind == RNode::Plus && n.subs[0].kind == RNode::Dot; }
@@ -324,6 +394,178 @@ void collect_literals(const RNode& raw, std::vector<std::string>& out) {
   }
 }
 
+
+// ---- Glushkov position automaton for the general class ----
+// Positions (atoms) are byte-consuming; first/last/follow sets are u64 masks
+// (<= 64 positions).  Multi-byte runes (Dot, negated classes) use a relaxed
+// UTF-8 shape: [00-7F] | [C0-DF][80-BF] | [E0-EF][80-BF]{2} | [F0-F4][80-BF]{3}
+// -- identical to Go on valid UTF-8 input (DESIGN.md notes the invalid-UTF-8
+// divergence).
+
+struct GAtom {
+  uint8_t set[32];
+};
+
+struct GBuild {
+  std::vector<GAtom> atoms;
+  std::vector<uint64_t> follow;
+  [[noreturn]] void overflow(const std::string& expr) {
+    fail("regex: NFA fallback for \"" + expr +
+         "\" needs more than 64 positions; simplify the pattern");
+  }
+};
+
+struct GInfo {
+  bool nullable;
+  uint64_t first, last;
+};
+
+static int g_add_atom(GBuild& b, const uint8_t* set, const std::string& expr) {
+  if (b.atoms.size() >= 64) b.overflow(expr);
+  GAtom a;
+  memcpy(a.set, set, 32);
+  b.atoms.push_back(a);
+  b.follow.push_back(0);
+  return int(b.atoms.size()) - 1;
+}
+
+static void g_range_set(uint8_t* set, int lo, int hi) {
+  for (int c = lo; c <= hi; c++) set[c >> 3] |= uint8_t(1) << (c & 7);
+}
+
+static GInfo g_cat(GBuild& b, GInfo x, GInfo y) {
+  // follow: last(x) -> first(y)
+  uint64_t m = x.last;
+  while (m) {
+    int i = __builtin_ctzll(m);
+    m &= m - 1;
+    b.follow[i] |= y.first;
+  }
+  GInfo r;
+  r.nullable = x.nullable && y.nullable;
+  r.first = x.first | (x.nullable ? y.first : 0);
+  r.last = y.last | (y.nullable ? x.last : 0);
+  return r;
+}
+
+static GInfo g_alt(GInfo x, GInfo y) {
+  return GInfo{x.nullable || y.nullable, x.first | y.first, x.last | y.last};
+}
+
+static void g_loop(GBuild& b, const GInfo& x) {
+  uint64_t m = x.last;
+  while (m) {
+    int i = __builtin_ctzll(m);
+    m &= m - 1;
+    b.follow[i] |= x.first;
+  }
+}
+
+// relaxed UTF-8 multi-byte rune
+static GInfo g_multibyte(GBuild& b, const std::string& expr) {
+  uint8_t cont[32] = {0}, l2[32] = {0}, l3[32] = {0}, l4[32] = {0};
+  g_range_set(cont, 0x80, 0xBF);
+  g_range_set(l2, 0xC0, 0xDF);
+  g_range_set(l3, 0xE0, 0xEF);
+  g_range_set(l4, 0xF0, 0xF4);
+  auto seq = [&](const uint8_t* lead, int ncont) {
+    int a = g_add_atom(b, lead, expr);
+    GInfo r{false, uint64_t(1) << a, uint64_t(1) << a};
+    for (int i = 0; i < ncont; i++) {
+      int c = g_add_atom(b, cont, expr);
+      r = g_cat(b, r, GInfo{false, uint64_t(1) << c, uint64_t(1) << c});
+    }
+    return r;
+  };
+  GInfo r = seq(l2, 1);
+  r = g_alt(r, seq(l3, 2));
+  r = g_alt(r, seq(l4, 3));
+  return r;
+}
+
+static GInfo g_build(GBuild& b, const RNode& n, const std::string& expr) {
+  switch (n.kind) {
+    case RNode::Empty:
+      return GInfo{true, 0, 0};
+    case RNode::Group:
+      return g_build(b, n.subs[0], expr);
+    case RNode::Lit: {
+      GInfo r{true, 0, 0};
+      for (unsigned char c : n.lit) {
+        uint8_t set[32] = {0};
+        set[c >> 3] = uint8_t(1) << (c & 7);
+        int a = g_add_atom(b, set, expr);
+        r = g_cat(b, r, GInfo{false, uint64_t(1) << a, uint64_t(1) << a});
+      }
+      return r;
+    }
+    case RNode::Dot: {
+      uint8_t ascii[32] = {0};
+      g_range_set(ascii, 0x00, 0x7F);  // (?s) DotNL: '.' matches any rune
+      int a = g_add_atom(b, ascii, expr);
+      GInfo r{false, uint64_t(1) << a, uint64_t(1) << a};
+      return g_alt(r, g_multibyte(b, expr));
+    }
+    case RNode::Class: {
+      int a = g_add_atom(b, n.cls, expr);
+      GInfo r{false, uint64_t(1) << a, uint64_t(1) << a};
+      if (n.cls_nonascii) r = g_alt(r, g_multibyte(b, expr));
+      return r;
+    }
+    case RNode::Concat: {
+      GInfo r{true, 0, 0};
+      for (const auto& sub : n.subs) r = g_cat(b, r, g_build(b, sub, expr));
+      return r;
+    }
+    case RNode::Alt: {
+      GInfo r = g_build(b, n.subs[0], expr);
+      for (size_t i = 1; i < n.subs.size(); i++) {
+        r = g_alt(r, g_build(b, n.subs[i], expr));
+      }
+      return r;
+    }
+    case RNode::Star: {
+      GInfo x = g_build(b, n.subs[0], expr);
+      g_loop(b, x);
+      return GInfo{true, x.first, x.last};
+    }
+    case RNode::Plus: {
+      GInfo x = g_build(b, n.subs[0], expr);
+      g_loop(b, x);
+      return x;
+    }
+    case RNode::Quest: {
+      GInfo x = g_build(b, n.subs[0], expr);
+      return GInfo{true, x.first, x.last};
+    }
+  }
+  fail("regex: unreachable node kind");
+}
+
+// blob: u16 nstates, u16 pad, u32 pad, u64 first, u64 last,
+//       u64 follow[nstates], u64 byte_table[256]
+static bytes g_serialize(const GBuild& b, const GInfo& root) {
+  bytes out;
+  uint16_t n = uint16_t(b.atoms.size());
+  out.push_back(uint8_t(n));
+  out.push_back(uint8_t(n >> 8));
+  out.resize(8, 0);
+  auto put64 = [&](uint64_t v) {
+    for (int i = 0; i < 8; i++) out.push_back(uint8_t(v >> (8 * i)));
+  };
+  put64(root.first);
+  put64(root.last);
+  for (uint16_t i = 0; i < n; i++) put64(b.follow[i]);
+  for (int c = 0; c < 256; c++) {
+    uint64_t m = 0;
+    for (uint16_t i = 0; i < n; i++) {
+      if ((b.atoms[i].set[c >> 3] >> (c & 7)) & 1) m |= uint64_t(1) << i;
+    }
+    put64(m);
+  }
+  return out;
+}
+
 }  // namespace
 
 RegexProg regex_compile(const std::string& expr) {
@@ -399,10 +641,15 @@ RegexProg regex_compile(const std::string& expr) {
 
   if (!re.is_only_prefix && !re.is_suffix_dot_star && !re.is_suffix_dot_plus &&
       re.substr_dot_star.empty() && re.substr_dot_plus.empty() && !re.has_or_values) {
-    fail("regex: pattern \"" + expr +
-         "\" falls outside the supported fast-path classes (literal / "
-         "alternation / prefix.* / prefix.+ / .+substr.+ / or-values); the "
-         "general NFA fallback is planned for round 2");
+    // General class: Glushkov NFA over the WHOLE original pattern, matched
+    // unanchored -- equivalent to Go's prefix-retry + anchored suffixRe loop
+    // (regex.go:186-211) for pure regexes.
+    GBuild b;
+    GInfo root = g_build(b, raw, expr);
+    re.has_nfa = true;
+    re.always_true = root.nullable;  // can match "" => unanchored matches all
+    re.nfa_blob = g_serialize(b, root);
+    re.prefix.clear();  // NFA matches the whole pattern; ignore the prefix
   }
   return re;
 }
@@ -463,8 +710,43 @@ static bool match_with_prefix(const RegexProg& re, strview s) {
   }
 }
 
+bool nfa_match(const uint8_t* blob, strview s) {
+  uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  auto rd64 = [&](size_t off) {
+    uint64_t v;
+    memcpy(&v, blob + off, 8);
+    return v;
+  };
+  const uint64_t first = rd64(8);
+  const uint64_t last = rd64(16);
+  const uint8_t* follow = blob + 24;
+  const uint8_t* table = blob + 24 + size_t(n) * 8;
+  uint64_t active = 0;
+  for (size_t i = 0; i < s.n; i++) {
+    uint64_t targets = first;
+    uint64_t m = active;
+    while (m) {
+      int x = __builtin_ctzll(m);
+      m &= m - 1;
+      uint64_t f;
+      memcpy(&f, follow + size_t(x) * 8, 8);
+      targets |= f;
+    }
+    uint64_t tb;
+    memcpy(&tb, table + size_t(uint8_t(s.p[i])) * 8, 8);
+    const uint64_t entered = targets & tb;
+    if (entered & last) return true;
+    active = entered;
+  }
+  return false;
+}
+
 bool regex_match(const RegexProg& re, strview s) {
   // Regex.MatchString (regex.go:86-98)
+  if (re.has_nfa) {
+    if (re.always_true) return true;
+    return nfa_match(re.nfa_blob.data(), s);
+  }
   if (re.is_only_prefix) {
     if (re.prefix.empty()) return true;
     return contains(s, re.prefix);

@@ -33,10 +33,23 @@ struct RegexProg {
   bool has_or_values = false;  // distinguishes empty-list from ["" ] etc.
   // Literals for bloom tokens (regex.go:101-124 GetLiterals).
   std::vector<std::string> literals;
+
+  // General-class fallback: Glushkov position automaton over bytes
+  // (replaces Go's suffixRe slow path, regex.go:148-151,188-191 — an
+  // unanchored whole-pattern existence match is equivalent to Go's
+  // prefix-retry + anchored-suffix loop for pure regexes).
+  bool has_nfa = false;
+  bool always_true = false;  // pattern matches the empty string
+  bytes nfa_blob;
 };
 
-// Compiles expr; throws vl::Error with a message naming the unsupported
-// construct when the pattern falls outside the fast-path classes.
+// NFA blob layout: u16 nstates, u16 pad, u32 pad, u64 first_mask,
+// u64 last_mask, u64 follow[nstates], u64 byte_table[256].
+bool nfa_match(const uint8_t* blob, strview s);
+
+// Compiles expr; patterns outside the fast-path classes compile to the NFA
+// fallback; throws vl::Error only for genuinely unsupported syntax
+// ({m,n} repetition, anchors, backreferences, >64 NFA positions).
 RegexProg regex_compile(const std::string& expr);
 
 // Regex.MatchString (regex.go:86-98,131-212).

@@ -592,9 +592,50 @@ __device__ bool d_regex_or_hasprefix(const DRegex& re, const A& a, long s0, long
   return false;
 }
 
+// Glushkov NFA executor (general regex class; see core/regex.cpp g_build).
+// blob: u16 nstates, pad to 8, u64 first, u64 last, u64 follow[n], u64 table[256]
+template <typename A>
+__device__ bool d_nfa_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
+  const uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  uint64_t first, last;
+  __builtin_memcpy(&first, blob + 8, 8);
+  __builtin_memcpy(&last, blob + 16, 8);
+  const uint8_t* follow = blob + 24;
+  const uint8_t* table = blob + 24 + size_t(n) * 8;
+  uint64_t active = 0;
+  for (long i = 0; i < sn; i++) {
+    uint64_t targets = first;
+    uint64_t m = active;
+    while (m) {
+      int x = __builtin_ctzll(m);
+      m &= m - 1;
+      uint64_t f;
+      __builtin_memcpy(&f, follow + size_t(x) * 8, 8);
+      targets |= f;
+    }
+    uint64_t tb;
+    __builtin_memcpy(&tb, table + size_t(a.u8(s0 + i)) * 8, 8);
+    const uint64_t entered = targets & tb;
+    if (entered & last) return true;
+    active = entered;
+  }
+  return false;
+}
+
 // Regex.MatchString (regex.go:86-212) over accessor bytes [s0, s0+sn).
 template <typename A>
 __device__ bool d_regex_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
+  if (blob[0] & kReAlways) return true;
+  if (blob[0] & kReNfa) {
+    // NFA blob sits after the (empty) prefix/substr/or-values header
+    DRegex hdr = d_regex_load(blob);
+    const uint8_t* p = hdr.ors;
+    for (int i = 0; i < hdr.n_or; i++) {
+      uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
+      p += 2 + len;
+    }
+    return d_nfa_match_at(p, a, s0, sn);
+  }
   DRegex re = d_regex_load(blob);
   if (re.flags & kReOnlyPrefix) {
     if (re.prefix_len == 0) return true;

@@ -102,6 +102,8 @@ enum : uint8_t {
   kReSubstrStar = 8,
   kReSubstrPlus = 16,
   kReHasOr = 32,
+  kReNfa = 64,      // general class: Glushkov NFA blob follows the or-values
+  kReAlways = 128,  // pattern matches the empty string => always true
 };
 
 }  // namespace vl

@@ -120,6 +120,8 @@ bytes serialize_regex(const RegexProg& re) {
   if (!re.substr_dot_star.empty()) flags |= kReSubstrStar;
   if (!re.substr_dot_plus.empty()) flags |= kReSubstrPlus;
   if (re.has_or_values) flags |= kReHasOr;
+  if (re.has_nfa) flags |= kReNfa;
+  if (re.always_true) flags |= kReAlways;
   const std::string& substr =
       !re.substr_dot_star.empty() ? re.substr_dot_star : re.substr_dot_plus;
   b.push_back(flags);
@@ -136,6 +138,7 @@ bytes serialize_regex(const RegexProg& re) {
     put16(v.size());
     b.insert(b.end(), v.begin(), v.end());
   }
+  if (re.has_nfa) b.insert(b.end(), re.nfa_blob.begin(), re.nfa_blob.end());
   return b;
 }
 
