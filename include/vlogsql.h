@@ -63,10 +63,16 @@ double vql_last_kernel_ms(void* stage);
 /* Copies result bitmaps to host (concatenated per-block u64 words). */
 int vql_fetch_bitmaps(void* stage, unsigned long long* out_words,
                       long long cap_words);
+/* Per-block matched-row counts of the last scan: the blockResult rowsLen
+ * popcount (block_result.go:403-413) and the `| stats count()` fast path
+ * (SURVEY.md §8f row 2). */
+int vql_fetch_block_hits(void* stage, unsigned long long* out,
+                         long long cap_blocks);
 
 /* Cold path (§8b vql_scan_batch): stage + scan + fetch + free in one call. */
 long long vql_scan_batch(void* part, void* filter, long block_lo, long block_hi,
-                         unsigned long long* out_words, long long cap_words);
+                         unsigned long long* out_words, long long cap_words,
+                         unsigned long long* out_popcounts);
 
 #ifdef __cplusplus
 }

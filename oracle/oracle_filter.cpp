@@ -321,6 +321,18 @@ static void apply_phrase(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
   }
 }
 
+// delegate used by filterSequence's len==1 ipv4/iso cases: evaluate the
+// phrase matcher with the sequence's single phrase + the sequence's tokens
+static void apply_phrase_for(const FilterNode& seq, const std::string& phrase,
+                             BlockCtx& ctx, Bitmap& bm) {
+  FilterNode tmp;
+  tmp.type = FilterNode::Phrase;
+  tmp.field = seq.field;
+  tmp.phrase = phrase;
+  tmp.token_hashes = seq.token_hashes;
+  apply_phrase(tmp, ctx, bm);
+}
+
 // ---- filterExact (filter_exact.go:178-235) ----
 
 static void apply_exact(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
@@ -609,6 +621,376 @@ static void apply_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
   }
 }
 
+
+// ---- filterPrefix (filter_prefix.go:58-316) ----
+
+static void apply_prefix(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  strview prefix(f.phrase);
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_prefix(strview(cv), prefix)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    bm.reset_bits();  // filter_prefix.go:73-78
+    return;
+  }
+  switch (ch.type) {
+    case ValueType::String: {
+      // matchStringByPrefix (filter_prefix.go:190-198)
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) { return match_prefix(v, prefix); });
+      return;
+    }
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) {
+        enc.push_back(match_prefix(strview(dv), prefix) ? 1 : 0);
+      }
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8:
+    case ValueType::Uint16:
+    case ValueType::Uint32:
+    case ValueType::Uint64: {
+      // matchUintNByPrefix (filter_prefix.go:200-285): no bloom gate
+      if (prefix.n == 0) return;
+      uint64_t n;
+      if (!try_parse_uint64(prefix, &n) || n > ch.max_value) {
+        bm.reset_bits();
+        return;
+      }
+      int width = ch.type == ValueType::Uint8 ? 1
+                  : ch.type == ValueType::Uint16 ? 2
+                  : ch.type == ValueType::Uint32 ? 4 : 8;
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string str = to_uint_string(v, width);
+        return match_prefix(strview(str), prefix);
+      });
+      return;
+    }
+    case ValueType::Int64: {
+      // matchInt64ByPrefix (filter_prefix.go:287-310)
+      if (prefix.n == 0) return;
+      if (!(prefix == std::string("-"))) {
+        int64_t n;
+        if (!try_parse_int64(prefix, &n) || n < int64_t(ch.min_value) ||
+            n > int64_t(ch.max_value)) {
+          bm.reset_bits();
+          return;
+        }
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string str;
+        format_int64(str, get_i64be_zigzag((const uint8_t*)v.p));
+        return match_prefix(strview(str), prefix);
+      });
+      return;
+    }
+    case ValueType::Float64: {
+      // matchFloat64ByPrefix (filter_prefix.go:148-176)
+      if (prefix.n == 0) return;
+      double ff;
+      bool ok = try_parse_float64_exact(prefix, &ff);
+      bool special = f.phrase == "." || f.phrase == "+" || f.phrase == "-" ||
+                     (prefix.n > 0 && (prefix.p[0] == 'e' || prefix.p[0] == 'E'));
+      if (!ok && !special) {
+        bm.reset_bits();
+        return;
+      }
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        uint64_t u = get_u64be((const uint8_t*)v.p);
+        double d;
+        memcpy(&d, &u, 8);
+        std::string str;
+        format_float64(str, d);
+        return match_prefix(strview(str), prefix);
+      });
+      return;
+    }
+    case ValueType::IPv4: {
+      // matchIPv4ByPrefix (filter_prefix.go:128-147)
+      if (prefix.n == 0) return;
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string str;
+        format_ipv4(str, get_u32be((const uint8_t*)v.p));
+        return match_prefix(strview(str), prefix);
+      });
+      return;
+    }
+    case ValueType::TimestampISO8601: {
+      // matchTimestampISO8601ByPrefix (filter_prefix.go:107-126)
+      if (prefix.n == 0) return;
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string str;
+        format_timestamp_iso8601(str, int64_t(get_u64be((const uint8_t*)v.p)));
+        return match_prefix(strview(str), prefix);
+      });
+      return;
+    }
+    default:
+      fail("unknown valueType in prefix filter");
+  }
+}
+
+// ---- filterExactPrefix (filter_exact_prefix.go:52-277) ----
+
+static void apply_exact_prefix(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  strview prefix(f.phrase);
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_exact_prefix(strview(cv), prefix)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (prefix.n > 0) bm.reset_bits();  // matchExactPrefix("", prefix)
+    return;
+  }
+  auto fmt_visit = [&](auto&& fmt) {
+    visit_values(ctx, ch, bm, [&](strview v) {
+      std::string str = fmt(v);
+      return match_exact_prefix(strview(str), prefix);
+    });
+  };
+  switch (ch.type) {
+    case ValueType::String: {
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm,
+                   [&](strview v) { return match_exact_prefix(v, prefix); });
+      return;
+    }
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) {
+        enc.push_back(match_exact_prefix(strview(dv), prefix) ? 1 : 0);
+      }
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8:
+    case ValueType::Uint16:
+    case ValueType::Uint32:
+    case ValueType::Uint64: {
+      // matchMinMaxExactPrefix (filter_exact_prefix.go:255-273)
+      if (prefix.n == 0) return;
+      if (!f.token_hashes.empty()) {
+        bm.reset_bits();
+        return;
+      }
+      uint64_t n;
+      if (!try_parse_uint64(prefix, &n) || n > ch.max_value) {
+        bm.reset_bits();
+        return;
+      }
+      int width = ch.type == ValueType::Uint8 ? 1
+                  : ch.type == ValueType::Uint16 ? 2
+                  : ch.type == ValueType::Uint32 ? 4 : 8;
+      fmt_visit([&](strview v) { return to_uint_string(v, width); });
+      return;
+    }
+    case ValueType::Int64: {
+      // matchInt64ByExactPrefix (filter_exact_prefix.go:226-253)
+      if (prefix.n == 0) return;
+      if (!f.token_hashes.empty()) {
+        bm.reset_bits();
+        return;
+      }
+      if (!(prefix == std::string("-"))) {
+        int64_t n;
+        if (!try_parse_int64(prefix, &n) || n > int64_t(ch.max_value) ||
+            n < int64_t(ch.min_value)) {
+          bm.reset_bits();
+          return;
+        }
+      }
+      fmt_visit([&](strview v) {
+        std::string str;
+        format_int64(str, get_i64be_zigzag((const uint8_t*)v.p));
+        return str;
+      });
+      return;
+    }
+    case ValueType::Float64: {
+      // matchFloat64ByExactPrefix (filter_exact_prefix.go:136-153)
+      if (prefix.n == 0) return;
+      if (f.token_hashes.size() > 2 * kBloomHashesCount ||
+          !match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      fmt_visit([&](strview v) {
+        uint64_t u = get_u64be((const uint8_t*)v.p);
+        double d;
+        memcpy(&d, &u, 8);
+        std::string str;
+        format_float64(str, d);
+        return str;
+      });
+      return;
+    }
+    case ValueType::IPv4: {
+      // matchIPv4ByExactPrefix (filter_exact_prefix.go:119-134)
+      if (prefix.n == 0) return;
+      if (f.phrase < "0" || f.phrase > "9" ||
+          f.token_hashes.size() > 3 * kBloomHashesCount ||
+          !match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      fmt_visit([&](strview v) {
+        std::string str;
+        format_ipv4(str, get_u32be((const uint8_t*)v.p));
+        return str;
+      });
+      return;
+    }
+    case ValueType::TimestampISO8601: {
+      // matchTimestampISO8601ByExactPrefix (filter_exact_prefix.go:102-117)
+      if (prefix.n == 0) return;
+      if (f.phrase < "0" || f.phrase > "9" ||
+          !match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      fmt_visit([&](strview v) {
+        std::string str;
+        format_timestamp_iso8601(str, int64_t(get_u64be((const uint8_t*)v.p)));
+        return str;
+      });
+      return;
+    }
+    default:
+      fail("unknown valueType in exact_prefix filter");
+  }
+}
+
+// ---- filterSequence (filter_sequence.go:84-269) ----
+
+static void apply_phrase_for(const FilterNode& seq, const std::string& phrase,
+                             BlockCtx& ctx, Bitmap& bm);
+
+static void apply_sequence(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  const auto& phrases = f.phrases;
+  if (phrases.empty()) return;  // filter_sequence.go:88-90
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_sequence(strview(cv), phrases)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!match_sequence(strview("", 0), phrases)) bm.reset_bits();
+    return;
+  }
+  switch (ch.type) {
+    case ValueType::String: {
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm,
+                   [&](strview v) { return match_sequence(v, phrases); });
+      return;
+    }
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) {
+        enc.push_back(match_sequence(strview(dv), phrases) ? 1 : 0);
+      }
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8:
+    case ValueType::Uint16:
+    case ValueType::Uint32:
+    case ValueType::Uint64:
+    case ValueType::Int64: {
+      // matchUintNBySequence / matchInt64BySequence (filter_sequence.go:219-258)
+      if (phrases.size() > 1) {
+        bm.reset_bits();
+        return;
+      }
+      if (ch.type == ValueType::Int64) {
+        match_int64_by_exact(ctx, ch, bm, strview(phrases[0]), f.token_hashes);
+      } else {
+        int width = ch.type == ValueType::Uint8 ? 1
+                    : ch.type == ValueType::Uint16 ? 2
+                    : ch.type == ValueType::Uint32 ? 4 : 8;
+        match_uint_by_exact(ctx, ch, bm, strview(phrases[0]), f.token_hashes, width);
+      }
+      return;
+    }
+    case ValueType::Float64: {
+      // matchFloat64BySequence (filter_sequence.go:179-196): always slow path
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        uint64_t u = get_u64be((const uint8_t*)v.p);
+        double d;
+        memcpy(&d, &u, 8);
+        std::string str;
+        format_float64(str, d);
+        return match_sequence(strview(str), phrases);
+      });
+      return;
+    }
+    case ValueType::IPv4:
+    case ValueType::TimestampISO8601: {
+      // len==1 delegates to the phrase matcher (filter_sequence.go:139-177)
+      if (phrases.size() == 1) {
+        apply_phrase_for(f, phrases[0], ctx, bm);
+        return;
+      }
+      if (!match_bloom_all(ctx, ch, f.token_hashes)) {
+        bm.reset_bits();
+        return;
+      }
+      bool is_ip = ch.type == ValueType::IPv4;
+      visit_values(ctx, ch, bm, [&](strview v) {
+        std::string str;
+        if (is_ip) {
+          format_ipv4(str, get_u32be((const uint8_t*)v.p));
+        } else {
+          format_timestamp_iso8601(str, int64_t(get_u64be((const uint8_t*)v.p)));
+        }
+        return match_sequence(strview(str), phrases);
+      });
+      return;
+    }
+    default:
+      fail("unknown valueType in sequence filter");
+  }
+}
+
 // ---- AND/OR bloom prefilters (filter_and.go:76-111, filter_or.go:80-115) ----
 
 static bool and_match_bloom(const FilterNode& f, BlockCtx& ctx) {
@@ -660,6 +1042,15 @@ void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
       return;
     case FilterNode::Regexp:
       apply_regexp(f, ctx, bm);
+      return;
+    case FilterNode::Prefix:
+      apply_prefix(f, ctx, bm);
+      return;
+    case FilterNode::ExactPrefix:
+      apply_exact_prefix(f, ctx, bm);
+      return;
+    case FilterNode::Sequence:
+      apply_sequence(f, ctx, bm);
       return;
     case FilterNode::Time:
       apply_time(f, ctx, bm);

@@ -168,6 +168,33 @@ TYPED_FILTERS = [
     '{"type":"regexp","field":"uni","re":"р.з"}',
     '{"type":"regexp","field":"u8","re":"1\\\\d"}',
     '{"type":"regexp","field":"ip","re":"10\\\\.\\\\d+\\\\.1?5"}',
+    # prefix / exact_prefix / sequence filters
+    '{"type":"prefix","field":"_msg","prefix":"log li"}',
+    '{"type":"prefix","field":"_msg","prefix":""}',
+    '{"type":"prefix","field":"lvl","prefix":"warn"}',
+    '{"type":"prefix","field":"u8","prefix":"1"}',
+    '{"type":"prefix","field":"u16","prefix":"10"}',
+    '{"type":"prefix","field":"i64","prefix":"-3"}',
+    '{"type":"prefix","field":"i64","prefix":"-"}',
+    '{"type":"prefix","field":"f64","prefix":"-18"}',
+    '{"type":"prefix","field":"ip","prefix":"10.2"}',
+    '{"type":"prefix","field":"iso","prefix":"2024-01-0"}',
+    '{"type":"exact_prefix","field":"_msg","prefix":"log line 1"}',
+    '{"type":"exact_prefix","field":"lvl","prefix":"ERR"}',
+    '{"type":"exact_prefix","field":"u8","prefix":"13"}',
+    '{"type":"exact_prefix","field":"i64","prefix":"-11"}',
+    '{"type":"exact_prefix","field":"f64","prefix":"0."}',
+    '{"type":"exact_prefix","field":"ip","prefix":"10.1"}',
+    '{"type":"exact_prefix","field":"iso","prefix":"2024-01-1"}',
+    '{"type":"sequence","field":"_msg","phrases":["log","level","took"]}',
+    '{"type":"sequence","field":"_msg","phrases":["level=error","13ms"]}',
+    '{"type":"sequence","field":"lvl","phrases":["warn"]}',
+    '{"type":"sequence","field":"u8","phrases":["13"]}',
+    '{"type":"sequence","field":"u8","phrases":["1","3"]}',
+    '{"type":"sequence","field":"ip","phrases":["10","35"]}',
+    '{"type":"sequence","field":"ip","phrases":["10.5.15.35"]}',
+    '{"type":"sequence","field":"iso","phrases":["2024","002Z"]}',
+    '{"type":"sequence","field":"f64","phrases":["18","625"]}',
     '{"type":"and","filters":['
     '{"type":"phrase","field":"lvl","phrase":"error"},'
     '{"type":"range","field":"u8","min":0,"max":50},'

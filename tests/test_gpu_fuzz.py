@@ -42,10 +42,10 @@ TYPED_REGEXES = [
 
 
 def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
+    leaf_kinds = ["phrase", "phrase", "exact", "regexp", "time", "range",
+                  "prefix", "exact_prefix", "sequence"]
     kind = rng.choice(
-        ["phrase", "phrase", "exact", "regexp", "time", "range", "and", "or",
-         "not"] if depth > 0 else
-        ["phrase", "phrase", "exact", "regexp", "time", "range"])
+        leaf_kinds + ["and", "or", "not"] if depth > 0 else leaf_kinds)
     if kind == "phrase":
         return {"type": "phrase",
                 "field": rng.choice(fields_str + fields_num),
@@ -54,6 +54,15 @@ def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
         return {"type": "exact",
                 "field": rng.choice(fields_str + fields_num),
                 "value": rng.choice(phrases)}
+    if kind in ("prefix", "exact_prefix"):
+        return {"type": kind,
+                "field": rng.choice(fields_str + fields_num),
+                "prefix": rng.choice(phrases)}
+    if kind == "sequence":
+        return {"type": "sequence",
+                "field": rng.choice(fields_str + fields_num),
+                "phrases": [rng.choice(phrases)
+                            for _ in range(rng.randrange(1, 4))]}
     if kind == "regexp":
         return {"type": "regexp", "field": rng.choice(fields_str),
                 "re": rng.choice(regexes)}

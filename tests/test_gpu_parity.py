@@ -73,11 +73,20 @@ def test_cold_scan_batch(gen_part):
     p = Part(gen_part)
     nwords = _nwords(p)
     buf = (ctypes.c_ulonglong * nwords)()
-    hits = lib.vql_scan_batch(part, filt, 0, -1, buf, nwords)
+    pops = (ctypes.c_ulonglong * p.blocks)()
+    hits = lib.vql_scan_batch(part, filt, 0, -1, buf, nwords, pops)
     assert hits >= 0, lib.vql_errstr().decode()
+    assert sum(pops) == hits
 
     orc = OracleScanner(gen_part)
     orc_hits, orc_bits = orc.scan(FILTERS[1], with_bitmaps=True)
+    # per-block popcounts from the oracle bitmaps (`| stats count()` parity)
+    off = 0
+    for b in range(p.blocks):
+        nw = (p.block_rows(b) + 63) // 64
+        words = orc_bits[off * 8:(off + nw) * 8]
+        assert pops[b] == bin(int.from_bytes(words, "little")).count("1")
+        off += nw
     orc.close()
     assert hits == orc_hits
     assert bytes(buf) == orc_bits

@@ -70,7 +70,12 @@ def load_product() -> ctypes.CDLL:
     lib.vql_scan_batch.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long,
                                    ctypes.c_long,
                                    ctypes.POINTER(ctypes.c_ulonglong),
-                                   ctypes.c_longlong]
+                                   ctypes.c_longlong,
+                                   ctypes.POINTER(ctypes.c_ulonglong)]
+    lib.vql_fetch_block_hits.restype = ctypes.c_int
+    lib.vql_fetch_block_hits.argtypes = [ctypes.c_void_p,
+                                         ctypes.POINTER(ctypes.c_ulonglong),
+                                         ctypes.c_longlong]
     _product = lib
     return lib
 
@@ -283,3 +288,10 @@ class Stage:
         if self.lib.vql_fetch_bitmaps(self.h, buf, nwords) != 0:
             raise RuntimeError(self.lib.vql_errstr().decode())
         return bytes(buf)[: nwords * 8]
+
+    def fetch_block_hits(self, nblocks):
+        """Per-block matched-row counts (`| stats count()` fast path)."""
+        buf = (ctypes.c_ulonglong * max(nblocks, 1))()
+        if self.lib.vql_fetch_block_hits(self.h, buf, nblocks) != 0:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+        return list(buf)[:nblocks]

@@ -53,6 +53,9 @@ std::vector<FieldTokens> common_tokens(const std::vector<FilterNode>& children,
       case FilterNode::Phrase:
       case FilterNode::Exact:
       case FilterNode::Regexp:
+      case FilterNode::Prefix:
+      case FilterNode::ExactPrefix:
+      case FilterNode::Sequence:
         merge(c.field, c.tokens);
         break;
       case FilterNode::Or:
@@ -135,6 +138,24 @@ FilterNode build(const JValue& v) {
     std::vector<std::string> lits;
     for (const auto& lit : n.re.literals) lits.push_back(skip_first_last_token(lit));
     n.tokens = tokenize_strings(lits);
+    n.token_hashes = probe_hashes(n.tokens);
+  } else if (type == "prefix" || type == "exact_prefix") {
+    n.type = type == "prefix" ? FilterNode::Prefix : FilterNode::ExactPrefix;
+    n.field = jget(v, "field").str;
+    n.phrase = jget(v, "prefix").str;
+    // filterPrefix/filterExactPrefix initTokens: getTokensSkipLast
+    // (filter_prefix.go:50-53, filter_exact_prefix.go:43-46)
+    n.tokens = get_tokens_skip_last(n.phrase);
+    n.token_hashes = probe_hashes(n.tokens);
+  } else if (type == "sequence") {
+    n.type = FilterNode::Sequence;
+    n.field = jget(v, "field").str;
+    for (const auto& pj : jget(v, "phrases").arr) {
+      // getNonEmptyPhrases (filter_sequence.go:57-66)
+      if (!pj.str.empty()) n.phrases.push_back(pj.str);
+    }
+    // filterSequence.initTokens (filter_sequence.go:51-55)
+    n.tokens = tokenize_strings(n.phrases);
     n.token_hashes = probe_hashes(n.tokens);
   } else if (type == "and" || type == "or") {
     n.type = type == "and" ? FilterNode::And : FilterNode::Or;

@@ -25,10 +25,14 @@ struct FieldTokens {
 };
 
 struct FilterNode {
-  enum Type { Phrase, Exact, Regexp, And, Or, Not, Time, Range, Noop } type;
+  enum Type {
+    Phrase, Exact, Regexp, And, Or, Not, Time, Range, Noop,
+    Prefix, ExactPrefix, Sequence
+  } type;
 
   std::string field;   // phrase/exact/regexp/range (as written in the query)
-  std::string phrase;  // Phrase: phrase; Exact: value
+  std::string phrase;  // Phrase: phrase; Exact: value; Prefix/ExactPrefix: prefix
+  std::vector<std::string> phrases;  // Sequence: non-empty phrases, in order
   RegexProg re;        // Regexp
   double min_f = 0, max_f = 0;    // Range
   int64_t min_ts = 0, max_ts = 0;  // Time

@@ -83,6 +83,61 @@ std::string skip_first_last_token(const std::string& s) {
   return std::string(p, n);
 }
 
+bool match_prefix(strview s, strview prefix) {
+  // filter_prefix.go:318-352
+  if (prefix.n == 0) return s.n > 0;
+  if (prefix.n > s.n) return false;
+  int sz;
+  uint32_t r = uint8_t(prefix.p[0]);
+  if (r >= 0x80) r = utf8_decode(prefix.p, prefix.n, &sz);
+  bool starts_with_token = is_token_rune(r);
+  long offset = 0;
+  for (;;) {
+    const char* found =
+        (const char*)memmem(s.p + offset, s.n - size_t(offset), prefix.p, prefix.n);
+    if (!found) return false;
+    offset = found - s.p;
+    if (starts_with_token && offset > 0) {
+      uint32_t rb = uint8_t(s.p[offset - 1]);
+      if (rb >= 0x80) rb = utf8_decode_last(s.p, size_t(offset), &sz);
+      if (rb == 0xFFFD || is_token_rune(rb)) {
+        offset++;
+        continue;
+      }
+    }
+    return true;
+  }
+}
+
+bool match_exact_prefix(strview s, strview prefix) {
+  return s.n >= prefix.n && memcmp(s.p, prefix.p, prefix.n) == 0;
+}
+
+bool match_sequence(strview s, const std::vector<std::string>& phrases) {
+  // filter_sequence.go:260-269
+  for (const auto& phrase : phrases) {
+    long n = get_phrase_pos(s, strview(phrase));
+    if (n < 0) return false;
+    s.p += n + phrase.size();
+    s.n -= size_t(n) + phrase.size();
+  }
+  return true;
+}
+
+std::vector<std::string> get_tokens_skip_last(const std::string& str) {
+  // filter_prefix.go:354-363
+  const char* p = str.data();
+  size_t n = str.size();
+  for (;;) {
+    if (n == 0) break;
+    int sz;
+    uint32_t r = utf8_decode_last(p, n, &sz);
+    if (!is_token_rune(r)) break;
+    n -= sz;
+  }
+  return tokenize_strings({std::string(p, n)});
+}
+
 bool match_string_by_all_tokens(strview v, const std::vector<std::string>& tokens) {
   for (const auto& t : tokens) {
     if (!match_phrase(v, strview(t))) return false;

@@ -20,6 +20,15 @@ long get_phrase_pos(strview s, strview phrase);
 // skipFirstLastToken (filter_regexp.go:53-69)
 std::string skip_first_last_token(const std::string& s);
 
+// matchPrefix (filter_prefix.go:318-352): empty prefix matches non-empty s.
+bool match_prefix(strview s, strview prefix);
+// matchExactPrefix = strings.HasPrefix (filter_exact_prefix.go:275-277)
+bool match_exact_prefix(strview s, strview prefix);
+// matchSequence (filter_sequence.go:260-269)
+bool match_sequence(strview s, const std::vector<std::string>& phrases);
+// getTokensSkipLast (filter_prefix.go:354-363)
+std::vector<std::string> get_tokens_skip_last(const std::string& s);
+
 // matchStringByAllTokens (filter_and.go:189-196)
 bool match_string_by_all_tokens(strview v, const std::vector<std::string>& tokens);
 // matchDictValuesByAllTokens (filter_and.go:198-208): match against

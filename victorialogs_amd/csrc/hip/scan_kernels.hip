@@ -164,16 +164,17 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
   return -1;
 }
 
-// getPhrasePos (filter_phrase.go:220-270) over accessor bytes [s0, s0+sn).
+// getPhrasePos (filter_phrase.go:220-270) over accessor bytes [s0, s0+sn);
+// returns the match position or -1.
 template <typename A>
-__device__ bool d_match_phrase_at(const A& a, long s0, long sn, const uint8_t* ph,
-                                  long phn, uint8_t flags) {
-  if (phn == 0) return sn == 0;  // filter_phrase.go:212-215
-  if (phn > sn) return false;
+__device__ long d_get_phrase_pos_at(const A& a, long s0, long sn, const uint8_t* ph,
+                                    long phn, uint8_t flags) {
+  if (phn == 0) return 0;
+  if (phn > sn) return -1;
   long pos = 0;
   for (;;) {
     long n = d_index_at(a, s0 + pos, sn - pos, ph, phn);
-    if (n < 0) return false;
+    if (n < 0) return -1;
     pos += n;
     if ((flags & kPhraseStartsToken) && pos > 0) {
       uint32_t rb = a.u8(s0 + pos - 1);
@@ -197,8 +198,73 @@ __device__ bool d_match_phrase_at(const A& a, long s0, long sn, const uint8_t* p
         continue;
       }
     }
+    return pos;
+  }
+}
+
+template <typename A>
+__device__ __forceinline__ bool d_match_phrase_at(const A& a, long s0, long sn,
+                                                  const uint8_t* ph, long phn,
+                                                  uint8_t flags) {
+  if (phn == 0) return sn == 0;  // filter_phrase.go:212-215
+  return d_get_phrase_pos_at(a, s0, sn, ph, phn, flags) >= 0;
+}
+
+// matchPrefix (filter_prefix.go:318-352): empty prefix matches non-empty s;
+// boundary check only at the start.
+template <typename A>
+__device__ bool d_match_prefix_at(const A& a, long s0, long sn, const uint8_t* pf,
+                                  long pfn, uint8_t flags) {
+  if (pfn == 0) return sn > 0;
+  if (pfn > sn) return false;
+  long off = 0;
+  for (;;) {
+    long n = d_index_at(a, s0 + off, sn - off, pf, pfn);
+    if (n < 0) return false;
+    off += n;
+    if ((flags & kPhraseStartsToken) && off > 0) {
+      uint32_t rb = a.u8(s0 + off - 1);
+      if (rb >= 0x80) {
+        int sz;
+        rb = d_utf8_decode_last(a, s0, off, &sz);
+      }
+      if (rb == 0xFFFD || d_is_token_rune(rb)) {
+        off++;
+        continue;
+      }
+    }
     return true;
   }
+}
+
+template <typename A>
+__device__ __forceinline__ bool d_has_prefix_bytes(const A& a, long s0, long sn,
+                                                   const uint8_t* pf, long pfn) {
+  if (pfn > sn) return false;
+  for (long i = 0; i < pfn; i++) {
+    if (a.u8(s0 + i) != pf[i]) return false;
+  }
+  return true;
+}
+
+// matchSequence (filter_sequence.go:260-269) over a serialized phrase list:
+// blob = u16 n, then per phrase { u16 len, u8 flags, bytes }.
+template <typename A>
+__device__ bool d_match_sequence_at(const A& a, long s0, long sn,
+                                    const uint8_t* blob) {
+  uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t* p = blob + 2;
+  for (uint16_t i = 0; i < n; i++) {
+    uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
+    uint8_t flags = p[2];
+    p += 3;
+    long pos = d_get_phrase_pos_at(a, s0, sn, p, len, flags);
+    if (pos < 0) return false;
+    s0 += pos + len;
+    sn -= pos + len;
+    p += len;
+  }
+  return true;
 }
 
 // ---- number/ip/timestamp formatting (device mirrors of values.cpp) ----
@@ -719,6 +785,12 @@ __device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
       double mx = __builtin_bit_cast(double, lb.vmax);
       return x >= mn && x <= mx;
     }
+    case kScanPrefixStr:
+      return d_match_prefix_at(a, s0, sn, lb.operand, lb.operand_len, lb.flags);
+    case kScanExactPrefixStr:
+      return d_has_prefix_bytes(a, s0, sn, lb.operand, lb.operand_len);
+    case kScanSeqStr:
+      return d_match_sequence_at(a, s0, sn, lb.operand);
     default:  // kScanRegexStr
       return d_regex_match_at(lb.operand, a, s0, sn);
   }
@@ -837,6 +909,49 @@ __device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
     }
+    case kScanPrefixFmt:
+    case kScanExactPrefixFmt:
+    case kScanSeqFmt: {
+      char buf[344];
+      int n;
+      switch (lb.flags >> 4) {
+        case kFmtU64: {
+          const uint8_t* p = lb.data + size_t(row) * lb.width;
+          uint64_t v;
+          switch (lb.width) {
+            case 1: v = p[0]; break;
+            case 2: v = d_get_u16be(p); break;
+            case 4: v = d_get_u32be(p); break;
+            default: v = d_get_u64be(p); break;
+          }
+          n = d_format_u64(buf, v);
+          break;
+        }
+        case kFmtI64: {
+          uint64_t u = d_get_u64be(lb.data + size_t(row) * 8);
+          n = d_format_i64(buf, int64_t(u >> 1) ^ (int64_t(u << 63) >> 63));
+          break;
+        }
+        case kFmtF64:
+          n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
+          break;
+        case kFmtIp:
+          n = d_format_ipv4(buf, d_get_u32be(lb.data + size_t(row) * 4));
+          break;
+        default:
+          n = d_format_iso8601(buf, int64_t(d_get_u64be(lb.data + size_t(row) * 8)));
+          break;
+      }
+      BufAcc a{(const uint8_t*)buf};
+      if (lb.kind == kScanPrefixFmt) {
+        return d_match_prefix_at(a, 0, n, lb.operand, lb.operand_len,
+                                 lb.flags & 15);
+      }
+      if (lb.kind == kScanExactPrefixFmt) {
+        return d_has_prefix_bytes(a, 0, n, lb.operand, lb.operand_len);
+      }
+      return d_match_sequence_at(a, 0, n, lb.operand);
+    }
     default:
       return false;
   }
@@ -844,7 +959,8 @@ __device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
 
 __device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
   return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr ||
-         kind == kScanRangeStr;
+         kind == kScanRangeStr || kind == kScanPrefixStr ||
+         kind == kScanExactPrefixStr || kind == kScanSeqStr;
 }
 
 // ---- the program kernel ----
@@ -1040,6 +1156,9 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
     unsigned long long s = 0;
     for (int k = 0; k < nwaves; k++) s += wave_sums[k];
     atomicAdd(hits, s);
+    // per-block popcount: the blockResult rowsLen / `| stats count()` fast
+    // path (block_result.go:403-413)
+    atomicAdd(blk.hits_out, s);
   }
 }
 

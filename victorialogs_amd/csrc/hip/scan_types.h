@@ -43,6 +43,21 @@ enum ScanKind : uint8_t {
   kScanPhraseF64 = 15,  // matchPhrase over Ryu-formatted float64 (filter_phrase.go:159-186)
   kScanRegexF64 = 16,   // regex over Ryu-formatted float64 (filter_regexp.go:155-166)
   kScanRangeStr = 17,   // matchRange via parseMathNumber per row (filter_range.go:261-265,369-372)
+  kScanPrefixStr = 18,      // matchPrefix over string rows (filter_prefix.go:318-352)
+  kScanExactPrefixStr = 19, // strings.HasPrefix (filter_exact_prefix.go:275-277)
+  kScanSeqStr = 20,         // matchSequence (filter_sequence.go:260-269)
+  kScanPrefixFmt = 21,      // matchPrefix over formatted value (fmt in flags>>4)
+  kScanExactPrefixFmt = 22, // HasPrefix over formatted value
+  kScanSeqFmt = 23,         // matchSequence over formatted value
+};
+
+// format source for the *Fmt kinds, stored in flags bits 4..7
+enum : uint8_t {
+  kFmtU64 = 1,   // BE uint of lb.width bytes -> decimal
+  kFmtI64 = 2,   // BE zig-zag int64 -> decimal
+  kFmtF64 = 3,   // BE float64 bits -> Ryu shortest 'f'
+  kFmtIp = 4,    // BE u32 -> dotted quad
+  kFmtIso = 5,   // BE u64 nsecs -> iso8601
 };
 
 // phrase flags
@@ -81,6 +96,7 @@ struct DevLeafBlock {
 
 struct DevBlock {
   uint64_t* bitmap_out;  // word 0 = rows [0,64) of the block
+  unsigned long long* hits_out;  // per-block matched-row counter
   uint32_t rows;
   uint32_t pad;
 };

@@ -22,6 +22,7 @@
 #include <vector>
 
 #include "core/filter.h"
+#include "core/bloom.h"
 #include "core/match.h"
 #include "core/tokenizer.h"
 #include "core/part_reader.h"
@@ -142,6 +143,23 @@ bytes serialize_regex(const RegexProg& re) {
   return b;
 }
 
+bytes serialize_phrases(const std::vector<std::string>& phrases);
+uint8_t phrase_flags_of(const std::string& phrase);
+
+bytes serialize_phrases(const std::vector<std::string>& phrases) {
+  // blob = u16 n, then per phrase { u16 len, u8 flags, bytes }
+  bytes b;
+  b.push_back(uint8_t(phrases.size()));
+  b.push_back(uint8_t(phrases.size() >> 8));
+  for (const auto& ph : phrases) {
+    b.push_back(uint8_t(ph.size()));
+    b.push_back(uint8_t(ph.size() >> 8));
+    b.push_back(phrase_flags_of(ph));
+    b.insert(b.end(), ph.begin(), ph.end());
+  }
+  return b;
+}
+
 uint8_t phrase_flags_of(const std::string& phrase) {
   // getPhrasePos boundary-rune precomputation (filter_phrase.go:228-238)
   if (phrase.empty()) return 0;
@@ -180,6 +198,7 @@ struct Stage {
   DevBlock* d_blocks = nullptr;
   DevChunk* d_chunks = nullptr;
   unsigned long long* d_hits = nullptr;
+  unsigned long long* d_block_hits = nullptr;
   uint64_t* d_bitmap = nullptr;
   size_t bitmap_words = 0;
   uint32_t nchunks = 0;
@@ -203,6 +222,7 @@ struct Stage {
     if (d_blocks) hipFree(d_blocks);
     if (d_chunks) hipFree(d_chunks);
     if (d_hits) hipFree(d_hits);
+    if (d_block_hits) hipFree(d_block_hits);
   }
 
   uint8_t* push(const void* src, size_t n, size_t align = 16) {
@@ -411,6 +431,33 @@ void stage_dict(DevLeafBlock& lb, BlockStageCtx& ctx, const ColumnHeader& ch,
   lb.kind = kScanDict;
   lb.dict_mask = mask;
   lb.data = sc.d_data;
+}
+
+// formats one encoded fixed-width value like the reference's to*String helpers
+std::string format_encoded(ValueType t, strview v) {
+  std::string s;
+  const uint8_t* p = (const uint8_t*)v.p;
+  switch (t) {
+    case ValueType::Uint8: format_uint64(s, p[0]); break;
+    case ValueType::Uint16: format_uint64(s, get_u16be(p)); break;
+    case ValueType::Uint32: format_uint64(s, get_u32be(p)); break;
+    case ValueType::Uint64: format_uint64(s, get_u64be(p)); break;
+    case ValueType::Int64: format_int64(s, get_i64be_zigzag(p)); break;
+    case ValueType::Float64: {
+      uint64_t u = get_u64be(p);
+      double d;
+      memcpy(&d, &u, 8);
+      format_float64(s, d);
+      break;
+    }
+    case ValueType::IPv4: format_ipv4(s, get_u32be(p)); break;
+    case ValueType::TimestampISO8601:
+      format_timestamp_iso8601(s, int64_t(get_u64be(p)));
+      break;
+    default:
+      fail("format_encoded: unexpected type");
+  }
+  return s;
 }
 
 // stage one (leaf, block) descriptor; mirrors the oracle's apply_* dispatch.
@@ -890,6 +937,398 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
       }
     }
 
+
+    case FilterNode::Prefix: {
+      // filterPrefix.applyToBlockSearch (filter_prefix.go:58-316)
+      strview prefix(f.phrase);
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = match_prefix(strview(cv), prefix) ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = kModeNone;  // filter_prefix.go:73-78
+        return;
+      }
+      auto scan_fmt = [&](uint8_t fmt, uint8_t width, bool bloom) {
+        const StagedStrCol& sc = ctx.stage_column(ch);
+        if (sc.is_const) {
+          std::string str = format_encoded(ch.type, strview(sc.const_value));
+          lb.mode = match_prefix(strview(str), prefix) ? kModeAll : kModeNone;
+          return;
+        }
+        lb.mode = kModeScan;
+        lb.kind = kScanPrefixFmt;
+        lb.width = width;
+        lb.flags = uint8_t((li.phrase_flags & 15) | (fmt << 4));
+        lb.operand = li.d_operand;
+        lb.operand_len = uint32_t(prefix.n);
+        lb.data = sc.d_data;
+        if (bloom) set_bloom_gate(lb, ctx, ch, li);
+      };
+      switch (ch.type) {
+        case ValueType::String: {
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            lb.mode = match_prefix(strview(sc.const_value), prefix) ? kModeAll
+                                                                    : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanPrefixStr;
+          lb.flags = li.phrase_flags;
+          lb.operand = li.d_operand;
+          lb.operand_len = uint32_t(prefix.n);
+          lb.data = sc.d_data;
+          lb.offsets = sc.d_offsets;
+          set_bloom_gate(lb, ctx, ch, li);
+          return;
+        }
+        case ValueType::Dict: {
+          uint32_t mask = dict_mask_of(
+              ch.dict, [&](strview dv) { return match_prefix(dv, prefix); });
+          stage_dict(lb, ctx, ch, mask);
+          return;
+        }
+        case ValueType::Uint8:
+        case ValueType::Uint16:
+        case ValueType::Uint32:
+        case ValueType::Uint64: {
+          // filter_prefix.go:200-285 (no bloom gate)
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          uint64_t n;
+          if (!try_parse_uint64(prefix, &n) || n > ch.max_value) {
+            lb.mode = kModeNone;
+            return;
+          }
+          uint8_t w = ch.type == ValueType::Uint8 ? 1
+                      : ch.type == ValueType::Uint16 ? 2
+                      : ch.type == ValueType::Uint32 ? 4 : 8;
+          scan_fmt(kFmtU64, w, false);
+          return;
+        }
+        case ValueType::Int64: {
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          if (f.phrase != "-") {
+            int64_t n;
+            if (!try_parse_int64(prefix, &n) || n < int64_t(ch.min_value) ||
+                n > int64_t(ch.max_value)) {
+              lb.mode = kModeNone;
+              return;
+            }
+          }
+          scan_fmt(kFmtI64, 8, false);
+          return;
+        }
+        case ValueType::Float64: {
+          // filter_prefix.go:148-176
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          double ff;
+          bool ok = try_parse_float64_exact(prefix, &ff);
+          bool special = f.phrase == "." || f.phrase == "+" || f.phrase == "-" ||
+                         prefix.p[0] == 'e' || prefix.p[0] == 'E';
+          if (!ok && !special) {
+            lb.mode = kModeNone;
+            return;
+          }
+          scan_fmt(kFmtF64, 8, true);
+          return;
+        }
+        case ValueType::IPv4:
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          scan_fmt(kFmtIp, 4, true);
+          return;
+        case ValueType::TimestampISO8601:
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          scan_fmt(kFmtIso, 8, true);
+          return;
+        default:
+          fail("unknown valueType while staging prefix filter");
+      }
+    }
+
+    case FilterNode::ExactPrefix: {
+      // filterExactPrefix.applyToBlockSearch (filter_exact_prefix.go:52-277)
+      strview prefix(f.phrase);
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = match_exact_prefix(strview(cv), prefix) ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = prefix.n > 0 ? kModeNone : kModeAll;
+        return;
+      }
+      auto scan_fmt = [&](uint8_t fmt, uint8_t width, bool bloom) {
+        const StagedStrCol& sc = ctx.stage_column(ch);
+        if (sc.is_const) {
+          std::string str = format_encoded(ch.type, strview(sc.const_value));
+          lb.mode =
+              match_exact_prefix(strview(str), prefix) ? kModeAll : kModeNone;
+          return;
+        }
+        lb.mode = kModeScan;
+        lb.kind = kScanExactPrefixFmt;
+        lb.width = width;
+        lb.flags = uint8_t(fmt << 4);
+        lb.operand = li.d_operand;
+        lb.operand_len = uint32_t(prefix.n);
+        lb.data = sc.d_data;
+        if (bloom) set_bloom_gate(lb, ctx, ch, li);
+      };
+      switch (ch.type) {
+        case ValueType::String: {
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            lb.mode = match_exact_prefix(strview(sc.const_value), prefix)
+                          ? kModeAll
+                          : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanExactPrefixStr;
+          lb.operand = li.d_operand;
+          lb.operand_len = uint32_t(prefix.n);
+          lb.data = sc.d_data;
+          lb.offsets = sc.d_offsets;
+          set_bloom_gate(lb, ctx, ch, li);
+          return;
+        }
+        case ValueType::Dict: {
+          uint32_t mask = dict_mask_of(ch.dict, [&](strview dv) {
+            return match_exact_prefix(dv, prefix);
+          });
+          stage_dict(lb, ctx, ch, mask);
+          return;
+        }
+        case ValueType::Uint8:
+        case ValueType::Uint16:
+        case ValueType::Uint32:
+        case ValueType::Uint64: {
+          // matchMinMaxExactPrefix (filter_exact_prefix.go:255-273)
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          if (!f.token_hashes.empty()) {
+            lb.mode = kModeNone;
+            return;
+          }
+          uint64_t n;
+          if (!try_parse_uint64(prefix, &n) || n > ch.max_value) {
+            lb.mode = kModeNone;
+            return;
+          }
+          uint8_t w = ch.type == ValueType::Uint8 ? 1
+                      : ch.type == ValueType::Uint16 ? 2
+                      : ch.type == ValueType::Uint32 ? 4 : 8;
+          scan_fmt(kFmtU64, w, false);
+          return;
+        }
+        case ValueType::Int64: {
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          if (!f.token_hashes.empty()) {
+            lb.mode = kModeNone;
+            return;
+          }
+          if (f.phrase != "-") {
+            int64_t n;
+            if (!try_parse_int64(prefix, &n) || n > int64_t(ch.max_value) ||
+                n < int64_t(ch.min_value)) {
+              lb.mode = kModeNone;
+              return;
+            }
+          }
+          scan_fmt(kFmtI64, 8, false);
+          return;
+        }
+        case ValueType::Float64: {
+          // filter_exact_prefix.go:136-153
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          if (f.token_hashes.size() > 2 * kBloomHashesCount) {
+            lb.mode = kModeNone;
+            return;
+          }
+          scan_fmt(kFmtF64, 8, true);
+          return;
+        }
+        case ValueType::IPv4: {
+          // filter_exact_prefix.go:119-134
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          if (f.phrase < "0" || f.phrase > "9" ||
+              f.token_hashes.size() > 3 * kBloomHashesCount) {
+            lb.mode = kModeNone;
+            return;
+          }
+          scan_fmt(kFmtIp, 4, true);
+          return;
+        }
+        case ValueType::TimestampISO8601: {
+          // filter_exact_prefix.go:102-117
+          if (prefix.n == 0) {
+            lb.mode = kModeAll;
+            return;
+          }
+          if (f.phrase < "0" || f.phrase > "9") {
+            lb.mode = kModeNone;
+            return;
+          }
+          scan_fmt(kFmtIso, 8, true);
+          return;
+        }
+        default:
+          fail("unknown valueType while staging exact_prefix filter");
+      }
+    }
+
+    case FilterNode::Sequence: {
+      // filterSequence.applyToBlockSearch (filter_sequence.go:84-258)
+      const auto& phrases = f.phrases;
+      if (phrases.empty()) {
+        lb.mode = kModeAll;
+        return;
+      }
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = match_sequence(strview(cv), phrases) ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode =
+            match_sequence(strview("", 0), phrases) ? kModeAll : kModeNone;
+        return;
+      }
+      switch (ch.type) {
+        case ValueType::String: {
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            lb.mode = match_sequence(strview(sc.const_value), phrases)
+                          ? kModeAll
+                          : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanSeqStr;
+          lb.operand = li.d_operand;
+          lb.operand_len = uint32_t(li.operand.size());
+          lb.data = sc.d_data;
+          lb.offsets = sc.d_offsets;
+          set_bloom_gate(lb, ctx, ch, li);
+          return;
+        }
+        case ValueType::Dict: {
+          uint32_t mask = dict_mask_of(
+              ch.dict, [&](strview dv) { return match_sequence(dv, phrases); });
+          stage_dict(lb, ctx, ch, mask);
+          return;
+        }
+        case ValueType::Uint8:
+        case ValueType::Uint16:
+        case ValueType::Uint32:
+        case ValueType::Uint64:
+        case ValueType::Int64: {
+          // filter_sequence.go:219-258: multi-phrase cannot match one number
+          if (phrases.size() > 1) {
+            lb.mode = kModeNone;
+            return;
+          }
+          bytes bin;
+          if (!exact_bin_value(ch, strview(phrases[0]), bin)) {
+            lb.mode = kModeNone;
+            return;
+          }
+          stage_eq_bin(lb, ctx, ch, li, st, bin);
+          return;
+        }
+        case ValueType::Float64: {
+          // matchFloat64BySequence (filter_sequence.go:179-196)
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            std::string str = format_encoded(ch.type, strview(sc.const_value));
+            lb.mode = match_sequence(strview(str), phrases) ? kModeAll
+                                                            : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanSeqFmt;
+          lb.width = 8;
+          lb.flags = uint8_t(kFmtF64 << 4);
+          lb.operand = li.d_operand;
+          lb.operand_len = uint32_t(li.operand.size());
+          lb.data = sc.d_data;
+          set_bloom_gate(lb, ctx, ch, li);
+          return;
+        }
+        case ValueType::IPv4:
+        case ValueType::TimestampISO8601: {
+          if (phrases.size() == 1) {
+            // delegate to the phrase matcher (filter_sequence.go:139-177)
+            FilterNode tmp;
+            tmp.type = FilterNode::Phrase;
+            tmp.field = f.field;
+            tmp.phrase = phrases[0];
+            tmp.token_hashes = f.token_hashes;
+            LeafInfo tmp_li;
+            tmp_li.node = &tmp;
+            tmp_li.cname = li.cname;
+            tmp_li.operand.assign(phrases[0].begin(), phrases[0].end());
+            tmp_li.phrase_flags = phrase_flags_of(phrases[0]);
+            tmp_li.d_hashes = li.d_hashes;
+            tmp_li.d_operand =
+                (const uint8_t*)st.push(tmp_li.operand.data(),
+                                        tmp_li.operand.size(), 8);
+            stage_leaf(tmp_li, ctx, st, lb);
+            return;
+          }
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            std::string str = format_encoded(ch.type, strview(sc.const_value));
+            lb.mode = match_sequence(strview(str), phrases) ? kModeAll
+                                                            : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanSeqFmt;
+          lb.width = ch.type == ValueType::IPv4 ? 4 : 8;
+          lb.flags = uint8_t((ch.type == ValueType::IPv4 ? kFmtIp : kFmtIso) << 4);
+          lb.operand = li.d_operand;
+          lb.operand_len = uint32_t(li.operand.size());
+          lb.data = sc.d_data;
+          set_bloom_gate(lb, ctx, ch, li);
+          return;
+        }
+        default:
+          fail("unknown valueType while staging sequence filter");
+      }
+    }
+
     default:
       fail("stage_leaf: non-leaf node");
   }
@@ -943,6 +1382,14 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
       case FilterNode::Regexp:
         li.operand = serialize_regex(li.node->re);
         break;
+      case FilterNode::Prefix:
+      case FilterNode::ExactPrefix:
+        li.operand.assign(li.node->phrase.begin(), li.node->phrase.end());
+        li.phrase_flags = phrase_flags_of(li.node->phrase);
+        break;
+      case FilterNode::Sequence:
+        li.operand = serialize_phrases(li.node->phrases);
+        break;
       default:
         break;
     }
@@ -980,6 +1427,7 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
     DevBlock& db = blocks_h[size_t(b - lo)];
     db.rows = uint32_t(bh.rows_count);
     db.bitmap_out = st->d_bitmap + st->block_word_off[size_t(b - lo)];
+    db.hits_out = nullptr;  // filled after d_block_hits is allocated
 
     for (int l = 0; l < nleaves; l++) {
       DevLeafBlock& lb = lbs_h[size_t(b - lo) * size_t(nleaves) + size_t(l)];
@@ -990,7 +1438,10 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
           case kScanPhraseStr:
           case kScanEqStr:
           case kScanRegexStr:
-          case kScanRangeStr: {
+          case kScanRangeStr:
+          case kScanPrefixStr:
+          case kScanExactPrefixStr:
+          case kScanSeqStr: {
             const StagedStrCol& sc = ctx.cols.at(leaf_infos[size_t(l)].cname);
             st->algo_bytes += sc.data_bytes + (bh.rows_count + 1) * 4;
             break;
@@ -1024,6 +1475,10 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
   HIP_CHECK(hipMalloc(&st->d_lbs, lbs_h.size() * sizeof(DevLeafBlock)));
   HIP_CHECK(hipMemcpy(st->d_lbs, lbs_h.data(), lbs_h.size() * sizeof(DevLeafBlock),
                       hipMemcpyHostToDevice));
+  HIP_CHECK(hipMalloc(&st->d_block_hits, size_t(nblocks) * 8));
+  for (long b = 0; b < nblocks; b++) {
+    blocks_h[size_t(b)].hits_out = st->d_block_hits + b;
+  }
   HIP_CHECK(hipMalloc(&st->d_blocks, blocks_h.size() * sizeof(DevBlock)));
   HIP_CHECK(hipMemcpy(st->d_blocks, blocks_h.data(),
                       blocks_h.size() * sizeof(DevBlock), hipMemcpyHostToDevice));
@@ -1038,6 +1493,8 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
 long long run_scan(Stage* st) {
   HIP_CHECK(hipSetDevice(st->device));
   HIP_CHECK(hipMemsetAsync(st->d_hits, 0, 8, st->stream));
+  HIP_CHECK(hipMemsetAsync(st->d_block_hits, 0, size_t(st->hi - st->lo) * 8,
+                           st->stream));
   HIP_CHECK(hipEventRecord(st->ev0, st->stream));
   HIP_CHECK(vql_launch_scan(st->d_ops, int(st->filter->ops.size()), st->d_lbs,
                             int(st->filter->leaves.size()), st->d_blocks,
@@ -1129,15 +1586,40 @@ int vql_fetch_bitmaps(void* s, unsigned long long* out_words, long long cap) {
   }
 }
 
+// Per-block matched-row counts of the last vql_scan_staged -- the
+// blockResult rowsLen popcount (block_result.go:403-413) and the
+// `| stats count()` fast path.  Returns 0 or -1.
+int vql_fetch_block_hits(void* s, unsigned long long* out, long long cap) {
+  try {
+    Stage* st = (Stage*)s;
+    long n = long(st->hi - st->lo);
+    if (cap < n) fail("vql_fetch_block_hits: buffer too small");
+    HIP_CHECK(hipSetDevice(st->device));
+    HIP_CHECK(hipMemcpy(out, st->d_block_hits, size_t(n) * 8,
+                        hipMemcpyDeviceToHost));
+    return 0;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
 // Cold path (SURVEY.md §8b vql_scan_batch): stage blocks [lo,hi), scan once,
 // fetch bitmaps, free.  Returns matched rows or -1.
 long long vql_scan_batch(void* part, void* filter, long lo, long hi,
-                         unsigned long long* out_words, long long cap) {
+                         unsigned long long* out_words, long long cap,
+                         unsigned long long* out_popcounts) {
   void* st = vql_stage(part, filter, 0, lo, hi);
   if (!st) return -1;
   long long hits = vql_scan_staged(st);
   if (hits >= 0 && out_words) {
     if (vql_fetch_bitmaps(st, out_words, cap) != 0) hits = -1;
+  }
+  if (hits >= 0 && out_popcounts) {
+    long n = hi < 0 ? vql_part_blocks(part) : hi;
+    if (vql_fetch_block_hits(st, out_popcounts, n - (lo < 0 ? 0 : lo)) != 0) {
+      hits = -1;
+    }
   }
   vql_stage_free(st);
   return hits;
