@@ -69,6 +69,16 @@ int vql_fetch_bitmaps(void* stage, unsigned long long* out_words,
 int vql_fetch_block_hits(void* stage, unsigned long long* out,
                          long long cap_blocks);
 
+/* blockResult materialization: after a scan, gathers
+ * the matched rows' values of `field` (decoded string form, like
+ * blockResult.getValues, block_result.go:306-478) into packed bytes +
+ * per-row byte offsets (nrows+1) + optional global row ids. */
+int vql_gather_sizes(void* stage, const char* field, unsigned long long* nrows,
+                     unsigned long long* nbytes);
+long long vql_gather(void* stage, const char* field, unsigned char* out_bytes,
+                     long long bytes_cap, unsigned long long* out_offs,
+                     long long offs_cap, unsigned long long* out_rowids);
+
 /* Cold path (§8b vql_scan_batch): stage + scan + fetch + free in one call. */
 long long vql_scan_batch(void* part, void* filter, long block_lo, long block_hi,
                          unsigned long long* out_words, long long cap_words,

@@ -76,6 +76,15 @@ def load_product() -> ctypes.CDLL:
     lib.vql_fetch_block_hits.argtypes = [ctypes.c_void_p,
                                          ctypes.POINTER(ctypes.c_ulonglong),
                                          ctypes.c_longlong]
+    lib.vql_gather_sizes.restype = ctypes.c_int
+    lib.vql_gather_sizes.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
+                                     ctypes.POINTER(ctypes.c_ulonglong),
+                                     ctypes.POINTER(ctypes.c_ulonglong)]
+    lib.vql_gather.restype = ctypes.c_longlong
+    lib.vql_gather.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
+                               ctypes.POINTER(ctypes.c_ubyte), ctypes.c_longlong,
+                               ctypes.POINTER(ctypes.c_ulonglong), ctypes.c_longlong,
+                               ctypes.POINTER(ctypes.c_ulonglong)]
     _product = lib
     return lib
 
@@ -288,6 +297,27 @@ class Stage:
         if self.lib.vql_fetch_bitmaps(self.h, buf, nwords) != 0:
             raise RuntimeError(self.lib.vql_errstr().decode())
         return bytes(buf)[: nwords * 8]
+
+    def gather(self, field, with_rowids=True):
+        """Gathers matched rows' decoded values of `field` (blockResult
+        materialization).  Returns (values: list[bytes], rowids)."""
+        nrows = ctypes.c_ulonglong()
+        nbytes = ctypes.c_ulonglong()
+        if self.lib.vql_gather_sizes(self.h, field.encode(), ctypes.byref(nrows),
+                                     ctypes.byref(nbytes)) != 0:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+        n, b = nrows.value, nbytes.value
+        bytes_buf = (ctypes.c_ubyte * max(b, 1))()
+        offs_buf = (ctypes.c_ulonglong * (n + 1))()
+        rowids_buf = (ctypes.c_ulonglong * max(n, 1))() if with_rowids else None
+        r = self.lib.vql_gather(self.h, field.encode(), bytes_buf, b, offs_buf,
+                                n + 1, rowids_buf)
+        if r < 0:
+            raise RuntimeError(self.lib.vql_errstr().decode())
+        raw = bytes(bytes_buf)[:b]
+        values = [raw[offs_buf[i]:offs_buf[i + 1]] for i in range(n)]
+        rowids = list(rowids_buf)[:n] if with_rowids else None
+        return values, rowids
 
     def fetch_block_hits(self, nblocks):
         """Per-block matched-row counts (`| stats count()` fast path)."""

@@ -1162,6 +1162,253 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
   }
 }
 
+
+// ---- gather kernels (blockResult materialization, §8f row 1) ----
+
+__device__ __forceinline__ int d_u64_declen(uint64_t v) {
+  int n = 1;
+  while (v >= 10) {
+    v /= 10;
+    n++;
+  }
+  return n;
+}
+
+// value byte length for one row (pass 1) -- must agree with d_gather_write
+__device__ uint32_t d_gather_len(const DevGatherCol& gc, uint32_t row) {
+  switch (gc.src) {
+    case kGatherStr:
+      return gc.offsets[row + 1] - gc.offsets[row];
+    case kGatherConst:
+      return gc.cval_len;
+    case kGatherDict: {
+      uint8_t code = gc.data[row];
+      return gc.dict_offs[code + 1] - gc.dict_offs[code];
+    }
+    case kGatherFmtU: {
+      const uint8_t* p = gc.data + size_t(row) * gc.width;
+      uint64_t v;
+      switch (gc.width) {
+        case 1: v = p[0]; break;
+        case 2: v = d_get_u16be(p); break;
+        case 4: v = d_get_u32be(p); break;
+        default: v = d_get_u64be(p); break;
+      }
+      return uint32_t(d_u64_declen(v));
+    }
+    case kGatherFmtI: {
+      uint64_t u = d_get_u64be(gc.data + size_t(row) * 8);
+      int64_t v = int64_t(u >> 1) ^ (int64_t(u << 63) >> 63);
+      if (v < 0) return uint32_t(1 + d_u64_declen(~uint64_t(v) + 1));
+      return uint32_t(d_u64_declen(uint64_t(v)));
+    }
+    case kGatherFmtF: {
+      char buf[344];
+      return uint32_t(vl_ryu::format_f64(buf, d_get_u64be(gc.data + size_t(row) * 8)));
+    }
+    case kGatherFmtIp: {
+      uint32_t ip = d_get_u32be(gc.data + size_t(row) * 4);
+      return uint32_t(d_u64_declen((ip >> 24) & 255) + d_u64_declen((ip >> 16) & 255) +
+                      d_u64_declen((ip >> 8) & 255) + d_u64_declen(ip & 255) + 3);
+    }
+    case kGatherFmtIso:
+      return 24;
+    default:  // kGatherMissing
+      return 0;
+  }
+}
+
+__device__ uint32_t d_gather_write(const DevGatherCol& gc, uint32_t row,
+                                   uint8_t* dst) {
+  switch (gc.src) {
+    case kGatherStr: {
+      uint32_t off = gc.offsets[row];
+      uint32_t len = gc.offsets[row + 1] - off;
+      for (uint32_t i = 0; i < len; i++) dst[i] = gc.data[off + i];
+      return len;
+    }
+    case kGatherConst: {
+      for (uint32_t i = 0; i < gc.cval_len; i++) dst[i] = gc.cval[i];
+      return gc.cval_len;
+    }
+    case kGatherDict: {
+      uint8_t code = gc.data[row];
+      uint32_t off = gc.dict_offs[code];
+      uint32_t len = gc.dict_offs[code + 1] - off;
+      for (uint32_t i = 0; i < len; i++) dst[i] = gc.dict_data[off + i];
+      return len;
+    }
+    case kGatherFmtU: {
+      const uint8_t* p = gc.data + size_t(row) * gc.width;
+      uint64_t v;
+      switch (gc.width) {
+        case 1: v = p[0]; break;
+        case 2: v = d_get_u16be(p); break;
+        case 4: v = d_get_u32be(p); break;
+        default: v = d_get_u64be(p); break;
+      }
+      return uint32_t(d_format_u64((char*)dst, v));
+    }
+    case kGatherFmtI: {
+      uint64_t u = d_get_u64be(gc.data + size_t(row) * 8);
+      return uint32_t(
+          d_format_i64((char*)dst, int64_t(u >> 1) ^ (int64_t(u << 63) >> 63)));
+    }
+    case kGatherFmtF:
+      return uint32_t(
+          vl_ryu::format_f64((char*)dst, d_get_u64be(gc.data + size_t(row) * 8)));
+    case kGatherFmtIp:
+      return uint32_t(d_format_ipv4((char*)dst, d_get_u32be(gc.data + size_t(row) * 4)));
+    case kGatherFmtIso:
+      return uint32_t(d_format_iso8601(
+          (char*)dst, int64_t(d_get_u64be(gc.data + size_t(row) * 8))));
+    default:
+      return 0;
+  }
+}
+
+__global__ __launch_bounds__(256) void gather_count_kernel(
+    const DevGatherCol* __restrict__ gcols, const DevBlock* __restrict__ blocks,
+    const DevChunk* __restrict__ chunks, DevChunkCount* __restrict__ out) {
+  __shared__ unsigned long long byte_sums[4];
+  __shared__ uint32_t row_sums[4];
+  const DevChunk ck = chunks[blockIdx.x];
+  const DevBlock blk = blocks[ck.block];
+  const DevGatherCol gc = gcols[ck.block];
+  const uint32_t r0 = ck.chunk * kChunkRows;
+  const uint32_t r1 = min(blk.rows, r0 + kChunkRows);
+  const uint32_t nwords = (r1 - r0 + 63) / 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int nwaves = blockDim.x >> 6;
+
+  unsigned long long my_bytes = 0;
+  uint32_t my_rows = 0;
+  for (uint32_t w = wave; w < nwords; w += nwaves) {
+    const uint64_t word = blk.bitmap_out[r0 / 64 + w];
+    const uint32_t row = r0 + w * 64 + lane;
+    unsigned long long len = 0;
+    if ((word >> lane) & 1) {
+      len = d_gather_len(gc, row);
+      my_rows++;
+    }
+    my_bytes += len;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    my_bytes += __shfl_down(my_bytes, off, 64);
+    my_rows += __shfl_down(my_rows, off, 64);
+  }
+  if (lane == 0) {
+    byte_sums[wave] = my_bytes;
+    row_sums[wave] = my_rows;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    unsigned long long b = 0;
+    uint32_t r = 0;
+    for (int k = 0; k < nwaves; k++) {
+      b += byte_sums[k];
+      r += row_sums[k];
+    }
+    out[blockIdx.x].rows = r;
+    out[blockIdx.x].bytes = b;
+  }
+}
+
+__global__ __launch_bounds__(256) void gather_copy_kernel(
+    const DevGatherCol* __restrict__ gcols, const DevBlock* __restrict__ blocks,
+    const DevChunk* __restrict__ chunks, const DevChunkBase* __restrict__ bases,
+    uint8_t* __restrict__ out_bytes, unsigned long long* __restrict__ out_offs,
+    unsigned long long* __restrict__ out_rowids) {
+  // per-word (row, byte) bases within the chunk, computed by thread 0
+  __shared__ uint32_t word_rows[kChunkWords];
+  __shared__ unsigned long long word_bytes[kChunkWords];
+  const DevChunk ck = chunks[blockIdx.x];
+  const DevBlock blk = blocks[ck.block];
+  const DevGatherCol gc = gcols[ck.block];
+  const DevChunkBase base = bases[blockIdx.x];
+  const uint32_t r0 = ck.chunk * kChunkRows;
+  const uint32_t r1 = min(blk.rows, r0 + kChunkRows);
+  const uint32_t nwords = (r1 - r0 + 63) / 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int nwaves = blockDim.x >> 6;
+
+  // per-word totals
+  for (uint32_t w = wave; w < nwords; w += nwaves) {
+    const uint64_t word = blk.bitmap_out[r0 / 64 + w];
+    const uint32_t row = r0 + w * 64 + lane;
+    unsigned long long len = ((word >> lane) & 1) ? d_gather_len(gc, row) : 0;
+    unsigned long long sum = len;
+    for (int off = 32; off > 0; off >>= 1) sum += __shfl_down(sum, off, 64);
+    if (lane == 0) {
+      word_rows[w] = uint32_t(__popcll(word));
+      word_bytes[w] = sum;
+    }
+  }
+  __syncthreads();
+  if (tid == 0) {
+    // exclusive scan over the chunk's words
+    uint32_t racc = 0;
+    unsigned long long bacc = 0;
+    for (uint32_t w = 0; w < nwords; w++) {
+      uint32_t r = word_rows[w];
+      unsigned long long b = word_bytes[w];
+      word_rows[w] = racc;
+      word_bytes[w] = bacc;
+      racc += r;
+      bacc += b;
+    }
+  }
+  __syncthreads();
+
+  for (uint32_t w = wave; w < nwords; w += nwaves) {
+    const uint64_t word = blk.bitmap_out[r0 / 64 + w];
+    const uint32_t row = r0 + w * 64 + lane;
+    const bool mine = (word >> lane) & 1;
+    unsigned long long len = mine ? d_gather_len(gc, row) : 0;
+    // inclusive scan of lengths over the wave
+    unsigned long long incl = len;
+    for (int off = 1; off < 64; off <<= 1) {
+      unsigned long long up = __shfl_up(incl, off, 64);
+      if (lane >= off) incl += up;
+    }
+    if (mine) {
+      const unsigned long long row_out =
+          base.row_base + word_rows[w] +
+          (unsigned long long)__popcll(word & ((uint64_t(1) << lane) - 1));
+      const unsigned long long byte_out = base.byte_base + word_bytes[w] + incl - len;
+      d_gather_write(gc, row, out_bytes + byte_out);
+      out_offs[row_out] = byte_out;
+      if (out_rowids) out_rowids[row_out] = base.gid_base + row;
+    }
+  }
+}
+
+extern "C" hipError_t vql_launch_gather_count(const DevGatherCol* gcols,
+                                              const DevBlock* blocks,
+                                              const DevChunk* chunks,
+                                              uint32_t nchunks, DevChunkCount* out,
+                                              hipStream_t stream) {
+  if (nchunks == 0) return hipSuccess;
+  hipLaunchKernelGGL(gather_count_kernel, dim3(nchunks), dim3(256), 0, stream,
+                     gcols, blocks, chunks, out);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t vql_launch_gather_copy(
+    const DevGatherCol* gcols, const DevBlock* blocks, const DevChunk* chunks,
+    uint32_t nchunks, const DevChunkBase* bases, uint8_t* out_bytes,
+    unsigned long long* out_offs, unsigned long long* out_rowids,
+    hipStream_t stream) {
+  if (nchunks == 0) return hipSuccess;
+  hipLaunchKernelGGL(gather_copy_kernel, dim3(nchunks), dim3(256), 0, stream,
+                     gcols, blocks, chunks, bases, out_bytes, out_offs, out_rowids);
+  return hipGetLastError();
+}
+
 // host-side launcher (called from vql_api.cpp)
 extern "C" hipError_t vql_launch_scan(const DevOp* ops, int nops,
                                       const DevLeafBlock* lbs, int nleaves,

@@ -122,4 +122,46 @@ enum : uint8_t {
   kReAlways = 128,  // pattern matches the empty string => always true
 };
 
+// ---- gather (blockResult materialization) ----
+// Compacts the matched rows' values of one column into packed bytes +
+// offsets + global row ids, mirroring blockResult.getValues semantics
+// (block_result.go:306-478): string columns return raw bytes, fixed-width
+// columns return their decoded string forms, dict columns the dict string.
+
+enum GatherSrc : uint8_t {
+  kGatherStr = 0,    // data+offsets
+  kGatherConst = 1,  // const value (operand ptr, const_len)
+  kGatherDict = 2,   // 1-byte codes + dict table (dict_data/dict_offs)
+  kGatherFmtU = 3,   // BE uint width w -> decimal
+  kGatherFmtI = 4,   // BE zig-zag i64 -> decimal
+  kGatherFmtF = 5,   // BE f64 bits -> Ryu 'f'
+  kGatherFmtIp = 6,  // BE u32 -> dotted quad
+  kGatherFmtIso = 7, // BE u64 -> iso8601
+  kGatherMissing = 8,  // column absent in the block -> empty values
+};
+
+struct DevGatherCol {
+  const uint8_t* data;
+  const uint32_t* offsets;
+  const uint8_t* dict_data;   // concatenated dict strings
+  const uint32_t* dict_offs;  // 9 entries
+  const uint8_t* cval;        // const value bytes
+  uint32_t cval_len;
+  uint8_t src;
+  uint8_t width;
+  uint8_t pad0, pad1;
+};
+
+struct DevChunkCount {
+  uint32_t rows;
+  uint32_t pad;
+  unsigned long long bytes;
+};
+
+struct DevChunkBase {
+  unsigned long long row_base;   // output row index of the chunk's first match
+  unsigned long long byte_base;  // output byte offset
+  unsigned long long gid_base;   // global row id of the chunk's row 0
+};
+
 }  // namespace vl

@@ -33,6 +33,14 @@ namespace vl {
 extern "C" hipError_t vql_launch_scan(const DevOp*, int, const DevLeafBlock*, int,
                                       const DevBlock*, const DevChunk*, uint32_t,
                                       unsigned long long*, hipStream_t);
+extern "C" hipError_t vql_launch_gather_count(const DevGatherCol*, const DevBlock*,
+                                              const DevChunk*, uint32_t,
+                                              DevChunkCount*, hipStream_t);
+extern "C" hipError_t vql_launch_gather_copy(const DevGatherCol*, const DevBlock*,
+                                             const DevChunk*, uint32_t,
+                                             const DevChunkBase*, uint8_t*,
+                                             unsigned long long*,
+                                             unsigned long long*, hipStream_t);
 }
 
 using namespace vl;
@@ -212,7 +220,22 @@ struct Stage {
   hipEvent_t ev0 = nullptr, ev1 = nullptr;
   double last_kernel_ms = 0;
 
+  // gather caches (blockResult materialization, §8f row 1)
+  struct GatherCtx {
+    DevGatherCol* d_gcols = nullptr;
+    DevChunkBase* d_bases = nullptr;
+    uint64_t nrows = 0, nbytes = 0;
+    bool sized = false;
+  };
+  std::map<std::string, GatherCtx> gathers;
+  std::vector<DevChunk> chunks_h;          // kept for gather passes
+  std::vector<uint32_t> block_rows_h;      // rows per staged block
+
   ~Stage() {
+    for (auto& kv : gathers) {
+      if (kv.second.d_gcols) hipFree(kv.second.d_gcols);
+      if (kv.second.d_bases) hipFree(kv.second.d_bases);
+    }
     if (ev0) hipEventDestroy(ev0);
     if (ev1) hipEventDestroy(ev1);
     if (stream) hipStreamDestroy(stream);
@@ -1468,6 +1491,10 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
   st->algo_bytes += st->bitmap_words * 8;
 
   st->nchunks = uint32_t(chunks_h.size());
+  st->chunks_h = chunks_h;
+  for (long b = lo; b < hi; b++) {
+    st->block_rows_h.push_back(uint32_t(part->bhs[size_t(b)].rows_count));
+  }
 
   HIP_CHECK(hipMalloc(&st->d_ops, filter->ops.size() * sizeof(DevOp)));
   HIP_CHECK(hipMemcpy(st->d_ops, filter->ops.data(),
@@ -1623,6 +1650,187 @@ long long vql_scan_batch(void* part, void* filter, long lo, long hi,
   }
   vql_stage_free(st);
   return hits;
+}
+
+}  // extern "C"
+
+namespace {
+
+// Builds (and caches) the per-block gather descriptors for `field` and runs
+// the count pass.  Mirrors blockResult.getValues source selection
+// (block_result.go:306-478): const columns repeat the const value, dict
+// columns look up the dict string, fixed-width columns decode to strings.
+Stage::GatherCtx& gather_prepare(Stage* st, const std::string& field) {
+  auto it = st->gathers.find(field);
+  if (it != st->gathers.end() && it->second.sized) return it->second;
+
+  HIP_CHECK(hipSetDevice(st->device));
+  Stage::GatherCtx& g = st->gathers[field];
+  std::string cname = canonical_field(field);
+  const long nblocks = st->hi - st->lo;
+
+  std::vector<DevGatherCol> gcols((size_t(nblocks)));
+  for (long b = 0; b < nblocks; b++) {
+    const BlockHeader& bh = st->part->bhs[size_t(st->lo + b)];
+    PartReader::BlockColumns bc;
+    st->part->pr.read_block_columns(bh, bc);
+    DevGatherCol& gc = gcols[size_t(b)];
+    memset(&gc, 0, sizeof(gc));
+
+    std::string cv;
+    if (st->part->pr.get_const_column(bc, cname, &cv)) {
+      gc.src = kGatherConst;
+      gc.cval_len = uint32_t(cv.size());
+      gc.cval = st->push(cv.data(), cv.size(), 8);
+      continue;
+    }
+    ColumnHeader ch;
+    if (!st->part->pr.get_column_header(bc, cname, &ch)) {
+      gc.src = kGatherMissing;
+      continue;
+    }
+    StringsBlockDec dec;
+    st->part->pr.read_values(ch, bh.rows_count, dec);
+    if (dec.is_const && ch.type == ValueType::String) {
+      gc.src = kGatherConst;
+      gc.cval_len = uint32_t(dec.data.size());
+      gc.cval = st->push(dec.data.data(), dec.data.size(), 8);
+      continue;
+    }
+    // non-const data payload
+    const uint8_t* d_data = st->push(dec.data.data(), dec.data.size(), 16);
+    st->reserve(16);
+    gc.data = d_data;
+    switch (ch.type) {
+      case ValueType::String: {
+        gc.src = kGatherStr;
+        gc.offsets = (const uint32_t*)st->push(dec.offsets.data(),
+                                               dec.offsets.size() * 4, 4);
+        break;
+      }
+      case ValueType::Dict: {
+        gc.src = kGatherDict;
+        bytes cat;
+        std::vector<uint32_t> doffs;
+        doffs.push_back(0);
+        for (const auto& dv : ch.dict) {
+          cat.insert(cat.end(), dv.begin(), dv.end());
+          doffs.push_back(uint32_t(cat.size()));
+        }
+        while (doffs.size() < 9) doffs.push_back(doffs.back());
+        gc.dict_data = st->push(cat.data(), cat.size(), 8);
+        gc.dict_offs = (const uint32_t*)st->push(doffs.data(), doffs.size() * 4, 4);
+        break;
+      }
+      case ValueType::Uint8: gc.src = kGatherFmtU; gc.width = 1; break;
+      case ValueType::Uint16: gc.src = kGatherFmtU; gc.width = 2; break;
+      case ValueType::Uint32: gc.src = kGatherFmtU; gc.width = 4; break;
+      case ValueType::Uint64: gc.src = kGatherFmtU; gc.width = 8; break;
+      case ValueType::Int64: gc.src = kGatherFmtI; gc.width = 8; break;
+      case ValueType::Float64: gc.src = kGatherFmtF; gc.width = 8; break;
+      case ValueType::IPv4: gc.src = kGatherFmtIp; gc.width = 4; break;
+      case ValueType::TimestampISO8601: gc.src = kGatherFmtIso; gc.width = 8; break;
+      default:
+        fail("gather: unknown valueType");
+    }
+  }
+  HIP_CHECK(hipMalloc(&g.d_gcols, gcols.size() * sizeof(DevGatherCol)));
+  HIP_CHECK(hipMemcpy(g.d_gcols, gcols.data(), gcols.size() * sizeof(DevGatherCol),
+                      hipMemcpyHostToDevice));
+
+  // count pass
+  DevChunkCount* d_counts = nullptr;
+  HIP_CHECK(hipMalloc(&d_counts, st->nchunks * sizeof(DevChunkCount)));
+  HIP_CHECK(vql_launch_gather_count(g.d_gcols, st->d_blocks, st->d_chunks,
+                                    st->nchunks, d_counts, st->stream));
+  std::vector<DevChunkCount> counts(st->nchunks);
+  HIP_CHECK(hipMemcpyAsync(counts.data(), d_counts,
+                           st->nchunks * sizeof(DevChunkCount),
+                           hipMemcpyDeviceToHost, st->stream));
+  HIP_CHECK(hipStreamSynchronize(st->stream));
+  HIP_CHECK(hipFree(d_counts));
+
+  // exclusive scan -> per-chunk bases; gid_base = first global row of block
+  std::vector<unsigned long long> block_row_base(size_t(st->hi - st->lo) + 1, 0);
+  for (size_t b = 0; b < st->block_rows_h.size(); b++) {
+    block_row_base[b + 1] = block_row_base[b] + st->block_rows_h[b];
+  }
+  std::vector<DevChunkBase> bases(st->nchunks);
+  unsigned long long racc = 0, bacc = 0;
+  for (uint32_t c = 0; c < st->nchunks; c++) {
+    bases[c].row_base = racc;
+    bases[c].byte_base = bacc;
+    bases[c].gid_base = block_row_base[st->chunks_h[c].block];
+    racc += counts[c].rows;
+    bacc += counts[c].bytes;
+  }
+  g.nrows = racc;
+  g.nbytes = bacc;
+  HIP_CHECK(hipMalloc(&g.d_bases, bases.size() * sizeof(DevChunkBase)));
+  HIP_CHECK(hipMemcpy(g.d_bases, bases.data(), bases.size() * sizeof(DevChunkBase),
+                      hipMemcpyHostToDevice));
+  g.sized = true;
+  return g;
+}
+
+}  // namespace
+
+extern "C" {
+
+// Sizes of the gather output for `field` over the CURRENT bitmaps (run a scan
+// first).  Returns 0 or -1.
+int vql_gather_sizes(void* s, const char* field, unsigned long long* nrows,
+                     unsigned long long* nbytes) {
+  try {
+    Stage* st = (Stage*)s;
+    Stage::GatherCtx& g = gather_prepare(st, field);
+    *nrows = g.nrows;
+    *nbytes = g.nbytes;
+    return 0;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
+// Gathers the matched rows' values of `field` into packed bytes (out_bytes),
+// per-row byte offsets (out_offs, nrows+1 entries) and optional global row
+// ids (out_rowids, nrows entries).  Returns the matched-row count or -1.
+long long vql_gather(void* s, const char* field, unsigned char* out_bytes,
+                     long long bytes_cap, unsigned long long* out_offs,
+                     long long offs_cap, unsigned long long* out_rowids) {
+  try {
+    Stage* st = (Stage*)s;
+    Stage::GatherCtx& g = gather_prepare(st, field);
+    if ((long long)g.nbytes > bytes_cap) fail("vql_gather: bytes buffer too small");
+    if ((long long)(g.nrows + 1) > offs_cap) fail("vql_gather: offsets buffer too small");
+    HIP_CHECK(hipSetDevice(st->device));
+    uint8_t* d_bytes = nullptr;
+    unsigned long long* d_offs = nullptr;
+    unsigned long long* d_rowids = nullptr;
+    HIP_CHECK(hipMalloc(&d_bytes, g.nbytes ? g.nbytes : 8));
+    HIP_CHECK(hipMalloc(&d_offs, (g.nrows + 1) * 8));
+    if (out_rowids) HIP_CHECK(hipMalloc(&d_rowids, g.nrows ? g.nrows * 8 : 8));
+    HIP_CHECK(vql_launch_gather_copy(g.d_gcols, st->d_blocks, st->d_chunks,
+                                     st->nchunks, g.d_bases, d_bytes, d_offs,
+                                     d_rowids, st->stream));
+    HIP_CHECK(hipStreamSynchronize(st->stream));
+    if (g.nbytes) {
+      HIP_CHECK(hipMemcpy(out_bytes, d_bytes, g.nbytes, hipMemcpyDeviceToHost));
+    }
+    HIP_CHECK(hipMemcpy(out_offs, d_offs, g.nrows * 8, hipMemcpyDeviceToHost));
+    out_offs[g.nrows] = g.nbytes;
+    if (out_rowids && g.nrows) {
+      HIP_CHECK(hipMemcpy(out_rowids, d_rowids, g.nrows * 8, hipMemcpyDeviceToHost));
+    }
+    hipFree(d_bytes);
+    hipFree(d_offs);
+    if (d_rowids) hipFree(d_rowids);
+    return (long long)g.nrows;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
 }
 
 }  // extern "C"
