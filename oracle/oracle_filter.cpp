@@ -2,6 +2,7 @@
 // reference function it restates.
 #include "oracle_filter.h"
 
+#include <algorithm>
 #include <cmath>
 #include <cstring>
 
@@ -192,6 +193,33 @@ static std::string to_uint_string(strview v, int width) {
   }
   std::string s;
   format_uint64(s, n);
+  return s;
+}
+
+// decoded string form of a fixed-width encoded value (to*String helpers)
+static std::string format_value(ValueType t, strview v) {
+  std::string s;
+  const uint8_t* p = (const uint8_t*)v.p;
+  switch (t) {
+    case ValueType::Uint8: format_uint64(s, p[0]); break;
+    case ValueType::Uint16: format_uint64(s, get_u16be(p)); break;
+    case ValueType::Uint32: format_uint64(s, get_u32be(p)); break;
+    case ValueType::Uint64: format_uint64(s, get_u64be(p)); break;
+    case ValueType::Int64: format_int64(s, get_i64be_zigzag(p)); break;
+    case ValueType::Float64: {
+      uint64_t u = get_u64be(p);
+      double d;
+      memcpy(&d, &u, 8);
+      format_float64(s, d);
+      break;
+    }
+    case ValueType::IPv4: format_ipv4(s, get_u32be(p)); break;
+    case ValueType::TimestampISO8601:
+      format_timestamp_iso8601(s, int64_t(get_u64be(p)));
+      break;
+    default:
+      fail("format_value: unexpected type");
+  }
   return s;
 }
 
@@ -991,6 +1019,574 @@ static void apply_sequence(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
   }
 }
 
+
+// ---- filterIn (filter_in.go:120-234) ----
+
+static bool in_set(const std::vector<std::string>& sorted_set, strview v) {
+  // binary search over the sorted value set
+  size_t lo = 0, hi = sorted_set.size();
+  while (lo < hi) {
+    size_t mid = (lo + hi) / 2;
+    const std::string& m = sorted_set[mid];
+    int c = memcmp(m.data(), v.p, std::min(m.size(), v.n));
+    if (c == 0) c = m.size() < v.n ? -1 : (m.size() > v.n ? 1 : 0);
+    if (c == 0) return true;
+    if (c < 0) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  return false;
+}
+
+static int bin_set_slot(ValueType t) {
+  switch (t) {
+    case ValueType::Uint8: return 0;
+    case ValueType::Uint16: return 1;
+    case ValueType::Uint32: return 2;
+    case ValueType::Uint64: return 3;
+    case ValueType::Int64: return 4;
+    case ValueType::Float64: return 5;
+    case ValueType::IPv4: return 6;
+    case ValueType::TimestampISO8601: return 7;
+    default: return -1;
+  }
+}
+
+// matchBloomFilterAnyTokenSet (filter_in.go:202-218)
+static bool match_bloom_any_token_set(const FilterNode& f, BlockCtx& ctx,
+                                      const ColumnHeader& ch) {
+  if (!match_bloom_all(ctx, ch, f.common_hashes)) return false;
+  if (f.set_hashes.size() > 1000 ||
+      f.set_hashes.size() > 10 * ctx.bh->rows_count) {
+    return true;
+  }
+  const auto& words = ctx.bloom(ch);
+  for (const auto& hs : f.set_hashes) {
+    if (bloom_contains_all(words.data(), words.size(), hs.data(), hs.size())) {
+      return true;
+    }
+  }
+  return false;
+}
+
+static void apply_in(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  if (f.values.empty()) {
+    bm.reset_bits();
+    return;
+  }
+  auto has = [&](strview v) {
+    for (const auto& s2 : f.values) {
+      if (strview(s2) == v) return true;
+    }
+    return false;
+  };
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!has(strview(cv))) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!has(strview("", 0))) bm.reset_bits();
+    return;
+  }
+  if (ch.type == ValueType::Dict) {
+    std::vector<uint8_t> enc;
+    for (const auto& dv : ch.dict) enc.push_back(has(strview(dv)) ? 1 : 0);
+    match_encoded_dict(ctx, ch, bm, enc);
+    return;
+  }
+  // matchAnyValue (filter_in.go:187-200)
+  const std::vector<std::string>* set;
+  std::vector<std::string> str_sorted;
+  if (ch.type == ValueType::String) {
+    str_sorted = f.values;
+    std::sort(str_sorted.begin(), str_sorted.end());
+    str_sorted.erase(std::unique(str_sorted.begin(), str_sorted.end()),
+                     str_sorted.end());
+    set = &str_sorted;
+  } else {
+    set = &f.bin_sets[size_t(bin_set_slot(ch.type))];
+  }
+  if (set->empty()) {
+    bm.reset_bits();
+    return;
+  }
+  if (!match_bloom_any_token_set(f, ctx, ch)) {
+    bm.reset_bits();
+    return;
+  }
+  visit_values(ctx, ch, bm, [&](strview v) { return in_set(*set, v); });
+}
+
+// ---- filterContainsAny (filter_contains_any.go:105-296) ----
+
+static bool match_any_phrase(strview v, const std::vector<std::string>& phrases) {
+  for (const auto& ph : phrases) {
+    if (match_phrase(v, strview(ph))) return true;
+  }
+  return false;
+}
+
+static void apply_contains_any(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  if (f.values.empty()) {
+    bm.reset_bits();
+    return;
+  }
+  for (const auto& v : f.values) {
+    if (v.empty()) return;  // empty value matches everything (:110-113)
+  }
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_any_phrase(strview(cv), f.values)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!match_any_phrase(strview("", 0), f.values)) bm.reset_bits();
+    return;
+  }
+  if (ch.type == ValueType::Dict) {
+    std::vector<uint8_t> enc;
+    for (const auto& dv : ch.dict) {
+      enc.push_back(match_any_phrase(strview(dv), f.values) ? 1 : 0);
+    }
+    match_encoded_dict(ctx, ch, bm, enc);
+    return;
+  }
+  if (ch.type == ValueType::Uint8 || ch.type == ValueType::Uint16 ||
+      ch.type == ValueType::Uint32 || ch.type == ValueType::Uint64) {
+    // uint columns use the exact binary sets (filter_contains_any.go:141-152)
+    const auto& set = f.bin_sets[size_t(bin_set_slot(ch.type))];
+    if (set.empty()) {
+      bm.reset_bits();
+      return;
+    }
+    if (!match_bloom_any_token_set(f, ctx, ch)) {
+      bm.reset_bits();
+      return;
+    }
+    visit_values(ctx, ch, bm, [&](strview v) { return in_set(set, v); });
+    return;
+  }
+  // common-token gate + per-phrase token-set survivor filter
+  // (matchValuesAnyPhrase, filter_contains_any.go:179-198)
+  if (!match_bloom_all(ctx, ch, f.common_hashes)) {
+    bm.reset_bits();
+    return;
+  }
+  std::vector<std::string> survivors;
+  {
+    const auto& words = ctx.bloom(ch);
+    for (size_t i = 0; i < f.values.size(); i++) {
+      if (bloom_contains_all(words.data(), words.size(), f.set_hashes[i].data(),
+                             f.set_hashes[i].size())) {
+        survivors.push_back(f.values[i]);
+      }
+    }
+  }
+  if (survivors.empty()) {
+    bm.reset_bits();
+    return;
+  }
+  visit_values(ctx, ch, bm, [&](strview v) {
+    if (ch.type == ValueType::String) return match_any_phrase(v, survivors);
+    std::string str = format_value(ch.type, v);
+    return match_any_phrase(strview(str), survivors);
+  });
+}
+
+// ---- filterContainsAll (filter_contains_all.go:123-321) ----
+
+static bool match_all_phrases(strview v, const std::vector<std::string>& phrases) {
+  for (const auto& ph : phrases) {
+    if (ph.empty()) continue;  // empty phrase matches everything (:310-321)
+    if (!match_phrase(v, strview(ph))) return false;
+  }
+  return true;
+}
+
+static void apply_contains_all(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  bool only_empty = f.values.size() == 1 && f.values[0].empty();
+  if (f.values.empty() || only_empty) return;  // :124-126
+
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_all_phrases(strview(cv), f.values)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!match_all_phrases(strview("", 0), f.values)) bm.reset_bits();
+    return;
+  }
+  if (ch.type == ValueType::Dict) {
+    std::vector<uint8_t> enc;
+    for (const auto& dv : ch.dict) {
+      enc.push_back(match_all_phrases(strview(dv), f.values) ? 1 : 0);
+    }
+    match_encoded_dict(ctx, ch, bm, enc);
+    return;
+  }
+  if (ch.type == ValueType::Uint8 || ch.type == ValueType::Uint16 ||
+      ch.type == ValueType::Uint32 || ch.type == ValueType::Uint64) {
+    // matchAllValues (filter_contains_all.go:183-204)
+    std::vector<std::string> distinct_nonempty;
+    for (const auto& v : f.values) {
+      if (!v.empty() &&
+          std::find(distinct_nonempty.begin(), distinct_nonempty.end(), v) ==
+              distinct_nonempty.end()) {
+        distinct_nonempty.push_back(v);
+      }
+    }
+    if (distinct_nonempty.empty()) return;
+    const auto& set = f.bin_sets[size_t(bin_set_slot(ch.type))];
+    if (distinct_nonempty.size() != 1 || set.size() != 1) {
+      bm.reset_bits();
+      return;
+    }
+    if (!match_bloom_all(ctx, ch, f.all_hashes)) {
+      bm.reset_bits();
+      return;
+    }
+    strview bin(set[0]);
+    visit_values(ctx, ch, bm, [&](strview v) { return v == bin; });
+    return;
+  }
+  if (!match_bloom_all(ctx, ch, f.all_hashes)) {
+    bm.reset_bits();
+    return;
+  }
+  visit_values(ctx, ch, bm, [&](strview v) {
+    if (ch.type == ValueType::String) return match_all_phrases(v, f.values);
+    std::string str = format_value(ch.type, v);
+    return match_all_phrases(strview(str), f.values);
+  });
+}
+
+// ---- filterStringRange (filter_string_range.go:47-230) ----
+
+static bool match_string_range(strview s, const std::string& mn,
+                               const std::string& mx) {
+  // matchStringRange: s >= min && s < max (plain byte order)
+  auto cmp = [](strview a, const std::string& b) {
+    int c = memcmp(a.p, b.data(), std::min(a.n, b.size()));
+    if (c != 0) return c;
+    return a.n < b.size() ? -1 : (a.n > b.size() ? 1 : 0);
+  };
+  return cmp(s, mn) >= 0 && cmp(s, mx) < 0;
+}
+
+static void apply_string_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  if (f.min_s > f.max_s) {
+    bm.reset_bits();
+    return;
+  }
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_string_range(strview(cv), f.min_s, f.max_s)) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!match_string_range(strview("", 0), f.min_s, f.max_s)) bm.reset_bits();
+    return;
+  }
+  // per-type prunes (filter_string_range.go:100-225)
+  switch (ch.type) {
+    case ValueType::String:
+    case ValueType::Dict:
+      break;
+    case ValueType::Int64:
+      if ((f.min_s != "-" && f.min_s > "9") || (f.max_s != "-" && f.max_s < "0")) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::Float64:
+      if (f.min_s > "9" || f.max_s < "+") {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    default:
+      if (f.min_s > "9" || f.max_s < "0") {
+        bm.reset_bits();
+        return;
+      }
+      break;
+  }
+  if (ch.type == ValueType::Dict) {
+    std::vector<uint8_t> enc;
+    for (const auto& dv : ch.dict) {
+      enc.push_back(match_string_range(strview(dv), f.min_s, f.max_s) ? 1 : 0);
+    }
+    match_encoded_dict(ctx, ch, bm, enc);
+    return;
+  }
+  visit_values(ctx, ch, bm, [&](strview v) {
+    if (ch.type == ValueType::String) {
+      return match_string_range(v, f.min_s, f.max_s);
+    }
+    std::string str = format_value(ch.type, v);
+    return match_string_range(strview(str), f.min_s, f.max_s);
+  });
+}
+
+// ---- filterIPv4Range (filter_ipv4_range.go:99-190) ----
+
+static void apply_ipv4_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  uint32_t mn = uint32_t(f.min_u), mx = uint32_t(f.max_u);
+  if (mn > mx) {
+    bm.reset_bits();
+    return;
+  }
+  auto match_str = [&](strview v) {
+    uint32_t ip;
+    if (!try_parse_ipv4(v, &ip)) return false;
+    return ip >= mn && ip <= mx;
+  };
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_str(strview(cv))) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    bm.reset_bits();
+    return;
+  }
+  switch (ch.type) {
+    case ValueType::String:
+      visit_values(ctx, ch, bm, match_str);
+      return;
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) enc.push_back(match_str(strview(dv)) ? 1 : 0);
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::IPv4: {
+      // matchIPv4ByRange (filter_ipv4_range.go:166-181)
+      if (ch.min_value > mx || ch.max_value < mn) {
+        bm.reset_bits();
+        return;
+      }
+      visit_values(ctx, ch, bm, [&](strview v) {
+        uint32_t ip = get_u32be((const uint8_t*)v.p);
+        return ip >= mn && ip <= mx;
+      });
+      return;
+    }
+    default:
+      bm.reset_bits();
+      return;
+  }
+}
+
+// ---- filterLenRange (filter_len_range.go:126-348) ----
+
+static uint64_t rune_count(strview s) {
+  uint64_t n = 0;
+  for (size_t i = 0; i < s.n; i++) {
+    if ((uint8_t(s.p[i]) & 0xC0) != 0x80) n++;
+  }
+  return n;
+}
+
+static void apply_len_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  uint64_t mn = f.min_u, mx = f.max_u;
+  if (mn > mx) {
+    bm.reset_bits();
+    return;
+  }
+  auto match_len = [&](strview v) {
+    uint64_t n = rune_count(v);
+    return n >= mn && n <= mx;
+  };
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!match_len(strview(cv))) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    if (!match_len(strview("", 0))) bm.reset_bits();
+    return;
+  }
+  // per-type prunes (filter_len_range.go:180-331)
+  auto minmax_len_ok = [&]() {
+    std::string s2;
+    format_uint64(s2, ch.min_value);
+    if (mx < s2.size()) return false;
+    s2.clear();
+    format_uint64(s2, ch.max_value);
+    return mn <= s2.size();
+  };
+  switch (ch.type) {
+    case ValueType::String:
+      visit_values(ctx, ch, bm, match_len);
+      return;
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) enc.push_back(match_len(strview(dv)) ? 1 : 0);
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    case ValueType::Uint8:
+      if (mn > 3 || mx == 0 || !minmax_len_ok()) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::Uint16:
+      if (mn > 5 || mx == 0 || !minmax_len_ok()) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::Uint32:
+      if (mn > 10 || mx == 0 || !minmax_len_ok()) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::Uint64:
+      if (mn > 20 || mx == 0 || !minmax_len_ok()) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::Int64:
+      if (mn > 20 || mx == 0) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::Float64:
+      if (mn > 24 || mx == 0) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::IPv4:
+      if (mn > 15 || mx < 7) {
+        bm.reset_bits();
+        return;
+      }
+      break;
+    case ValueType::TimestampISO8601:
+      // matchTimestampISO8601ByLenRange (filter_len_range.go:180-185)
+      if (mn > 24 || mx < 24) bm.reset_bits();
+      return;
+    default:
+      fail("unknown valueType in len_range filter");
+  }
+  visit_values(ctx, ch, bm, [&](strview v) {
+    std::string str = format_value(ch.type, v);
+    return match_len(strview(str));
+  });
+}
+
+// ---- filterDayRange / filterWeekRange (filter_day_range.go:126-139,
+//      filter_week_range.go:128-141) ----
+
+static constexpr int64_t kNsPerDay = 24LL * 3600 * 1000000000;
+
+static int64_t floor_mod(int64_t a, int64_t m) {
+  int64_t r = a % m;
+  return r;  // Go's % (truncated); day offsets of pre-1970 stamps are negative
+}
+
+static void apply_day_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  int64_t start = int64_t(f.min_u), end = int64_t(f.max_u);
+  if (start > end) {
+    bm.reset_bits();
+    return;
+  }
+  if (start == 0 && end == kNsPerDay - 1) return;
+  const auto& ts = ctx.get_timestamps();
+  bm.for_each_set_bit([&](uint64_t idx) {
+    int64_t off = floor_mod(ts[idx] - f.tz_offset, kNsPerDay);
+    return off >= start && off <= end;
+  });
+}
+
+static void apply_week_range(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  int64_t start = int64_t(f.min_u), end = int64_t(f.max_u);
+  if (start > end) {
+    bm.reset_bits();
+    return;
+  }
+  if (start == 0 /*Sunday*/ && end == 6 /*Saturday*/) return;
+  const auto& ts = ctx.get_timestamps();
+  bm.for_each_set_bit([&](uint64_t idx) {
+    // Go time.Weekday: days since epoch + 4 (1970-01-01 = Thursday), floor
+    int64_t t = ts[idx] - f.tz_offset;
+    int64_t days = t / kNsPerDay;
+    if (t % kNsPerDay < 0) days--;
+    int64_t wd = (days + 4) % 7;
+    if (wd < 0) wd += 7;
+    return wd >= start && wd <= end;
+  });
+}
+
+// ---- filterValueType (filter_value_type.go:44-67) ----
+
+static const char* value_type_name(ValueType t) {
+  switch (t) {
+    case ValueType::String: return "string";
+    case ValueType::Dict: return "dict";
+    case ValueType::Uint8: return "uint8";
+    case ValueType::Uint16: return "uint16";
+    case ValueType::Uint32: return "uint32";
+    case ValueType::Uint64: return "uint64";
+    case ValueType::Int64: return "int64";
+    case ValueType::Float64: return "float64";
+    case ValueType::IPv4: return "ipv4";
+    case ValueType::TimestampISO8601: return "iso8601";
+    default: return "unknown";
+  }
+}
+
+static void apply_value_type(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  std::string name = canonical_field(f.field);
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (f.min_s != "const") bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    bm.reset_bits();
+    return;
+  }
+  if (f.min_s != value_type_name(ch.type)) bm.reset_bits();
+}
+
+// ---- filterStreamID (filter_stream_id.go:127-143) ----
+
+static void apply_stream_id(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  if (f.stream_ids.empty()) {
+    bm.reset_bits();
+    return;
+  }
+  const StreamID& sid = ctx.bh->stream_id;
+  uint64_t tp = uint64_t(sid.account_id) << 32 | sid.project_id;
+  for (const auto& id : f.stream_ids) {
+    if (id[0] == tp && id[1] == sid.id_hi && id[2] == sid.id_lo) return;
+  }
+  bm.reset_bits();
+}
+
 // ---- AND/OR bloom prefilters (filter_and.go:76-111, filter_or.go:80-115) ----
 
 static bool and_match_bloom(const FilterNode& f, BlockCtx& ctx) {
@@ -1051,6 +1647,36 @@ void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
       return;
     case FilterNode::Sequence:
       apply_sequence(f, ctx, bm);
+      return;
+    case FilterNode::In:
+      apply_in(f, ctx, bm);
+      return;
+    case FilterNode::ContainsAny:
+      apply_contains_any(f, ctx, bm);
+      return;
+    case FilterNode::ContainsAll:
+      apply_contains_all(f, ctx, bm);
+      return;
+    case FilterNode::StringRange:
+      apply_string_range(f, ctx, bm);
+      return;
+    case FilterNode::IPv4Range:
+      apply_ipv4_range(f, ctx, bm);
+      return;
+    case FilterNode::LenRange:
+      apply_len_range(f, ctx, bm);
+      return;
+    case FilterNode::DayRange:
+      apply_day_range(f, ctx, bm);
+      return;
+    case FilterNode::WeekRange:
+      apply_week_range(f, ctx, bm);
+      return;
+    case FilterNode::ValueTypeFilter:
+      apply_value_type(f, ctx, bm);
+      return;
+    case FilterNode::StreamIdFilter:
+      apply_stream_id(f, ctx, bm);
       return;
     case FilterNode::Time:
       apply_time(f, ctx, bm);

@@ -61,6 +61,10 @@ def typed_part(tmp_path_factory):
                          "3d2h", "100KB", "", "nan?"][i % 10] for i in range(rows)]},
                     {"name": "uni", "values": [
                         ("раз два три" if i % 3 == 0 else "foo bar") for i in range(rows)]},
+                    # stays ValueType string (a few unparseable entries)
+                    {"name": "ipstr", "values": [
+                        ("n/a" if i % 9 == 8 else f"172.16.{i % 256}.{(i * 5) % 256}")
+                        for i in range(rows)]},
                 ],
             },
             {
@@ -121,6 +125,21 @@ FILTERS = [
     '{"type":"or","filters":['
     '{"type":"phrase","field":"dict_0","phrase":"warn"},'
     '{"type":"phrase","field":"dict_1","phrase":"debug"}]}]}',
+    # set/range filter family on the generated part
+    '{"type":"in","field":"dict_0","values":["error","fatal","nosuch"]}',
+    '{"type":"in","field":"u8_0","values":["17","42","9999"]}',
+    '{"type":"in","field":"host","values":["host_0","host_3"]}',
+    '{"type":"in","field":"_msg","values":[]}',
+    '{"type":"contains_any","field":"_msg","values":["stream 0","uuid"]}',
+    '{"type":"contains_all","field":"_msg","values":["message","stream","worker"]}',
+    '{"type":"string_range","field":"dict_0","min":"e","max":"g"}',
+    '{"type":"string_range","field":"u8_0","min":"1","max":"20"}',
+    '{"type":"len_range","field":"_msg","min":100,"max":200}',
+    '{"type":"len_range","field":"u16_0","min":3,"max":3}',
+    '{"type":"day_range","start":0,"end":43200000000000}',
+    '{"type":"week_range","start":0,"end":3}',
+    '{"type":"value_type","field":"u8_0","value_type":"uint8"}',
+    '{"type":"value_type","field":"dict_0","value_type":"dict"}',
 ]
 
 TYPED_FILTERS = [
@@ -202,4 +221,81 @@ TYPED_FILTERS = [
     '{"type":"or","filters":['
     '{"type":"phrase","field":"_msg","phrase":"level=warn"},'
     '{"type":"phrase","field":"_msg","phrase":"other stream row 7"}]}',
+    # ---- set/range filter family (filter_in.go, filter_contains_any.go,
+    #      filter_contains_all.go, filter_string_range.go, filter_ipv4_range.go,
+    #      filter_len_range.go, filter_day_range.go, filter_week_range.go,
+    #      filter_value_type.go, filter_stream_id.go) ----
+    '{"type":"in","field":"lvl","values":["warn","fatal"]}',
+    '{"type":"in","field":"lvl","values":["nosuch"]}',
+    '{"type":"in","field":"u8","values":["13","200","5"]}',
+    '{"type":"in","field":"u16","values":["300","1000","999999"]}',
+    '{"type":"in","field":"u32","values":["171512","70000"]}',
+    '{"type":"in","field":"u64","values":["5099999901"]}',
+    '{"type":"in","field":"i64","values":["-37","0","37"]}',
+    '{"type":"in","field":"f64","values":["-18.625","0.125","1e300"]}',
+    '{"type":"in","field":"ip","values":["10.5.15.35","10.9.27.63"]}',
+    '{"type":"in","field":"iso","values":["2024-01-03T02:02:06.002Z"]}',
+    '{"type":"in","field":"_msg","values":["log line 1 level=info took 1ms"]}',
+    '{"type":"in","field":"constcol","values":["fixed value 42"]}',
+    '{"type":"in","field":"missing_col","values":["a",""]}',
+    '{"type":"contains_any","field":"_msg","values":["level=error","13ms"]}',
+    '{"type":"contains_any","field":"lvl","values":["warn","fatal"]}',
+    '{"type":"contains_any","field":"u8","values":["13","5"]}',
+    '{"type":"contains_any","field":"f64","values":["18.625","625"]}',
+    '{"type":"contains_any","field":"ip","values":["10.5","192.168"]}',
+    '{"type":"contains_any","field":"iso","values":["002Z"]}',
+    '{"type":"contains_any","field":"i64","values":["-37"]}',
+    '{"type":"contains_all","field":"_msg","values":["level=error","took"]}',
+    '{"type":"contains_all","field":"_msg","values":["log","line","7"]}',
+    '{"type":"contains_all","field":"u8","values":["13","13"]}',
+    '{"type":"contains_all","field":"u8","values":["13","5"]}',
+    '{"type":"contains_all","field":"f64","values":["18","625"]}',
+    '{"type":"contains_all","field":"ip","values":["10","35"]}',
+    '{"type":"contains_all","field":"iso","values":["2024","002Z"]}',
+    '{"type":"string_range","field":"lvl","min":"e","max":"warn"}',
+    '{"type":"string_range","field":"u8","min":"10","max":"20"}',
+    '{"type":"string_range","field":"i64","min":"-5","max":"3"}',
+    '{"type":"string_range","field":"f64","min":"-1","max":"1"}',
+    '{"type":"string_range","field":"ip","min":"10.1","max":"10.3"}',
+    '{"type":"string_range","field":"iso","min":"2024-01-1","max":"2024-01-2"}',
+    '{"type":"string_range","field":"_msg","min":"log line 5","max":"log line 7"}',
+    '{"type":"string_range","field":"lvl","min":"z","max":"a"}',
+    '{"type":"ipv4_range","field":"ip","min":167777280,"max":167790000}',
+    '{"type":"ipv4_range","field":"ip","min":168099840,"max":168165375}',
+    '{"type":"ipv4_range","field":"ipstr","min":2886729728,"max":2886733823}',
+    '{"type":"in","field":"ipstr","values":["172.16.3.15","n/a"]}',
+    '{"type":"len_range","field":"ipstr","min":3,"max":3}',
+    '{"type":"string_range","field":"ipstr","min":"172.16.1","max":"172.16.2"}',
+    '{"type":"ipv4_range","field":"_msg","min":3232235776,"max":3232236031}',
+    '{"type":"ipv4_range","field":"lvl","min":0,"max":4294967295}',
+    '{"type":"len_range","field":"_msg","min":30,"max":33}',
+    '{"type":"len_range","field":"lvl","min":4,"max":4}',
+    '{"type":"len_range","field":"u8","min":1,"max":1}',
+    '{"type":"len_range","field":"u16","min":4,"max":4}',
+    '{"type":"len_range","field":"i64","min":1,"max":2}',
+    '{"type":"len_range","field":"f64","min":1,"max":4}',
+    '{"type":"len_range","field":"ip","min":10,"max":11}',
+    '{"type":"len_range","field":"iso","min":24,"max":24}',
+    '{"type":"len_range","field":"iso","min":1,"max":23}',
+    '{"type":"len_range","field":"uni","min":7,"max":11}',
+    '{"type":"day_range","start":80000000000000,"end":80000150000000}',
+    '{"type":"day_range","start":0,"end":86399999999999}',
+    '{"type":"day_range","start":79000000000000,"end":82000000000000,'
+    '"offset":1000000000000}',
+    '{"type":"week_range","start":2,"end":3}',
+    '{"type":"week_range","start":3,"end":6}',
+    '{"type":"week_range","start":3,"end":3,"offset":-90000000000000}',
+    '{"type":"value_type","field":"lvl","value_type":"dict"}',
+    '{"type":"value_type","field":"u8","value_type":"uint8"}',
+    '{"type":"value_type","field":"constcol","value_type":"const"}',
+    '{"type":"value_type","field":"f64","value_type":"float64"}',
+    '{"type":"value_type","field":"missing_col","value_type":"string"}',
+    '{"type":"stream_id","ids":[{"account":0,"project":0,"hi":"1","lo":"1"}]}',
+    '{"type":"stream_id","ids":[{"account":0,"project":0,"hi":"0","lo":"1"},'
+    '{"account":0,"project":0,"hi":"9","lo":"9"}]}',
+    '{"type":"stream_id","ids":[]}',
+    '{"type":"and","filters":['
+    '{"type":"in","field":"lvl","values":["error","fatal"]},'
+    '{"type":"len_range","field":"_msg","min":1,"max":40},'
+    '{"type":"week_range","start":2,"end":2}]}',
 ]

@@ -43,7 +43,10 @@ TYPED_REGEXES = [
 
 def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
     leaf_kinds = ["phrase", "phrase", "exact", "regexp", "time", "range",
-                  "prefix", "exact_prefix", "sequence"]
+                  "prefix", "exact_prefix", "sequence",
+                  "in", "contains_any", "contains_all", "string_range",
+                  "ipv4_range", "len_range", "day_range", "week_range",
+                  "value_type"]
     kind = rng.choice(
         leaf_kinds + ["and", "or", "not"] if depth > 0 else leaf_kinds)
     if kind == "phrase":
@@ -58,6 +61,46 @@ def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
         return {"type": kind,
                 "field": rng.choice(fields_str + fields_num),
                 "prefix": rng.choice(phrases)}
+    if kind in ("in", "contains_any", "contains_all"):
+        return {"type": kind,
+                "field": rng.choice(fields_str + fields_num),
+                "values": [rng.choice(phrases)
+                           for _ in range(rng.randrange(0, 4))]}
+    if kind == "string_range":
+        a, b = rng.choice(phrases), rng.choice(phrases)
+        return {"type": "string_range",
+                "field": rng.choice(fields_str + fields_num),
+                "min": a, "max": b}
+    if kind == "ipv4_range":
+        a = rng.randrange(0, 2**32)
+        b = a + rng.randrange(-100, 2**24)
+        return {"type": "ipv4_range",
+                "field": rng.choice(fields_str + fields_num),
+                "min": a, "max": min(b, 2**32 - 1)}
+    if kind == "len_range":
+        a = rng.randrange(0, 30)
+        return {"type": "len_range",
+                "field": rng.choice(fields_str + fields_num),
+                "min": a, "max": a + rng.randrange(0, 30)}
+    if kind == "day_range":
+        a = rng.randrange(0, 86400 * 10**9)
+        d = {"type": "day_range", "start": a,
+             "end": min(a + rng.randrange(0, 10**13), 86400 * 10**9 - 1)}
+        if rng.random() < 0.5:
+            d["offset"] = rng.randrange(-14, 14) * 3600 * 10**9
+        return d
+    if kind == "week_range":
+        a = rng.randrange(0, 7)
+        d = {"type": "week_range", "start": a, "end": rng.randrange(a, 7)}
+        if rng.random() < 0.5:
+            d["offset"] = rng.randrange(-14, 14) * 3600 * 10**9
+        return d
+    if kind == "value_type":
+        return {"type": "value_type",
+                "field": rng.choice(fields_str + fields_num),
+                "value_type": rng.choice(
+                    ["string", "dict", "uint8", "uint16", "uint32", "uint64",
+                     "int64", "float64", "ipv4", "iso8601", "const"])}
     if kind == "sequence":
         return {"type": "sequence",
                 "field": rng.choice(fields_str + fields_num),

@@ -1,6 +1,7 @@
 #include "filter.h"
 
 #include <algorithm>
+#include <cstring>
 #include <cmath>
 #include <cstdlib>
 #include <map>
@@ -8,6 +9,7 @@
 #include "bloom.h"
 #include "json.h"
 #include "match.h"
+#include "values.h"
 #include "tokenizer.h"
 
 namespace vl {
@@ -173,6 +175,152 @@ FilterNode build(const JValue& v) {
     n.field = jget(v, "field").str;
     n.min_f = jget(v, "min").num;
     n.max_f = jget(v, "max").num;
+  } else if (type == "in" || type == "contains_any" || type == "contains_all") {
+    n.type = type == "in" ? FilterNode::In
+             : type == "contains_any" ? FilterNode::ContainsAny
+                                      : FilterNode::ContainsAll;
+    n.field = jget(v, "field").str;
+    for (const auto& vj : jget(v, "values").arr) n.values.push_back(vj.str);
+    if (n.type == FilterNode::ContainsAll) {
+      // getTokensHashesAll (in_values.go:94-102): tokenizeHashes over all
+      // values, then appendHashesHashes
+      std::vector<strview> vs;
+      for (const auto& s2 : n.values) vs.push_back(strview(s2));
+      for (uint64_t th : tokenize_hashes(vs)) {
+        append_hash_hashes(n.all_hashes, th);
+      }
+    } else {
+      // getTokensHashesAny (in_values.go:104-125): per-value token sets with
+      // the common tokens factored out
+      std::vector<std::vector<std::string>> token_sets;
+      for (const auto& s2 : n.values) {
+        token_sets.push_back(tokenize_strings({s2}));
+      }
+      std::vector<std::string> common;
+      if (!token_sets.empty()) {
+        common = token_sets[0];
+        for (size_t i = 1; i < token_sets.size() && !common.empty(); i++) {
+          std::vector<std::string> kept;
+          for (const auto& t : common) {
+            if (std::find(token_sets[i].begin(), token_sets[i].end(), t) !=
+                token_sets[i].end()) {
+              kept.push_back(t);
+            }
+          }
+          common = std::move(kept);
+        }
+      }
+      if (!common.empty()) {
+        for (auto& ts : token_sets) {
+          std::vector<std::string> kept;
+          for (auto& t : ts) {
+            if (std::find(common.begin(), common.end(), t) == common.end()) {
+              kept.push_back(std::move(t));
+            }
+          }
+          ts = std::move(kept);
+        }
+      }
+      n.common_hashes = probe_hashes(common);
+      for (const auto& ts : token_sets) {
+        n.set_hashes.push_back(probe_hashes(ts));
+      }
+    }
+    if (n.type != FilterNode::Noop) {  // In/ContainsAny/ContainsAll
+      // per-type binary value sets (in_values.go:141-315), sorted for the
+      // device binary search; ContainsAny/ContainsAll use the uint slots
+      // (filter_contains_any.go:141-152, filter_contains_all.go:183-204)
+      n.bin_sets.resize(8);
+      for (const auto& s2 : n.values) {
+        strview sv(s2);
+        uint64_t u;
+        if (try_parse_uint64(sv, &u)) {
+          bytes b;
+          if (u < (1 << 8)) {
+            b.push_back(uint8_t(u));
+            n.bin_sets[0].emplace_back((const char*)b.data(), b.size());
+          }
+          b.clear();
+          if (u < (1 << 16)) {
+            put_u16be(b, uint16_t(u));
+            n.bin_sets[1].emplace_back((const char*)b.data(), b.size());
+          }
+          b.clear();
+          if (u < (uint64_t(1) << 32)) {
+            put_u32be(b, uint32_t(u));
+            n.bin_sets[2].emplace_back((const char*)b.data(), b.size());
+          }
+          b.clear();
+          put_u64be(b, u);
+          n.bin_sets[3].emplace_back((const char*)b.data(), b.size());
+        }
+        int64_t i64v;
+        if (try_parse_int64(sv, &i64v)) {
+          bytes b;
+          put_i64be_zigzag(b, i64v);
+          n.bin_sets[4].emplace_back((const char*)b.data(), b.size());
+        }
+        double f;
+        if (try_parse_float64_exact(sv, &f)) {
+          bytes b;
+          uint64_t fu;
+          memcpy(&fu, &f, 8);
+          put_u64be(b, fu);
+          n.bin_sets[5].emplace_back((const char*)b.data(), b.size());
+        }
+        uint32_t ip;
+        if (try_parse_ipv4(sv, &ip)) {
+          bytes b;
+          put_u32be(b, ip);
+          n.bin_sets[6].emplace_back((const char*)b.data(), b.size());
+        }
+        int64_t ts;
+        if (try_parse_timestamp_iso8601(sv, &ts)) {
+          bytes b;
+          put_u64be(b, uint64_t(ts));
+          n.bin_sets[7].emplace_back((const char*)b.data(), b.size());
+        }
+      }
+      for (auto& bs2 : n.bin_sets) {
+        std::sort(bs2.begin(), bs2.end());
+        bs2.erase(std::unique(bs2.begin(), bs2.end()), bs2.end());
+      }
+    }
+  } else if (type == "string_range") {
+    n.type = FilterNode::StringRange;
+    n.field = jget(v, "field").str;
+    n.min_s = jget(v, "min").str;
+    n.max_s = jget(v, "max").str;
+  } else if (type == "ipv4_range") {
+    n.type = FilterNode::IPv4Range;
+    n.field = jget(v, "field").str;
+    n.min_u = uint64_t(jget(v, "min").num);
+    n.max_u = uint64_t(jget(v, "max").num);
+  } else if (type == "len_range") {
+    n.type = FilterNode::LenRange;
+    n.field = jget(v, "field").str;
+    n.min_u = uint64_t(jget(v, "min").num);
+    n.max_u = uint64_t(jget(v, "max").num);
+  } else if (type == "day_range" || type == "week_range") {
+    n.type = type == "day_range" ? FilterNode::DayRange : FilterNode::WeekRange;
+    n.min_u = uint64_t(jget(v, "start").num);
+    n.max_u = uint64_t(jget(v, "end").num);
+    auto it = v.obj.find("offset");
+    if (it != v.obj.end()) n.tz_offset = int64_t(it->second.num);
+  } else if (type == "value_type") {
+    n.type = FilterNode::ValueTypeFilter;
+    n.field = jget(v, "field").str;
+    n.min_s = jget(v, "value_type").str;
+  } else if (type == "stream_id") {
+    n.type = FilterNode::StreamIdFilter;
+    for (const auto& sj : jget(v, "ids").arr) {
+      uint64_t acct = uint64_t(jget(sj, "account").num);
+      uint64_t proj = uint64_t(jget(sj, "project").num);
+      // hi/lo as decimal strings to keep full u64 precision through JSON
+      uint64_t hi = strtoull(jget(sj, "hi").str.c_str(), nullptr, 10);
+      uint64_t lo = strtoull(jget(sj, "lo").str.c_str(), nullptr, 10);
+      n.stream_ids.push_back({acct << 32 | proj, hi, lo});
+    }
   } else if (type == "noop") {
     n.type = FilterNode::Noop;
   } else {

@@ -9,6 +9,7 @@
 // filter_or.go:122-193).
 #pragma once
 
+#include <array>
 #include <memory>
 #include <string>
 #include <vector>
@@ -27,7 +28,9 @@ struct FieldTokens {
 struct FilterNode {
   enum Type {
     Phrase, Exact, Regexp, And, Or, Not, Time, Range, Noop,
-    Prefix, ExactPrefix, Sequence
+    Prefix, ExactPrefix, Sequence,
+    In, ContainsAny, ContainsAll, StringRange, IPv4Range, LenRange,
+    DayRange, WeekRange, ValueTypeFilter, StreamIdFilter
   } type;
 
   std::string field;   // phrase/exact/regexp/range (as written in the query)
@@ -37,6 +40,20 @@ struct FilterNode {
   double min_f = 0, max_f = 0;    // Range
   int64_t min_ts = 0, max_ts = 0;  // Time
   std::vector<FilterNode> children;  // And/Or (n), Not (1)
+  std::vector<std::string> values;   // In/ContainsAny/ContainsAll
+  std::string min_s, max_s;          // StringRange; ValueTypeFilter: min_s=type
+  uint64_t min_u = 0, max_u = 0;     // IPv4Range/LenRange; Day/WeekRange: start/end
+  int64_t tz_offset = 0;             // Day/WeekRange offset (nsecs)
+  std::vector<std::array<uint64_t, 3>> stream_ids;  // {acct<<32|proj, hi, lo}
+
+  // In/ContainsAny: inValues token structures (in_values.go:94-125)
+  std::vector<uint64_t> common_hashes;             // probe hashes, common tokens
+  std::vector<std::vector<uint64_t>> set_hashes;   // per-value probe hashes
+  // ContainsAll: tokensHashesAll (in_values.go:94-102)
+  std::vector<uint64_t> all_hashes;
+  // In: per-type binary value sets, sorted (in_values.go:141-315); index by
+  // width slot: [0]=u8 [1]=u16 [2]=u32 [3]=u64 [4]=i64 [5]=f64 [6]=ipv4 [7]=iso
+  std::vector<std::vector<std::string>> bin_sets;
 
   // compiled:
   std::vector<std::string> tokens;

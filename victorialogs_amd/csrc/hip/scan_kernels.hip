@@ -603,6 +603,176 @@ __device__ double d_parse_math_number(const A& a, long s0, long sn) {
   return __builtin_nan("");
 }
 
+
+// value in a sorted string set: blob = u32 n, u32 offs[n+1], bytes
+template <typename A>
+__device__ bool d_in_sorted_str(const uint8_t* blob, const A& a, long s0, long sn) {
+  uint32_t n;
+  __builtin_memcpy(&n, blob, 4);
+  const uint8_t* offs = blob + 4;
+  const uint8_t* data = blob + 4 + size_t(n + 1) * 4;
+  auto off_at = [&](uint32_t i) {
+    uint32_t o;
+    __builtin_memcpy(&o, offs + size_t(i) * 4, 4);
+    return o;
+  };
+  uint32_t lo = 0, hi = n;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi) / 2;
+    uint32_t mo = off_at(mid), ml = off_at(mid + 1) - mo;
+    // lexicographic compare set[mid] vs row
+    int c = 0;
+    long k = 0;
+    long lim = ml < uint32_t(sn) ? ml : uint32_t(sn);
+    for (; k < lim; k++) {
+      uint8_t cb = data[mo + k], rb = a.u8(s0 + k);
+      if (cb != rb) {
+        c = cb < rb ? -1 : 1;
+        break;
+      }
+    }
+    if (c == 0) c = long(ml) < sn ? -1 : (long(ml) > sn ? 1 : 0);
+    if (c == 0) return true;
+    if (c < 0) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  return false;
+}
+
+// fixed-width value in a sorted packed binary set (bytewise order == value
+// order for BE encodings)
+__device__ bool d_in_sorted_bin(const uint8_t* vals, uint32_t n, uint8_t width,
+                                const uint8_t* p) {
+  uint32_t lo = 0, hi = n;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi) / 2;
+    const uint8_t* m = vals + size_t(mid) * width;
+    int c = 0;
+    for (int k = 0; k < width; k++) {
+      if (m[k] != p[k]) {
+        c = m[k] < p[k] ? -1 : 1;
+        break;
+      }
+    }
+    if (c == 0) return true;
+    if (c < 0) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  return false;
+}
+
+// matchAnyPhrase / matchAllPhrases over a serialized phrase list (the
+// sequence blob layout: u16 n, { u16 len, u8 flags, bytes })
+template <typename A>
+__device__ bool d_match_any_phrase_at(const A& a, long s0, long sn,
+                                      const uint8_t* blob) {
+  uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t* p = blob + 2;
+  for (uint16_t i = 0; i < n; i++) {
+    uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
+    uint8_t flags = p[2];
+    p += 3;
+    if (len == 0 ? sn == 0 : d_get_phrase_pos_at(a, s0, sn, p, len, flags) >= 0) {
+      return true;
+    }
+    p += len;
+  }
+  return false;
+}
+
+template <typename A>
+__device__ bool d_match_all_phrases_at(const A& a, long s0, long sn,
+                                       const uint8_t* blob) {
+  uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t* p = blob + 2;
+  for (uint16_t i = 0; i < n; i++) {
+    uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
+    uint8_t flags = p[2];
+    p += 3;
+    if (len != 0 && d_get_phrase_pos_at(a, s0, sn, p, len, flags) < 0) {
+      return false;  // empty phrases match everything (filter_contains_all.go:312-315)
+    }
+    p += len;
+  }
+  return true;
+}
+
+// matchStringRange: s >= min && s < max; blob = u32 minlen, u32 maxlen, bytes
+template <typename A>
+__device__ bool d_string_range_at(const A& a, long s0, long sn,
+                                  const uint8_t* blob) {
+  uint32_t mn_len, mx_len;
+  __builtin_memcpy(&mn_len, blob, 4);
+  __builtin_memcpy(&mx_len, blob + 4, 4);
+  const uint8_t* mn = blob + 8;
+  const uint8_t* mx = mn + mn_len;
+  auto cmp = [&](const uint8_t* b, uint32_t bl) {
+    long lim = long(bl) < sn ? long(bl) : sn;
+    for (long k = 0; k < lim; k++) {
+      uint8_t rb = a.u8(s0 + k);
+      if (rb != b[k]) return rb < b[k] ? -1 : 1;
+    }
+    return sn < long(bl) ? -1 : (sn > long(bl) ? 1 : 0);
+  };
+  return cmp(mn, mn_len) >= 0 && cmp(mx, mx_len) < 0;
+}
+
+template <typename A>
+__device__ uint64_t d_rune_count(const A& a, long s0, long sn) {
+  uint64_t n = 0;
+  for (long i = 0; i < sn; i++) {
+    if ((a.u8(s0 + i) & 0xC0) != 0x80) n++;
+  }
+  return n;
+}
+
+// tryParseIPv4 over an accessor (values_encoder.go:675-730)
+template <typename A>
+__device__ bool d_try_parse_ipv4(const A& a, long s0, long sn, uint32_t* out) {
+  if (sn < 7 || sn > 15) return false;
+  int dots = 0;
+  for (long i = 0; i < sn; i++) {
+    if (a.u8(s0 + i) == '.') dots++;
+  }
+  if (dots != 3) return false;
+  uint32_t ip = 0;
+  for (int oct = 0; oct < 4; oct++) {
+    long seg_end = s0 + sn;
+    if (oct < 3) {
+      long j = s0;
+      while (j < s0 + sn && a.u8(j) != '.') j++;
+      seg_end = j;
+    }
+    long len = seg_end - s0;
+    if (len <= 0 || len > 3) return false;
+    // tryParseDateUint64 two-digit fast path quirk: only first char checked
+    uint32_t v = 0;
+    if (len == 2) {
+      uint8_t c0 = a.u8(s0);
+      if (c0 < '0' || c0 > '9') return false;
+      v = 10 * uint32_t(c0 - '0') + uint32_t(uint8_t(a.u8(s0 + 1) - '0'));
+    } else {
+      for (long k = 0; k < len; k++) {
+        uint8_t c = a.u8(s0 + k);
+        if (c < '0' || c > '9') return false;
+        v = v * 10 + (c - '0');
+      }
+    }
+    if (v > 255) return false;
+    ip = ip << 8 | v;
+    s0 = seg_end + 1;
+    sn -= len + 1;
+  }
+  *out = ip;
+  return true;
+}
+
 // ---- regex fast paths on serialized blob (regex.go:86-212) ----
 
 struct DRegex {
@@ -791,6 +961,23 @@ __device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
       return d_has_prefix_bytes(a, s0, sn, lb.operand, lb.operand_len);
     case kScanSeqStr:
       return d_match_sequence_at(a, s0, sn, lb.operand);
+    case kScanInStr:
+      return d_in_sorted_str(lb.operand, a, s0, sn);
+    case kScanAnyPhraseStr:
+      return d_match_any_phrase_at(a, s0, sn, lb.operand);
+    case kScanAllPhrasesStr:
+      return d_match_all_phrases_at(a, s0, sn, lb.operand);
+    case kScanStrRange:
+      return d_string_range_at(a, s0, sn, lb.operand);
+    case kScanIPv4RangeStr: {
+      uint32_t ip;
+      if (!d_try_parse_ipv4(a, s0, sn, &ip)) return false;
+      return ip >= uint32_t(lb.vmin) && ip <= uint32_t(lb.vmax);
+    }
+    case kScanLenRangeStr: {
+      uint64_t n = d_rune_count(a, s0, sn);
+      return n >= lb.vmin && n <= lb.vmax;
+    }
     default:  // kScanRegexStr
       return d_regex_match_at(lb.operand, a, s0, sn);
   }
@@ -909,9 +1096,39 @@ __device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
     }
+    case kScanInBin:
+      return d_in_sorted_bin(lb.operand, lb.operand_len / lb.width, lb.width,
+                             lb.data + size_t(row) * lb.width);
+    case kScanIPv4RangeBin: {
+      const uint8_t* p = lb.data + size_t(row) * 4;
+      uint32_t ip = uint32_t(p[0]) << 24 | uint32_t(p[1]) << 16 |
+                    uint32_t(p[2]) << 8 | p[3];
+      return ip >= uint32_t(lb.vmin) && ip <= uint32_t(lb.vmax);
+    }
+    case kScanDayRange: {
+      long long off_tz;
+      __builtin_memcpy(&off_tz, lb.operand, 8);
+      int64_t off = (lb.ts[row] - off_tz) % (24LL * 3600 * 1000000000);
+      return off >= int64_t(lb.vmin) && off <= int64_t(lb.vmax);
+    }
+    case kScanWeekRange: {
+      long long off_tz;
+      __builtin_memcpy(&off_tz, lb.operand, 8);
+      const int64_t day = 24LL * 3600 * 1000000000;
+      int64_t t = lb.ts[row] - off_tz;
+      int64_t days = t / day;
+      if (t % day < 0) days--;
+      int64_t wd = (days + 4) % 7;
+      if (wd < 0) wd += 7;
+      return wd >= int64_t(lb.vmin) && wd <= int64_t(lb.vmax);
+    }
     case kScanPrefixFmt:
     case kScanExactPrefixFmt:
-    case kScanSeqFmt: {
+    case kScanSeqFmt:
+    case kScanAnyPhraseFmt:
+    case kScanAllPhrasesFmt:
+    case kScanStrRangeFmt:
+    case kScanLenRangeFmt: {
       char buf[344];
       int n;
       switch (lb.flags >> 4) {
@@ -943,14 +1160,25 @@ __device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
           break;
       }
       BufAcc a{(const uint8_t*)buf};
-      if (lb.kind == kScanPrefixFmt) {
-        return d_match_prefix_at(a, 0, n, lb.operand, lb.operand_len,
-                                 lb.flags & 15);
+      switch (lb.kind) {
+        case kScanPrefixFmt:
+          return d_match_prefix_at(a, 0, n, lb.operand, lb.operand_len,
+                                   lb.flags & 15);
+        case kScanExactPrefixFmt:
+          return d_has_prefix_bytes(a, 0, n, lb.operand, lb.operand_len);
+        case kScanAnyPhraseFmt:
+          return d_match_any_phrase_at(a, 0, n, lb.operand);
+        case kScanAllPhrasesFmt:
+          return d_match_all_phrases_at(a, 0, n, lb.operand);
+        case kScanStrRangeFmt:
+          return d_string_range_at(a, 0, n, lb.operand);
+        case kScanLenRangeFmt: {
+          uint64_t rc = d_rune_count(a, 0, n);
+          return rc >= lb.vmin && rc <= lb.vmax;
+        }
+        default:
+          return d_match_sequence_at(a, 0, n, lb.operand);
       }
-      if (lb.kind == kScanExactPrefixFmt) {
-        return d_has_prefix_bytes(a, 0, n, lb.operand, lb.operand_len);
-      }
-      return d_match_sequence_at(a, 0, n, lb.operand);
     }
     default:
       return false;
@@ -960,7 +1188,10 @@ __device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
 __device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
   return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr ||
          kind == kScanRangeStr || kind == kScanPrefixStr ||
-         kind == kScanExactPrefixStr || kind == kScanSeqStr;
+         kind == kScanExactPrefixStr || kind == kScanSeqStr ||
+         kind == kScanInStr || kind == kScanAnyPhraseStr ||
+         kind == kScanAllPhrasesStr || kind == kScanStrRange ||
+         kind == kScanIPv4RangeStr || kind == kScanLenRangeStr;
 }
 
 // ---- the program kernel ----

@@ -49,6 +49,20 @@ enum ScanKind : uint8_t {
   kScanPrefixFmt = 21,      // matchPrefix over formatted value (fmt in flags>>4)
   kScanExactPrefixFmt = 22, // HasPrefix over formatted value
   kScanSeqFmt = 23,         // matchSequence over formatted value
+  kScanInStr = 24,          // value in sorted string set (filter_in.go:187-200)
+  kScanInBin = 25,          // fixed-width value in sorted binary set
+  kScanAnyPhraseStr = 26,   // matchAnyPhrase (filter_contains_any.go:293-300)
+  kScanAllPhrasesStr = 27,  // matchAllPhrases (filter_contains_all.go:310-321)
+  kScanAnyPhraseFmt = 28,   // matchAnyPhrase over formatted value
+  kScanAllPhrasesFmt = 29,  // matchAllPhrases over formatted value
+  kScanStrRange = 30,       // s >= min && s < max (filter_string_range.go:225-229)
+  kScanStrRangeFmt = 31,    // formatted value string range
+  kScanIPv4RangeStr = 32,   // parse ipv4 from string row, range compare
+  kScanLenRangeStr = 33,    // rune-count range (filter_len_range.go:333-336)
+  kScanLenRangeFmt = 34,    // formatted value length range
+  kScanDayRange = 35,       // (ts - offset) % day in [start,end] (filter_day_range.go)
+  kScanWeekRange = 36,      // weekday(ts - offset) in [start,end] (filter_week_range.go)
+  kScanIPv4RangeBin = 37,   // BE u32 in [vmin,vmax] (filter_ipv4_range.go:166-181)
 };
 
 // format source for the *Fmt kinds, stored in flags bits 4..7

@@ -289,6 +289,7 @@ struct BlockStageCtx {
   PartReader::BlockColumns bc;
   std::map<std::string, StagedStrCol> cols;
   std::map<std::string, StagedBloom> blooms;
+  std::map<std::string, std::vector<uint64_t>> host_blooms;
   const int64_t* d_ts = nullptr;
 
   const StagedStrCol& stage_column(const ColumnHeader& ch) {
@@ -311,11 +312,20 @@ struct BlockStageCtx {
     return cols.emplace(ch.name, std::move(sc)).first->second;
   }
 
+  const std::vector<uint64_t>& bloom_host(const ColumnHeader& ch) {
+    auto it = host_blooms.find(ch.name);
+    if (it == host_blooms.end()) {
+      std::vector<uint64_t> words;
+      pr->read_bloom(ch, words);
+      it = host_blooms.emplace(ch.name, std::move(words)).first;
+    }
+    return it->second;
+  }
+
   const StagedBloom& stage_bloom(const ColumnHeader& ch) {
     auto it = blooms.find(ch.name);
     if (it != blooms.end()) return it->second;
-    std::vector<uint64_t> words;
-    pr->read_bloom(ch, words);
+    const std::vector<uint64_t>& words = bloom_host(ch);
     StagedBloom sb;
     sb.nwords = uint32_t(words.size());
     sb.d_words = (const uint64_t*)st->push(words.data(), words.size() * 8, 8);
@@ -365,6 +375,43 @@ bool stage_eq_bin(DevLeafBlock& lb, BlockStageCtx& ctx, const ColumnHeader& ch,
   lb.data = sc.d_data;
   set_bloom_gate(lb, ctx, ch, li);
   return true;
+}
+
+// Binary equality without a device bloom gate (the gate already ran on the
+// host for the set-family filters).
+void stage_eq_bin_nogate(DevLeafBlock& lb, BlockStageCtx& ctx,
+                         const ColumnHeader& ch, Stage& st, const bytes& bin) {
+  const StagedStrCol& sc = ctx.stage_column(ch);
+  if (sc.is_const) {
+    lb.mode = sc.const_value.size() == bin.size() &&
+                      memcmp(sc.const_value.data(), bin.data(), bin.size()) == 0
+                  ? kModeAll
+                  : kModeNone;
+    return;
+  }
+  lb.mode = kModeScan;
+  lb.kind = kScanEqBin;
+  lb.width = uint8_t(bin.size());
+  lb.operand = (const uint8_t*)st.push(bin.data(), bin.size(), 8);
+  lb.operand_len = uint32_t(bin.size());
+  lb.data = sc.d_data;
+}
+
+bool in_sorted_bin_host(const std::vector<std::string>& sorted_set, strview v) {
+  size_t lo = 0, hi = sorted_set.size();
+  while (lo < hi) {
+    size_t mid = (lo + hi) / 2;
+    const std::string& m = sorted_set[mid];
+    int c = memcmp(m.data(), v.p, std::min(m.size(), v.n));
+    if (c == 0) c = m.size() < v.n ? -1 : (m.size() > v.n ? 1 : 0);
+    if (c == 0) return true;
+    if (c < 0) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  return false;
 }
 
 // Returns the parsed+pruned binary value for exact matches on binary columns,
@@ -481,6 +528,151 @@ std::string format_encoded(ValueType t, strview v) {
       fail("format_encoded: unexpected type");
   }
   return s;
+}
+
+
+// ---- helpers for the set/range filter family ----
+
+// sorted string set blob: u32 n, u32 offs[n+1], bytes (kScanInStr layout)
+bytes serialize_str_set(const std::vector<std::string>& sorted_set) {
+  bytes b;
+  auto put32 = [&](uint32_t v) {
+    b.push_back(uint8_t(v));
+    b.push_back(uint8_t(v >> 8));
+    b.push_back(uint8_t(v >> 16));
+    b.push_back(uint8_t(v >> 24));
+  };
+  put32(uint32_t(sorted_set.size()));
+  uint32_t off = 0;
+  for (const auto& v : sorted_set) {
+    put32(off);
+    off += uint32_t(v.size());
+  }
+  put32(off);
+  for (const auto& v : sorted_set) b.insert(b.end(), v.begin(), v.end());
+  return b;
+}
+
+bool h_match_any_phrase(strview v, const std::vector<std::string>& phrases) {
+  for (const auto& ph : phrases) {
+    if (match_phrase(v, strview(ph))) return true;
+  }
+  return false;
+}
+
+bool h_match_all_phrases(strview v, const std::vector<std::string>& phrases) {
+  for (const auto& ph : phrases) {
+    if (ph.empty()) continue;  // filter_contains_all.go:310-321
+    if (!match_phrase(v, strview(ph))) return false;
+  }
+  return true;
+}
+
+bool h_match_string_range(strview s, const std::string& mn,
+                          const std::string& mx) {
+  // matchStringRange (filter_string_range.go:225-229)
+  auto cmp = [](strview a, const std::string& b) {
+    int c = memcmp(a.p, b.data(), std::min(a.n, b.size()));
+    if (c != 0) return c;
+    return a.n < b.size() ? -1 : (a.n > b.size() ? 1 : 0);
+  };
+  return cmp(s, mn) >= 0 && cmp(s, mx) < 0;
+}
+
+uint64_t h_rune_count(strview s) {
+  uint64_t n = 0;
+  for (size_t i = 0; i < s.n; i++) {
+    if ((uint8_t(s.p[i]) & 0xC0) != 0x80) n++;
+  }
+  return n;
+}
+
+bool h_in_values(const std::vector<std::string>& values, strview v) {
+  for (const auto& s2 : values) {
+    if (strview(s2) == v) return true;
+  }
+  return false;
+}
+
+bool h_bloom_all(BlockStageCtx& ctx, const ColumnHeader& ch,
+                 const std::vector<uint64_t>& hashes) {
+  if (hashes.empty()) return true;
+  const auto& words = ctx.bloom_host(ch);
+  return bloom_contains_all(words.data(), words.size(), hashes.data(),
+                            hashes.size());
+}
+
+// matchBloomFilterAnyTokenSet (filter_in.go:202-218); block-level — a bloom
+// miss implies zero matching rows, so host gating is result-identical
+bool h_bloom_any_token_set(const FilterNode& f, BlockStageCtx& ctx,
+                           const ColumnHeader& ch) {
+  if (!h_bloom_all(ctx, ch, f.common_hashes)) return false;
+  if (f.set_hashes.size() > 1000 ||
+      f.set_hashes.size() > 10 * ctx.bh->rows_count) {
+    return true;
+  }
+  const auto& words = ctx.bloom_host(ch);
+  for (const auto& hs : f.set_hashes) {
+    if (bloom_contains_all(words.data(), words.size(), hs.data(), hs.size())) {
+      return true;
+    }
+  }
+  return false;
+}
+
+int bin_set_slot(ValueType t) {
+  switch (t) {
+    case ValueType::Uint8: return 0;
+    case ValueType::Uint16: return 1;
+    case ValueType::Uint32: return 2;
+    case ValueType::Uint64: return 3;
+    case ValueType::Int64: return 4;
+    case ValueType::Float64: return 5;
+    case ValueType::IPv4: return 6;
+    case ValueType::TimestampISO8601: return 7;
+    default: return -1;
+  }
+}
+
+const char* value_type_name(ValueType t) {
+  // filterValueType names (filter_value_type.go:44-67)
+  switch (t) {
+    case ValueType::String: return "string";
+    case ValueType::Dict: return "dict";
+    case ValueType::Uint8: return "uint8";
+    case ValueType::Uint16: return "uint16";
+    case ValueType::Uint32: return "uint32";
+    case ValueType::Uint64: return "uint64";
+    case ValueType::Int64: return "int64";
+    case ValueType::Float64: return "float64";
+    case ValueType::IPv4: return "ipv4";
+    case ValueType::TimestampISO8601: return "iso8601";
+    default: return "unknown";
+  }
+}
+
+uint8_t fmt_of_type(ValueType t) {
+  switch (t) {
+    case ValueType::Uint8:
+    case ValueType::Uint16:
+    case ValueType::Uint32:
+    case ValueType::Uint64: return kFmtU64;
+    case ValueType::Int64: return kFmtI64;
+    case ValueType::Float64: return kFmtF64;
+    case ValueType::IPv4: return kFmtIp;
+    case ValueType::TimestampISO8601: return kFmtIso;
+    default: return 0;
+  }
+}
+
+uint8_t width_of_type(ValueType t) {
+  switch (t) {
+    case ValueType::Uint8: return 1;
+    case ValueType::Uint16: return 2;
+    case ValueType::Uint32: return 4;
+    case ValueType::IPv4: return 4;
+    default: return 8;
+  }
 }
 
 // stage one (leaf, block) descriptor; mirrors the oracle's apply_* dispatch.
@@ -1352,6 +1544,567 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
       }
     }
 
+
+    case FilterNode::In: {
+      // filterIn (filter_in.go:120-234)
+      if (f.values.empty()) {
+        lb.mode = kModeNone;
+        return;
+      }
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = h_in_values(f.values, strview(cv)) ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = h_in_values(f.values, strview("", 0)) ? kModeAll : kModeNone;
+        return;
+      }
+      if (ch.type == ValueType::Dict) {
+        uint32_t mask = dict_mask_of(
+            ch.dict, [&](strview dv) { return h_in_values(f.values, dv); });
+        stage_dict(lb, ctx, ch, mask);
+        return;
+      }
+      if (ch.type == ValueType::String) {
+        if (!h_bloom_any_token_set(f, ctx, ch)) {
+          lb.mode = kModeNone;
+          return;
+        }
+        const StagedStrCol& sc = ctx.stage_column(ch);
+        if (sc.is_const) {
+          lb.mode = h_in_values(f.values, strview(sc.const_value)) ? kModeAll
+                                                                   : kModeNone;
+          return;
+        }
+        lb.mode = kModeScan;
+        lb.kind = kScanInStr;
+        lb.operand = li.d_operand;  // sorted deduped string-set blob
+        lb.operand_len = uint32_t(li.operand.size());
+        lb.data = sc.d_data;
+        lb.offsets = sc.d_offsets;
+        return;
+      }
+      const auto& set = f.bin_sets[size_t(bin_set_slot(ch.type))];
+      if (set.empty()) {
+        lb.mode = kModeNone;
+        return;
+      }
+      if (!h_bloom_any_token_set(f, ctx, ch)) {
+        lb.mode = kModeNone;
+        return;
+      }
+      const StagedStrCol& sc = ctx.stage_column(ch);
+      uint8_t w = width_of_type(ch.type);
+      if (sc.is_const) {
+        lb.mode = in_sorted_bin_host(set, strview(sc.const_value)) ? kModeAll
+                                                                   : kModeNone;
+        return;
+      }
+      bytes packed;
+      for (const auto& v : set) packed.insert(packed.end(), v.begin(), v.end());
+      lb.mode = kModeScan;
+      lb.kind = kScanInBin;
+      lb.width = w;
+      lb.operand = (const uint8_t*)st.push(packed.data(), packed.size(), 8);
+      lb.operand_len = uint32_t(packed.size());
+      lb.data = sc.d_data;
+      return;
+    }
+
+    case FilterNode::ContainsAny: {
+      // filterContainsAny (filter_contains_any.go:105-296)
+      if (f.values.empty()) {
+        lb.mode = kModeNone;
+        return;
+      }
+      for (const auto& v : f.values) {
+        if (v.empty()) {
+          lb.mode = kModeAll;  // empty value matches everything (:110-113)
+          return;
+        }
+      }
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = h_match_any_phrase(strview(cv), f.values) ? kModeAll
+                                                            : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = h_match_any_phrase(strview("", 0), f.values) ? kModeAll
+                                                               : kModeNone;
+        return;
+      }
+      if (ch.type == ValueType::Dict) {
+        uint32_t mask = dict_mask_of(ch.dict, [&](strview dv) {
+          return h_match_any_phrase(dv, f.values);
+        });
+        stage_dict(lb, ctx, ch, mask);
+        return;
+      }
+      if (ch.type == ValueType::Uint8 || ch.type == ValueType::Uint16 ||
+          ch.type == ValueType::Uint32 || ch.type == ValueType::Uint64) {
+        // uint columns match whole-value only (filter_contains_any.go:141-152)
+        const auto& set = f.bin_sets[size_t(bin_set_slot(ch.type))];
+        if (set.empty() || !h_bloom_any_token_set(f, ctx, ch)) {
+          lb.mode = kModeNone;
+          return;
+        }
+        const StagedStrCol& sc = ctx.stage_column(ch);
+        uint8_t w = width_of_type(ch.type);
+        if (sc.is_const) {
+          lb.mode = in_sorted_bin_host(set, strview(sc.const_value)) ? kModeAll
+                                                                     : kModeNone;
+          return;
+        }
+        bytes packed;
+        for (const auto& v : set) packed.insert(packed.end(), v.begin(), v.end());
+        lb.mode = kModeScan;
+        lb.kind = kScanInBin;
+        lb.width = w;
+        lb.operand = (const uint8_t*)st.push(packed.data(), packed.size(), 8);
+        lb.operand_len = uint32_t(packed.size());
+        lb.data = sc.d_data;
+        return;
+      }
+      // common-token gate + per-value token-set survivor filter
+      // (matchValuesAnyPhrase, filter_contains_any.go:179-198); survivor
+      // filtering is result-identical: a bloom miss for value i means no row
+      // contains all of i's tokens, so phrase i matches no row
+      if (!h_bloom_all(ctx, ch, f.common_hashes)) {
+        lb.mode = kModeNone;
+        return;
+      }
+      std::vector<std::string> survivors;
+      {
+        const auto& words = ctx.bloom_host(ch);
+        for (size_t i = 0; i < f.values.size(); i++) {
+          if (bloom_contains_all(words.data(), words.size(),
+                                 f.set_hashes[i].data(),
+                                 f.set_hashes[i].size())) {
+            survivors.push_back(f.values[i]);
+          }
+        }
+      }
+      if (survivors.empty()) {
+        lb.mode = kModeNone;
+        return;
+      }
+      const StagedStrCol& sc = ctx.stage_column(ch);
+      if (sc.is_const) {
+        std::string str = ch.type == ValueType::String
+                              ? sc.const_value
+                              : format_encoded(ch.type, strview(sc.const_value));
+        lb.mode = h_match_any_phrase(strview(str), survivors) ? kModeAll
+                                                              : kModeNone;
+        return;
+      }
+      bytes blob = serialize_phrases(survivors);
+      lb.mode = kModeScan;
+      lb.operand = (const uint8_t*)st.push(blob.data(), blob.size(), 8);
+      lb.operand_len = uint32_t(blob.size());
+      lb.data = sc.d_data;
+      if (ch.type == ValueType::String) {
+        lb.kind = kScanAnyPhraseStr;
+        lb.offsets = sc.d_offsets;
+      } else {
+        lb.kind = kScanAnyPhraseFmt;
+        lb.width = width_of_type(ch.type);
+        lb.flags = uint8_t(fmt_of_type(ch.type) << 4);
+      }
+      return;
+    }
+
+    case FilterNode::ContainsAll: {
+      // filterContainsAll (filter_contains_all.go:123-321)
+      bool only_empty = f.values.size() == 1 && f.values[0].empty();
+      if (f.values.empty() || only_empty) {
+        lb.mode = kModeAll;
+        return;
+      }
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = h_match_all_phrases(strview(cv), f.values) ? kModeAll
+                                                             : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = h_match_all_phrases(strview("", 0), f.values) ? kModeAll
+                                                                : kModeNone;
+        return;
+      }
+      if (ch.type == ValueType::Dict) {
+        uint32_t mask = dict_mask_of(ch.dict, [&](strview dv) {
+          return h_match_all_phrases(dv, f.values);
+        });
+        stage_dict(lb, ctx, ch, mask);
+        return;
+      }
+      if (ch.type == ValueType::Uint8 || ch.type == ValueType::Uint16 ||
+          ch.type == ValueType::Uint32 || ch.type == ValueType::Uint64) {
+        // matchAllValues (filter_contains_all.go:183-204)
+        std::vector<std::string> distinct;
+        for (const auto& v : f.values) {
+          if (!v.empty() &&
+              std::find(distinct.begin(), distinct.end(), v) == distinct.end()) {
+            distinct.push_back(v);
+          }
+        }
+        if (distinct.empty()) {
+          lb.mode = kModeAll;
+          return;
+        }
+        const auto& set = f.bin_sets[size_t(bin_set_slot(ch.type))];
+        if (distinct.size() != 1 || set.size() != 1 ||
+            !h_bloom_all(ctx, ch, f.all_hashes)) {
+          lb.mode = kModeNone;
+          return;
+        }
+        bytes bin(set[0].begin(), set[0].end());
+        stage_eq_bin_nogate(lb, ctx, ch, st, bin);
+        return;
+      }
+      if (!h_bloom_all(ctx, ch, f.all_hashes)) {
+        lb.mode = kModeNone;
+        return;
+      }
+      const StagedStrCol& sc = ctx.stage_column(ch);
+      if (sc.is_const) {
+        std::string str = ch.type == ValueType::String
+                              ? sc.const_value
+                              : format_encoded(ch.type, strview(sc.const_value));
+        lb.mode = h_match_all_phrases(strview(str), f.values) ? kModeAll
+                                                              : kModeNone;
+        return;
+      }
+      lb.mode = kModeScan;
+      lb.operand = li.d_operand;  // serialized phrase list
+      lb.operand_len = uint32_t(li.operand.size());
+      lb.data = sc.d_data;
+      if (ch.type == ValueType::String) {
+        lb.kind = kScanAllPhrasesStr;
+        lb.offsets = sc.d_offsets;
+      } else {
+        lb.kind = kScanAllPhrasesFmt;
+        lb.width = width_of_type(ch.type);
+        lb.flags = uint8_t(fmt_of_type(ch.type) << 4);
+      }
+      return;
+    }
+
+    case FilterNode::StringRange: {
+      // filterStringRange (filter_string_range.go:47-230)
+      if (f.min_s > f.max_s) {
+        lb.mode = kModeNone;
+        return;
+      }
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = h_match_string_range(strview(cv), f.min_s, f.max_s)
+                      ? kModeAll
+                      : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = h_match_string_range(strview("", 0), f.min_s, f.max_s)
+                      ? kModeAll
+                      : kModeNone;
+        return;
+      }
+      // per-type prunes (filter_string_range.go:100-225)
+      switch (ch.type) {
+        case ValueType::String:
+        case ValueType::Dict:
+          break;
+        case ValueType::Int64:
+          if ((f.min_s != "-" && f.min_s > "9") ||
+              (f.max_s != "-" && f.max_s < "0")) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::Float64:
+          if (f.min_s > "9" || f.max_s < "+") {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        default:
+          if (f.min_s > "9" || f.max_s < "0") {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+      }
+      if (ch.type == ValueType::Dict) {
+        uint32_t mask = dict_mask_of(ch.dict, [&](strview dv) {
+          return h_match_string_range(dv, f.min_s, f.max_s);
+        });
+        stage_dict(lb, ctx, ch, mask);
+        return;
+      }
+      const StagedStrCol& sc = ctx.stage_column(ch);
+      if (sc.is_const) {
+        std::string str = ch.type == ValueType::String
+                              ? sc.const_value
+                              : format_encoded(ch.type, strview(sc.const_value));
+        lb.mode = h_match_string_range(strview(str), f.min_s, f.max_s)
+                      ? kModeAll
+                      : kModeNone;
+        return;
+      }
+      lb.mode = kModeScan;
+      lb.operand = li.d_operand;  // u32 minlen, u32 maxlen, min, max
+      lb.operand_len = uint32_t(li.operand.size());
+      lb.data = sc.d_data;
+      if (ch.type == ValueType::String) {
+        lb.kind = kScanStrRange;
+        lb.offsets = sc.d_offsets;
+      } else {
+        lb.kind = kScanStrRangeFmt;
+        lb.width = width_of_type(ch.type);
+        lb.flags = uint8_t(fmt_of_type(ch.type) << 4);
+      }
+      return;
+    }
+
+    case FilterNode::IPv4Range: {
+      // filterIPv4Range (filter_ipv4_range.go:99-190)
+      uint32_t mn = uint32_t(f.min_u), mx = uint32_t(f.max_u);
+      if (mn > mx) {
+        lb.mode = kModeNone;
+        return;
+      }
+      auto match_str = [&](strview v) {
+        uint32_t ip;
+        if (!try_parse_ipv4(v, &ip)) return false;
+        return ip >= mn && ip <= mx;
+      };
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = match_str(strview(cv)) ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = kModeNone;
+        return;
+      }
+      switch (ch.type) {
+        case ValueType::String: {
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            lb.mode = match_str(strview(sc.const_value)) ? kModeAll : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanIPv4RangeStr;
+          lb.vmin = mn;
+          lb.vmax = mx;
+          lb.data = sc.d_data;
+          lb.offsets = sc.d_offsets;
+          return;
+        }
+        case ValueType::Dict: {
+          uint32_t mask = dict_mask_of(ch.dict, match_str);
+          stage_dict(lb, ctx, ch, mask);
+          return;
+        }
+        case ValueType::IPv4: {
+          // matchIPv4ByRange (filter_ipv4_range.go:166-181)
+          if (ch.min_value > mx || ch.max_value < mn) {
+            lb.mode = kModeNone;
+            return;
+          }
+          const StagedStrCol& sc = ctx.stage_column(ch);
+          if (sc.is_const) {
+            uint32_t ip = get_u32be((const uint8_t*)sc.const_value.data());
+            lb.mode = (ip >= mn && ip <= mx) ? kModeAll : kModeNone;
+            return;
+          }
+          lb.mode = kModeScan;
+          lb.kind = kScanIPv4RangeBin;
+          lb.width = 4;
+          lb.vmin = mn;
+          lb.vmax = mx;
+          lb.data = sc.d_data;
+          return;
+        }
+        default:
+          lb.mode = kModeNone;
+          return;
+      }
+    }
+
+    case FilterNode::LenRange: {
+      // filterLenRange (filter_len_range.go:126-348)
+      uint64_t mn = f.min_u, mx = f.max_u;
+      if (mn > mx) {
+        lb.mode = kModeNone;
+        return;
+      }
+      auto match_len = [&](strview v) {
+        uint64_t n = h_rune_count(v);
+        return n >= mn && n <= mx;
+      };
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = match_len(strview(cv)) ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = match_len(strview("", 0)) ? kModeAll : kModeNone;
+        return;
+      }
+      // per-type prunes (filter_len_range.go:180-331)
+      auto minmax_len_ok = [&]() {
+        std::string s2;
+        format_uint64(s2, ch.min_value);
+        if (mx < s2.size()) return false;
+        s2.clear();
+        format_uint64(s2, ch.max_value);
+        return mn <= s2.size();
+      };
+      switch (ch.type) {
+        case ValueType::String:
+          break;
+        case ValueType::Dict: {
+          uint32_t mask = dict_mask_of(ch.dict, match_len);
+          stage_dict(lb, ctx, ch, mask);
+          return;
+        }
+        case ValueType::Uint8:
+          if (mn > 3 || mx == 0 || !minmax_len_ok()) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::Uint16:
+          if (mn > 5 || mx == 0 || !minmax_len_ok()) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::Uint32:
+          if (mn > 10 || mx == 0 || !minmax_len_ok()) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::Uint64:
+          if (mn > 20 || mx == 0 || !minmax_len_ok()) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::Int64:
+          if (mn > 20 || mx == 0) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::Float64:
+          if (mn > 24 || mx == 0) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::IPv4:
+          if (mn > 15 || mx < 7) {
+            lb.mode = kModeNone;
+            return;
+          }
+          break;
+        case ValueType::TimestampISO8601:
+          // formatted length is always 24 (filter_len_range.go:180-185)
+          lb.mode = (mn <= 24 && mx >= 24) ? kModeAll : kModeNone;
+          return;
+        default:
+          fail("unknown valueType while staging len_range filter");
+      }
+      const StagedStrCol& sc = ctx.stage_column(ch);
+      if (sc.is_const) {
+        std::string str = ch.type == ValueType::String
+                              ? sc.const_value
+                              : format_encoded(ch.type, strview(sc.const_value));
+        lb.mode = match_len(strview(str)) ? kModeAll : kModeNone;
+        return;
+      }
+      lb.mode = kModeScan;
+      lb.vmin = mn;
+      lb.vmax = mx;
+      lb.data = sc.d_data;
+      if (ch.type == ValueType::String) {
+        lb.kind = kScanLenRangeStr;
+        lb.offsets = sc.d_offsets;
+      } else {
+        lb.kind = kScanLenRangeFmt;
+        lb.width = width_of_type(ch.type);
+        lb.flags = uint8_t(fmt_of_type(ch.type) << 4);
+      }
+      return;
+    }
+
+    case FilterNode::DayRange:
+    case FilterNode::WeekRange: {
+      // filterDayRange / filterWeekRange (filter_day_range.go:126-139,
+      // filter_week_range.go:128-141)
+      int64_t start = int64_t(f.min_u), end = int64_t(f.max_u);
+      if (start > end) {
+        lb.mode = kModeNone;
+        return;
+      }
+      const bool is_day = f.type == FilterNode::DayRange;
+      const int64_t full_end = is_day ? 24LL * 3600 * 1000000000 - 1 : 6;
+      if (start == 0 && end == full_end) {
+        lb.mode = kModeAll;
+        return;
+      }
+      lb.mode = kModeScan;
+      lb.kind = is_day ? kScanDayRange : kScanWeekRange;
+      lb.ts = ctx.stage_timestamps();
+      lb.vmin = uint64_t(start);
+      lb.vmax = uint64_t(end);
+      lb.operand = li.d_operand;  // 8-byte tz offset
+      lb.operand_len = 8;
+      return;
+    }
+
+    case FilterNode::ValueTypeFilter: {
+      // filterValueType (filter_value_type.go:44-67)
+      std::string cv = const_val();
+      if (!cv.empty()) {
+        lb.mode = f.min_s == "const" ? kModeAll : kModeNone;
+        return;
+      }
+      ColumnHeader ch;
+      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
+        lb.mode = kModeNone;
+        return;
+      }
+      lb.mode = f.min_s == value_type_name(ch.type) ? kModeAll : kModeNone;
+      return;
+    }
+
+    case FilterNode::StreamIdFilter: {
+      // filterStreamID block gate (filter_stream_id.go:127-143)
+      const StreamID& sid = ctx.bh->stream_id;
+      uint64_t tp = uint64_t(sid.account_id) << 32 | sid.project_id;
+      for (const auto& id : f.stream_ids) {
+        if (id[0] == tp && id[1] == sid.id_hi && id[2] == sid.id_lo) {
+          lb.mode = kModeAll;
+          return;
+        }
+      }
+      lb.mode = kModeNone;
+      return;
+    }
+
     default:
       fail("stage_leaf: non-leaf node");
   }
@@ -1413,6 +2166,39 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
       case FilterNode::Sequence:
         li.operand = serialize_phrases(li.node->phrases);
         break;
+      case FilterNode::In: {
+        std::vector<std::string> sset = li.node->values;
+        std::sort(sset.begin(), sset.end());
+        sset.erase(std::unique(sset.begin(), sset.end()), sset.end());
+        li.operand = serialize_str_set(sset);
+        break;
+      }
+      case FilterNode::ContainsAll:
+        li.operand = serialize_phrases(li.node->values);
+        break;
+      case FilterNode::StringRange: {
+        bytes b;
+        auto put32 = [&](uint32_t v) {
+          b.push_back(uint8_t(v));
+          b.push_back(uint8_t(v >> 8));
+          b.push_back(uint8_t(v >> 16));
+          b.push_back(uint8_t(v >> 24));
+        };
+        put32(uint32_t(li.node->min_s.size()));
+        put32(uint32_t(li.node->max_s.size()));
+        b.insert(b.end(), li.node->min_s.begin(), li.node->min_s.end());
+        b.insert(b.end(), li.node->max_s.begin(), li.node->max_s.end());
+        li.operand = std::move(b);
+        break;
+      }
+      case FilterNode::DayRange:
+      case FilterNode::WeekRange: {
+        bytes b(8);
+        int64_t off = li.node->tz_offset;
+        memcpy(b.data(), &off, 8);
+        li.operand = std::move(b);
+        break;
+      }
       default:
         break;
     }
@@ -1464,7 +2250,13 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
           case kScanRangeStr:
           case kScanPrefixStr:
           case kScanExactPrefixStr:
-          case kScanSeqStr: {
+          case kScanSeqStr:
+          case kScanInStr:
+          case kScanAnyPhraseStr:
+          case kScanAllPhrasesStr:
+          case kScanStrRange:
+          case kScanIPv4RangeStr:
+          case kScanLenRangeStr: {
             const StagedStrCol& sc = ctx.cols.at(leaf_infos[size_t(l)].cname);
             st->algo_bytes += sc.data_bytes + (bh.rows_count + 1) * 4;
             break;
