@@ -8,6 +8,7 @@
 
 #include "../victorialogs_amd/csrc/core/bloom.h"
 #include "../victorialogs_amd/csrc/core/match.h"
+#include "../victorialogs_amd/csrc/core/unicode_case.h"
 #include "../victorialogs_amd/csrc/core/values.h"
 
 namespace vl {
@@ -1628,6 +1629,61 @@ static bool or_match_bloom(const FilterNode& f, BlockCtx& ctx) {
   return false;
 }
 
+
+// ---- filterAnyCasePhrase / filterAnyCasePrefix
+//      (filter_any_case_phrase.go:85-158, filter_any_case_prefix.go:90-160) ----
+
+static void apply_phrase(const FilterNode& f, BlockCtx& ctx, Bitmap& bm);
+static void apply_prefix(const FilterNode& f, BlockCtx& ctx, Bitmap& bm);
+
+static void apply_any_case(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  const bool is_ph = f.type == FilterNode::AnyCasePhrase;
+  std::string name = canonical_field(f.field);
+  strview lower(f.min_s);
+  auto host_match = [&](strview v) {
+    return is_ph ? match_any_case_phrase(v, lower)
+                 : match_any_case_prefix(v, lower);
+  };
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    if (!host_match(strview(cv))) bm.reset_bits();
+    return;
+  }
+  ColumnHeader ch;
+  if (!ctx.column_header(name, &ch)) {
+    // phrase matches missing columns only when empty; prefix never does
+    if (!is_ph || lower.n > 0) bm.reset_bits();
+    return;
+  }
+  switch (ch.type) {
+    case ValueType::String:
+      visit_values(ctx, ch, bm, host_match);
+      return;
+    case ValueType::Dict: {
+      std::vector<uint8_t> enc;
+      for (const auto& dv : ch.dict) enc.push_back(host_match(strview(dv)) ? 1 : 0);
+      match_encoded_dict(ctx, ch, bm, enc);
+      return;
+    }
+    default: {
+      // numeric/ip/iso: same matchers as phrase/prefix with the lowercase
+      // (iso: uppercase) pattern (filter_any_case_phrase.go:116-137)
+      const bool iso = ch.type == ValueType::TimestampISO8601;
+      FilterNode tmp;
+      tmp.type = is_ph ? FilterNode::Phrase : FilterNode::Prefix;
+      tmp.field = f.field;
+      tmp.phrase = iso ? f.max_s : f.min_s;
+      tmp.token_hashes = iso ? f.all_hashes : f.token_hashes;
+      if (is_ph) {
+        apply_phrase(tmp, ctx, bm);
+      } else {
+        apply_prefix(tmp, ctx, bm);
+      }
+      return;
+    }
+  }
+}
+
 void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
   switch (f.type) {
     case FilterNode::Phrase:
@@ -1677,6 +1733,10 @@ void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
       return;
     case FilterNode::StreamIdFilter:
       apply_stream_id(f, ctx, bm);
+      return;
+    case FilterNode::AnyCasePhrase:
+    case FilterNode::AnyCasePrefix:
+      apply_any_case(f, ctx, bm);
       return;
     case FilterNode::Time:
       apply_time(f, ctx, bm);

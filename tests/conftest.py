@@ -294,6 +294,27 @@ TYPED_FILTERS = [
     '{"type":"stream_id","ids":[{"account":0,"project":0,"hi":"0","lo":"1"},'
     '{"account":0,"project":0,"hi":"9","lo":"9"}]}',
     '{"type":"stream_id","ids":[]}',
+    # ---- any-case filters (filter_any_case_phrase.go, filter_any_case_prefix.go) ----
+    '{"type":"any_case_phrase","field":"lvl","phrase":"ERROR"}',
+    '{"type":"any_case_phrase","field":"lvl","phrase":"error"}',
+    '{"type":"any_case_phrase","field":"_msg","phrase":"LEVEL=WARN"}',
+    '{"type":"any_case_phrase","field":"_msg","phrase":"Other STREAM"}',
+    '{"type":"any_case_phrase","field":"uni","phrase":"ДВА"}',
+    '{"type":"any_case_phrase","field":"uni","phrase":"FOO"}',
+    '{"type":"any_case_phrase","field":"u8","phrase":"13"}',
+    '{"type":"any_case_phrase","field":"f64","phrase":"-18.625"}',
+    '{"type":"any_case_phrase","field":"ip","phrase":"10.5.15.35"}',
+    '{"type":"any_case_phrase","field":"iso","phrase":"2024-01-03t02:02:06.002z"}',
+    '{"type":"any_case_phrase","field":"constcol","phrase":"FIXED Value"}',
+    '{"type":"any_case_phrase","field":"missing_col","phrase":""}',
+    '{"type":"any_case_prefix","field":"lvl","prefix":"ERR"}',
+    '{"type":"any_case_prefix","field":"lvl","prefix":"err"}',
+    '{"type":"any_case_prefix","field":"_msg","prefix":"LOG Line"}',
+    '{"type":"any_case_prefix","field":"uni","prefix":"РАЗ"}',
+    '{"type":"any_case_prefix","field":"u8","prefix":"1"}',
+    '{"type":"any_case_prefix","field":"iso","prefix":"2024-01-0"}',
+    '{"type":"any_case_prefix","field":"mix","prefix":"1.5g"}',
+    '{"type":"any_case_prefix","field":"missing_col","prefix":""}',
     '{"type":"and","filters":['
     '{"type":"in","field":"lvl","values":["error","fatal"]},'
     '{"type":"len_range","field":"_msg","min":1,"max":40},'

@@ -28,7 +28,7 @@ GEN_REGEXES = [
 
 TYPED_PHRASES = [
     "level=error", "took", "row", "warn", "ERROR", "13", "два", "value",
-    "1.5", "200ms", "absent", "",
+    "1.5", "200ms", "absent", "", "ДВА", "Foo", "LEVEL",
 ]
 TYPED_FIELDS_STR = ["_msg", "lvl", "uni", "mix", "constcol", "nope"]
 TYPED_FIELDS_NUM = ["u8", "u16", "u32", "u64", "i64", "f64", "ip", "iso"]
@@ -46,7 +46,7 @@ def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
                   "prefix", "exact_prefix", "sequence",
                   "in", "contains_any", "contains_all", "string_range",
                   "ipv4_range", "len_range", "day_range", "week_range",
-                  "value_type"]
+                  "value_type", "any_case_phrase", "any_case_prefix"]
     kind = rng.choice(
         leaf_kinds + ["and", "or", "not"] if depth > 0 else leaf_kinds)
     if kind == "phrase":
@@ -57,6 +57,14 @@ def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
         return {"type": "exact",
                 "field": rng.choice(fields_str + fields_num),
                 "value": rng.choice(phrases)}
+    if kind in ("any_case_phrase",):
+        return {"type": kind,
+                "field": rng.choice(fields_str + fields_num),
+                "phrase": rng.choice(phrases + [p.upper() for p in phrases])}
+    if kind == "any_case_prefix":
+        return {"type": kind,
+                "field": rng.choice(fields_str + fields_num),
+                "prefix": rng.choice(phrases + [p.title() for p in phrases])}
     if kind in ("prefix", "exact_prefix"):
         return {"type": kind,
                 "field": rng.choice(fields_str + fields_num),

@@ -9,6 +9,7 @@
 #include "bloom.h"
 #include "json.h"
 #include "match.h"
+#include "unicode_case.h"
 #include "values.h"
 #include "tokenizer.h"
 
@@ -321,6 +322,22 @@ FilterNode build(const JValue& v) {
       uint64_t lo = strtoull(jget(sj, "lo").str.c_str(), nullptr, 10);
       n.stream_ids.push_back({acct << 32 | proj, hi, lo});
     }
+  } else if (type == "any_case_phrase" || type == "any_case_prefix") {
+    // filterAnyCasePhrase / filterAnyCasePrefix
+    // (filter_any_case_phrase.go:19-62, filter_any_case_prefix.go:21-66)
+    bool is_phrase = type == "any_case_phrase";
+    n.type = is_phrase ? FilterNode::AnyCasePhrase : FilterNode::AnyCasePrefix;
+    n.field = jget(v, "field").str;
+    n.phrase = jget(v, is_phrase ? "phrase" : "prefix").str;
+    n.min_s = to_lower_str(strview(n.phrase));
+    n.max_s = to_upper_str(strview(n.phrase));
+    std::vector<std::string> toks =
+        is_phrase ? tokenize_strings({n.phrase}) : get_tokens_skip_last(n.phrase);
+    n.tokens = toks;
+    n.token_hashes = probe_hashes(toks);
+    std::vector<std::string> toks_up;
+    for (const auto& t : toks) toks_up.push_back(to_upper_str(strview(t)));
+    n.all_hashes = probe_hashes(toks_up);
   } else if (type == "noop") {
     n.type = FilterNode::Noop;
   } else {

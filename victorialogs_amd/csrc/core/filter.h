@@ -30,7 +30,8 @@ struct FilterNode {
     Phrase, Exact, Regexp, And, Or, Not, Time, Range, Noop,
     Prefix, ExactPrefix, Sequence,
     In, ContainsAny, ContainsAll, StringRange, IPv4Range, LenRange,
-    DayRange, WeekRange, ValueTypeFilter, StreamIdFilter
+    DayRange, WeekRange, ValueTypeFilter, StreamIdFilter,
+    AnyCasePhrase, AnyCasePrefix
   } type;
 
   std::string field;   // phrase/exact/regexp/range (as written in the query)
@@ -42,6 +43,7 @@ struct FilterNode {
   std::vector<FilterNode> children;  // And/Or (n), Not (1)
   std::vector<std::string> values;   // In/ContainsAny/ContainsAll
   std::string min_s, max_s;          // StringRange; ValueTypeFilter: min_s=type
+                                     // AnyCase*: min_s=lowercase, max_s=uppercase
   uint64_t min_u = 0, max_u = 0;     // IPv4Range/LenRange; Day/WeekRange: start/end
   int64_t tz_offset = 0;             // Day/WeekRange offset (nsecs)
   std::vector<std::array<uint64_t, 3>> stream_ids;  // {acct<<32|proj, hi, lo}
@@ -50,6 +52,7 @@ struct FilterNode {
   std::vector<uint64_t> common_hashes;             // probe hashes, common tokens
   std::vector<std::vector<uint64_t>> set_hashes;   // per-value probe hashes
   // ContainsAll: tokensHashesAll (in_values.go:94-102)
+  // AnyCase*: uppercase token hashes (filter_any_case_phrase.go:53-62)
   std::vector<uint64_t> all_hashes;
   // In: per-type binary value sets, sorted (in_values.go:141-315); index by
   // width slot: [0]=u8 [1]=u16 [2]=u32 [3]=u64 [4]=i64 [5]=f64 [6]=ipv4 [7]=iso

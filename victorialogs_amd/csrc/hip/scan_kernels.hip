@@ -83,6 +83,27 @@ struct TileAcc {
   __device__ __forceinline__ uint8_t u8(long off) const { return tile[swz(off)]; }
 };
 
+// Byte-wise ASCII tolower view over another accessor.  Only correct for
+// ASCII bytes; rows containing non-ASCII bytes are resolved on the host via
+// the override bitmaps (DevLeafBlock.ovr_mask), so corruption of >=0x80
+// bytes is harmless.  The SWAR form is carry-safe across bytes.
+template <typename A>
+struct LowerAcc {
+  A a;
+  __device__ __forceinline__ uint64_t u64a(long off) const {
+    uint64_t x = a.u64a(off);
+    uint64_t low7 = x & 0x7F7F7F7F7F7F7F7FULL;
+    uint64_t ge_a = low7 + 0x3F3F3F3F3F3F3F3FULL;   // high bit: byte >= 0x41
+    uint64_t ge_z1 = low7 + 0x2525252525252525ULL;  // high bit: byte >= 0x5B
+    uint64_t is_az = ge_a & ~ge_z1 & ~x & 0x8080808080808080ULL;
+    return x | (is_az >> 2);
+  }
+  __device__ __forceinline__ uint8_t u8(long off) const {
+    uint8_t c = a.u8(off);
+    return uint8_t(c - 'A') < 26 ? uint8_t(c + 0x20) : c;
+  }
+};
+
 // Go utf8.DecodeRuneInString semantics (0xFFFD,1 on invalid), reading bytes
 // [s, s+n) of the accessor at base offset `off`.
 template <typename A>
@@ -978,6 +999,23 @@ __device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
       uint64_t n = d_rune_count(a, s0, sn);
       return n >= lb.vmin && n <= lb.vmax;
     }
+    case kScanAnyCasePhraseStr: {
+      // matchAnyCasePhrase (filter_any_case_phrase.go:159-181); operand is
+      // the lowercase phrase, flags from the lowercase phrase
+      if (lb.operand_len == 0) return sn == 0;
+      if (long(lb.operand_len) > sn) return false;
+      LowerAcc<A> la{a};
+      return d_get_phrase_pos_at(la, s0, sn, lb.operand, lb.operand_len,
+                                 lb.flags & 15) >= 0;
+    }
+    case kScanAnyCasePrefixStr: {
+      // matchAnyCasePrefix (filter_any_case_prefix.go:161-183)
+      if (lb.operand_len == 0) return sn > 0;
+      if (long(lb.operand_len) > sn) return false;
+      LowerAcc<A> la{a};
+      return d_match_prefix_at(la, s0, sn, lb.operand, lb.operand_len,
+                               lb.flags & 15);
+    }
     default:  // kScanRegexStr
       return d_regex_match_at(lb.operand, a, s0, sn);
   }
@@ -1191,7 +1229,8 @@ __device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
          kind == kScanExactPrefixStr || kind == kScanSeqStr ||
          kind == kScanInStr || kind == kScanAnyPhraseStr ||
          kind == kScanAllPhrasesStr || kind == kScanStrRange ||
-         kind == kScanIPv4RangeStr || kind == kScanLenRangeStr;
+         kind == kScanIPv4RangeStr || kind == kScanLenRangeStr ||
+         kind == kScanAnyCasePhraseStr || kind == kScanAnyCasePrefixStr;
 }
 
 // ---- the program kernel ----
@@ -1333,6 +1372,12 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
             } else {
               GlobalAcc a{lb.data};
               pred = d_eval_string_row(lb, a, s, e_fix - s);
+            }
+            if (lb.ovr_mask != nullptr) {
+              const uint32_t row = g0 + uint32_t(lane);
+              if ((lb.ovr_mask[row >> 6] >> (row & 63)) & 1) {
+                pred = ((lb.ovr_val[row >> 6] >> (row & 63)) & 1) != 0;
+              }
             }
           }
           const uint64_t word = __ballot(pred);

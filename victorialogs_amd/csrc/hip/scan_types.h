@@ -63,6 +63,8 @@ enum ScanKind : uint8_t {
   kScanDayRange = 35,       // (ts - offset) % day in [start,end] (filter_day_range.go)
   kScanWeekRange = 36,      // weekday(ts - offset) in [start,end] (filter_week_range.go)
   kScanIPv4RangeBin = 37,   // BE u32 in [vmin,vmax] (filter_ipv4_range.go:166-181)
+  kScanAnyCasePhraseStr = 38,  // matchAnyCasePhrase, ASCII rows (filter_any_case_phrase.go:159-181)
+  kScanAnyCasePrefixStr = 39,  // matchAnyCasePrefix, ASCII rows (filter_any_case_prefix.go:161-183)
 };
 
 // format source for the *Fmt kinds, stored in flags bits 4..7
@@ -105,6 +107,11 @@ struct DevLeafBlock {
   const uint8_t* data;     // column payload (bytes / fixed-width / codes)
   const uint32_t* offsets; // strings: u32[rows+1]
   const int64_t* ts;       // kScanTsRange
+  // any-case kinds: rows with non-ASCII bytes are resolved on the host
+  // (stringsutil.AppendLowercase needs full Unicode simple case mapping);
+  // bit set in ovr_mask => the row's result is the ovr_val bit
+  const uint64_t* ovr_mask;
+  const uint64_t* ovr_val;
   uint64_t vmin, vmax;     // range bounds (bit pattern for kScanRangeF)
 };
 
