@@ -39,8 +39,12 @@ long long vql_part_rows(void* part);
 long vql_block_rows(void* part, long block);
 
 /* Compiles a JSON filter tree (same shapes filter_test.go:34-59 builds
- * programmatically).  Node types: phrase, exact, regexp, and, or, not, time,
- * range, noop. */
+ * programmatically).  Node types: phrase, exact, regexp, prefix,
+ * exact_prefix, sequence, any_case_phrase, any_case_prefix, in,
+ * contains_any, contains_all, string_range, ipv4_range, len_range,
+ * day_range, week_range, value_type, stream_id, and, or, not, time, range,
+ * noop — every applyToBlockSearch filter except filter_stream (needs
+ * indexdb). */
 void* vql_compile_filter(const char* json);
 void vql_free_filter(void* filter);
 
