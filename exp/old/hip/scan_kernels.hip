@@ -20,7 +20,6 @@
 // (BASELINE.json north star).
 #include <hip/hip_runtime.h>
 
-#include "../core/parse_float.h"
 #include "../core/ryu.h"
 #include "scan_types.h"
 
@@ -82,27 +81,6 @@ struct TileAcc {
     return *(const uint64_t*)(tile + swz(off));
   }
   __device__ __forceinline__ uint8_t u8(long off) const { return tile[swz(off)]; }
-};
-
-// Byte-wise ASCII tolower view over another accessor.  Only correct for
-// ASCII bytes; rows containing non-ASCII bytes are resolved on the host via
-// the override bitmaps (DevLeafBlock.ovr_mask), so corruption of >=0x80
-// bytes is harmless.  The SWAR form is carry-safe across bytes.
-template <typename A>
-struct LowerAcc {
-  A a;
-  __device__ __forceinline__ uint64_t u64a(long off) const {
-    uint64_t x = a.u64a(off);
-    uint64_t low7 = x & 0x7F7F7F7F7F7F7F7FULL;
-    uint64_t ge_a = low7 + 0x3F3F3F3F3F3F3F3FULL;   // high bit: byte >= 0x41
-    uint64_t ge_z1 = low7 + 0x2525252525252525ULL;  // high bit: byte >= 0x5B
-    uint64_t is_az = ge_a & ~ge_z1 & ~x & 0x8080808080808080ULL;
-    return x | (is_az >> 2);
-  }
-  __device__ __forceinline__ uint8_t u8(long off) const {
-    uint8_t c = a.u8(off);
-    return uint8_t(c - 'A') < 26 ? uint8_t(c + 0x20) : c;
-  }
 };
 
 // Go utf8.DecodeRuneInString semantics (0xFFFD,1 on invalid), reading bytes
@@ -615,324 +593,14 @@ __device__ bool d_try_parse_bytes(const A& a, long s0, long sn, long long* out) 
 
 // parseMathNumber subset (pipe_math.go:1066-1080; same legs as the host
 // parse_math_number in values.cpp -- float, duration, bytes; others NaN)
-// tryParseDateUint64 subset used by RFC3339 parsing (values_encoder.go:552+):
-// plain decimal digits, with the reference's 2-digit fast-path quirk of
-// checking only the first char (mirrored in host try_parse_date_uint64)
-template <typename A>
-__device__ bool d_parse_date_u64(const A& a, long s0, long sn, uint64_t* out) {
-  if (sn == 0 || sn > 18) return false;
-  if (sn == 2) {
-    uint8_t c0 = a.u8(s0);
-    if (c0 < '0' || c0 > '9') return false;
-    *out = 10 * uint64_t(c0 - '0') + uint64_t(uint8_t(a.u8(s0 + 1) - '0'));
-    return true;
-  }
-  uint64_t v = 0;
-  for (long i = 0; i < sn; i++) {
-    uint8_t c = a.u8(s0 + i);
-    if (c < '0' || c > '9') return false;
-    v = v * 10 + (c - '0');
-  }
-  *out = v;
-  return true;
-}
-
-__device__ inline int64_t d_days_from_civil(int64_t y, int m, int64_t d) {
-  y -= m <= 2;
-  int64_t era = (y >= 0 ? y : y - 399) / 400;
-  int64_t yoe = y - era * 400;
-  int64_t doy = (153 * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1;
-  int64_t doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
-  return era * 146097 + doe - 719468;
-}
-
-// tryParseTimestampSecs (values_encoder.go:469-550); consumes the leading
-// "YYYY-MM-DD[T ]hh:mm:ss", returns seconds + the consumed length
-template <typename A>
-__device__ bool d_parse_ts_secs(const A& a, long s0, long sn, int64_t* secs,
-                                long* consumed) {
-  if (sn < 19) return false;
-  uint64_t n;
-  if (a.u8(s0 + 4) != '-') return false;
-  if (!d_parse_date_u64(a, s0, 4, &n) || n < 1677 || n > 2262) return false;
-  int64_t year = int64_t(n);
-  long i = 5;
-  if (a.u8(s0 + i + 2) != '-') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n)) return false;
-  int64_t month = int64_t(n);
-  i += 3;
-  uint8_t delim = a.u8(s0 + i + 2);
-  if (delim != 'T' && delim != ' ') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n)) return false;
-  int64_t day = int64_t(n);
-  i += 3;
-  if (a.u8(s0 + i + 2) != ':') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n) || n > 60) return false;
-  int64_t hour = int64_t(n);
-  i += 3;
-  if (a.u8(s0 + i + 2) != ':') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n) || n > 60) return false;
-  int64_t minute = int64_t(n);
-  i += 3;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n) || n > 60) return false;
-  int64_t sec = int64_t(n);
-  i += 2;
-  // Go time.Date normalization: month/day may overflow their ranges
-  int64_t ym = (month - 1);
-  int64_t yadd = ym >= 0 ? ym / 12 : -((-ym + 11) / 12);
-  int64_t mo = ym - yadd * 12 + 1;
-  int64_t days = d_days_from_civil(year + yadd, int(mo), day);
-  *secs = days * 86400 + hour * 3600 + minute * 60 + sec;
-  *consumed = i;
-  return true;
-}
-
-// TryParseTimestampRFC3339Nano (values_encoder.go:340-381); no-timezone
-// inputs use a zero local offset (the runtime boxes are UTC)
-template <typename A>
-__device__ bool d_parse_rfc3339(const A& a, long s0, long sn, int64_t* out) {
-  if (sn < 19) return false;
-  int64_t secs;
-  long used;
-  if (!d_parse_ts_secs(a, s0, sn, &secs, &used)) return false;
-  long i = s0 + used, n = sn - used;
-  int64_t nsecs = secs * 1000000000;
-  // parseTimezoneOffset (values_encoder.go:383-406)
-  if (n > 0 && a.u8(i + n - 1) == 'Z') {
-    n--;
-  } else {
-    long tz = -1;
-    for (long k = n - 1; k >= 0; k--) {
-      uint8_t c = a.u8(i + k);
-      if (c == '+' || c == '-') {
-        tz = k;
-        break;
-      }
-    }
-    if (tz >= 0) {
-      long on = n - tz - 1;
-      if (on != 5 || a.u8(i + tz + 3) != ':') return false;
-      uint64_t hh, mm;
-      if (!d_parse_date_u64(a, i + tz + 1, 2, &hh) || hh > 24) return false;
-      if (!d_parse_date_u64(a, i + tz + 4, 2, &mm) || mm > 60) return false;
-      int64_t off = int64_t(hh) * 3600000000000LL + int64_t(mm) * 60000000000LL;
-      if (a.u8(i + tz) == '-') off = -off;
-      nsecs -= off;
-      n = tz;
-    }
-  }
-  if (n == 0) {
-    *out = nsecs;
-    return true;
-  }
-  if (a.u8(i) == '.') {
-    i++;
-    n--;
-  }
-  if (n > 9) return false;
-  uint64_t frac;
-  if (!d_parse_date_u64(a, i, n, &frac)) return false;
-  for (long k = n; k < 9; k++) frac *= 10;
-  *out = nsecs + int64_t(frac);
-  return true;
-}
-
-template <typename A>
-struct AccReader {
-  const A* a;
-  long s0;
-  __device__ uint8_t u8(long i) const { return a->u8(s0 + i); }
-};
-
 template <typename A>
 __device__ double d_parse_math_number(const A& a, long s0, long sn) {
   double f;
   if (sn > 0 && d_try_parse_float64(a, s0, sn, &f)) return f;
   long long v;
-  if (sn > 0 && d_try_parse_duration(a, s0, sn, &v)) return double(v);
-  if (sn > 0 && d_try_parse_bytes(a, s0, sn, &v)) return double(v);
-  AccReader<A> r{&a, s0};
-  if (sn > 0 && vl_pf::pf_is_likely_number(r, sn)) {
-    double d;
-    if (vl_pf::go_parse_float(r, sn, &d)) return d;
-    int64_t iv;
-    if (vl_pf::go_parse_int0(r, sn, &iv)) return double(iv);
-  }
-  int64_t ts;
-  if (d_parse_rfc3339(a, s0, sn, &ts)) return double(ts);
-  uint32_t ip;
-  if (d_try_parse_ipv4(a, s0, sn, &ip)) return double(ip);
+  if (d_try_parse_duration(a, s0, sn, &v)) return double(v);
+  if (d_try_parse_bytes(a, s0, sn, &v)) return double(v);
   return __builtin_nan("");
-}
-
-
-// value in a sorted string set: blob = u32 n, u32 offs[n+1], bytes
-template <typename A>
-__device__ bool d_in_sorted_str(const uint8_t* blob, const A& a, long s0, long sn) {
-  uint32_t n;
-  __builtin_memcpy(&n, blob, 4);
-  const uint8_t* offs = blob + 4;
-  const uint8_t* data = blob + 4 + size_t(n + 1) * 4;
-  auto off_at = [&](uint32_t i) {
-    uint32_t o;
-    __builtin_memcpy(&o, offs + size_t(i) * 4, 4);
-    return o;
-  };
-  uint32_t lo = 0, hi = n;
-  while (lo < hi) {
-    uint32_t mid = (lo + hi) / 2;
-    uint32_t mo = off_at(mid), ml = off_at(mid + 1) - mo;
-    // lexicographic compare set[mid] vs row
-    int c = 0;
-    long k = 0;
-    long lim = ml < uint32_t(sn) ? ml : uint32_t(sn);
-    for (; k < lim; k++) {
-      uint8_t cb = data[mo + k], rb = a.u8(s0 + k);
-      if (cb != rb) {
-        c = cb < rb ? -1 : 1;
-        break;
-      }
-    }
-    if (c == 0) c = long(ml) < sn ? -1 : (long(ml) > sn ? 1 : 0);
-    if (c == 0) return true;
-    if (c < 0) {
-      lo = mid + 1;
-    } else {
-      hi = mid;
-    }
-  }
-  return false;
-}
-
-// fixed-width value in a sorted packed binary set (bytewise order == value
-// order for BE encodings)
-__device__ bool d_in_sorted_bin(const uint8_t* vals, uint32_t n, uint8_t width,
-                                const uint8_t* p) {
-  uint32_t lo = 0, hi = n;
-  while (lo < hi) {
-    uint32_t mid = (lo + hi) / 2;
-    const uint8_t* m = vals + size_t(mid) * width;
-    int c = 0;
-    for (int k = 0; k < width; k++) {
-      if (m[k] != p[k]) {
-        c = m[k] < p[k] ? -1 : 1;
-        break;
-      }
-    }
-    if (c == 0) return true;
-    if (c < 0) {
-      lo = mid + 1;
-    } else {
-      hi = mid;
-    }
-  }
-  return false;
-}
-
-// matchAnyPhrase / matchAllPhrases over a serialized phrase list (the
-// sequence blob layout: u16 n, { u16 len, u8 flags, bytes })
-template <typename A>
-__device__ bool d_match_any_phrase_at(const A& a, long s0, long sn,
-                                      const uint8_t* blob) {
-  uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
-  const uint8_t* p = blob + 2;
-  for (uint16_t i = 0; i < n; i++) {
-    uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
-    uint8_t flags = p[2];
-    p += 3;
-    if (len == 0 ? sn == 0 : d_get_phrase_pos_at(a, s0, sn, p, len, flags) >= 0) {
-      return true;
-    }
-    p += len;
-  }
-  return false;
-}
-
-template <typename A>
-__device__ bool d_match_all_phrases_at(const A& a, long s0, long sn,
-                                       const uint8_t* blob) {
-  uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
-  const uint8_t* p = blob + 2;
-  for (uint16_t i = 0; i < n; i++) {
-    uint16_t len = uint16_t(p[0]) | uint16_t(p[1]) << 8;
-    uint8_t flags = p[2];
-    p += 3;
-    if (len != 0 && d_get_phrase_pos_at(a, s0, sn, p, len, flags) < 0) {
-      return false;  // empty phrases match everything (filter_contains_all.go:312-315)
-    }
-    p += len;
-  }
-  return true;
-}
-
-// matchStringRange: s >= min && s < max; blob = u32 minlen, u32 maxlen, bytes
-template <typename A>
-__device__ bool d_string_range_at(const A& a, long s0, long sn,
-                                  const uint8_t* blob) {
-  uint32_t mn_len, mx_len;
-  __builtin_memcpy(&mn_len, blob, 4);
-  __builtin_memcpy(&mx_len, blob + 4, 4);
-  const uint8_t* mn = blob + 8;
-  const uint8_t* mx = mn + mn_len;
-  auto cmp = [&](const uint8_t* b, uint32_t bl) {
-    long lim = long(bl) < sn ? long(bl) : sn;
-    for (long k = 0; k < lim; k++) {
-      uint8_t rb = a.u8(s0 + k);
-      if (rb != b[k]) return rb < b[k] ? -1 : 1;
-    }
-    return sn < long(bl) ? -1 : (sn > long(bl) ? 1 : 0);
-  };
-  return cmp(mn, mn_len) >= 0 && cmp(mx, mx_len) < 0;
-}
-
-template <typename A>
-__device__ uint64_t d_rune_count(const A& a, long s0, long sn) {
-  uint64_t n = 0;
-  for (long i = 0; i < sn; i++) {
-    if ((a.u8(s0 + i) & 0xC0) != 0x80) n++;
-  }
-  return n;
-}
-
-// tryParseIPv4 over an accessor (values_encoder.go:675-730)
-template <typename A>
-__device__ bool d_try_parse_ipv4(const A& a, long s0, long sn, uint32_t* out) {
-  if (sn < 7 || sn > 15) return false;
-  int dots = 0;
-  for (long i = 0; i < sn; i++) {
-    if (a.u8(s0 + i) == '.') dots++;
-  }
-  if (dots != 3) return false;
-  uint32_t ip = 0;
-  for (int oct = 0; oct < 4; oct++) {
-    long seg_end = s0 + sn;
-    if (oct < 3) {
-      long j = s0;
-      while (j < s0 + sn && a.u8(j) != '.') j++;
-      seg_end = j;
-    }
-    long len = seg_end - s0;
-    if (len <= 0 || len > 3) return false;
-    // tryParseDateUint64 two-digit fast path quirk: only first char checked
-    uint32_t v = 0;
-    if (len == 2) {
-      uint8_t c0 = a.u8(s0);
-      if (c0 < '0' || c0 > '9') return false;
-      v = 10 * uint32_t(c0 - '0') + uint32_t(uint8_t(a.u8(s0 + 1) - '0'));
-    } else {
-      for (long k = 0; k < len; k++) {
-        uint8_t c = a.u8(s0 + k);
-        if (c < '0' || c > '9') return false;
-        v = v * 10 + (c - '0');
-      }
-    }
-    if (v > 255) return false;
-    ip = ip << 8 | v;
-    s0 = seg_end + 1;
-    sn -= len + 1;
-  }
-  *out = ip;
-  return true;
 }
 
 // ---- regex fast paths on serialized blob (regex.go:86-212) ----
@@ -1096,14 +764,20 @@ __device__ __forceinline__ uint16_t d_get_u16be(const uint8_t* p) {
   return uint16_t(p[0]) << 8 | p[1];
 }
 
-// Cold string kinds, kept OUT of the hot scan loop: inlining every matcher
-// into the workgroup loop grew the loop body past the instruction cache and
-// cost ~18% on the phrase workload (798 vs 677 us/launch, profiles/r01b).
+// String-kind predicate over an accessor (tile or global).
 template <typename A>
-__device__ __noinline__ bool d_eval_string_row_cold(const DevLeafBlock& lb,
-                                                    const A& a, long s0,
-                                                    long sn) {
+__device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
+                                  long sn) {
   switch (lb.kind) {
+    case kScanPhraseStr:
+      return d_match_phrase_at(a, s0, sn, lb.operand, lb.operand_len, lb.flags);
+    case kScanEqStr: {
+      if (sn != long(lb.operand_len)) return false;
+      for (long k = 0; k < sn; k++) {
+        if (a.u8(s0 + k) != lb.operand[k]) return false;
+      }
+      return true;
+    }
     case kScanRangeStr: {
       // matchRange (filter_range.go:369-372)
       double x = d_parse_math_number(a, s0, sn);
@@ -1117,69 +791,39 @@ __device__ __noinline__ bool d_eval_string_row_cold(const DevLeafBlock& lb,
       return d_has_prefix_bytes(a, s0, sn, lb.operand, lb.operand_len);
     case kScanSeqStr:
       return d_match_sequence_at(a, s0, sn, lb.operand);
-    case kScanInStr:
-      return d_in_sorted_str(lb.operand, a, s0, sn);
-    case kScanAnyPhraseStr:
-      return d_match_any_phrase_at(a, s0, sn, lb.operand);
-    case kScanAllPhrasesStr:
-      return d_match_all_phrases_at(a, s0, sn, lb.operand);
-    case kScanStrRange:
-      return d_string_range_at(a, s0, sn, lb.operand);
-    case kScanIPv4RangeStr: {
-      uint32_t ip;
-      if (!d_try_parse_ipv4(a, s0, sn, &ip)) return false;
-      return ip >= uint32_t(lb.vmin) && ip <= uint32_t(lb.vmax);
-    }
-    case kScanLenRangeStr: {
-      uint64_t n = d_rune_count(a, s0, sn);
-      return n >= lb.vmin && n <= lb.vmax;
-    }
-    case kScanAnyCasePhraseStr: {
-      // matchAnyCasePhrase (filter_any_case_phrase.go:159-181); operand is
-      // the lowercase phrase, flags from the lowercase phrase
-      if (lb.operand_len == 0) return sn == 0;
-      if (long(lb.operand_len) > sn) return false;
-      LowerAcc<A> la{a};
-      return d_get_phrase_pos_at(la, s0, sn, lb.operand, lb.operand_len,
-                                 lb.flags & 15) >= 0;
-    }
-    case kScanAnyCasePrefixStr: {
-      // matchAnyCasePrefix (filter_any_case_prefix.go:161-183)
-      if (lb.operand_len == 0) return sn > 0;
-      if (long(lb.operand_len) > sn) return false;
-      LowerAcc<A> la{a};
-      return d_match_prefix_at(la, s0, sn, lb.operand, lb.operand_len,
-                               lb.flags & 15);
-    }
     default:  // kScanRegexStr
       return d_regex_match_at(lb.operand, a, s0, sn);
   }
 }
 
-// String-kind predicate over an accessor (tile or global); the two kinds the
-// steady-state workloads hammer stay inline, everything else is a call.
-template <typename A>
-__device__ __forceinline__ bool d_eval_string_row(const DevLeafBlock& lb,
-                                                  const A& a, long s0,
-                                                  long sn) {
-  if (lb.kind == kScanPhraseStr) {
-    return d_match_phrase_at(a, s0, sn, lb.operand, lb.operand_len, lb.flags);
-  }
-  if (lb.kind == kScanEqStr) {
-    if (sn != long(lb.operand_len)) return false;
-    for (long k = 0; k < sn; k++) {
-      if (a.u8(s0 + k) != lb.operand[k]) return false;
-    }
-    return true;
-  }
-  return d_eval_string_row_cold(lb, a, s0, sn);
-}
-
-// Cold fixed-width kinds (formatters, regex, parsers) behind a call so the
-// hot scan loop stays small (see d_eval_string_row_cold).
-__device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
-                                                   uint32_t row) {
+// Fixed-width / dict / timestamp predicate (coalesced global reads).
+__device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
   switch (lb.kind) {
+    case kScanEqBin: {
+      const uint8_t* p = lb.data + size_t(row) * lb.width;
+      switch (lb.width) {
+        case 1: return p[0] == lb.operand[0];
+        case 2: return p[0] == lb.operand[0] && p[1] == lb.operand[1];
+        case 4: {
+          uint32_t a, b;
+          __builtin_memcpy(&a, p, 4);
+          __builtin_memcpy(&b, lb.operand, 4);
+          return a == b;
+        }
+        default: {
+          uint64_t a, b;
+          __builtin_memcpy(&a, p, 8);
+          __builtin_memcpy(&b, lb.operand, 8);
+          return a == b;
+        }
+      }
+    }
+    case kScanDict:
+      return (lb.dict_mask >> lb.data[row]) & 1;
+    case kScanTsRange: {
+      int64_t v = lb.ts[row];
+      return v >= int64_t(lb.vmin) && v <= int64_t(lb.vmax);
+    }
     case kScanRangeU: {
       const uint8_t* p = lb.data + size_t(row) * lb.width;
       uint64_t v;
@@ -1265,39 +909,9 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
     }
-    case kScanInBin:
-      return d_in_sorted_bin(lb.operand, lb.operand_len / lb.width, lb.width,
-                             lb.data + size_t(row) * lb.width);
-    case kScanIPv4RangeBin: {
-      const uint8_t* p = lb.data + size_t(row) * 4;
-      uint32_t ip = uint32_t(p[0]) << 24 | uint32_t(p[1]) << 16 |
-                    uint32_t(p[2]) << 8 | p[3];
-      return ip >= uint32_t(lb.vmin) && ip <= uint32_t(lb.vmax);
-    }
-    case kScanDayRange: {
-      long long off_tz;
-      __builtin_memcpy(&off_tz, lb.operand, 8);
-      int64_t off = (lb.ts[row] - off_tz) % (24LL * 3600 * 1000000000);
-      return off >= int64_t(lb.vmin) && off <= int64_t(lb.vmax);
-    }
-    case kScanWeekRange: {
-      long long off_tz;
-      __builtin_memcpy(&off_tz, lb.operand, 8);
-      const int64_t day = 24LL * 3600 * 1000000000;
-      int64_t t = lb.ts[row] - off_tz;
-      int64_t days = t / day;
-      if (t % day < 0) days--;
-      int64_t wd = (days + 4) % 7;
-      if (wd < 0) wd += 7;
-      return wd >= int64_t(lb.vmin) && wd <= int64_t(lb.vmax);
-    }
     case kScanPrefixFmt:
     case kScanExactPrefixFmt:
-    case kScanSeqFmt:
-    case kScanAnyPhraseFmt:
-    case kScanAllPhrasesFmt:
-    case kScanStrRangeFmt:
-    case kScanLenRangeFmt: {
+    case kScanSeqFmt: {
       char buf[344];
       int n;
       switch (lb.flags >> 4) {
@@ -1329,25 +943,14 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
           break;
       }
       BufAcc a{(const uint8_t*)buf};
-      switch (lb.kind) {
-        case kScanPrefixFmt:
-          return d_match_prefix_at(a, 0, n, lb.operand, lb.operand_len,
-                                   lb.flags & 15);
-        case kScanExactPrefixFmt:
-          return d_has_prefix_bytes(a, 0, n, lb.operand, lb.operand_len);
-        case kScanAnyPhraseFmt:
-          return d_match_any_phrase_at(a, 0, n, lb.operand);
-        case kScanAllPhrasesFmt:
-          return d_match_all_phrases_at(a, 0, n, lb.operand);
-        case kScanStrRangeFmt:
-          return d_string_range_at(a, 0, n, lb.operand);
-        case kScanLenRangeFmt: {
-          uint64_t rc = d_rune_count(a, 0, n);
-          return rc >= lb.vmin && rc <= lb.vmax;
-        }
-        default:
-          return d_match_sequence_at(a, 0, n, lb.operand);
+      if (lb.kind == kScanPrefixFmt) {
+        return d_match_prefix_at(a, 0, n, lb.operand, lb.operand_len,
+                                 lb.flags & 15);
       }
+      if (lb.kind == kScanExactPrefixFmt) {
+        return d_has_prefix_bytes(a, 0, n, lb.operand, lb.operand_len);
+      }
+      return d_match_sequence_at(a, 0, n, lb.operand);
     }
     default:
       return false;
@@ -1357,149 +960,7 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
 __device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
   return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr ||
          kind == kScanRangeStr || kind == kScanPrefixStr ||
-         kind == kScanExactPrefixStr || kind == kScanSeqStr ||
-         kind == kScanInStr || kind == kScanAnyPhraseStr ||
-         kind == kScanAllPhrasesStr || kind == kScanStrRange ||
-         kind == kScanIPv4RangeStr || kind == kScanLenRangeStr ||
-         kind == kScanAnyCasePhraseStr || kind == kScanAnyCasePrefixStr;
-}
-
-// Fixed-width / dict / timestamp predicate (coalesced global reads); the
-// steady-state kinds stay inline, the rest is a call.
-__device__ __forceinline__ bool d_eval_fixed_row(const DevLeafBlock& lb,
-                                                 uint32_t row) {
-  switch (lb.kind) {
-    case kScanEqBin: {
-      const uint8_t* p = lb.data + size_t(row) * lb.width;
-      switch (lb.width) {
-        case 1: return p[0] == lb.operand[0];
-        case 2: return p[0] == lb.operand[0] && p[1] == lb.operand[1];
-        case 4: {
-          uint32_t a, b;
-          __builtin_memcpy(&a, p, 4);
-          __builtin_memcpy(&b, lb.operand, 4);
-          return a == b;
-        }
-        default: {
-          uint64_t a, b;
-          __builtin_memcpy(&a, p, 8);
-          __builtin_memcpy(&b, lb.operand, 8);
-          return a == b;
-        }
-      }
-    }
-    case kScanDict:
-      return (lb.dict_mask >> lb.data[row]) & 1;
-    case kScanTsRange: {
-      int64_t v = lb.ts[row];
-      return v >= int64_t(lb.vmin) && v <= int64_t(lb.vmax);
-    }
-    default:
-      return d_eval_fixed_row_cold(lb, row);
-  }
-}
-
-
-// The per-wave LDS-tiled string scan loop, templated on the row predicate so
-// the hot filters get their own clone with ONLY their matcher inlined (the
-// compiler stopped unswitching the kind dispatch out of this loop once the
-// kind count grew, costing ~13% on the phrase workload).
-template <typename EvalFn>
-__device__ __forceinline__ void d_string_tile_loop(
-    const DevLeafBlock& lb, uint8_t* wtile, uint64_t* out, uint32_t r0,
-    uint32_t r1, uint32_t nwords, int lane, int wave, int nwaves,
-    EvalFn eval) {
-  typedef uint32_t v4u __attribute__((ext_vector_type(4)));
-  v4u* dst = (v4u*)wtile;
-  uint32_t wd = wave;
-  // Offsets pipeline: each lane holds offsets[g0+lane] (clamped to g1);
-  // a lane's row end is the next lane's start (shfl), lane ng-1's end is
-  // the clamped value itself.  The next group's offsets are prefetched
-  // while this group's tile copy is in flight.
-  uint32_t o_lane = 0, o_end = 0;
-  if (wd < nwords) {
-    o_lane = lb.offsets[min(r0 + wd * 64 + uint32_t(lane), r1)];
-    if (lane == 0) o_end = lb.offsets[min(r0 + wd * 64 + 64, r1)];
-  }
-  while (wd < nwords) {
-    const uint32_t g0 = r0 + wd * 64;
-    const uint32_t g1 = min(g0 + 64, r1);
-    const uint32_t ng = g1 - g0;
-    const uint32_t byte0 =
-        uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
-    const uint32_t byte1 = uint32_t(__shfl(int(o_end), 0, 64));
-    const uint32_t nbytes = byte1 - byte0;
-    const bool use_tile = nbytes <= kWaveTileBytes;
-    if (use_tile) {
-      const v4u* src = (const v4u*)(lb.data + byte0);
-      const uint32_t n16 = (nbytes + 15) >> 4;
-      uint32_t k = lane;
-      // 8-deep batches: 8 independent loads in flight per lane
-      for (; k + 448 < n16; k += 512) {
-        // nt loads: each byte is read once per kernel; keep L2 for
-        // the offsets/bitmap traffic (cdna guide: nt-weights row)
-        v4u a0 = __builtin_nontemporal_load(src + k);
-        v4u a1 = __builtin_nontemporal_load(src + k + 64);
-        v4u a2 = __builtin_nontemporal_load(src + k + 128);
-        v4u a3 = __builtin_nontemporal_load(src + k + 192);
-        v4u a4 = __builtin_nontemporal_load(src + k + 256);
-        v4u a5 = __builtin_nontemporal_load(src + k + 320);
-        v4u a6 = __builtin_nontemporal_load(src + k + 384);
-        v4u a7 = __builtin_nontemporal_load(src + k + 448);
-        dst[k ^ ((k >> 4) & 15)] = a0;
-        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
-        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
-        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
-        dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
-        dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
-        dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
-        dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
-      }
-      for (; k + 192 < n16; k += 256) {
-        v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
-            a3 = src[k + 192];
-        dst[k ^ ((k >> 4) & 15)] = a0;
-        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
-        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
-        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
-      }
-      for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
-    }
-    // prefetch next group's offsets while the copy is in flight
-    const uint32_t next_wd = wd + nwaves;
-    uint32_t o_next = 0, o_end_next = 0;
-    if (next_wd < nwords) {
-      o_next = lb.offsets[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
-      if (lane == 0) o_end_next = lb.offsets[min(r0 + next_wd * 64 + 64, r1)];
-    }
-    if (use_tile) {
-      // every lane's ds_writes must land before cross-lane reads below
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    }
-    bool pred = false;
-    if (uint32_t(lane) < ng) {
-      const long s = o_lane;
-      const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
-      const long e_fix = uint32_t(lane) == ng - 1 ? long(byte1) : e;
-      if (use_tile) {
-        TileAcc a{wtile};
-        pred = eval(a, s - byte0, e_fix - s);
-      } else {
-        GlobalAcc a{lb.data};
-        pred = eval(a, s, e_fix - s);
-      }
-    }
-    uint64_t word = __ballot(pred);
-    if (lb.ovr_mask != nullptr) {
-      // host-resolved rows (non-ASCII any-case): merge at word level
-      const uint64_t mw = lb.ovr_mask[wd], vw = lb.ovr_val[wd];
-      word = (word & ~mw) | (vw & mw);
-    }
-    if (lane == 0) out[wd] = word;
-    o_lane = o_next;
-    o_end = o_end_next;
-    wd = next_wd;
-  }
+         kind == kScanExactPrefixStr || kind == kScanSeqStr;
 }
 
 // ---- the program kernel ----
@@ -1558,20 +1019,96 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
       }
 
       if (d_is_string_kind(lb.kind)) {
+        // Per-wave LDS-tiled string scan: each wave independently copies the
+        // bytes of a 64-row group into its private tile (coalesced 16 B
+        // loads) and matches lane-per-row; one wavefront ballot = one
+        // bitmap word.  No workgroup barriers on this path.
+        typedef uint32_t v4u __attribute__((ext_vector_type(4)));
         uint8_t* wtile = tile + wave * kWaveTileBytes;
-        if (lb.kind == kScanPhraseStr) {
-          // hot clone: only the phrase matcher in the loop body
-          d_string_tile_loop(lb, wtile, out, r0, r1, nwords, lane, wave,
-                             nwaves, [&](const auto& a, long s0, long sn) {
-                               return d_match_phrase_at(a, s0, sn, lb.operand,
-                                                        lb.operand_len,
-                                                        lb.flags);
-                             });
-        } else {
-          d_string_tile_loop(lb, wtile, out, r0, r1, nwords, lane, wave,
-                             nwaves, [&](const auto& a, long s0, long sn) {
-                               return d_eval_string_row(lb, a, s0, sn);
-                             });
+        v4u* dst = (v4u*)wtile;
+        uint32_t wd = wave;
+        // Offsets pipeline: each lane holds offsets[g0+lane] (clamped to g1);
+        // a lane's row end is the next lane's start (shfl), lane ng-1's end is
+        // the clamped value itself.  The next group's offsets are prefetched
+        // while this group's tile copy is in flight.
+        uint32_t o_lane = 0, o_end = 0;
+        if (wd < nwords) {
+          o_lane = lb.offsets[min(r0 + wd * 64 + uint32_t(lane), r1)];
+          if (lane == 0) o_end = lb.offsets[min(r0 + wd * 64 + 64, r1)];
+        }
+        while (wd < nwords) {
+          const uint32_t g0 = r0 + wd * 64;
+          const uint32_t g1 = min(g0 + 64, r1);
+          const uint32_t ng = g1 - g0;
+          const uint32_t byte0 =
+              uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
+          const uint32_t byte1 = uint32_t(__shfl(int(o_end), 0, 64));
+          const uint32_t nbytes = byte1 - byte0;
+          const bool use_tile = nbytes <= kWaveTileBytes;
+          if (use_tile) {
+            const v4u* src = (const v4u*)(lb.data + byte0);
+            const uint32_t n16 = (nbytes + 15) >> 4;
+            uint32_t k = lane;
+            // 8-deep batches: 8 independent loads in flight per lane
+            for (; k + 448 < n16; k += 512) {
+              // nt loads: each byte is read once per kernel; keep L2 for
+              // the offsets/bitmap traffic (cdna guide: nt-weights row)
+              v4u a0 = __builtin_nontemporal_load(src + k);
+              v4u a1 = __builtin_nontemporal_load(src + k + 64);
+              v4u a2 = __builtin_nontemporal_load(src + k + 128);
+              v4u a3 = __builtin_nontemporal_load(src + k + 192);
+              v4u a4 = __builtin_nontemporal_load(src + k + 256);
+              v4u a5 = __builtin_nontemporal_load(src + k + 320);
+              v4u a6 = __builtin_nontemporal_load(src + k + 384);
+              v4u a7 = __builtin_nontemporal_load(src + k + 448);
+              dst[k ^ ((k >> 4) & 15)] = a0;
+              dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+              dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+              dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+              dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
+              dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
+              dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
+              dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
+            }
+            for (; k + 192 < n16; k += 256) {
+              v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
+                  a3 = src[k + 192];
+              dst[k ^ ((k >> 4) & 15)] = a0;
+              dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+              dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+              dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+            }
+            for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
+          }
+          // prefetch next group's offsets while the copy is in flight
+          const uint32_t next_wd = wd + nwaves;
+          uint32_t o_next = 0, o_end_next = 0;
+          if (next_wd < nwords) {
+            o_next = lb.offsets[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
+            if (lane == 0) o_end_next = lb.offsets[min(r0 + next_wd * 64 + 64, r1)];
+          }
+          if (use_tile) {
+            // every lane's ds_writes must land before cross-lane reads below
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          }
+          bool pred = false;
+          if (uint32_t(lane) < ng) {
+            const long s = o_lane;
+            const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
+            const long e_fix = uint32_t(lane) == ng - 1 ? long(byte1) : e;
+            if (use_tile) {
+              TileAcc a{wtile};
+              pred = d_eval_string_row(lb, a, s - byte0, e_fix - s);
+            } else {
+              GlobalAcc a{lb.data};
+              pred = d_eval_string_row(lb, a, s, e_fix - s);
+            }
+          }
+          const uint64_t word = __ballot(pred);
+          if (lane == 0) out[wd] = word;
+          o_lane = o_next;
+          o_end = o_end_next;
+          wd = next_wd;
         }
       } else {
         for (uint32_t w = wave; w < nwords; w += nwaves) {

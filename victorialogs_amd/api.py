@@ -12,7 +12,9 @@ _ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def lib_path() -> str:
-    return os.path.join(_ROOT, "victorialogs_amd", "libvlogsql.so")
+    # VQL_LIB overrides for A/B kernel experiments (same box, two builds)
+    return os.environ.get("VQL_LIB") or os.path.join(
+        _ROOT, "victorialogs_amd", "libvlogsql.so")
 
 
 def oracle_lib_path() -> str:

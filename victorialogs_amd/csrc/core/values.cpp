@@ -1,5 +1,7 @@
 #include "values.h"
 
+#include "parse_float.h"
+
 #include <cmath>
 #include <cstdio>
 #include <cstdlib>
@@ -171,7 +173,7 @@ static int64_t days_from_civil(int64_t y, int m /*1..12*/, int64_t d) {
 
 // tryParseTimestampSecs (values_encoder.go:469-550)
 static bool try_parse_timestamp_secs(strview s, int64_t* secs_out, strview* tail) {
-  if (s.n < 17) return false;
+  if (s.n < 19) return false;
   if (s.p[4] != '-') return false;
   uint64_t n;
   if (!try_parse_date_uint64(strview(s.p, 4), &n) || n < 1677 || n > 2262) return false;
@@ -371,14 +373,91 @@ bool try_parse_bytes(strview s, int64_t* out) {
   return true;
 }
 
+// parseTimezoneOffset + tryParseHHMM (values_encoder.go:383-426); local
+// timezone (no suffix) treated as UTC (both runtime containers are UTC)
+static bool parse_tz_offset(strview s, int64_t* off, strview* prefix) {
+  if (s.n > 0 && s.p[s.n - 1] == 'Z') {
+    *off = 0;
+    *prefix = strview(s.p, s.n - 1);
+    return true;
+  }
+  long n = -1;
+  for (long i = long(s.n) - 1; i >= 0; i--) {
+    if (s.p[i] == '+' || s.p[i] == '-') {
+      n = i;
+      break;
+    }
+  }
+  if (n < 0) {
+    *off = 0;  // GetLocalTimezoneOffsetNsecs; UTC here
+    *prefix = s;
+    return true;
+  }
+  strview os(s.p + n + 1, s.n - size_t(n) - 1);
+  if (os.n != 5 || os.p[2] != ':') return false;
+  uint64_t hh, mm;
+  if (!try_parse_date_uint64(strview(os.p, 2), &hh) || hh > 24) return false;
+  if (!try_parse_date_uint64(strview(os.p + 3, 2), &mm) || mm > 60) return false;
+  int64_t v = int64_t(hh) * 3600000000000LL + int64_t(mm) * 60000000000LL;
+  *off = s.p[n] == '-' ? -v : v;
+  *prefix = strview(s.p, size_t(n));
+  return true;
+}
+
+bool try_parse_timestamp_rfc3339(strview s, int64_t* out) {
+  // TryParseTimestampRFC3339Nano (values_encoder.go:340-381)
+  if (s.n < 19) return false;
+  int64_t secs;
+  strview tail;
+  if (!try_parse_timestamp_secs(s, &secs, &tail)) return false;
+  s = tail;
+  int64_t nsecs = secs * 1000000000;
+  int64_t off;
+  strview prefix;
+  if (!parse_tz_offset(s, &off, &prefix)) return false;
+  nsecs -= off;
+  s = prefix;
+  if (s.n == 0) {
+    *out = nsecs;
+    return true;
+  }
+  if (s.p[0] == '.') {
+    s.p++;
+    s.n--;
+  }
+  size_t digits = s.n;
+  if (digits > 9) return false;
+  uint64_t n64;
+  if (!try_parse_date_uint64(s, &n64)) return false;
+  for (size_t k = digits; k < 9; k++) n64 *= 10;
+  *out = nsecs + int64_t(n64);
+  return true;
+}
+
+struct SvReader {
+  const char* p;
+  uint8_t u8(long i) const { return uint8_t(p[i]); }
+};
+
 double parse_math_number(strview s) {
-  // parseMathNumber subset (pipe_math.go:1066-1080, block_result.go:2710-2735)
+  // parseMathNumber (pipe_math.go:1066-1080) over tryParseNumber
+  // (block_result.go:2710-2737)
   double f;
   if (s.n > 0 && try_parse_float64(s, &f)) return f;
   int64_t nsecs;
-  if (try_parse_duration(s, &nsecs)) return double(nsecs);
+  if (s.n > 0 && try_parse_duration(s, &nsecs)) return double(nsecs);
   int64_t b;
-  if (try_parse_bytes(s, &b)) return double(b);
+  if (s.n > 0 && try_parse_bytes(s, &b)) return double(b);
+  SvReader r{s.p};
+  if (s.n > 0 && vl_pf::pf_is_likely_number(r, long(s.n))) {
+    double d;
+    if (vl_pf::go_parse_float(r, long(s.n), &d)) return d;
+    int64_t iv;
+    if (vl_pf::go_parse_int0(r, long(s.n), &iv)) return double(iv);
+  }
+  if (try_parse_timestamp_rfc3339(s, &nsecs)) return double(nsecs);
+  uint32_t ip;
+  if (try_parse_ipv4(s, &ip)) return double(ip);
   return NAN;
 }
 

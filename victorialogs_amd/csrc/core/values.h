@@ -37,6 +37,9 @@ bool try_parse_float64(strview s, double* out);               // :779-781 (non-e
 bool try_parse_ipv4(strview s, uint32_t* out);                // :675-730
 bool try_parse_timestamp_iso8601(strview s, int64_t* out);    // :428-466
 bool try_parse_duration(strview s, int64_t* out);             // :990-1061
+// TryParseTimestampRFC3339Nano (values_encoder.go:340-381); no-timezone
+// inputs use a zero local offset (both runtime boxes are UTC; documented)
+bool try_parse_timestamp_rfc3339(strview s, int64_t* out);
 bool try_parse_bytes(strview s, int64_t* out);                // :855-966
 
 // parseMathNumber subset (pipe_math.go:1066-1080 / block_result.go:2710-2735):
