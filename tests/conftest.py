@@ -61,6 +61,12 @@ def typed_part(tmp_path_factory):
                          "3d2h", "100KB", "", "nan?"][i % 10] for i in range(rows)]},
                     {"name": "uni", "values": [
                         ("раз два три" if i % 3 == 0 else "foo bar") for i in range(rows)]},
+                    # parseMathNumber legs: scientific/hex/rfc3339/ipv4
+                    {"name": "mix2", "values": [
+                        ["1e5", "1.5e-3", "-2E2", "0x1F", "0o17", "0b101",
+                         "2024-01-01T00:00:00Z", "2024-01-01T01:00:00+01:00",
+                         "10.0.0.1", "1_000", "inf", "0x1.8p1"][i % 12]
+                        for i in range(rows)]},
                     # stays ValueType string (a few unparseable entries)
                     {"name": "ipstr", "values": [
                         ("n/a" if i % 9 == 8 else f"172.16.{i % 256}.{(i * 5) % 256}")
@@ -265,6 +271,10 @@ TYPED_FILTERS = [
     '{"type":"ipv4_range","field":"ipstr","min":2886729728,"max":2886733823}',
     '{"type":"in","field":"ipstr","values":["172.16.3.15","n/a"]}',
     '{"type":"len_range","field":"ipstr","min":3,"max":3}',
+    '{"type":"range","field":"mix2","min":0,"max":200000}',
+    '{"type":"range","field":"mix2","min":-300,"max":40}',
+    '{"type":"range","field":"mix2","min":1.7e18,"max":1.8e18}',
+    '{"type":"range","field":"mix2","min":167772161,"max":167772161}',
     '{"type":"string_range","field":"ipstr","min":"172.16.1","max":"172.16.2"}',
     '{"type":"ipv4_range","field":"_msg","min":3232235776,"max":3232236031}',
     '{"type":"ipv4_range","field":"lvl","min":0,"max":4294967295}',

@@ -352,7 +352,7 @@ VL_PF_HD bool go_parse_float(const R& r, long n, double* out) {
     // normalize into pf_round192 form: P = mant << 64
     int lz2 = pf_clz64(mant);
     uint64_t t2 = mant << lz2;
-    double d = pf_round192(t2, truncated ? 1 : 0, 0, e2 - lz2 - 64);
+    double d = pf_round192(t2, truncated ? 1 : 0, 0, e2 - lz2 - 128);
     *out = neg ? -d : d;
     return (d != pf_bits(0x7FF0000000000000ULL));
   }
