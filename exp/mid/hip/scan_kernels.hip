@@ -20,7 +20,6 @@
 // (BASELINE.json north star).
 #include <hip/hip_runtime.h>
 
-#include "../core/parse_float.h"
 #include "../core/ryu.h"
 #include "scan_types.h"
 
@@ -82,27 +81,6 @@ struct TileAcc {
     return *(const uint64_t*)(tile + swz(off));
   }
   __device__ __forceinline__ uint8_t u8(long off) const { return tile[swz(off)]; }
-};
-
-// Byte-wise ASCII tolower view over another accessor.  Only correct for
-// ASCII bytes; rows containing non-ASCII bytes are resolved on the host via
-// the override bitmaps (DevLeafBlock.ovr_mask), so corruption of >=0x80
-// bytes is harmless.  The SWAR form is carry-safe across bytes.
-template <typename A>
-struct LowerAcc {
-  A a;
-  __device__ __forceinline__ uint64_t u64a(long off) const {
-    uint64_t x = a.u64a(off);
-    uint64_t low7 = x & 0x7F7F7F7F7F7F7F7FULL;
-    uint64_t ge_a = low7 + 0x3F3F3F3F3F3F3F3FULL;   // high bit: byte >= 0x41
-    uint64_t ge_z1 = low7 + 0x2525252525252525ULL;  // high bit: byte >= 0x5B
-    uint64_t is_az = ge_a & ~ge_z1 & ~x & 0x8080808080808080ULL;
-    return x | (is_az >> 2);
-  }
-  __device__ __forceinline__ uint8_t u8(long off) const {
-    uint8_t c = a.u8(off);
-    return uint8_t(c - 'A') < 26 ? uint8_t(c + 0x20) : c;
-  }
 };
 
 // Go utf8.DecodeRuneInString semantics (0xFFFD,1 on invalid), reading bytes
@@ -615,153 +593,13 @@ __device__ bool d_try_parse_bytes(const A& a, long s0, long sn, long long* out) 
 
 // parseMathNumber subset (pipe_math.go:1066-1080; same legs as the host
 // parse_math_number in values.cpp -- float, duration, bytes; others NaN)
-// tryParseDateUint64 subset used by RFC3339 parsing (values_encoder.go:552+):
-// plain decimal digits, with the reference's 2-digit fast-path quirk of
-// checking only the first char (mirrored in host try_parse_date_uint64)
-template <typename A>
-__device__ bool d_parse_date_u64(const A& a, long s0, long sn, uint64_t* out) {
-  if (sn == 0 || sn > 18) return false;
-  if (sn == 2) {
-    uint8_t c0 = a.u8(s0);
-    if (c0 < '0' || c0 > '9') return false;
-    *out = 10 * uint64_t(c0 - '0') + uint64_t(uint8_t(a.u8(s0 + 1) - '0'));
-    return true;
-  }
-  uint64_t v = 0;
-  for (long i = 0; i < sn; i++) {
-    uint8_t c = a.u8(s0 + i);
-    if (c < '0' || c > '9') return false;
-    v = v * 10 + (c - '0');
-  }
-  *out = v;
-  return true;
-}
-
-__device__ inline int64_t d_days_from_civil(int64_t y, int m, int64_t d) {
-  y -= m <= 2;
-  int64_t era = (y >= 0 ? y : y - 399) / 400;
-  int64_t yoe = y - era * 400;
-  int64_t doy = (153 * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1;
-  int64_t doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
-  return era * 146097 + doe - 719468;
-}
-
-// tryParseTimestampSecs (values_encoder.go:469-550); consumes the leading
-// "YYYY-MM-DD[T ]hh:mm:ss", returns seconds + the consumed length
-template <typename A>
-__device__ bool d_parse_ts_secs(const A& a, long s0, long sn, int64_t* secs,
-                                long* consumed) {
-  if (sn < 19) return false;
-  uint64_t n;
-  if (a.u8(s0 + 4) != '-') return false;
-  if (!d_parse_date_u64(a, s0, 4, &n) || n < 1677 || n > 2262) return false;
-  int64_t year = int64_t(n);
-  long i = 5;
-  if (a.u8(s0 + i + 2) != '-') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n)) return false;
-  int64_t month = int64_t(n);
-  i += 3;
-  uint8_t delim = a.u8(s0 + i + 2);
-  if (delim != 'T' && delim != ' ') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n)) return false;
-  int64_t day = int64_t(n);
-  i += 3;
-  if (a.u8(s0 + i + 2) != ':') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n) || n > 60) return false;
-  int64_t hour = int64_t(n);
-  i += 3;
-  if (a.u8(s0 + i + 2) != ':') return false;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n) || n > 60) return false;
-  int64_t minute = int64_t(n);
-  i += 3;
-  if (!d_parse_date_u64(a, s0 + i, 2, &n) || n > 60) return false;
-  int64_t sec = int64_t(n);
-  i += 2;
-  // Go time.Date normalization: month/day may overflow their ranges
-  int64_t ym = (month - 1);
-  int64_t yadd = ym >= 0 ? ym / 12 : -((-ym + 11) / 12);
-  int64_t mo = ym - yadd * 12 + 1;
-  int64_t days = d_days_from_civil(year + yadd, int(mo), day);
-  *secs = days * 86400 + hour * 3600 + minute * 60 + sec;
-  *consumed = i;
-  return true;
-}
-
-// TryParseTimestampRFC3339Nano (values_encoder.go:340-381); no-timezone
-// inputs use a zero local offset (the runtime boxes are UTC)
-template <typename A>
-__device__ bool d_parse_rfc3339(const A& a, long s0, long sn, int64_t* out) {
-  if (sn < 19) return false;
-  int64_t secs;
-  long used;
-  if (!d_parse_ts_secs(a, s0, sn, &secs, &used)) return false;
-  long i = s0 + used, n = sn - used;
-  int64_t nsecs = secs * 1000000000;
-  // parseTimezoneOffset (values_encoder.go:383-406)
-  if (n > 0 && a.u8(i + n - 1) == 'Z') {
-    n--;
-  } else {
-    long tz = -1;
-    for (long k = n - 1; k >= 0; k--) {
-      uint8_t c = a.u8(i + k);
-      if (c == '+' || c == '-') {
-        tz = k;
-        break;
-      }
-    }
-    if (tz >= 0) {
-      long on = n - tz - 1;
-      if (on != 5 || a.u8(i + tz + 3) != ':') return false;
-      uint64_t hh, mm;
-      if (!d_parse_date_u64(a, i + tz + 1, 2, &hh) || hh > 24) return false;
-      if (!d_parse_date_u64(a, i + tz + 4, 2, &mm) || mm > 60) return false;
-      int64_t off = int64_t(hh) * 3600000000000LL + int64_t(mm) * 60000000000LL;
-      if (a.u8(i + tz) == '-') off = -off;
-      nsecs -= off;
-      n = tz;
-    }
-  }
-  if (n == 0) {
-    *out = nsecs;
-    return true;
-  }
-  if (a.u8(i) == '.') {
-    i++;
-    n--;
-  }
-  if (n > 9) return false;
-  uint64_t frac;
-  if (!d_parse_date_u64(a, i, n, &frac)) return false;
-  for (long k = n; k < 9; k++) frac *= 10;
-  *out = nsecs + int64_t(frac);
-  return true;
-}
-
-template <typename A>
-struct AccReader {
-  const A* a;
-  long s0;
-  __device__ uint8_t u8(long i) const { return a->u8(s0 + i); }
-};
-
 template <typename A>
 __device__ double d_parse_math_number(const A& a, long s0, long sn) {
   double f;
   if (sn > 0 && d_try_parse_float64(a, s0, sn, &f)) return f;
   long long v;
-  if (sn > 0 && d_try_parse_duration(a, s0, sn, &v)) return double(v);
-  if (sn > 0 && d_try_parse_bytes(a, s0, sn, &v)) return double(v);
-  AccReader<A> r{&a, s0};
-  if (sn > 0 && vl_pf::pf_is_likely_number(r, sn)) {
-    double d;
-    if (vl_pf::go_parse_float(r, sn, &d)) return d;
-    int64_t iv;
-    if (vl_pf::go_parse_int0(r, sn, &iv)) return double(iv);
-  }
-  int64_t ts;
-  if (d_parse_rfc3339(a, s0, sn, &ts)) return double(ts);
-  uint32_t ip;
-  if (d_try_parse_ipv4(a, s0, sn, &ip)) return double(ip);
+  if (d_try_parse_duration(a, s0, sn, &v)) return double(v);
+  if (d_try_parse_bytes(a, s0, sn, &v)) return double(v);
   return __builtin_nan("");
 }
 
@@ -1096,14 +934,20 @@ __device__ __forceinline__ uint16_t d_get_u16be(const uint8_t* p) {
   return uint16_t(p[0]) << 8 | p[1];
 }
 
-// Cold string kinds, kept OUT of the hot scan loop: inlining every matcher
-// into the workgroup loop grew the loop body past the instruction cache and
-// cost ~18% on the phrase workload (798 vs 677 us/launch, profiles/r01b).
+// String-kind predicate over an accessor (tile or global).
 template <typename A>
-__device__ __noinline__ bool d_eval_string_row_cold(const DevLeafBlock& lb,
-                                                    const A& a, long s0,
-                                                    long sn) {
+__device__ bool d_eval_string_row(const DevLeafBlock& lb, const A& a, long s0,
+                                  long sn) {
   switch (lb.kind) {
+    case kScanPhraseStr:
+      return d_match_phrase_at(a, s0, sn, lb.operand, lb.operand_len, lb.flags);
+    case kScanEqStr: {
+      if (sn != long(lb.operand_len)) return false;
+      for (long k = 0; k < sn; k++) {
+        if (a.u8(s0 + k) != lb.operand[k]) return false;
+      }
+      return true;
+    }
     case kScanRangeStr: {
       // matchRange (filter_range.go:369-372)
       double x = d_parse_math_number(a, s0, sn);
@@ -1134,52 +978,39 @@ __device__ __noinline__ bool d_eval_string_row_cold(const DevLeafBlock& lb,
       uint64_t n = d_rune_count(a, s0, sn);
       return n >= lb.vmin && n <= lb.vmax;
     }
-    case kScanAnyCasePhraseStr: {
-      // matchAnyCasePhrase (filter_any_case_phrase.go:159-181); operand is
-      // the lowercase phrase, flags from the lowercase phrase
-      if (lb.operand_len == 0) return sn == 0;
-      if (long(lb.operand_len) > sn) return false;
-      LowerAcc<A> la{a};
-      return d_get_phrase_pos_at(la, s0, sn, lb.operand, lb.operand_len,
-                                 lb.flags & 15) >= 0;
-    }
-    case kScanAnyCasePrefixStr: {
-      // matchAnyCasePrefix (filter_any_case_prefix.go:161-183)
-      if (lb.operand_len == 0) return sn > 0;
-      if (long(lb.operand_len) > sn) return false;
-      LowerAcc<A> la{a};
-      return d_match_prefix_at(la, s0, sn, lb.operand, lb.operand_len,
-                               lb.flags & 15);
-    }
     default:  // kScanRegexStr
       return d_regex_match_at(lb.operand, a, s0, sn);
   }
 }
 
-// String-kind predicate over an accessor (tile or global); the two kinds the
-// steady-state workloads hammer stay inline, everything else is a call.
-template <typename A>
-__device__ __forceinline__ bool d_eval_string_row(const DevLeafBlock& lb,
-                                                  const A& a, long s0,
-                                                  long sn) {
-  if (lb.kind == kScanPhraseStr) {
-    return d_match_phrase_at(a, s0, sn, lb.operand, lb.operand_len, lb.flags);
-  }
-  if (lb.kind == kScanEqStr) {
-    if (sn != long(lb.operand_len)) return false;
-    for (long k = 0; k < sn; k++) {
-      if (a.u8(s0 + k) != lb.operand[k]) return false;
-    }
-    return true;
-  }
-  return d_eval_string_row_cold(lb, a, s0, sn);
-}
-
-// Cold fixed-width kinds (formatters, regex, parsers) behind a call so the
-// hot scan loop stays small (see d_eval_string_row_cold).
-__device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
-                                                   uint32_t row) {
+// Fixed-width / dict / timestamp predicate (coalesced global reads).
+__device__ bool d_eval_fixed_row(const DevLeafBlock& lb, uint32_t row) {
   switch (lb.kind) {
+    case kScanEqBin: {
+      const uint8_t* p = lb.data + size_t(row) * lb.width;
+      switch (lb.width) {
+        case 1: return p[0] == lb.operand[0];
+        case 2: return p[0] == lb.operand[0] && p[1] == lb.operand[1];
+        case 4: {
+          uint32_t a, b;
+          __builtin_memcpy(&a, p, 4);
+          __builtin_memcpy(&b, lb.operand, 4);
+          return a == b;
+        }
+        default: {
+          uint64_t a, b;
+          __builtin_memcpy(&a, p, 8);
+          __builtin_memcpy(&b, lb.operand, 8);
+          return a == b;
+        }
+      }
+    }
+    case kScanDict:
+      return (lb.dict_mask >> lb.data[row]) & 1;
+    case kScanTsRange: {
+      int64_t v = lb.ts[row];
+      return v >= int64_t(lb.vmin) && v <= int64_t(lb.vmax);
+    }
     case kScanRangeU: {
       const uint8_t* p = lb.data + size_t(row) * lb.width;
       uint64_t v;
@@ -1355,155 +1186,12 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
 }
 
 __device__ __forceinline__ bool d_is_string_kind(uint8_t kind) {
-  constexpr uint64_t mask =
-      (uint64_t(1) << kScanPhraseStr) | (uint64_t(1) << kScanEqStr) |
-      (uint64_t(1) << kScanRegexStr) | (uint64_t(1) << kScanRangeStr) |
-      (uint64_t(1) << kScanPrefixStr) | (uint64_t(1) << kScanExactPrefixStr) |
-      (uint64_t(1) << kScanSeqStr) | (uint64_t(1) << kScanInStr) |
-      (uint64_t(1) << kScanAnyPhraseStr) | (uint64_t(1) << kScanAllPhrasesStr) |
-      (uint64_t(1) << kScanStrRange) | (uint64_t(1) << kScanIPv4RangeStr) |
-      (uint64_t(1) << kScanLenRangeStr) | (uint64_t(1) << kScanAnyCasePhraseStr) |
-      (uint64_t(1) << kScanAnyCasePrefixStr);
-  return kind < 64 && ((mask >> kind) & 1);
-}
-
-// Fixed-width / dict / timestamp predicate (coalesced global reads); the
-// steady-state kinds stay inline, the rest is a call.
-__device__ __forceinline__ bool d_eval_fixed_row(const DevLeafBlock& lb,
-                                                 uint32_t row) {
-  switch (lb.kind) {
-    case kScanEqBin: {
-      const uint8_t* p = lb.data + size_t(row) * lb.width;
-      switch (lb.width) {
-        case 1: return p[0] == lb.operand[0];
-        case 2: return p[0] == lb.operand[0] && p[1] == lb.operand[1];
-        case 4: {
-          uint32_t a, b;
-          __builtin_memcpy(&a, p, 4);
-          __builtin_memcpy(&b, lb.operand, 4);
-          return a == b;
-        }
-        default: {
-          uint64_t a, b;
-          __builtin_memcpy(&a, p, 8);
-          __builtin_memcpy(&b, lb.operand, 8);
-          return a == b;
-        }
-      }
-    }
-    case kScanDict:
-      return (lb.dict_mask >> lb.data[row]) & 1;
-    case kScanTsRange: {
-      int64_t v = lb.ts[row];
-      return v >= int64_t(lb.vmin) && v <= int64_t(lb.vmax);
-    }
-    default:
-      return d_eval_fixed_row_cold(lb, row);
-  }
-}
-
-
-// The per-wave LDS-tiled string scan loop, templated on the row predicate so
-// the hot filters get their own clone with ONLY their matcher inlined (the
-// compiler stopped unswitching the kind dispatch out of this loop once the
-// kind count grew, costing ~13% on the phrase workload).
-template <bool kOvr, typename EvalFn>
-__device__ __forceinline__ void d_string_tile_loop(
-    const DevLeafBlock& lb, uint8_t* wtile, uint64_t* out, uint32_t r0,
-    uint32_t r1, uint32_t nwords, int lane, int wave, int nwaves,
-    EvalFn eval) {
-  typedef uint32_t v4u __attribute__((ext_vector_type(4)));
-  v4u* dst = (v4u*)wtile;
-  uint32_t wd = wave;
-  // Offsets pipeline: each lane holds offsets[g0+lane] (clamped to g1);
-  // a lane's row end is the next lane's start (shfl), lane ng-1's end is
-  // the clamped value itself.  The next group's offsets are prefetched
-  // while this group's tile copy is in flight.
-  uint32_t o_lane = 0, o_end = 0;
-  if (wd < nwords) {
-    o_lane = lb.offsets[min(r0 + wd * 64 + uint32_t(lane), r1)];
-    if (lane == 0) o_end = lb.offsets[min(r0 + wd * 64 + 64, r1)];
-  }
-  while (wd < nwords) {
-    const uint32_t g0 = r0 + wd * 64;
-    const uint32_t g1 = min(g0 + 64, r1);
-    const uint32_t ng = g1 - g0;
-    const uint32_t byte0 =
-        uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
-    const uint32_t byte1 = uint32_t(__shfl(int(o_end), 0, 64));
-    const uint32_t nbytes = byte1 - byte0;
-    const bool use_tile = nbytes <= kWaveTileBytes;
-    if (use_tile) {
-      const v4u* src = (const v4u*)(lb.data + byte0);
-      const uint32_t n16 = (nbytes + 15) >> 4;
-      uint32_t k = lane;
-      // 8-deep batches: 8 independent loads in flight per lane
-      for (; k + 448 < n16; k += 512) {
-        // nt loads: each byte is read once per kernel; keep L2 for
-        // the offsets/bitmap traffic (cdna guide: nt-weights row)
-        v4u a0 = __builtin_nontemporal_load(src + k);
-        v4u a1 = __builtin_nontemporal_load(src + k + 64);
-        v4u a2 = __builtin_nontemporal_load(src + k + 128);
-        v4u a3 = __builtin_nontemporal_load(src + k + 192);
-        v4u a4 = __builtin_nontemporal_load(src + k + 256);
-        v4u a5 = __builtin_nontemporal_load(src + k + 320);
-        v4u a6 = __builtin_nontemporal_load(src + k + 384);
-        v4u a7 = __builtin_nontemporal_load(src + k + 448);
-        dst[k ^ ((k >> 4) & 15)] = a0;
-        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
-        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
-        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
-        dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
-        dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
-        dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
-        dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
-      }
-      for (; k + 192 < n16; k += 256) {
-        v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
-            a3 = src[k + 192];
-        dst[k ^ ((k >> 4) & 15)] = a0;
-        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
-        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
-        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
-      }
-      for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
-    }
-    // prefetch next group's offsets while the copy is in flight
-    const uint32_t next_wd = wd + nwaves;
-    uint32_t o_next = 0, o_end_next = 0;
-    if (next_wd < nwords) {
-      o_next = lb.offsets[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
-      if (lane == 0) o_end_next = lb.offsets[min(r0 + next_wd * 64 + 64, r1)];
-    }
-    if (use_tile) {
-      // every lane's ds_writes must land before cross-lane reads below
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    }
-    bool pred = false;
-    if (uint32_t(lane) < ng) {
-      const long s = o_lane;
-      const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
-      const long e_fix = uint32_t(lane) == ng - 1 ? long(byte1) : e;
-      if (use_tile) {
-        TileAcc a{wtile};
-        pred = eval(a, s - byte0, e_fix - s);
-      } else {
-        GlobalAcc a{lb.data};
-        pred = eval(a, s, e_fix - s);
-      }
-    }
-    uint64_t word = __ballot(pred);
-    if (kOvr && lb.hashes != nullptr) {
-      // host-resolved rows (non-ASCII any-case, stored in the unused bloom
-      // gate fields): merge at word level
-      const uint64_t mw = lb.hashes[wd], vw = lb.bloom[wd];
-      word = (word & ~mw) | (vw & mw);
-    }
-    if (lane == 0) out[wd] = word;
-    o_lane = o_next;
-    o_end = o_end_next;
-    wd = next_wd;
-  }
+  return kind == kScanPhraseStr || kind == kScanEqStr || kind == kScanRegexStr ||
+         kind == kScanRangeStr || kind == kScanPrefixStr ||
+         kind == kScanExactPrefixStr || kind == kScanSeqStr ||
+         kind == kScanInStr || kind == kScanAnyPhraseStr ||
+         kind == kScanAllPhrasesStr || kind == kScanStrRange ||
+         kind == kScanIPv4RangeStr || kind == kScanLenRangeStr;
 }
 
 // ---- the program kernel ----
@@ -1562,21 +1250,96 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
       }
 
       if (d_is_string_kind(lb.kind)) {
+        // Per-wave LDS-tiled string scan: each wave independently copies the
+        // bytes of a 64-row group into its private tile (coalesced 16 B
+        // loads) and matches lane-per-row; one wavefront ballot = one
+        // bitmap word.  No workgroup barriers on this path.
+        typedef uint32_t v4u __attribute__((ext_vector_type(4)));
         uint8_t* wtile = tile + wave * kWaveTileBytes;
-        if (lb.kind == kScanPhraseStr) {
-          // hot clone: only the phrase matcher in the loop body
-          d_string_tile_loop<false>(lb, wtile, out, r0, r1, nwords, lane, wave,
-                                    nwaves, [&](const auto& a, long s0, long sn) {
-                                      return d_match_phrase_at(a, s0, sn,
-                                                               lb.operand,
-                                                               lb.operand_len,
-                                                               lb.flags);
-                                    });
-        } else {
-          d_string_tile_loop<true>(lb, wtile, out, r0, r1, nwords, lane, wave,
-                                   nwaves, [&](const auto& a, long s0, long sn) {
-                                     return d_eval_string_row(lb, a, s0, sn);
-                                   });
+        v4u* dst = (v4u*)wtile;
+        uint32_t wd = wave;
+        // Offsets pipeline: each lane holds offsets[g0+lane] (clamped to g1);
+        // a lane's row end is the next lane's start (shfl), lane ng-1's end is
+        // the clamped value itself.  The next group's offsets are prefetched
+        // while this group's tile copy is in flight.
+        uint32_t o_lane = 0, o_end = 0;
+        if (wd < nwords) {
+          o_lane = lb.offsets[min(r0 + wd * 64 + uint32_t(lane), r1)];
+          if (lane == 0) o_end = lb.offsets[min(r0 + wd * 64 + 64, r1)];
+        }
+        while (wd < nwords) {
+          const uint32_t g0 = r0 + wd * 64;
+          const uint32_t g1 = min(g0 + 64, r1);
+          const uint32_t ng = g1 - g0;
+          const uint32_t byte0 =
+              uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
+          const uint32_t byte1 = uint32_t(__shfl(int(o_end), 0, 64));
+          const uint32_t nbytes = byte1 - byte0;
+          const bool use_tile = nbytes <= kWaveTileBytes;
+          if (use_tile) {
+            const v4u* src = (const v4u*)(lb.data + byte0);
+            const uint32_t n16 = (nbytes + 15) >> 4;
+            uint32_t k = lane;
+            // 8-deep batches: 8 independent loads in flight per lane
+            for (; k + 448 < n16; k += 512) {
+              // nt loads: each byte is read once per kernel; keep L2 for
+              // the offsets/bitmap traffic (cdna guide: nt-weights row)
+              v4u a0 = __builtin_nontemporal_load(src + k);
+              v4u a1 = __builtin_nontemporal_load(src + k + 64);
+              v4u a2 = __builtin_nontemporal_load(src + k + 128);
+              v4u a3 = __builtin_nontemporal_load(src + k + 192);
+              v4u a4 = __builtin_nontemporal_load(src + k + 256);
+              v4u a5 = __builtin_nontemporal_load(src + k + 320);
+              v4u a6 = __builtin_nontemporal_load(src + k + 384);
+              v4u a7 = __builtin_nontemporal_load(src + k + 448);
+              dst[k ^ ((k >> 4) & 15)] = a0;
+              dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+              dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+              dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+              dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
+              dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
+              dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
+              dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
+            }
+            for (; k + 192 < n16; k += 256) {
+              v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128],
+                  a3 = src[k + 192];
+              dst[k ^ ((k >> 4) & 15)] = a0;
+              dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+              dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+              dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+            }
+            for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
+          }
+          // prefetch next group's offsets while the copy is in flight
+          const uint32_t next_wd = wd + nwaves;
+          uint32_t o_next = 0, o_end_next = 0;
+          if (next_wd < nwords) {
+            o_next = lb.offsets[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
+            if (lane == 0) o_end_next = lb.offsets[min(r0 + next_wd * 64 + 64, r1)];
+          }
+          if (use_tile) {
+            // every lane's ds_writes must land before cross-lane reads below
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          }
+          bool pred = false;
+          if (uint32_t(lane) < ng) {
+            const long s = o_lane;
+            const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
+            const long e_fix = uint32_t(lane) == ng - 1 ? long(byte1) : e;
+            if (use_tile) {
+              TileAcc a{wtile};
+              pred = d_eval_string_row(lb, a, s - byte0, e_fix - s);
+            } else {
+              GlobalAcc a{lb.data};
+              pred = d_eval_string_row(lb, a, s, e_fix - s);
+            }
+          }
+          const uint64_t word = __ballot(pred);
+          if (lane == 0) out[wd] = word;
+          o_lane = o_next;
+          o_end = o_end_next;
+          wd = next_wd;
         }
       } else {
         for (uint32_t w = wave; w < nwords; w += nwaves) {

@@ -63,8 +63,6 @@ enum ScanKind : uint8_t {
   kScanDayRange = 35,       // (ts - offset) % day in [start,end] (filter_day_range.go)
   kScanWeekRange = 36,      // weekday(ts - offset) in [start,end] (filter_week_range.go)
   kScanIPv4RangeBin = 37,   // BE u32 in [vmin,vmax] (filter_ipv4_range.go:166-181)
-  kScanAnyCasePhraseStr = 38,  // matchAnyCasePhrase, ASCII rows (filter_any_case_phrase.go:159-181)
-  kScanAnyCasePrefixStr = 39,  // matchAnyCasePrefix, ASCII rows (filter_any_case_prefix.go:161-183)
 };
 
 // format source for the *Fmt kinds, stored in flags bits 4..7
@@ -96,12 +94,7 @@ struct DevLeafBlock {
   uint8_t kind;
   uint8_t width;  // fixed-width kinds: 1/2/4/8
   uint8_t flags;
-  // bloom gate probe hashes; 0 = no gate.  The any-case string kinds carry
-  // no bloom gate and reuse `hashes`/`bloom` as the host-resolved override
-  // bitmaps instead (rows with non-ASCII bytes: bit set in hashes=>ovr_mask
-  // means the row result is the bloom=>ovr_val bit) — keeps the descriptor
-  // at 96 B, which the hot scan loop is sensitive to.
-  uint32_t nhashes;
+  uint32_t nhashes;        // bloom gate probe hashes; 0 = no gate
   uint32_t bloom_words;
   uint32_t dict_mask;      // kScanDict: bit i set if dict value i matches
   uint32_t operand_len;

@@ -25,7 +25,6 @@
 #include "core/bloom.h"
 #include "core/match.h"
 #include "core/tokenizer.h"
-#include "core/unicode_case.h"
 #include "core/part_reader.h"
 #include "core/values.h"
 #include "hip/scan_types.h"
@@ -2106,111 +2105,6 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
       return;
     }
 
-
-    case FilterNode::AnyCasePhrase:
-    case FilterNode::AnyCasePrefix: {
-      // filterAnyCasePhrase / filterAnyCasePrefix
-      // (filter_any_case_phrase.go:85-138, filter_any_case_prefix.go:90-140)
-      const bool is_ph = f.type == FilterNode::AnyCasePhrase;
-      strview lower(f.min_s);
-      auto host_match = [&](strview v) {
-        return is_ph ? match_any_case_phrase(v, lower)
-                     : match_any_case_prefix(v, lower);
-      };
-      std::string cv = const_val();
-      if (!cv.empty()) {
-        lb.mode = host_match(strview(cv)) ? kModeAll : kModeNone;
-        return;
-      }
-      ColumnHeader ch;
-      if (!pr.get_column_header(ctx.bc, li.cname, &ch)) {
-        // phrase: empty phrase matches missing columns; prefix: never
-        lb.mode = is_ph && lower.n == 0 ? kModeAll : kModeNone;
-        return;
-      }
-      switch (ch.type) {
-        case ValueType::String: {
-          const StagedStrCol& sc = ctx.stage_column(ch);
-          if (sc.is_const) {
-            lb.mode = host_match(strview(sc.const_value)) ? kModeAll : kModeNone;
-            return;
-          }
-          // rows with non-ASCII bytes need the full Unicode lowercase
-          // mapping (AppendLowercase slow path) — resolve them on the host
-          // at stage time; the kernel handles the ASCII rows
-          StringsBlockDec dec;
-          pr.read_values(ch, ctx.bh->rows_count, dec);
-          const uint32_t rows = uint32_t(ctx.bh->rows_count);
-          const size_t nw = (rows + 63) / 64;
-          std::vector<uint64_t> mask(nw, 0), val(nw, 0);
-          bool any = false;
-          for (uint32_t r = 0; r < rows; r++) {
-            const char* vp = (const char*)dec.data.data() + dec.offsets[r];
-            const size_t vn = dec.offsets[r + 1] - dec.offsets[r];
-            bool ascii = true;
-            for (size_t k = 0; k < vn; k++) {
-              if (uint8_t(vp[k]) >= 0x80) {
-                ascii = false;
-                break;
-              }
-            }
-            if (ascii) continue;
-            any = true;
-            mask[r >> 6] |= uint64_t(1) << (r & 63);
-            if (host_match(strview(vp, vn))) {
-              val[r >> 6] |= uint64_t(1) << (r & 63);
-            }
-          }
-          lb.mode = kModeScan;
-          lb.kind = is_ph ? kScanAnyCasePhraseStr : kScanAnyCasePrefixStr;
-          lb.flags = li.phrase_flags;  // flags of the lowercase phrase
-          lb.operand = li.d_operand;
-          lb.operand_len = uint32_t(f.min_s.size());
-          lb.data = sc.d_data;
-          lb.offsets = sc.d_offsets;
-          if (any) {
-            // override bitmaps ride in the unused bloom-gate fields
-            // (scan_types.h DevLeafBlock comment); nhashes stays 0 so the
-            // bloom gate never fires for these kinds
-            lb.hashes = (const uint64_t*)st.push(mask.data(), nw * 8, 8);
-            lb.bloom = (const uint64_t*)st.push(val.data(), nw * 8, 8);
-          }
-          return;
-        }
-        case ValueType::Dict: {
-          uint32_t m = dict_mask_of(ch.dict, host_match);
-          stage_dict(lb, ctx, ch, m);
-          return;
-        }
-        default: {
-          // numeric/ip/iso types: the stored values never contain letters in
-          // mixed case, so the reference delegates to the exact/prefix
-          // matchers with the lowercase (iso: uppercase) phrase
-          const bool iso = ch.type == ValueType::TimestampISO8601;
-          FilterNode tmp;
-          tmp.type = is_ph ? FilterNode::Phrase : FilterNode::Prefix;
-          tmp.field = f.field;
-          tmp.phrase = iso ? f.max_s : f.min_s;
-          tmp.token_hashes = iso ? f.all_hashes : f.token_hashes;
-          LeafInfo tmp_li;
-          tmp_li.node = &tmp;
-          tmp_li.cname = li.cname;
-          tmp_li.operand.assign(tmp.phrase.begin(), tmp.phrase.end());
-          tmp_li.phrase_flags = phrase_flags_of(tmp.phrase);
-          if (!tmp_li.operand.empty()) {
-            tmp_li.d_operand = (const uint8_t*)st.push(tmp_li.operand.data(),
-                                                       tmp_li.operand.size(), 8);
-          }
-          if (!tmp.token_hashes.empty()) {
-            tmp_li.d_hashes = (const uint64_t*)st.push(
-                tmp.token_hashes.data(), tmp.token_hashes.size() * 8, 8);
-          }
-          stage_leaf(tmp_li, ctx, st, lb);
-          return;
-        }
-      }
-    }
-
     default:
       fail("stage_leaf: non-leaf node");
   }
@@ -2305,11 +2199,6 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
         li.operand = std::move(b);
         break;
       }
-      case FilterNode::AnyCasePhrase:
-      case FilterNode::AnyCasePrefix:
-        li.operand.assign(li.node->min_s.begin(), li.node->min_s.end());
-        li.phrase_flags = phrase_flags_of(li.node->min_s);
-        break;
       default:
         break;
     }
@@ -2367,9 +2256,7 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
           case kScanAllPhrasesStr:
           case kScanStrRange:
           case kScanIPv4RangeStr:
-          case kScanLenRangeStr:
-          case kScanAnyCasePhraseStr:
-          case kScanAnyCasePrefixStr: {
+          case kScanLenRangeStr: {
             const StagedStrCol& sc = ctx.cols.at(leaf_infos[size_t(l)].cname);
             st->algo_bytes += sc.data_bytes + (bh.rows_count + 1) * 4;
             break;
