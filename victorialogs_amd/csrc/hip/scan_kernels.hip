@@ -1407,9 +1407,15 @@ __device__ __forceinline__ bool d_eval_fixed_row(const DevLeafBlock& lb,
 // the hot filters get their own clone with ONLY their matcher inlined (the
 // compiler stopped unswitching the kind dispatch out of this loop once the
 // kind count grew, costing ~13% on the phrase workload).
+// NB: every descriptor field is passed BY VALUE — capturing DevLeafBlock by
+// reference made the compiler re-load lb.data/lb.offsets from memory inside
+// the group loop under SGPR-spill pressure (global_load_dwordx2 + vmcnt(0)
+// per iteration, ~7%% of the phrase workload).
 template <bool kOvr, typename EvalFn>
 __device__ __forceinline__ void d_string_tile_loop(
-    const DevLeafBlock& lb, uint8_t* wtile, uint64_t* out, uint32_t r0,
+    const uint8_t* __restrict__ col_data, const uint32_t* __restrict__ col_offs,
+    const uint64_t* __restrict__ ovr_mask, const uint64_t* __restrict__ ovr_val,
+    uint8_t* wtile, uint64_t* out, uint32_t r0,
     uint32_t r1, uint32_t nwords, int lane, int wave, int nwaves,
     EvalFn eval) {
   typedef uint32_t v4u __attribute__((ext_vector_type(4)));
@@ -1421,8 +1427,8 @@ __device__ __forceinline__ void d_string_tile_loop(
   // while this group's tile copy is in flight.
   uint32_t o_lane = 0, o_end = 0;
   if (wd < nwords) {
-    o_lane = lb.offsets[min(r0 + wd * 64 + uint32_t(lane), r1)];
-    if (lane == 0) o_end = lb.offsets[min(r0 + wd * 64 + 64, r1)];
+    o_lane = col_offs[min(r0 + wd * 64 + uint32_t(lane), r1)];
+    if (lane == 0) o_end = col_offs[min(r0 + wd * 64 + 64, r1)];
   }
   while (wd < nwords) {
     const uint32_t g0 = r0 + wd * 64;
@@ -1434,10 +1440,46 @@ __device__ __forceinline__ void d_string_tile_loop(
     const uint32_t nbytes = byte1 - byte0;
     const bool use_tile = nbytes <= kWaveTileBytes;
     if (use_tile) {
-      const v4u* src = (const v4u*)(lb.data + byte0);
+      const v4u* src = (const v4u*)(col_data + byte0);
       const uint32_t n16 = (nbytes + 15) >> 4;
       uint32_t k = lane;
-      // 8-deep batches: 8 independent loads in flight per lane
+      // 16-deep batches: a 64-row group of ~256 B rows is 1024 slots, so one
+      // batch puts the whole group's loads in flight per lane (256 B/lane)
+      for (; k + 960 < n16; k += 1024) {
+        v4u b0 = __builtin_nontemporal_load(src + k);
+        v4u b1 = __builtin_nontemporal_load(src + k + 64);
+        v4u b2 = __builtin_nontemporal_load(src + k + 128);
+        v4u b3 = __builtin_nontemporal_load(src + k + 192);
+        v4u b4 = __builtin_nontemporal_load(src + k + 256);
+        v4u b5 = __builtin_nontemporal_load(src + k + 320);
+        v4u b6 = __builtin_nontemporal_load(src + k + 384);
+        v4u b7 = __builtin_nontemporal_load(src + k + 448);
+        v4u b8 = __builtin_nontemporal_load(src + k + 512);
+        v4u b9 = __builtin_nontemporal_load(src + k + 576);
+        v4u b10 = __builtin_nontemporal_load(src + k + 640);
+        v4u b11 = __builtin_nontemporal_load(src + k + 704);
+        v4u b12 = __builtin_nontemporal_load(src + k + 768);
+        v4u b13 = __builtin_nontemporal_load(src + k + 832);
+        v4u b14 = __builtin_nontemporal_load(src + k + 896);
+        v4u b15 = __builtin_nontemporal_load(src + k + 960);
+        dst[k ^ ((k >> 4) & 15)] = b0;
+        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = b1;
+        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = b2;
+        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = b3;
+        dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = b4;
+        dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = b5;
+        dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = b6;
+        dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = b7;
+        dst[(k + 512) ^ (((k + 512) >> 4) & 15)] = b8;
+        dst[(k + 576) ^ (((k + 576) >> 4) & 15)] = b9;
+        dst[(k + 640) ^ (((k + 640) >> 4) & 15)] = b10;
+        dst[(k + 704) ^ (((k + 704) >> 4) & 15)] = b11;
+        dst[(k + 768) ^ (((k + 768) >> 4) & 15)] = b12;
+        dst[(k + 832) ^ (((k + 832) >> 4) & 15)] = b13;
+        dst[(k + 896) ^ (((k + 896) >> 4) & 15)] = b14;
+        dst[(k + 960) ^ (((k + 960) >> 4) & 15)] = b15;
+      }
+      // 8-deep batches for mid-size remainders
       for (; k + 448 < n16; k += 512) {
         // nt loads: each byte is read once per kernel; keep L2 for
         // the offsets/bitmap traffic (cdna guide: nt-weights row)
@@ -1472,8 +1514,8 @@ __device__ __forceinline__ void d_string_tile_loop(
     const uint32_t next_wd = wd + nwaves;
     uint32_t o_next = 0, o_end_next = 0;
     if (next_wd < nwords) {
-      o_next = lb.offsets[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
-      if (lane == 0) o_end_next = lb.offsets[min(r0 + next_wd * 64 + 64, r1)];
+      o_next = col_offs[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
+      if (lane == 0) o_end_next = col_offs[min(r0 + next_wd * 64 + 64, r1)];
     }
     if (use_tile) {
       // every lane's ds_writes must land before cross-lane reads below
@@ -1488,15 +1530,15 @@ __device__ __forceinline__ void d_string_tile_loop(
         TileAcc a{wtile};
         pred = eval(a, s - byte0, e_fix - s);
       } else {
-        GlobalAcc a{lb.data};
+        GlobalAcc a{col_data};
         pred = eval(a, s, e_fix - s);
       }
     }
     uint64_t word = __ballot(pred);
-    if (kOvr && lb.hashes != nullptr) {
+    if (kOvr && ovr_mask != nullptr) {
       // host-resolved rows (non-ASCII any-case, stored in the unused bloom
       // gate fields): merge at word level
-      const uint64_t mw = lb.hashes[wd], vw = lb.bloom[wd];
+      const uint64_t mw = ovr_mask[wd], vw = ovr_val[wd];
       word = (word & ~mw) | (vw & mw);
     }
     if (lane == 0) out[wd] = word;
@@ -1563,17 +1605,27 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
 
       if (d_is_string_kind(lb.kind)) {
         uint8_t* wtile = tile + wave * kWaveTileBytes;
+        const uint8_t* col_data = lb.data;
+        const uint32_t* col_offs = lb.offsets;
+        const uint8_t* op_ptr = lb.operand;
+        const uint32_t op_len = lb.operand_len;
+        const uint8_t op_flags = lb.flags;
         if (lb.kind == kScanPhraseStr) {
           // hot clone: only the phrase matcher in the loop body
-          d_string_tile_loop<false>(lb, wtile, out, r0, r1, nwords, lane, wave,
-                                    nwaves, [&](const auto& a, long s0, long sn) {
+          d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                    wtile, out, r0, r1, nwords, lane, wave,
+                                    nwaves, [=](const auto& a, long s0, long sn) {
                                       return d_match_phrase_at(a, s0, sn,
-                                                               lb.operand,
-                                                               lb.operand_len,
-                                                               lb.flags);
+                                                               op_ptr, op_len,
+                                                               op_flags);
                                     });
         } else {
-          d_string_tile_loop<true>(lb, wtile, out, r0, r1, nwords, lane, wave,
+          const bool anycase = lb.kind == kScanAnyCasePhraseStr ||
+                               lb.kind == kScanAnyCasePrefixStr;
+          d_string_tile_loop<true>(col_data, col_offs,
+                                   anycase ? lb.hashes : nullptr,
+                                   anycase ? lb.bloom : nullptr,
+                                   wtile, out, r0, r1, nwords, lane, wave,
                                    nwaves, [&](const auto& a, long s0, long sn) {
                                      return d_eval_string_row(lb, a, s0, sn);
                                    });
