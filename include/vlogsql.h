@@ -83,6 +83,15 @@ long long vql_gather(void* stage, const char* field, unsigned char* out_bytes,
                      long long bytes_cap, unsigned long long* out_offs,
                      long long offs_cap, unsigned long long* out_rowids);
 
+/* GPU ingest-side bloom build: builds the marshaled
+ * bloom-filter bytes for one column block (the write path's tokenizeHashes +
+ * bloomFilterMarshalHashes, lib/logstorage/block.go:160-168), bit-identical
+ * to the CPU writer.  offsets = u32[rows+1] over the concatenated value
+ * bytes.  Returns the marshaled length; fills `out` when cap suffices. */
+long long vql_bloom_build(const unsigned char* data, long long nbytes,
+                          const unsigned int* offsets, long long rows,
+                          int device, unsigned char* out, long long cap);
+
 /* Cold path (§8b vql_scan_batch): stage + scan + fetch + free in one call. */
 long long vql_scan_batch(void* part, void* filter, long block_lo, long block_hi,
                          unsigned long long* out_words, long long cap_words,

@@ -203,6 +203,30 @@ int orc_parse_iso8601(const char* s, long sn, long long* out) {
   return 1;
 }
 
+// CPU reference for the GPU ingest-side bloom build (test infrastructure):
+// exactly the write path's tokenizeHashes + bloomFilterMarshalHashes
+// (block.go:160-168 via part_writer.cpp:162).
+long long orc_bloom_build(const unsigned char* data, long long nbytes,
+                          const unsigned int* offsets, long long rows,
+                          unsigned char* out, long long cap) {
+  try {
+    (void)nbytes;
+    std::vector<strview> vs;
+    vs.reserve(size_t(rows));
+    for (long long i = 0; i < rows; i++) {
+      vs.emplace_back((const char*)data + offsets[i],
+                      size_t(offsets[i + 1] - offsets[i]));
+    }
+    bytes bf = bloom_marshal_hashes(tokenize_hashes(vs));
+    long long n = (long long)bf.size();
+    if (n <= cap) memcpy(out, bf.data(), bf.size());
+    return n;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
 // Writes a custom part from a JSON fixture (filter_test.go fixture style):
 // {"blocks":[{"stream":0,"timestamps":[...],
 //             "columns":[{"name":"x","values":["a","b",...]}]}]}

@@ -327,3 +327,25 @@ class Stage:
         if self.lib.vql_fetch_block_hits(self.h, buf, nblocks) != 0:
             raise RuntimeError(self.lib.vql_errstr().decode())
         return list(buf)[:nblocks]
+
+
+def gpu_bloom_build(values, device=0):
+    """GPU ingest-side bloom build (vql_bloom_build): returns the marshaled
+    bloom bytes for a list of value byte-strings, bit-identical to the CPU
+    writer's column bloom."""
+    lib = load_product()
+    lib.vql_bloom_build.restype = ctypes.c_longlong
+    data = b"".join(values)
+    offs = [0]
+    for v in values:
+        offs.append(offs[-1] + len(v))
+    offs_arr = (ctypes.c_uint32 * len(offs))(*offs)
+    buf = ctypes.create_string_buffer(max(len(data) * 2 + 1024, 1 << 16))
+    n = lib.vql_bloom_build(data, len(data), offs_arr, len(values), device,
+                            ctypes.cast(buf, ctypes.POINTER(ctypes.c_ubyte)),
+                            len(buf))
+    if n < 0:
+        raise RuntimeError(lib.vql_errstr().decode())
+    if n > len(buf):
+        raise RuntimeError("bloom output larger than buffer")
+    return buf.raw[:n]

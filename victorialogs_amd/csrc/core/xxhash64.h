@@ -12,6 +12,13 @@
 #include <cstdint>
 #include <cstring>
 
+// usable from HIP device code too (GPU ingest-side bloom build)
+#if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
+#define VL_XX_HD __host__ __device__
+#else
+#define VL_XX_HD
+#endif
+
 namespace vl {
 
 namespace xx {
@@ -21,24 +28,24 @@ constexpr uint64_t P3 = 1609587929392839161ULL;
 constexpr uint64_t P4 = 9650029242287828579ULL;
 constexpr uint64_t P5 = 2870177450012600261ULL;
 
-inline uint64_t rotl(uint64_t x, int r) { return (x << r) | (x >> (64 - r)); }
-inline uint64_t rd64(const uint8_t* p) {
+VL_XX_HD inline uint64_t rotl(uint64_t x, int r) { return (x << r) | (x >> (64 - r)); }
+VL_XX_HD inline uint64_t rd64(const uint8_t* p) {
   uint64_t v;
   memcpy(&v, p, 8);  // little-endian host
   return v;
 }
-inline uint32_t rd32(const uint8_t* p) {
+VL_XX_HD inline uint32_t rd32(const uint8_t* p) {
   uint32_t v;
   memcpy(&v, p, 4);
   return v;
 }
-inline uint64_t round_(uint64_t acc, uint64_t input) {
+VL_XX_HD inline uint64_t round_(uint64_t acc, uint64_t input) {
   acc += input * P2;
   acc = rotl(acc, 31);
   acc *= P1;
   return acc;
 }
-inline uint64_t merge_round(uint64_t acc, uint64_t val) {
+VL_XX_HD inline uint64_t merge_round(uint64_t acc, uint64_t val) {
   val = round_(0, val);
   acc ^= val;
   acc = acc * P1 + P4;
@@ -46,7 +53,7 @@ inline uint64_t merge_round(uint64_t acc, uint64_t val) {
 }
 }  // namespace xx
 
-inline uint64_t xxhash64(const void* data, size_t len) {
+VL_XX_HD inline uint64_t xxhash64(const void* data, size_t len) {
   using namespace xx;
   const uint8_t* p = (const uint8_t*)data;
   const uint8_t* end = p + len;

@@ -1942,3 +1942,121 @@ extern "C" hipError_t vql_launch_scan(const DevOp* ops, int nops,
 }
 
 }  // namespace vl
+
+// ---- GPU ingest-side bloom build ----
+// Restates the write path column.mustWriteTo bloom leg (block.go:160-168):
+// tokenizeHashes (hash_tokenizer.go:68-166, hash-level dedup) ->
+// bloomFilterMarshalHashes (bloomfilter.go:49-55,83-121).  Bloom bits are
+// idempotent, so the kernel inserts every token occurrence's XXH64 into a
+// global open-addressing set (the dedup gives the unique-hash COUNT that
+// sizes the filter), then a second kernel sets the 6 chained probe bits per
+// unique hash.
+
+#include "../core/xxhash64.h"
+
+namespace vl {
+
+// one lane per row: tokenize + insert token hashes into the set
+__global__ __launch_bounds__(256) void bloom_tokenize_kernel(
+    const uint8_t* __restrict__ data, const uint32_t* __restrict__ offsets,
+    uint32_t rows, unsigned long long* __restrict__ slots, uint32_t cap_mask,
+    unsigned long long* __restrict__ unique_count, int* __restrict__ overflow) {
+  const uint32_t row = blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= rows) return;
+  const uint8_t* p = data + offsets[row];
+  const long n = long(offsets[row + 1]) - long(offsets[row]);
+  bool ascii = true;
+  for (long i = 0; i < n; i++) {
+    if (p[i] >= 0x80) {
+      ascii = false;
+      break;
+    }
+  }
+  long i = 0;
+  while (i < n) {
+    long start, end;
+    if (ascii) {
+      while (i < n && !d_is_token_char(p[i])) i++;
+      start = i;
+      while (i < n && d_is_token_char(p[i])) i++;
+      end = i;
+    } else {
+      GlobalAcc a{p};
+      while (i < n) {
+        int sz;
+        uint32_t r = d_utf8_decode(a, i, n - i, &sz);
+        if (d_is_token_rune(r)) break;
+        i += sz;
+      }
+      start = i;
+      while (i < n) {
+        int sz;
+        uint32_t r = d_utf8_decode(a, i, n - i, &sz);
+        if (!d_is_token_rune(r)) break;
+        i += sz;
+      }
+      end = i;
+    }
+    if (end <= start) break;
+    const uint64_t h = vl::xxhash64(p + start, size_t(end - start));
+    // open addressing, linear probing; slot 0 == empty (h==0 never occurs in
+    // practice; documented)
+    uint32_t idx = uint32_t(h) & cap_mask;
+    for (uint32_t probes = 0;; probes++) {
+      if (probes > cap_mask) {
+        *overflow = 1;
+        return;
+      }
+      unsigned long long prev = atomicCAS(&slots[idx], 0ULL, h);
+      if (prev == 0) {
+        atomicAdd(unique_count, 1ULL);
+        break;
+      }
+      if (prev == h) break;
+      idx = (idx + 1) & cap_mask;
+    }
+  }
+}
+
+// one lane per slot: 6 chained probe hashes per unique token hash
+// (appendHashesHashes, bloomfilter.go:159-169), bits set with atomicOr
+__global__ __launch_bounds__(256) void bloom_setbits_kernel(
+    const unsigned long long* __restrict__ slots, uint32_t cap,
+    unsigned long long* __restrict__ bits, uint64_t max_bits) {
+  const uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= cap) return;
+  uint64_t h = slots[i];
+  if (h == 0) return;
+  uint64_t buf = h;
+  for (int k = 0; k < 6; k++) {
+    uint64_t hk = vl::xxhash64(&buf, 8);
+    buf++;
+    uint64_t idx = hk % max_bits;
+    atomicOr(&bits[idx >> 6], 1ULL << (idx & 63));
+  }
+}
+
+extern "C" int vql_launch_bloom_tokenize(const void* data, const void* offsets,
+                                         unsigned rows, void* slots,
+                                         unsigned cap_mask, void* unique_count,
+                                         void* overflow, void* stream) {
+  dim3 grid((rows + 255) / 256), block(256);
+  hipLaunchKernelGGL(bloom_tokenize_kernel, grid, block, 0,
+                     (hipStream_t)stream, (const uint8_t*)data,
+                     (const uint32_t*)offsets, rows,
+                     (unsigned long long*)slots, cap_mask,
+                     (unsigned long long*)unique_count, (int*)overflow);
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+extern "C" int vql_launch_bloom_setbits(const void* slots, unsigned cap,
+                                        void* bits, unsigned long long max_bits,
+                                        void* stream) {
+  dim3 grid((cap + 255) / 256), block(256);
+  hipLaunchKernelGGL(bloom_setbits_kernel, grid, block, 0, (hipStream_t)stream,
+                     (const unsigned long long*)slots, cap,
+                     (unsigned long long*)bits, max_bits);
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+}  // namespace vl

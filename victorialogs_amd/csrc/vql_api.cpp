@@ -2538,6 +2538,104 @@ int vql_fetch_block_hits(void* s, unsigned long long* out, long long cap) {
 
 // Cold path (SURVEY.md §8b vql_scan_batch): stage blocks [lo,hi), scan once,
 // fetch bitmaps, free.  Returns matched rows or -1.
+// GPU ingest-side bloom build: marshaled bloom bytes
+// for one column block, bit-identical to the write path's
+// tokenizeHashes + bloomFilterMarshalHashes (block.go:160-168).  `offsets` is
+// u32[rows+1] over the concatenated value bytes.  Returns the marshaled
+// length (BE u64 words); fills `out` when cap suffices; -1 on error.
+extern "C" int vql_launch_bloom_tokenize(const void*, const void*, unsigned,
+                                         void*, unsigned, void*, void*, void*);
+extern "C" int vql_launch_bloom_setbits(const void*, unsigned, void*,
+                                        unsigned long long, void*);
+
+extern "C" long long vql_bloom_build(const unsigned char* data,
+                                     long long nbytes,
+                                     const unsigned int* offsets,
+                                     long long rows, int device,
+                                     unsigned char* out, long long cap) {
+  try {
+    HIP_CHECK(hipSetDevice(device));
+    uint8_t* d_data = nullptr;
+    uint32_t* d_offs = nullptr;
+    unsigned long long* d_slots = nullptr;
+    unsigned long long* d_unique = nullptr;
+    int* d_overflow = nullptr;
+    uint64_t* d_bits = nullptr;
+    auto cleanup = [&]() {
+      if (d_data) hipFree(d_data);
+      if (d_offs) hipFree(d_offs);
+      if (d_slots) hipFree(d_slots);
+      if (d_unique) hipFree(d_unique);
+      if (d_overflow) hipFree(d_overflow);
+      if (d_bits) hipFree(d_bits);
+    };
+    HIP_CHECK(hipMalloc(&d_data, size_t(nbytes) + 16));
+    HIP_CHECK(hipMemcpy(d_data, data, size_t(nbytes), hipMemcpyHostToDevice));
+    HIP_CHECK(hipMalloc(&d_offs, (size_t(rows) + 1) * 4));
+    HIP_CHECK(hipMemcpy(d_offs, offsets, (size_t(rows) + 1) * 4,
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMalloc(&d_unique, 8));
+    HIP_CHECK(hipMalloc(&d_overflow, 4));
+
+    uint64_t cap_slots = 1024;
+    while (cap_slots < uint64_t(nbytes) / 2 + 1024) cap_slots <<= 1;
+    // start smaller for realistic corpora; grow on overflow
+    uint64_t try_cap = 1 << 16;
+    unsigned long long unique = 0;
+    for (;;) {
+      if (try_cap > cap_slots) try_cap = cap_slots;
+      HIP_CHECK(hipMalloc(&d_slots, try_cap * 8));
+      HIP_CHECK(hipMemset(d_slots, 0, try_cap * 8));
+      HIP_CHECK(hipMemset(d_unique, 0, 8));
+      HIP_CHECK(hipMemset(d_overflow, 0, 4));
+      if (vql_launch_bloom_tokenize(d_data, d_offs, unsigned(rows), d_slots,
+                                    unsigned(try_cap - 1), d_unique,
+                                    d_overflow, nullptr) != 0) {
+        cleanup();
+        fail("bloom tokenize launch failed");
+      }
+      int overflow = 0;
+      HIP_CHECK(hipMemcpy(&overflow, d_overflow, 4, hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(&unique, d_unique, 8, hipMemcpyDeviceToHost));
+      // keep the table below half full so dedup probing stays short
+      if (!overflow && unique * 2 <= try_cap) break;
+      hipFree(d_slots);
+      d_slots = nullptr;
+      if (try_cap >= cap_slots) {
+        cleanup();
+        fail("bloom hash set overflow");
+      }
+      try_cap <<= 2;
+    }
+    const uint64_t bits_count = unique * 16;  // bloomFilterBitsPerItem
+    const uint64_t words = (bits_count + 63) / 64;
+    const long long out_len = (long long)(words * 8);
+    if (words > 0 && out_len <= cap) {
+      HIP_CHECK(hipMalloc(&d_bits, words * 8));
+      HIP_CHECK(hipMemset(d_bits, 0, words * 8));
+      if (vql_launch_bloom_setbits(d_slots, unsigned(try_cap), d_bits,
+                                   words * 64, nullptr) != 0) {
+        cleanup();
+        fail("bloom setbits launch failed");
+      }
+      std::vector<uint64_t> host_words(words);
+      HIP_CHECK(hipMemcpy(host_words.data(), d_bits, words * 8,
+                          hipMemcpyDeviceToHost));
+      for (uint64_t i = 0; i < words; i++) {
+        uint64_t w = host_words[i];
+        for (int b = 7; b >= 0; b--) {
+          out[i * 8 + (7 - b)] = uint8_t(w >> (b * 8));  // BE marshal
+        }
+      }
+    }
+    cleanup();
+    return out_len;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
 long long vql_scan_batch(void* part, void* filter, long lo, long hi,
                          unsigned long long* out_words, long long cap,
                          unsigned long long* out_popcounts) {
