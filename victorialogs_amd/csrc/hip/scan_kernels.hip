@@ -1619,6 +1619,13 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
                                                                op_ptr, op_len,
                                                                op_flags);
                                     });
+        } else if (lb.kind == kScanRegexStr) {
+          // second hot clone: regex fast paths + NFA (BASELINE config 3)
+          d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                    wtile, out, r0, r1, nwords, lane, wave,
+                                    nwaves, [=](const auto& a, long s0, long sn) {
+                                      return d_regex_match_at(op_ptr, a, s0, sn);
+                                    });
         } else {
           const bool anycase = lb.kind == kScanAnyCasePhraseStr ||
                                lb.kind == kScanAnyCasePrefixStr;
