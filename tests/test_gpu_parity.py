@@ -117,3 +117,34 @@ def test_unsupported_regex_fails_at_compile(typed_part):
     never fall back to CPU silently."""
     with pytest.raises(RuntimeError, match="not supported|fast-path"):
         Filter('{"type":"regexp","field":"_msg","re":"a{2,3}"}')
+
+
+def test_multichunk_block(tmp_path):
+    """Blocks larger than kChunkRows (8192) split into multiple workgroup
+    chunks; bitmap words must still land at the right block offsets."""
+    from victorialogs_amd import write_custom_part
+
+    rows = 20000  # 3 chunks (8192 + 8192 + 3616)
+    spec = {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i * 1000 for i in range(rows)],
+        "columns": [
+            {"name": "_msg", "values": [
+                f"entry {i} kind={'odd' if i % 2 else 'even'}"
+                for i in range(rows)]},
+            {"name": "lvl", "values": [
+                ["debug", "info", "warn", "error"][i % 4] for i in range(rows)]},
+            {"name": "num", "values": [str(i % 977) for i in range(rows)]},
+        ],
+    }]}
+    d = str(tmp_path / "bigblock")
+    write_custom_part(d, spec)
+    for f in [
+        '{"type":"phrase","field":"_msg","phrase":"kind=odd"}',
+        '{"type":"phrase","field":"lvl","phrase":"error"}',
+        '{"type":"range","field":"num","min":100,"max":200}',
+        '{"type":"and","filters":[{"type":"phrase","field":"_msg","phrase":"entry"},'
+        '{"type":"not","filter":{"type":"phrase","field":"lvl","phrase":"info"}}]}',
+        '{"type":"time","min":1700000000005000000,"max":1700000000015000000}',
+    ]:
+        assert_parity(d, f)
