@@ -33,7 +33,9 @@ namespace vl {
 // word per group) with no workgroup barriers: the wave copies its group's
 // bytes with coalesced 16-byte loads (4-deep batches keep ~4 KiB per wave in
 // flight), then matches lane-per-row from the tile.
-constexpr uint32_t kWaveTileBytes = 16896;  // 64 rows x <=264 B avg
+constexpr uint32_t kWaveTileBytes = 17408;  // 1088 slots: 64 rows x <=272 B,
+// a multiple of 64*16 B so the copy loop can round every group up to full
+// 64-slot strides (no single-lane straggler load per group)
 constexpr uint32_t kNumWaves = 4;
 
 __device__ __forceinline__ bool d_is_token_char(uint8_t c) {
@@ -1441,7 +1443,9 @@ __device__ __forceinline__ void d_string_tile_loop(
     const bool use_tile = nbytes <= kWaveTileBytes;
     if (use_tile) {
       const v4u* src = (const v4u*)(col_data + byte0);
-      const uint32_t n16 = (nbytes + 15) >> 4;
+      // round up to a full 64-slot stride: all 64 lanes always load together
+      // (the staging arena leaves >=1 KiB of slack after each column)
+      const uint32_t n16 = (((nbytes + 15) >> 4) + 63) & ~63u;
       uint32_t k = lane;
       // 16-deep batches: a 64-row group of ~256 B rows is 1024 slots, so one
       // batch puts the whole group's loads in flight per lane (256 B/lane)

@@ -305,7 +305,9 @@ struct BlockStageCtx {
       sc.const_value.assign((const char*)dec.data.data(), dec.data.size());
     } else {
       sc.d_data = st->push(dec.data.data(), dec.data.size());
-      st->reserve(16);  // tiled 16 B copy may read up to 15 B past the data end
+      // the tile copy rounds groups up to full 64-slot (1 KiB) strides and
+      // may read up to ~1 KiB past the group end
+      st->reserve(1024);
       sc.d_offsets = (const uint32_t*)st->push(dec.offsets.data(),
                                                dec.offsets.size() * 4, 4);
       sc.data_bytes = dec.data.size();
