@@ -203,6 +203,18 @@ int orc_parse_iso8601(const char* s, long sn, long long* out) {
   return 1;
 }
 
+// TryParseTimestampRFC3339Nano / parseMathNumber probes for golden tests
+long long orc_parse_rfc3339(const char* s, long sn, long long* out) {
+  int64_t v;
+  if (!try_parse_timestamp_rfc3339(strview(s, size_t(sn)), &v)) return -1;
+  *out = v;
+  return 0;
+}
+
+double orc_parse_math_number(const char* s, long sn) {
+  return parse_math_number(strview(s, size_t(sn)));
+}
+
 // CPU reference for the GPU ingest-side bloom build (test infrastructure):
 // exactly the write path's tokenizeHashes + bloomFilterMarshalHashes
 // (block.go:160-168 via part_writer.cpp:162).

@@ -160,3 +160,76 @@ def test_parse_iso8601_rejects(lib):
     for bad in [b"2023-11-14T22:13:20Z", b"2023/11/14T22:13:20.123Z",
                 b"1500-01-01T00:00:00.000Z", b"2023-13-40T22:13:20.12Z"]:
         assert lib.orc_parse_iso8601(bad, len(bad), ctypes.byref(out)) == 0
+
+
+def test_rfc3339_parse_table(lib):
+    """TryParseTimestampRFC3339Nano success/failure tables
+    (values_encoder_test.go:151-236), expected nsecs via Python datetime."""
+    import ctypes
+    from datetime import datetime, timezone
+
+    lib.orc_parse_rfc3339.restype = ctypes.c_longlong
+
+    def parse(s):
+        out = ctypes.c_longlong()
+        r = lib.orc_parse_rfc3339(s.encode(), len(s), ctypes.byref(out))
+        return out.value if r == 0 else None
+
+    def py_ns(iso, frac_ns=0):
+        dt = datetime.strptime(iso, "%Y-%m-%dT%H:%M:%S").replace(
+            tzinfo=timezone.utc)
+        return int(dt.timestamp()) * 10**9 + frac_ns
+
+    ok = [
+        ("2023-01-15T23:45:51Z", py_ns("2023-01-15T23:45:51")),
+        ("2023-01-15T23:45:51.1Z", py_ns("2023-01-15T23:45:51", 100000000)),
+        ("2023-01-15T23:45:51.123456789Z",
+         py_ns("2023-01-15T23:45:51", 123456789)),
+        ("1677-09-21T00:12:44Z", py_ns("1677-09-21T00:12:44")),
+        ("2262-04-11T23:47:15.999999999Z",
+         py_ns("2262-04-11T23:47:15", 999999999)),
+        # timezone offsets (values_encoder_test.go:186-188)
+        ("2023-01-16T00:45:51+01:00", py_ns("2023-01-15T23:45:51")),
+        ("2023-01-16T00:45:51.123-01:00",
+         py_ns("2023-01-16T01:45:51", 123000000)),
+        # SQL datetime delimiter
+        ("2023-01-16 00:45:51+01:00", py_ns("2023-01-15T23:45:51")),
+        ("2023-01-16 00:45:51.123-01:00",
+         py_ns("2023-01-16T01:45:51", 123000000)),
+    ]
+    for s, want in ok:
+        assert parse(s) == want, s
+
+    bad = ["", "foobar", "2023-01-15T22:15:51.Z", "1676-09-21T00:12:43Z",
+           "2263-04-11T23:47:17Z", "1677-09-21T00:12:43.999999999Z",
+           "2262-04-11T23:47:16Z", "YYYY-04-11T23:47:17Z",
+           "2023-MM-11T23:47:17Z", "2023-01-DDT23:47:17Z"]
+    for s in bad:
+        assert parse(s) is None, s
+
+
+def test_parse_math_number_legs(lib):
+    """parseMathNumber legs (pipe_math.go:1066-1080, block_result.go:2710-2752)."""
+    import ctypes
+    import math
+
+    lib.orc_parse_math_number.restype = ctypes.c_double
+
+    def pm(s):
+        return lib.orc_parse_math_number(s.encode(), len(s))
+
+    cases = [
+        ("123", 123.0), ("-1.5", -1.5), ("1e5", 1e5), ("1.5e-3", 0.0015),
+        ("0x1F", 31.0), ("0o17", 15.0), ("0b101", 5.0), ("017", 17.0),
+        ("1_000", 1000.0), ("inf", math.inf), ("-Inf", -math.inf),
+        ("1.5KB", 1500.0), ("2h30m", 9e12), ("100ms", 1e8),
+        ("10.0.0.1", 167772161.0),
+        ("2024-01-01T00:00:00Z", 1704067200e9),
+        ("0x1.8p1", 3.0),
+        ("9007199254740993", 9007199254740992.0),  # tryParseFloat64 leg, lossy
+    ]
+    for s, want in cases:
+        got = pm(s)
+        assert got == want, f"{s}: got {got} want {want}"
+    for s in ["abc", "", "1.2.3", "12:34", "nan?"]:
+        assert math.isnan(pm(s)), s
