@@ -1552,124 +1552,6 @@ __device__ __forceinline__ void d_string_tile_loop(
   }
 }
 
-
-// Software-pipelined variant of the tile loop for the hottest kinds: group
-// g+1's global loads are issued into registers BEFORE group g's match runs,
-// so the wave's own HBM latency overlaps its match work (on top of the
-// 8-waves/CU interleaving).  17 v4u registers stage up to 1088 slots.
-template <typename EvalFn>
-__device__ __forceinline__ void d_string_tile_loop_pipe(
-    const uint8_t* __restrict__ col_data, const uint32_t* __restrict__ col_offs,
-    uint8_t* wtile, uint64_t* out, uint32_t r0, uint32_t r1, uint32_t nwords,
-    int lane, int wave, int nwaves, EvalFn eval) {
-  typedef uint32_t v4u __attribute__((ext_vector_type(4)));
-  v4u* dst = (v4u*)wtile;
-  uint32_t wd = wave;
-  if (wd >= nwords) return;
-  v4u regs[16];  // 1024 slots; the unaligned tail (<=64 slots) copies direct
-
-  auto off_of = [&](uint32_t g) {
-    return col_offs[min(r0 + g * 64 + uint32_t(lane), r1)];
-  };
-  auto off_end_of = [&](uint32_t g) {
-    return lane == 0 ? col_offs[min(r0 + g * 64 + 64, r1)] : 0;
-  };
-
-  // offsets for wd and wd+nwaves in flight
-  uint32_t o_lane = off_of(wd), o_end = off_end_of(wd);
-  uint32_t o_next = 0, o_end_next = 0;
-  if (wd + nwaves < nwords) {
-    o_next = off_of(wd + nwaves);
-    o_end_next = off_end_of(wd + nwaves);
-  }
-
-  uint32_t cur_b0 = uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
-  uint32_t cur_b1 = uint32_t(__shfl(int(o_end), 0, 64));
-  uint32_t cur_nb = cur_b1 - cur_b0;
-  bool cur_tile = cur_nb <= kWaveTileBytes;
-  uint32_t cur_n16 = (((cur_nb + 15) >> 4) + 63) & ~63u;
-
-  auto issue = [&](uint32_t b0, uint32_t n16) {
-    const v4u* src = (const v4u*)(col_data + b0);
-    const uint32_t lim = n16 < 1024 ? n16 : 1024;
-#pragma unroll
-    for (int j = 0; j < 16; j++) {
-      const uint32_t k = uint32_t(lane) + uint32_t(j) * 64;
-      if (k < lim) regs[j] = __builtin_nontemporal_load(src + k);
-    }
-  };
-  if (cur_tile) issue(cur_b0, cur_n16);
-
-  while (wd < nwords) {
-    const uint32_t next_wd = wd + nwaves;
-    const uint32_t nn_wd = next_wd + nwaves;
-    // prefetch offsets two groups ahead
-    uint32_t o_nn = 0, o_end_nn = 0;
-    if (nn_wd < nwords) {
-      o_nn = off_of(nn_wd);
-      o_end_nn = off_end_of(nn_wd);
-    }
-    // stage the current group's registers into LDS (counted vmcnt waits on
-    // exactly these loads, which have been in flight since last iteration)
-    if (cur_tile) {
-      const uint32_t lim = cur_n16 < 1024 ? cur_n16 : 1024;
-#pragma unroll
-      for (int j = 0; j < 16; j++) {
-        const uint32_t k = uint32_t(lane) + uint32_t(j) * 64;
-        if (k < lim) dst[k ^ ((k >> 4) & 15)] = regs[j];
-      }
-      // unaligned tail beyond the staged 1024 slots: direct copy (rare)
-      const v4u* src = (const v4u*)(col_data + cur_b0);
-      for (uint32_t k = uint32_t(lane) + 1024; k < cur_n16; k += 64) {
-        dst[k ^ ((k >> 4) & 15)] = src[k];
-      }
-    }
-    // next group's parameters (its offsets landed an iteration ago) + loads
-    uint32_t nx_b0 = 0, nx_b1 = 0, nx_n16 = 0;
-    bool nx_tile = false;
-    if (next_wd < nwords) {
-      nx_b0 = uint32_t(__builtin_amdgcn_readfirstlane(int(o_next))) & ~15u;
-      nx_b1 = uint32_t(__shfl(int(o_end_next), 0, 64));
-      const uint32_t nb = nx_b1 - nx_b0;
-      nx_tile = nb <= kWaveTileBytes;
-      nx_n16 = (((nb + 15) >> 4) + 63) & ~63u;
-      if (nx_tile) issue(nx_b0, nx_n16);
-    }
-    if (cur_tile) {
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    }
-    // match the current group
-    const uint32_t g0 = r0 + wd * 64;
-    const uint32_t g1 = min(g0 + 64, r1);
-    const uint32_t ng = g1 - g0;
-    bool pred = false;
-    if (uint32_t(lane) < ng) {
-      const long s = o_lane;
-      const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
-      const long e_fix = uint32_t(lane) == ng - 1 ? long(cur_b1) : e;
-      if (cur_tile) {
-        TileAcc a{wtile};
-        pred = eval(a, s - cur_b0, e_fix - s);
-      } else {
-        GlobalAcc a{col_data};
-        pred = eval(a, s, e_fix - s);
-      }
-    }
-    const uint64_t word = __ballot(pred);
-    if (lane == 0) out[wd] = word;
-    // rotate pipeline state
-    wd = next_wd;
-    o_lane = o_next;
-    o_end = o_end_next;
-    o_next = o_nn;
-    o_end_next = o_end_nn;
-    cur_b0 = nx_b0;
-    cur_b1 = nx_b1;
-    cur_n16 = nx_n16;
-    cur_tile = nx_tile;
-  }
-}
-
 // ---- the program kernel ----
 
 __global__ __launch_bounds__(256) void scan_program_kernel(
@@ -1734,19 +1616,20 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
         const uint8_t op_flags = lb.flags;
         if (lb.kind == kScanPhraseStr) {
           // hot clone: only the phrase matcher in the loop body
-          d_string_tile_loop_pipe(col_data, col_offs, wtile, out, r0, r1,
-                                  nwords, lane, wave, nwaves,
-                                  [=](const auto& a, long s0, long sn) {
-                                    return d_match_phrase_at(a, s0, sn, op_ptr,
-                                                             op_len, op_flags);
-                                  });
+          d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                    wtile, out, r0, r1, nwords, lane, wave,
+                                    nwaves, [=](const auto& a, long s0, long sn) {
+                                      return d_match_phrase_at(a, s0, sn,
+                                                               op_ptr, op_len,
+                                                               op_flags);
+                                    });
         } else if (lb.kind == kScanRegexStr) {
           // second hot clone: regex fast paths + NFA (BASELINE config 3)
-          d_string_tile_loop_pipe(col_data, col_offs, wtile, out, r0, r1,
-                                  nwords, lane, wave, nwaves,
-                                  [=](const auto& a, long s0, long sn) {
-                                    return d_regex_match_at(op_ptr, a, s0, sn);
-                                  });
+          d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                    wtile, out, r0, r1, nwords, lane, wave,
+                                    nwaves, [=](const auto& a, long s0, long sn) {
+                                      return d_regex_match_at(op_ptr, a, s0, sn);
+                                    });
         } else {
           const bool anycase = lb.kind == kScanAnyCasePhraseStr ||
                                lb.kind == kScanAnyCasePrefixStr;
