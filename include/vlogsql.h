@@ -92,6 +92,22 @@ long long vql_bloom_build(const unsigned char* data, long long nbytes,
                           const unsigned int* offsets, long long rows,
                           int device, unsigned char* out, long long cap);
 
+/* Whole-query driver (§8b vql_scan_query): shards `nparts` parts across
+ * `ngpus` devices (round-robin by part, the reference's independent-worker
+ * batching, storage_search.go:1035-1067), stages + scans each shard on its
+ * device concurrently, and reduces the per-device matched-row counters (an
+ * in-process sum; across processes the same reduction is RCCL all_reduce —
+ * bench.py's multi-rank path).  Fills stats{matched_rows, bytes_scanned,
+ * rows_scanned, elapsed_ms} and returns matched rows, or -1. */
+typedef struct {
+  unsigned long long matched_rows;
+  unsigned long long rows_scanned;
+  unsigned long long bytes_scanned; /* algorithmic bytes over all passes */
+  double elapsed_ms;
+} vql_stats;
+long long vql_scan_query(void** parts, int nparts, void* filter, int ngpus,
+                         vql_stats* stats);
+
 /* Cold path (§8b vql_scan_batch): stage + scan + fetch + free in one call. */
 long long vql_scan_batch(void* part, void* filter, long block_lo, long block_hi,
                          unsigned long long* out_words, long long cap_words,

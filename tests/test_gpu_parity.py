@@ -148,3 +148,44 @@ def test_multichunk_block(tmp_path):
         '{"type":"time","min":1700000000005000000,"max":1700000000015000000}',
     ]:
         assert_parity(d, f)
+
+
+def test_scan_query_driver(gen_part, typed_part):
+    """vql_scan_query (§8b): multi-part whole-query driver on one device must
+    equal the sum of per-part oracle scans."""
+    import ctypes
+
+    from victorialogs_amd import OracleScanner, Part, Filter
+    from victorialogs_amd.api import load_product
+
+    class VqlStats(ctypes.Structure):
+        _fields_ = [("matched_rows", ctypes.c_ulonglong),
+                    ("rows_scanned", ctypes.c_ulonglong),
+                    ("bytes_scanned", ctypes.c_ulonglong),
+                    ("elapsed_ms", ctypes.c_double)]
+
+    lib = load_product()
+    lib.vql_scan_query.restype = ctypes.c_longlong
+
+    fjson = '{"type":"phrase","field":"_msg","phrase":"stream"}'
+    parts = [Part(gen_part), Part(typed_part)]
+    filt = Filter(fjson)
+    arr = (ctypes.c_void_p * 2)(parts[0].h, parts[1].h)
+    st = VqlStats()
+    n = lib.vql_scan_query(arr, 2, filt.h, 1, ctypes.byref(st))
+    assert n >= 0, lib.vql_errstr().decode()
+
+    want = 0
+    rows = 0
+    for d in (gen_part, typed_part):
+        sc = OracleScanner(d)
+        h, _ = sc.scan(fjson)
+        want += h
+        rows += sum(sc.block_rows(i) for i in range(sc.blocks))
+        sc.close()
+    assert n == want == st.matched_rows
+    assert st.rows_scanned == rows
+    assert st.bytes_scanned > 0 and st.elapsed_ms > 0
+    filt.close()
+    for p in parts:
+        p.close()

@@ -17,7 +17,9 @@
 #include <functional>
 #include <map>
 #include <memory>
+#include <chrono>
 #include <mutex>
+#include <thread>
 #include <string>
 #include <vector>
 
@@ -2540,6 +2542,69 @@ int vql_fetch_block_hits(void* s, unsigned long long* out, long long cap) {
 
 // Cold path (SURVEY.md §8b vql_scan_batch): stage blocks [lo,hi), scan once,
 // fetch bitmaps, free.  Returns matched rows or -1.
+// Whole-query driver (SURVEY.md §8b vql_scan_query): parts round-robin over
+// devices, one staging+scan thread per device, counters summed in-process.
+struct VqlStatsC {
+  unsigned long long matched_rows;
+  unsigned long long rows_scanned;
+  unsigned long long bytes_scanned;
+  double elapsed_ms;
+};
+
+extern "C" long long vql_scan_query(void** parts, int nparts, void* filter,
+                                    int ngpus, VqlStatsC* stats) {
+  try {
+    if (nparts <= 0 || ngpus <= 0) fail("vql_scan_query: bad arguments");
+    auto& vf = *(std::shared_ptr<VqlFilter>*)filter;
+    std::vector<std::thread> threads;
+    std::vector<unsigned long long> hits(size_t(ngpus), 0);
+    std::vector<unsigned long long> rows(size_t(ngpus), 0);
+    std::vector<unsigned long long> bytes(size_t(ngpus), 0);
+    std::vector<std::string> errs{std::vector<std::string>::size_type(ngpus)};
+    auto t0 = std::chrono::steady_clock::now();
+    for (int d = 0; d < ngpus; d++) {
+      threads.emplace_back([&, d]() {
+        try {
+          for (int i = d; i < nparts; i += ngpus) {
+            Stage* st = build_stage((VqlPart*)parts[i], vf, d, 0, -1);
+            std::unique_ptr<Stage> guard(st);
+            long long h2 = run_scan(st);
+            if (h2 < 0) fail("scan failed");
+            hits[size_t(d)] += (unsigned long long)h2;
+            rows[size_t(d)] += st->rows;
+            bytes[size_t(d)] += st->algo_bytes;
+          }
+        } catch (const std::exception& e) {
+          errs[size_t(d)] = e.what();
+        }
+      });
+    }
+    for (auto& t : threads) t.join();
+    for (const auto& e : errs) {
+      if (!e.empty()) fail("vql_scan_query: " + e);
+    }
+    double ms = std::chrono::duration<double, std::milli>(
+                    std::chrono::steady_clock::now() - t0)
+                    .count();
+    unsigned long long h = 0, r = 0, b = 0;
+    for (int d = 0; d < ngpus; d++) {
+      h += hits[size_t(d)];
+      r += rows[size_t(d)];
+      b += bytes[size_t(d)];
+    }
+    if (stats) {
+      stats->matched_rows = h;
+      stats->rows_scanned = r;
+      stats->bytes_scanned = b;
+      stats->elapsed_ms = ms;
+    }
+    return (long long)h;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
 // GPU ingest-side bloom build: marshaled bloom bytes
 // for one column block, bit-identical to the write path's
 // tokenizeHashes + bloomFilterMarshalHashes (block.go:160-168).  `offsets` is
