@@ -166,6 +166,9 @@ def test_scan_query_driver(gen_part, typed_part):
 
     lib = load_product()
     lib.vql_scan_query.restype = ctypes.c_longlong
+    lib.vql_scan_query.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_void_p]
 
     fjson = '{"type":"phrase","field":"_msg","phrase":"stream"}'
     parts = [Part(gen_part), Part(typed_part)]
