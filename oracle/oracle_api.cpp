@@ -203,6 +203,28 @@ int orc_parse_iso8601(const char* s, long sn, long long* out) {
   return 1;
 }
 
+// tryParseDuration / tryParseBytes / tryParseIPv4 probes for golden tests
+long long orc_parse_duration(const char* s, long sn, long long* out) {
+  int64_t v;
+  if (!try_parse_duration(strview(s, size_t(sn)), &v)) return -1;
+  *out = v;
+  return 0;
+}
+
+long long orc_parse_bytes(const char* s, long sn, long long* out) {
+  int64_t v;
+  if (!try_parse_bytes(strview(s, size_t(sn)), &v)) return -1;
+  *out = v;
+  return 0;
+}
+
+long long orc_parse_ipv4(const char* s, long sn, unsigned int* out) {
+  uint32_t v;
+  if (!try_parse_ipv4(strview(s, size_t(sn)), &v)) return -1;
+  *out = v;
+  return 0;
+}
+
 // TryParseTimestampRFC3339Nano / parseMathNumber probes for golden tests
 long long orc_parse_rfc3339(const char* s, long sn, long long* out) {
   int64_t v;

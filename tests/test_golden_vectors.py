@@ -233,3 +233,95 @@ def test_parse_math_number_legs(lib):
         assert got == want, f"{s}: got {got} want {want}"
     for s in ["abc", "", "1.2.3", "12:34", "nan?"]:
         assert math.isnan(pm(s)), s
+
+
+def _probe(lib, fn, s, ctype):
+    import ctypes
+    out = ctype()
+    b = s.encode()
+    r = getattr(lib, fn)(b, len(b), ctypes.byref(out))
+    return out.value if r == 0 else None
+
+
+def test_duration_parse_table(lib):
+    """tryParseDuration tables (values_encoder_test.go:316-393)."""
+    import ctypes
+    ns, us, ms = 1, 10**3, 10**6
+    sec, minute, hour = 10**9, 60 * 10**9, 3600 * 10**9
+    day, week = 24 * 3600 * 10**9, 7 * 24 * 3600 * 10**9
+    year = 365 * 24 * 3600 * 10**9  # values_encoder.go:1131
+
+    def dur(s):
+        return _probe(lib, "orc_parse_duration", s, ctypes.c_longlong)
+
+    ok = [
+        ("0s", 0), ("0.0w0d0h0s0.0ms", 0), ("-0.0w0.00d0h0s0.0000ms", 0),
+        ("-0w", 0), ("1s", sec), ("1.5ms", int(1.5 * ms)), ("1µs", us),
+        ("1ns", 1), ("1h", hour), ("0.001h", int(0.001 * hour)),
+        ("0.05h", int(0.05 * hour)), ("1.5d", int(1.5 * day)),
+        ("1.5w", int(1.5 * week)), ("2.5y", int(2.5 * year)),
+        ("1h5m35s", hour + 5 * minute + 35 * sec),
+        ("1m5.123456789s", minute + int(5.123456789 * sec)),
+        ("1h5m", hour + 5 * minute),
+        ("1.1h5m2.5s3_456ns", int(1.1 * hour) + 5 * minute + int(2.5 * sec)
+         + 3456),
+        ("-1h5m3s", -(hour + 5 * minute + 3 * sec)),
+        ("9_223_372_036_854_775_807ns", 2**63 - 1),
+        ("9223372036854775807ns", 2**63 - 1),
+        ("-9223372036854775808ns", -2**63 + 1),
+        ("15_223_372_036_854_775_808ns", 2**63 - 1),  # clamped
+        ("-15_223_372_036_854_775_808ns", -2**63 + 1),
+    ]
+    for s, want in ok:
+        assert dur(s) == want, f"{s}: got {dur(s)} want {want}"
+    for s in ["", "2", "2.5", "foobar", "1foo", "1soo", "3.43e", "3.43es",
+              " 2s", "2s ", "2s 3ms"]:
+        assert dur(s) is None, s
+
+
+def test_bytes_parse_table(lib):
+    """tryParseBytes tables (values_encoder_test.go:416-507)."""
+    import ctypes
+
+    def b(s):
+        return _probe(lib, "orc_parse_bytes", s, ctypes.c_longlong)
+
+    ok = [
+        ("1_500", 1500), ("2.5B", 2),
+        ("1.5K", 1500), ("1.5M", 1500000), ("1.5G", 1500000000),
+        ("1.5T", 1500000000000),
+        ("1.5KB", 1500), ("1.5MB", 1500000), ("1.5GB", 1500000000),
+        ("1.5TB", 1500000000000),
+        ("1.5Ki", int(1.5 * 2**10)), ("1.5Mi", int(1.5 * 2**20)),
+        ("1.5Gi", int(1.5 * 2**30)), ("1.5Ti", int(1.5 * 2**40)),
+        ("1.5KiB", int(1.5 * 2**10)), ("1.5MiB", int(1.5 * 2**20)),
+        ("1.5GiB", int(1.5 * 2**30)), ("1.5TiB", int(1.5 * 2**40)),
+        ("1MiB500KiB200B", 2**20 + 500 * 2**10 + 200),
+        ("9_223_372_036_854_775_807", 2**63 - 1),
+        ("9223372036854775807B", 2**63 - 1),
+        ("-9223372036854775808B", -2**63 + 1),
+        ("15_223_372_036_854_775_808", 2**63 - 1),
+        ("-15_223_372_036_854_775_808", -2**63 + 1),
+    ]
+    for s, want in ok:
+        assert b(s) == want, f"{s}: got {b(s)} want {want}"
+    for s in ["", "foobar", "123q", "123qs", "123qsb", "123sqsb", "123s5qsb",
+              "1b", "1k", "1m", "1g", "1t", "1kb", "1mb", "1gb", "1tb"]:
+        assert b(s) is None, s
+
+
+def test_ipv4_parse_table(lib):
+    """tryParseIPv4 tables (values_encoder_test.go:100-149)."""
+    import ctypes
+
+    def ip(s):
+        return _probe(lib, "orc_parse_ipv4", s, ctypes.c_uint)
+
+    for s in ["0.0.0.0", "1.2.3.4", "255.255.255.255", "127.0.0.1"]:
+        a, b, c, d = (int(x) for x in s.split("."))
+        assert ip(s) == (a << 24) | (b << 16) | (c << 8) | d, s
+    for s in ["", "foo", "a.b.c.d", "127.0.0.x", "127.0.x.0", "127.x.0.0",
+              "x.0.0.0", "127.127.127.256", "127.127.256.127",
+              "127.256.127.127", "256.127.127.127", "-1.127.127.127",
+              "127.-1.127.127", "127.127.-1.127", "127.127.127.-1"]:
+        assert ip(s) is None, s
