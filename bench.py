@@ -162,7 +162,8 @@ def main():
     t0 = time.time()
     parts = [Part(d) for d in part_dirs]
     filt = Filter(filter_json)
-    stages = [Stage(p, filt, device=local_rank) for p in parts]
+    # one multi-part stage: the whole pass is ONE kernel launch
+    stages = [Stage(parts, filt, device=local_rank)]
     staged_bytes = sum(s.staged_bytes for s in stages)
     algo_bytes = sum(s.algo_bytes for s in stages)
     rows = sum(s.rows for s in stages)

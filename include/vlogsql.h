@@ -53,6 +53,10 @@ void vql_free_filter(void* filter);
  * block_hi < 0 means "all blocks". */
 void* vql_stage(void* part, void* filter, int device, long block_lo,
                 long block_hi);
+/* Stages every block of nparts parts into ONE context/launch (the worker
+ * batching of storage_search.go:1035-1067 is part-agnostic).  Bitmaps are
+ * concatenated in (part, block) order. */
+void* vql_stage_parts(void** parts, int nparts, void* filter, int device);
 void vql_stage_free(void* stage);
 long long vql_stage_bytes(void* stage);      /* bytes resident in HBM */
 long long vql_stage_algo_bytes(void* stage); /* algorithmic bytes per pass */

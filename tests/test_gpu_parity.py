@@ -192,3 +192,35 @@ def test_scan_query_driver(gen_part, typed_part):
     filt.close()
     for p in parts:
         p.close()
+
+
+def test_multipart_stage(gen_part, typed_part):
+    """vql_stage_parts: one launch over two parts must equal the per-part
+    oracle bitmaps concatenated in (part, block) order."""
+    from victorialogs_amd import Filter, OracleScanner, Part, Stage
+
+    fjson = '{"type":"phrase","field":"_msg","phrase":"stream"}'
+    parts = [Part(gen_part), Part(typed_part)]
+    filt = Filter(fjson)
+    st = Stage(parts, filt, device=0)
+    hits = st.scan()
+
+    want_words = b""
+    want_hits = 0
+    for d in (gen_part, typed_part):
+        sc = OracleScanner(d)
+        h, words = sc.scan(fjson, with_bitmaps=True)
+        want_hits += h
+        want_words += words
+        sc.close()
+    assert hits == want_hits
+    got = st.fetch_bitmaps()
+    assert got == want_words
+    # gather across parts: global rowids keep increasing over the part seam
+    _, rowids = st.gather("_msg")
+    assert len(rowids) == hits
+    assert all(a < b for a, b in zip(rowids, rowids[1:]))
+    st.close()
+    filt.close()
+    for p in parts:
+        p.close()

@@ -259,11 +259,22 @@ class Filter:
 
 
 class Stage:
-    """Staged (HBM-resident) scan context: decode-once, scan-many."""
+    """Staged (HBM-resident) scan context: decode-once, scan-many.
 
-    def __init__(self, part: Part, filt: Filter, device=0, lo=0, hi=-1):
+    `part` may be a single Part or a list of Parts — a multi-part stage scans
+    every block of every part in ONE kernel launch (vql_stage_parts)."""
+
+    def __init__(self, part, filt: Filter, device=0, lo=0, hi=-1):
         self.lib = load_product()
-        self.h = self.lib.vql_stage(part.h, filt.h, device, lo, hi)
+        if isinstance(part, (list, tuple)):
+            arr = (ctypes.c_void_p * len(part))(*[p.h for p in part])
+            self.lib.vql_stage_parts.restype = ctypes.c_void_p
+            self.lib.vql_stage_parts.argtypes = [
+                ctypes.POINTER(ctypes.c_void_p), ctypes.c_int,
+                ctypes.c_void_p, ctypes.c_int]
+            self.h = self.lib.vql_stage_parts(arr, len(part), filt.h, device)
+        else:
+            self.h = self.lib.vql_stage(part.h, filt.h, device, lo, hi)
         if not self.h:
             raise RuntimeError(self.lib.vql_errstr().decode())
 

@@ -196,10 +196,14 @@ struct LeafInfo {
 };
 
 struct Stage {
-  VqlPart* part;
+  VqlPart* part;                      // first part (compat)
   std::shared_ptr<VqlFilter> filter;  // keep alive
   int device = 0;
-  long lo = 0, hi = 0;
+  long lo = 0, hi = 0;                // single-part ranges (compat)
+  // every staged block: (part, block index) — multi-part stages scan all of
+  // them in ONE kernel launch (the reference worker model batches blocks
+  // without caring which part they came from, storage_search.go:1035-1067)
+  std::vector<std::pair<VqlPart*, long>> block_refs;
 
   uint8_t* arena = nullptr;
   size_t arena_cap = 0, arena_used = 0;
@@ -2220,32 +2224,36 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
   }
 }
 
-Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
-                   long lo, long hi) {
+Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
+                        std::shared_ptr<VqlFilter> filter, int device) {
+  if (refs.empty()) fail("build_stage: no blocks");
   auto st = std::make_unique<Stage>();
-  st->part = part;
+  st->part = refs[0].first;
   st->filter = filter;
   st->device = device;
-  if (hi < 0 || size_t(hi) > part->bhs.size()) hi = long(part->bhs.size());
-  if (lo < 0) lo = 0;
-  st->lo = lo;
-  st->hi = hi;
+  st->lo = 0;
+  st->hi = long(refs.size());
+  st->block_refs = std::move(refs);
 
   HIP_CHECK(hipSetDevice(device));
   HIP_CHECK(hipStreamCreate(&st->stream));
   HIP_CHECK(hipEventCreate(&st->ev0));
   HIP_CHECK(hipEventCreate(&st->ev1));
 
-  const long nblocks = hi - lo;
+  const long nblocks = long(st->block_refs.size());
   const int nleaves = int(filter->leaves.size());
+  auto bh_of = [&](long i) -> const BlockHeader& {
+    return st->block_refs[size_t(i)].first->bhs[size_t(
+        st->block_refs[size_t(i)].second)];
+  };
 
   // Arena estimate: decoded column data is bounded by the part's uncompressed
   // size; add offsets (4 B/row/string-col), timestamps (8 B/row), blooms and
   // the output bitmaps, with slack.
   uint64_t rows = 0;
   uint64_t est = 64 << 20;
-  for (long b = lo; b < hi; b++) {
-    const BlockHeader& bh = part->bhs[size_t(b)];
+  for (long b = 0; b < nblocks; b++) {
+    const BlockHeader& bh = bh_of(b);
     rows += bh.rows_count;
     est += bh.uncompressed_size_bytes + bh.rows_count * 24 + (2 << 20);
   }
@@ -2328,10 +2336,9 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
 
   // bitmap output buffer
   st->block_word_off.resize(size_t(nblocks) + 1, 0);
-  for (long b = lo; b < hi; b++) {
-    st->block_word_off[size_t(b - lo) + 1] =
-        st->block_word_off[size_t(b - lo)] +
-        (part->bhs[size_t(b)].rows_count + 63) / 64;
+  for (long b = 0; b < nblocks; b++) {
+    st->block_word_off[size_t(b) + 1] =
+        st->block_word_off[size_t(b)] + (bh_of(b).rows_count + 63) / 64;
   }
   st->bitmap_words = size_t(st->block_word_off.back());
   st->d_bitmap = (uint64_t*)st->reserve(st->bitmap_words * 8, 8);
@@ -2340,21 +2347,22 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
   std::vector<DevBlock> blocks_h((size_t(nblocks)));
   std::vector<DevLeafBlock> lbs_h(size_t(nblocks) * size_t(nleaves));
   std::vector<DevChunk> chunks_h;
-  for (long b = lo; b < hi; b++) {
-    const BlockHeader& bh = part->bhs[size_t(b)];
+  for (long b = 0; b < nblocks; b++) {
+    VqlPart* bp = st->block_refs[size_t(b)].first;
+    const BlockHeader& bh = bh_of(b);
     BlockStageCtx ctx;
     ctx.st = st.get();
-    ctx.pr = &part->pr;
+    ctx.pr = &bp->pr;
     ctx.bh = &bh;
-    part->pr.read_block_columns(bh, ctx.bc);
+    bp->pr.read_block_columns(bh, ctx.bc);
 
-    DevBlock& db = blocks_h[size_t(b - lo)];
+    DevBlock& db = blocks_h[size_t(b)];
     db.rows = uint32_t(bh.rows_count);
-    db.bitmap_out = st->d_bitmap + st->block_word_off[size_t(b - lo)];
+    db.bitmap_out = st->d_bitmap + st->block_word_off[size_t(b)];
     db.hits_out = nullptr;  // filled after d_block_hits is allocated
 
     for (int l = 0; l < nleaves; l++) {
-      DevLeafBlock& lb = lbs_h[size_t(b - lo) * size_t(nleaves) + size_t(l)];
+      DevLeafBlock& lb = lbs_h[size_t(b) * size_t(nleaves) + size_t(l)];
       stage_leaf(leaf_infos[size_t(l)], ctx, *st, lb);
       if (lb.mode == kModeScan) {
         // algorithmic bytes one pass must read for this leaf
@@ -2393,7 +2401,7 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
     // chunks
     uint32_t nch = uint32_t((bh.rows_count + kChunkRows - 1) / kChunkRows);
     for (uint32_t c = 0; c < nch; c++) {
-      chunks_h.push_back(DevChunk{uint32_t(b - lo), c});
+      chunks_h.push_back(DevChunk{uint32_t(b), c});
     }
   }
   // bitmap write traffic
@@ -2401,8 +2409,8 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
 
   st->nchunks = uint32_t(chunks_h.size());
   st->chunks_h = chunks_h;
-  for (long b = lo; b < hi; b++) {
-    st->block_rows_h.push_back(uint32_t(part->bhs[size_t(b)].rows_count));
+  for (long b = 0; b < nblocks; b++) {
+    st->block_rows_h.push_back(uint32_t(bh_of(b).rows_count));
   }
 
   HIP_CHECK(hipMalloc(&st->d_ops, filter->ops.size() * sizeof(DevOp)));
@@ -2426,11 +2434,24 @@ Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
   return st.release();
 }
 
+Stage* build_stage(VqlPart* part, std::shared_ptr<VqlFilter> filter, int device,
+                   long lo, long hi) {
+  if (hi < 0 || size_t(hi) > part->bhs.size()) hi = long(part->bhs.size());
+  if (lo < 0) lo = 0;
+  std::vector<std::pair<VqlPart*, long>> refs;
+  refs.reserve(size_t(hi - lo));
+  for (long b = lo; b < hi; b++) refs.emplace_back(part, b);
+  Stage* st = build_stage_refs(std::move(refs), std::move(filter), device);
+  st->lo = lo;
+  st->hi = hi;
+  return st;
+}
+
 long long run_scan(Stage* st) {
   HIP_CHECK(hipSetDevice(st->device));
   HIP_CHECK(hipMemsetAsync(st->d_hits, 0, 8, st->stream));
-  HIP_CHECK(hipMemsetAsync(st->d_block_hits, 0, size_t(st->hi - st->lo) * 8,
-                           st->stream));
+  HIP_CHECK(hipMemsetAsync(st->d_block_hits, 0,
+                           st->block_refs.size() * 8, st->stream));
   HIP_CHECK(hipEventRecord(st->ev0, st->stream));
   HIP_CHECK(vql_launch_scan(st->d_ops, int(st->filter->ops.size()), st->d_lbs,
                             int(st->filter->leaves.size()), st->d_blocks,
@@ -2488,6 +2509,24 @@ void* vql_stage(void* part, void* filter, int device, long block_lo, long block_
     return nullptr;
   }
 }
+
+/* Stages ALL blocks of `nparts` parts into one context: one kernel launch
+ * scans everything (the reference batches blocks regardless of their part,
+ * storage_search.go:1035-1067). */
+void* vql_stage_parts(void** parts, int nparts, void* filter, int device) {
+  try {
+    std::vector<std::pair<VqlPart*, long>> refs;
+    for (int i = 0; i < nparts; i++) {
+      auto* p = (VqlPart*)parts[i];
+      for (long b = 0; b < long(p->bhs.size()); b++) refs.emplace_back(p, b);
+    }
+    return build_stage_refs(std::move(refs),
+                            *(std::shared_ptr<VqlFilter>*)filter, device);
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return nullptr;
+  }
+}
 void vql_stage_free(void* s) { delete (Stage*)s; }
 long long vql_stage_bytes(void* s) { return (long long)((Stage*)s)->staged_bytes; }
 long long vql_stage_algo_bytes(void* s) {
@@ -2528,7 +2567,7 @@ int vql_fetch_bitmaps(void* s, unsigned long long* out_words, long long cap) {
 int vql_fetch_block_hits(void* s, unsigned long long* out, long long cap) {
   try {
     Stage* st = (Stage*)s;
-    long n = long(st->hi - st->lo);
+    long n = long(st->block_refs.size());
     if (cap < n) fail("vql_fetch_block_hits: buffer too small");
     HIP_CHECK(hipSetDevice(st->device));
     HIP_CHECK(hipMemcpy(out, st->d_block_hits, size_t(n) * 8,
@@ -2565,8 +2604,15 @@ extern "C" long long vql_scan_query(void** parts, int nparts, void* filter,
     for (int d = 0; d < ngpus; d++) {
       threads.emplace_back([&, d]() {
         try {
+          std::vector<std::pair<VqlPart*, long>> refs;
           for (int i = d; i < nparts; i += ngpus) {
-            Stage* st = build_stage((VqlPart*)parts[i], vf, d, 0, -1);
+            auto* p = (VqlPart*)parts[i];
+            for (long b = 0; b < long(p->bhs.size()); b++) {
+              refs.emplace_back(p, b);
+            }
+          }
+          if (!refs.empty()) {
+            Stage* st = build_stage_refs(std::move(refs), vf, d);
             std::unique_ptr<Stage> guard(st);
             long long h2 = run_scan(st);
             if (h2 < 0) fail("scan failed");
@@ -2737,30 +2783,32 @@ Stage::GatherCtx& gather_prepare(Stage* st, const std::string& field) {
   HIP_CHECK(hipSetDevice(st->device));
   Stage::GatherCtx& g = st->gathers[field];
   std::string cname = canonical_field(field);
-  const long nblocks = st->hi - st->lo;
+  const long nblocks = long(st->block_refs.size());
 
   std::vector<DevGatherCol> gcols((size_t(nblocks)));
   for (long b = 0; b < nblocks; b++) {
-    const BlockHeader& bh = st->part->bhs[size_t(st->lo + b)];
+    VqlPart* bpart = st->block_refs[size_t(b)].first;
+    const BlockHeader& bh =
+        bpart->bhs[size_t(st->block_refs[size_t(b)].second)];
     PartReader::BlockColumns bc;
-    st->part->pr.read_block_columns(bh, bc);
+    bpart->pr.read_block_columns(bh, bc);
     DevGatherCol& gc = gcols[size_t(b)];
     memset(&gc, 0, sizeof(gc));
 
     std::string cv;
-    if (st->part->pr.get_const_column(bc, cname, &cv)) {
+    if (bpart->pr.get_const_column(bc, cname, &cv)) {
       gc.src = kGatherConst;
       gc.cval_len = uint32_t(cv.size());
       gc.cval = st->push(cv.data(), cv.size(), 8);
       continue;
     }
     ColumnHeader ch;
-    if (!st->part->pr.get_column_header(bc, cname, &ch)) {
+    if (!bpart->pr.get_column_header(bc, cname, &ch)) {
       gc.src = kGatherMissing;
       continue;
     }
     StringsBlockDec dec;
-    st->part->pr.read_values(ch, bh.rows_count, dec);
+    bpart->pr.read_values(ch, bh.rows_count, dec);
     if (dec.is_const && ch.type == ValueType::String) {
       gc.src = kGatherConst;
       gc.cval_len = uint32_t(dec.data.size());
@@ -2821,7 +2869,7 @@ Stage::GatherCtx& gather_prepare(Stage* st, const std::string& field) {
   HIP_CHECK(hipFree(d_counts));
 
   // exclusive scan -> per-chunk bases; gid_base = first global row of block
-  std::vector<unsigned long long> block_row_base(size_t(st->hi - st->lo) + 1, 0);
+  std::vector<unsigned long long> block_row_base(st->block_refs.size() + 1, 0);
   for (size_t b = 0; b < st->block_rows_h.size(); b++) {
     block_row_base[b + 1] = block_row_base[b] + st->block_rows_h[b];
   }
