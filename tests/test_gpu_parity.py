@@ -214,7 +214,7 @@ def test_multipart_stage(gen_part, typed_part):
         want_words += words
         sc.close()
     assert hits == want_hits
-    got = st.fetch_bitmaps()
+    got = st.fetch_bitmaps(len(want_words) // 8)
     assert got == want_words
     # gather across parts: global rowids keep increasing over the part seam
     _, rowids = st.gather("_msg")
