@@ -185,6 +185,14 @@ TYPED_FILTERS = [
     '{"type":"regexp","field":"f64","re":"-1"}',
     '{"type":"range","field":"mix","min":0,"max":10000}',
     '{"type":"range","field":"mix","min":-10,"max":1.6e9}',
+    # {m,n} repetition and top-level anchors (expanded into the NFA)
+    '{"type":"regexp","field":"_msg","re":"to{1,2}k"}',
+    '{"type":"regexp","field":"u8","re":"1{2}"}',
+    '{"type":"regexp","field":"_msg","re":"^log line [0-9]{1,3} "}',
+    '{"type":"regexp","field":"lvl","re":"^(warn|error)$"}',
+    '{"type":"regexp","field":"u8","re":"^[0-9]{2}$"}',
+    '{"type":"regexp","field":"mix","re":"ms$"}',
+    '{"type":"regexp","field":"ip","re":"^10\\\\.2"}',
     # general regex class (Glushkov NFA on device)
     '{"type":"regexp","field":"_msg","re":"took \\\\d+ms"}',
     '{"type":"regexp","field":"_msg","re":"level=[a-z]+ took"}',

@@ -41,9 +41,9 @@ def test_filter_compile_errors():
         '{"type":"phrase"}',                       # missing fields
         '{"type":"wat","field":"x","phrase":"y"}', # unknown type
         'not json at all',
-        '{"type":"regexp","field":"x","re":"a{2,3}"}',  # {m,n} unsupported
         '{"type":"regexp","field":"x","re":"a\\\\b"}',  # word-boundary assertion
-        '{"type":"regexp","field":"x","re":"^foo"}',    # anchors unsupported
+        '{"type":"regexp","field":"x","re":"a{2000}"}', # repeat count too big
+        '{"type":"regexp","field":"x","re":"(a|^b)c"}', # mid-pattern anchor
     ]
     for f in bad:
         h = lib.vql_compile_filter(f.encode())

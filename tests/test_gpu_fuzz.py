@@ -24,6 +24,8 @@ GEN_REGEXES = [
     # NFA class
     "stream \\d+ and", "u64=\\d?\\d", "ip=[0-9]+\\.[0-9]+", "pa?d=",
     "[^;]*uuid", "w\\w+ker",
+    "^message", "worker 0;$", "^message.*uuid", "s{1,2}am", "e{2}",
+    "^[a-z ]{10,30}for", "o{1,3}",
 ]
 
 TYPED_PHRASES = [
@@ -38,6 +40,8 @@ TYPED_REGEXES = [
     # NFA class
     "took \\d+ms", "l[a-z]+e\\d", "ро?w", "[0-9]+\\.[0-9]+", "\\dms|GiB",
     "2024-\\d+-0[1-5]T",
+    "^log", "ms$", "^other stream row \\d{1,2};", "l{2}", "[0-9]{3}",
+    "^(debug|info|warn|error)$",
 ]
 
 

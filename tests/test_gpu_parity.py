@@ -116,7 +116,7 @@ def test_unsupported_regex_fails_at_compile(typed_part):
     """Unsupported constructs must raise a clear error at compile time,
     never fall back to CPU silently."""
     with pytest.raises(RuntimeError, match="not supported|fast-path"):
-        Filter('{"type":"regexp","field":"_msg","re":"a{2,3}"}')
+        Filter('{"type":"regexp","field":"_msg","re":"a\\\\bb"}')
 
 
 def test_multichunk_block(tmp_path):

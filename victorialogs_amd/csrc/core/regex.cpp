@@ -60,7 +60,84 @@ struct Parser {
           q.subs.push_back(std::move(atom));
           atom = std::move(q);
         } else if (c == '{') {
-          err("{m,n} repetition is not supported");
+          // {m}, {m,}, {m,n}: expanded by cloning (Go regexp/syntax
+          // semantics; a '{' that is not a valid repetition is a literal)
+          size_t save = pos;
+          pos++;
+          long m = -1, nrep = -2;  // nrep -2: no comma; -1: open-ended
+          long v = 0;
+          bool any = false;
+          while (!eof() && peek() >= '0' && peek() <= '9') {
+            v = v * 10 + (peek() - '0');
+            if (v > 100000) break;
+            any = true;
+            pos++;
+          }
+          if (any) m = v;
+          if (!eof() && peek() == ',') {
+            pos++;
+            v = 0;
+            bool any2 = false;
+            while (!eof() && peek() >= '0' && peek() <= '9') {
+              v = v * 10 + (peek() - '0');
+              if (v > 100000) break;
+              any2 = true;
+              pos++;
+            }
+            nrep = any2 ? v : -1;
+          }
+          if (m < 0 || eof() || peek() != '}') {
+            pos = save;  // literal '{'
+            break;
+          }
+          pos++;
+          if (nrep == -2) nrep = m;
+          if (m > 1000 || nrep > 1000) err("invalid repeat count");
+          if (nrep >= 0 && nrep < m) err("invalid repeat count");
+          const long max_copies = nrep < 0 ? m + 1 : nrep;
+          if (max_copies > 64) {
+            err("repetition too large for the 64-position NFA");
+          }
+          // X{m,n} = X^m (X (X ... )?)? ; X{m,} = X^m X*
+          RNode expanded;
+          expanded.kind = RNode::Concat;
+          for (long k = 0; k < m; k++) expanded.subs.push_back(atom);
+          if (nrep < 0) {
+            RNode st;
+            st.kind = RNode::Star;
+            st.subs.push_back(atom);
+            expanded.subs.push_back(std::move(st));
+          } else if (nrep > m) {
+            RNode opt;
+            opt.kind = RNode::Empty;
+            for (long k = nrep; k > m; k--) {
+              RNode q;
+              q.kind = RNode::Quest;
+              if (opt.kind == RNode::Empty) {
+                q.subs.push_back(atom);
+              } else {
+                RNode cc;
+                cc.kind = RNode::Concat;
+                cc.subs.push_back(atom);
+                cc.subs.push_back(std::move(opt));
+                q.subs.push_back(std::move(cc));
+              }
+              opt = std::move(q);
+            }
+            expanded.subs.push_back(std::move(opt));
+          }
+          if (expanded.subs.size() == 1) {
+            atom = std::move(expanded.subs[0]);
+          } else if (expanded.subs.empty()) {
+            expanded.kind = RNode::Empty;
+            atom = std::move(expanded);
+          } else {
+            atom = std::move(expanded);
+          }
+          if (!eof() && (peek() == '*' || peek() == '+' || peek() == '?' ||
+                         peek() == '{')) {
+            err("nested repetition operator");
+          }
         } else {
           break;
         }
@@ -544,11 +621,16 @@ static GInfo g_build(GBuild& b, const RNode& n, const std::string& expr) {
 
 // blob: u16 nstates, u16 pad, u32 pad, u64 first, u64 last,
 //       u64 follow[nstates], u64 byte_table[256]
-static bytes g_serialize(const GBuild& b, const GInfo& root) {
+static bytes g_serialize(const GBuild& b, const GInfo& root,
+                         bool a_start = false, bool a_end = false,
+                         bool nullable = false) {
   bytes out;
   uint16_t n = uint16_t(b.atoms.size());
   out.push_back(uint8_t(n));
   out.push_back(uint8_t(n >> 8));
+  // byte 2: anchor flags (1 = ^ anchored, 2 = $ anchored, 4 = nullable root)
+  out.push_back(uint8_t((a_start ? 1 : 0) | (a_end ? 2 : 0) |
+                        (nullable ? 4 : 0)));
   out.resize(8, 0);
   auto put64 = [&](uint64_t v) {
     for (int i = 0; i < 8; i++) out.push_back(uint8_t(v >> (8 * i)));
@@ -569,9 +651,31 @@ static bytes g_serialize(const GBuild& b, const GInfo& root) {
 }  // namespace
 
 RegexProg regex_compile(const std::string& expr) {
-  Parser p(expr);
+  // Top-level anchors: a leading '^' / trailing unescaped '$' anchor the
+  // whole pattern (Go regexp semantics without multiline).  Anchors inside
+  // the pattern (alternation branches, groups) remain unsupported and error
+  // in the parser.
+  std::string body = expr;
+  bool a_start = false, a_end = false;
+  if (!body.empty() && body[0] == '^') {
+    a_start = true;
+    body.erase(body.begin());
+  }
+  if (!body.empty() && body.back() == '$') {
+    size_t bs = 0;
+    while (bs + 1 < body.size() && body[body.size() - 2 - bs] == '\\') bs++;
+    if (bs % 2 == 0) {
+      a_end = true;
+      body.pop_back();
+    }
+  }
+  Parser p(body);
   RNode raw = p.parse_alt();
   if (!p.eof()) p.err("unexpected )");
+  if ((a_start || a_end) && raw.kind == RNode::Alt) {
+    fail("regex: anchors with top-level alternation are not supported in " +
+         expr);
+  }
 
   RegexProg re;
   re.expr = expr;
@@ -639,16 +743,28 @@ RegexProg regex_compile(const std::string& expr) {
     re.has_or_values = true;
   }
 
-  if (!re.is_only_prefix && !re.is_suffix_dot_star && !re.is_suffix_dot_plus &&
-      re.substr_dot_star.empty() && re.substr_dot_plus.empty() && !re.has_or_values) {
+  if (a_start || a_end ||
+      (!re.is_only_prefix && !re.is_suffix_dot_star && !re.is_suffix_dot_plus &&
+       re.substr_dot_star.empty() && re.substr_dot_plus.empty() &&
+       !re.has_or_values)) {
     // General class: Glushkov NFA over the WHOLE original pattern, matched
     // unanchored -- equivalent to Go's prefix-retry + anchored suffixRe loop
-    // (regex.go:186-211) for pure regexes.
+    // (regex.go:186-211) for pure regexes.  Anchored patterns always take
+    // this path (the fast-path classes assume unanchored semantics).
     GBuild b;
     GInfo root = g_build(b, raw, expr);
     re.has_nfa = true;
-    re.always_true = root.nullable;  // can match "" => unanchored matches all
-    re.nfa_blob = g_serialize(b, root);
+    // nullable root matches "": unanchored or half-anchored => every string
+    // has an empty prefix/suffix match; both-anchored only matches ""
+    re.always_true = root.nullable && !(a_start && a_end);
+    re.is_only_prefix = false;
+    re.is_suffix_dot_star = false;
+    re.is_suffix_dot_plus = false;
+    re.substr_dot_star.clear();
+    re.substr_dot_plus.clear();
+    re.has_or_values = false;
+    re.or_values.clear();
+    re.nfa_blob = g_serialize(b, root, a_start, a_end, root.nullable);
     re.prefix.clear();  // NFA matches the whole pattern; ignore the prefix
   }
   return re;
@@ -712,6 +828,8 @@ static bool match_with_prefix(const RegexProg& re, strview s) {
 
 bool nfa_match(const uint8_t* blob, strview s) {
   uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t anchors = blob[2];
+  const bool a_start = anchors & 1, a_end = anchors & 2;
   auto rd64 = [&](size_t off) {
     uint64_t v;
     memcpy(&v, blob + off, 8);
@@ -721,9 +839,11 @@ bool nfa_match(const uint8_t* blob, strview s) {
   const uint64_t last = rd64(16);
   const uint8_t* follow = blob + 24;
   const uint8_t* table = blob + 24 + size_t(n) * 8;
+  if (s.n == 0) return (anchors & 4) != 0;  // nullable root matches ""
   uint64_t active = 0;
   for (size_t i = 0; i < s.n; i++) {
-    uint64_t targets = first;
+    // '^' anchored: new matches may start only at offset 0
+    uint64_t targets = (a_start && i > 0) ? 0 : first;
     uint64_t m = active;
     while (m) {
       int x = __builtin_ctzll(m);
@@ -735,10 +855,11 @@ bool nfa_match(const uint8_t* blob, strview s) {
     uint64_t tb;
     memcpy(&tb, table + size_t(uint8_t(s.p[i])) * 8, 8);
     const uint64_t entered = targets & tb;
-    if (entered & last) return true;
+    if (!a_end && (entered & last)) return true;
     active = entered;
   }
-  return false;
+  // '$' anchored: accept only with a final position active at string end
+  return a_end && (active & last) != 0;
 }
 
 bool regex_match(const RegexProg& re, strview s) {

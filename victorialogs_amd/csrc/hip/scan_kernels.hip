@@ -997,14 +997,17 @@ __device__ bool d_regex_or_hasprefix(const DRegex& re, const A& a, long s0, long
 template <typename A>
 __device__ bool d_nfa_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
   const uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t anchors = blob[2];  // 1 = '^', 2 = '$', 4 = nullable root
+  const bool a_start = anchors & 1, a_end = anchors & 2;
   uint64_t first, last;
   __builtin_memcpy(&first, blob + 8, 8);
   __builtin_memcpy(&last, blob + 16, 8);
   const uint8_t* follow = blob + 24;
   const uint8_t* table = blob + 24 + size_t(n) * 8;
+  if (sn == 0) return (anchors & 4) != 0;
   uint64_t active = 0;
   for (long i = 0; i < sn; i++) {
-    uint64_t targets = first;
+    uint64_t targets = (a_start && i > 0) ? 0 : first;
     uint64_t m = active;
     while (m) {
       int x = __builtin_ctzll(m);
@@ -1016,10 +1019,10 @@ __device__ bool d_nfa_match_at(const uint8_t* blob, const A& a, long s0, long sn
     uint64_t tb;
     __builtin_memcpy(&tb, table + size_t(a.u8(s0 + i)) * 8, 8);
     const uint64_t entered = targets & tb;
-    if (entered & last) return true;
+    if (!a_end && (entered & last)) return true;
     active = entered;
   }
-  return false;
+  return a_end && (active & last) != 0;
 }
 
 // Regex.MatchString (regex.go:86-212) over accessor bytes [s0, s0+sn).
