@@ -175,8 +175,10 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
       hit &= hit - 1;
       long pos = w + k;
       if (pos > last) return -1;
+      // verify from byte 0: the SWAR zero-scan's borrow cascade can flag a
+      // byte equal to c0^1 right after a true candidate ("101" vs "11")
       bool eq = true;
-      for (long i = 1; i < subn; i++) {
+      for (long i = 0; i < subn; i++) {
         if (a.u8(pos + i) != sub[i]) {
           eq = false;
           break;
