@@ -178,9 +178,16 @@ def main():
             kms += s.last_kernel_ms
         return hits, kms
 
-    # warmup
+    # warmup + full-size correctness property: the phrase workload's filter
+    # matches every generated row by construction, so the device count must
+    # equal the staged row count (a size-independent invariant at the full
+    # 100M-row config; parity at oracle-checkable sizes lives in tests/)
     for _ in range(args.warmup):
         hits, _ = one_step()
+    if args.workload == "phrase" and hits != rows:
+        raise SystemExit(
+            f"correctness check failed: phrase workload matched {hits} of "
+            f"{rows} rows (expected all)")
 
     # timed region: barrier + sync on both sides, MAX over ranks
     if distributed:

@@ -12,6 +12,7 @@
 #include "../victorialogs_amd/csrc/core/match.h"
 #include "../victorialogs_amd/csrc/core/part_writer.h"
 #include "../victorialogs_amd/csrc/core/tokenizer.h"
+#include "../victorialogs_amd/csrc/core/unicode_case.h"
 #include "../victorialogs_amd/csrc/core/values.h"
 #include "../victorialogs_amd/csrc/core/xxhash64.h"
 
@@ -201,6 +202,21 @@ int orc_parse_iso8601(const char* s, long sn, long long* out) {
   if (!try_parse_timestamp_iso8601(strview(s, size_t(sn)), &v)) return 0;
   *out = v;
   return 1;
+}
+
+// Unicode simple case mapping probes (strings.ToLower/ToUpper restatement)
+long orc_to_lower(const char* s, long sn, char* out, long cap) {
+  std::string r = to_lower_str(strview(s, size_t(sn)));
+  long n = long(r.size()) < cap ? long(r.size()) : cap;
+  memcpy(out, r.data(), size_t(n));
+  return long(r.size());
+}
+
+long orc_to_upper(const char* s, long sn, char* out, long cap) {
+  std::string r = to_upper_str(strview(s, size_t(sn)));
+  long n = long(r.size()) < cap ? long(r.size()) : cap;
+  memcpy(out, r.data(), size_t(n));
+  return long(r.size());
 }
 
 // tryParseDuration / tryParseBytes / tryParseIPv4 probes for golden tests

@@ -325,3 +325,33 @@ def test_ipv4_parse_table(lib):
               "127.256.127.127", "256.127.127.127", "-1.127.127.127",
               "127.-1.127.127", "127.127.-1.127", "127.127.127.-1"]:
         assert ip(s) is None, s
+
+
+def test_unicode_simple_case_mapping(lib):
+    """Go unicode.ToLower/ToUpper SIMPLE mappings (strings.ToLower semantics)
+    including the special cases where the full mapping is multi-rune."""
+    import ctypes
+
+    lib.orc_to_lower.restype = ctypes.c_long
+    lib.orc_to_upper.restype = ctypes.c_long
+
+    def low(s):
+        b = s.encode()
+        buf = ctypes.create_string_buffer(256)
+        n = lib.orc_to_lower(b, len(b), buf, 256)
+        return buf.raw[:n].decode()
+
+    def up(s):
+        b = s.encode()
+        buf = ctypes.create_string_buffer(256)
+        n = lib.orc_to_upper(b, len(b), buf, 256)
+        return buf.raw[:n].decode()
+
+    assert low("HeLLo WORLD") == "hello world"
+    assert low("ПрИвет") == "привет"
+    assert low("İ") == "i"          # U+0130 simple mapping (not "i̇")
+    assert low("ẞ") == "ß"          # U+1E9E -> U+00DF
+    assert up("привет") == "ПРИВЕТ"
+    assert up("ß") == "ß"           # simple mapping keeps ß (not "SS")
+    assert up("ﬁ") == "ﬁ"           # ligature: identity simple mapping
+    assert low("ÀÉÎÕÜ") == "àéîõü"

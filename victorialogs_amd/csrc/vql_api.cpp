@@ -240,19 +240,19 @@ struct Stage {
 
   ~Stage() {
     for (auto& kv : gathers) {
-      if (kv.second.d_gcols) hipFree(kv.second.d_gcols);
-      if (kv.second.d_bases) hipFree(kv.second.d_bases);
+      if (kv.second.d_gcols) (void)hipFree(kv.second.d_gcols);
+      if (kv.second.d_bases) (void)hipFree(kv.second.d_bases);
     }
-    if (ev0) hipEventDestroy(ev0);
-    if (ev1) hipEventDestroy(ev1);
-    if (stream) hipStreamDestroy(stream);
-    if (arena) hipFree(arena);
-    if (d_ops) hipFree(d_ops);
-    if (d_lbs) hipFree(d_lbs);
-    if (d_blocks) hipFree(d_blocks);
-    if (d_chunks) hipFree(d_chunks);
-    if (d_hits) hipFree(d_hits);
-    if (d_block_hits) hipFree(d_block_hits);
+    if (ev0) (void)hipEventDestroy(ev0);
+    if (ev1) (void)hipEventDestroy(ev1);
+    if (stream) (void)hipStreamDestroy(stream);
+    if (arena) (void)hipFree(arena);
+    if (d_ops) (void)hipFree(d_ops);
+    if (d_lbs) (void)hipFree(d_lbs);
+    if (d_blocks) (void)hipFree(d_blocks);
+    if (d_chunks) (void)hipFree(d_chunks);
+    if (d_hits) (void)hipFree(d_hits);
+    if (d_block_hits) (void)hipFree(d_block_hits);
   }
 
   uint8_t* push(const void* src, size_t n, size_t align = 16) {
@@ -2255,7 +2255,10 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
   for (long b = 0; b < nblocks; b++) {
     const BlockHeader& bh = bh_of(b);
     rows += bh.rows_count;
-    est += bh.uncompressed_size_bytes + bh.rows_count * 24 + (2 << 20);
+    // slack per block covers blooms + offsets + bitmap (measured ~200 KB
+    // for 8192-row blocks; 2 MB/block over-allocated ~24 GB on the 100M-row
+    // multi-part stage)
+    est += bh.uncompressed_size_bytes + bh.rows_count * 24 + (512 << 10);
   }
   st->rows = rows;
   st->arena_cap = size_t(est);
@@ -2675,12 +2678,12 @@ extern "C" long long vql_bloom_build(const unsigned char* data,
     int* d_overflow = nullptr;
     uint64_t* d_bits = nullptr;
     auto cleanup = [&]() {
-      if (d_data) hipFree(d_data);
-      if (d_offs) hipFree(d_offs);
-      if (d_slots) hipFree(d_slots);
-      if (d_unique) hipFree(d_unique);
-      if (d_overflow) hipFree(d_overflow);
-      if (d_bits) hipFree(d_bits);
+      if (d_data) (void)hipFree(d_data);
+      if (d_offs) (void)hipFree(d_offs);
+      if (d_slots) (void)hipFree(d_slots);
+      if (d_unique) (void)hipFree(d_unique);
+      if (d_overflow) (void)hipFree(d_overflow);
+      if (d_bits) (void)hipFree(d_bits);
     };
     HIP_CHECK(hipMalloc(&d_data, size_t(nbytes) + 16));
     HIP_CHECK(hipMemcpy(d_data, data, size_t(nbytes), hipMemcpyHostToDevice));
@@ -2712,7 +2715,7 @@ extern "C" long long vql_bloom_build(const unsigned char* data,
       HIP_CHECK(hipMemcpy(&unique, d_unique, 8, hipMemcpyDeviceToHost));
       // keep the table below half full so dedup probing stays short
       if (!overflow && unique * 2 <= try_cap) break;
-      hipFree(d_slots);
+      (void)hipFree(d_slots);
       d_slots = nullptr;
       if (try_cap >= cap_slots) {
         cleanup();
