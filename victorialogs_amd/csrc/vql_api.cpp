@@ -2944,9 +2944,9 @@ long long vql_gather(void* s, const char* field, unsigned char* out_bytes,
     if (out_rowids && g.nrows) {
       HIP_CHECK(hipMemcpy(out_rowids, d_rowids, g.nrows * 8, hipMemcpyDeviceToHost));
     }
-    hipFree(d_bytes);
-    hipFree(d_offs);
-    if (d_rowids) hipFree(d_rowids);
+    (void)hipFree(d_bytes);
+    (void)hipFree(d_offs);
+    if (d_rowids) (void)hipFree(d_rowids);
     return (long long)g.nrows;
   } catch (const std::exception& e) {
     g_err = e.what();
