@@ -12,6 +12,7 @@
 #include "../victorialogs_amd/csrc/core/match.h"
 #include "../victorialogs_amd/csrc/core/part_writer.h"
 #include "../victorialogs_amd/csrc/core/tokenizer.h"
+#include "../victorialogs_amd/csrc/core/regex.h"
 #include "../victorialogs_amd/csrc/core/unicode_case.h"
 #include "../victorialogs_amd/csrc/core/values.h"
 #include "../victorialogs_amd/csrc/core/xxhash64.h"
@@ -202,6 +203,19 @@ int orc_parse_iso8601(const char* s, long sn, long long* out) {
   if (!try_parse_timestamp_iso8601(strview(s, size_t(sn)), &v)) return 0;
   *out = v;
   return 1;
+}
+
+// Regex probe for differential fuzzing against an independent engine:
+// compiles `pat` (fast paths + Glushkov NFA) and matches `s` unanchored.
+// Returns 1 match, 0 no match, -1 compile-reject.
+long orc_regex_match(const char* pat, long pn, const char* s, long sn) {
+  try {
+    RegexProg re = regex_compile(std::string(pat, size_t(pn)));
+    return regex_match(re, strview(s, size_t(sn))) ? 1 : 0;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
 }
 
 // Unicode simple case mapping probes (strings.ToLower/ToUpper restatement)
