@@ -560,6 +560,39 @@ static GInfo g_multibyte(GBuild& b, const std::string& expr) {
   return r;
 }
 
+// Re-compress alternations of single-byte literals/classes into ONE byte-set
+// position before the Glushkov build: parse expands small positive classes
+// into Alt-of-chars for the or-values classification, which would otherwise
+// cost one NFA position per character ("[a-z]{3}" = 78 positions instead
+// of 3).
+static RNode compress_classes(const RNode& n) {
+  RNode out;
+  out.kind = n.kind;
+  out.lit = n.lit;
+  memcpy(out.cls, n.cls, sizeof(out.cls));
+  out.cls_nonascii = n.cls_nonascii;
+  out.subs.reserve(n.subs.size());
+  for (const auto& sub : n.subs) out.subs.push_back(compress_classes(sub));
+  if (out.kind != RNode::Alt) return out;
+  uint8_t set[32] = {0};
+  for (const auto& sub : out.subs) {
+    if (sub.kind == RNode::Lit && sub.lit.size() == 1 &&
+        uint8_t(sub.lit[0]) < 0x80) {
+      uint8_t c = uint8_t(sub.lit[0]);
+      set[c >> 3] |= uint8_t(1) << (c & 7);
+    } else if (sub.kind == RNode::Class && !sub.cls_nonascii) {
+      for (int i = 0; i < 32; i++) set[i] |= sub.cls[i];
+    } else {
+      return out;  // not a pure single-byte alternation
+    }
+  }
+  RNode cn;
+  cn.kind = RNode::Class;
+  memcpy(cn.cls, set, 32);
+  cn.cls_nonascii = false;
+  return cn;
+}
+
 static GInfo g_build(GBuild& b, const RNode& n, const std::string& expr) {
   switch (n.kind) {
     case RNode::Empty:
@@ -752,7 +785,8 @@ RegexProg regex_compile(const std::string& expr) {
     // (regex.go:186-211) for pure regexes.  Anchored patterns always take
     // this path (the fast-path classes assume unanchored semantics).
     GBuild b;
-    GInfo root = g_build(b, raw, expr);
+    RNode packed = compress_classes(raw);
+    GInfo root = g_build(b, packed, expr);
     re.has_nfa = true;
     // nullable root matches "": unanchored or half-anchored => every string
     // has an empty prefix/suffix match; both-anchored only matches ""
