@@ -205,6 +205,73 @@ int orc_parse_iso8601(const char* s, long sn, long long* out) {
   return 1;
 }
 
+// valuesEncoder.encode probe (values_encoder.go:109-154): encodes the
+// newline-joined values and returns "<type> <min> <max>" + a decode
+// round-trip verdict, for the TestValuesEncoder table port.
+long orc_encode_values(const char* joined, long jn, char* out, long cap) {
+  try {
+    std::vector<std::string> values;
+    const char* p = joined;
+    const char* end = joined + jn;
+    while (p < end) {
+      const char* nl = (const char*)memchr(p, '\n', size_t(end - p));
+      if (!nl) nl = end;
+      values.emplace_back(p, nl);
+      p = nl + 1;
+    }
+    if (jn == 0) values.clear();
+    EncodedColumn ec;
+    encode_values(ec, values);
+    // decode round trip through format_value-equivalents
+    bool rt = true;
+    if (ec.type == ValueType::String) {
+      for (size_t i = 0; i < values.size(); i++) {
+        if (!(ec.values[i] == strview(values[i]))) rt = false;
+      }
+    } else if (ec.type == ValueType::Dict) {
+      for (size_t i = 0; i < values.size(); i++) {
+        uint8_t id = uint8_t(ec.values[i].p[0]);
+        if (id >= ec.dict.size() || ec.dict[id] != values[i]) rt = false;
+      }
+    } else {
+      for (size_t i = 0; i < values.size(); i++) {
+        std::string str;
+        const uint8_t* q = (const uint8_t*)ec.values[i].p;
+        switch (ec.type) {
+          case ValueType::Uint8: format_uint64(str, q[0]); break;
+          case ValueType::Uint16: format_uint64(str, get_u16be(q)); break;
+          case ValueType::Uint32: format_uint64(str, get_u32be(q)); break;
+          case ValueType::Uint64: format_uint64(str, get_u64be(q)); break;
+          case ValueType::Int64: format_int64(str, get_i64be_zigzag(q)); break;
+          case ValueType::Float64: {
+            uint64_t u = get_u64be(q);
+            double d;
+            memcpy(&d, &u, 8);
+            format_float64(str, d);
+            break;
+          }
+          case ValueType::IPv4: format_ipv4(str, get_u32be(q)); break;
+          case ValueType::TimestampISO8601:
+            format_timestamp_iso8601(str, int64_t(get_u64be(q)));
+            break;
+          default: rt = false;
+        }
+        if (str != values[i]) rt = false;
+      }
+    }
+    char buf[96];
+    int n = snprintf(buf, sizeof(buf), "%d %llu %llu %d", int(ec.type),
+                     (unsigned long long)ec.min_value,
+                     (unsigned long long)ec.max_value, rt ? 1 : 0);
+    long m = n < cap ? n : cap;
+    memcpy(out, buf, size_t(m));
+    return n;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
 // Regex probe for differential fuzzing against an independent engine:
 // compiles `pat` (fast paths + Glushkov NFA) and matches `s` unanchored.
 // Returns 1 match, 0 no match, -1 compile-reject.

@@ -355,3 +355,47 @@ def test_unicode_simple_case_mapping(lib):
     assert up("ß") == "ß"           # simple mapping keeps ß (not "SS")
     assert up("ﬁ") == "ﬁ"           # ligature: identity simple mapping
     assert low("ÀÉÎÕÜ") == "àéîõü"
+
+
+def test_values_encoder_type_selection(lib):
+    """TestValuesEncoder table (values_encoder_test.go:11-99): encode-type
+    selection + min/max + decode round-trip.  Type ids per values.h
+    (matching values_encoder.go constants)."""
+    import ctypes
+    import math
+
+    def enc(values):
+        joined = "\n".join(values).encode()
+        buf = ctypes.create_string_buffer(128)
+        n = lib.orc_encode_values(joined, len(joined), buf, 128)
+        assert n > 0, lib.orc_errstr().decode()
+        t, mn, mx, rt = buf.raw[:n].decode().split()
+        assert rt == "1", f"decode round-trip failed for {values[:4]}..."
+        return int(t), int(mn), int(mx)
+
+    STRING, DICT = 1, 2
+    U8, U16, U32, U64, F64, IPV4, ISO = 3, 4, 5, 6, 7, 8, 9
+
+    n = 9  # maxDictLen + 1
+    assert enc(["value_%d" % i for i in range(n)]) == (STRING, 0, 0)
+    assert enc(["foobar"]) == (DICT, 0, 0)
+    assert enc(["foo", "bar"]) == (DICT, 0, 0)
+    assert enc(["1", "2foo"]) == (DICT, 0, 0)
+    assert enc(["%d" % (i + 1) for i in range(n)]) == (U8, 1, n)
+    assert enc(["%d" % ((i + 1) << 8) for i in range(n)]) == (U16, 1 << 8, n << 8)
+    assert enc(["%d" % ((i + 1) << 16) for i in range(n)]) == (U32, 1 << 16, n << 16)
+    assert enc(["%d" % ((i + 1) << 32) for i in range(n)]) == (U64, 1 << 32, n << 32)
+    # Go's %g prints the SHORTEST round-trip form (Python repr equivalent);
+    # with these inputs the reference's FMA-based tryParseFloat64Exact
+    # reconstructs the identical doubles.  (Six-digit renderings like
+    # "2.23607" are 1 ULP lossy in the reference itself — values_encoder.go
+    # math.FMA path — and our restatement mirrors that bit-for-bit.)
+    def gofmt(v):  # Go %g: shortest, integral floats without ".0"
+        r = repr(v)
+        return r[:-2] if r.endswith(".0") else r
+
+    t, mn, mx = enc([gofmt(math.sqrt(i + 1)) for i in range(n)])
+    assert (t, mn, mx) == (F64, 4607182418800017408, 4613937818241073152)
+    assert enc(["1.2.3.%d" % i for i in range(n)]) == (IPV4, 16909056, 16909064)
+    t, mn, mx = enc(["2011-04-19T03:44:01.%03dZ" % i for i in range(n)])
+    assert (t, mn, mx) == (ISO, 1303184641000000000, 1303184641008000000)
