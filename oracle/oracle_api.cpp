@@ -373,7 +373,7 @@ int orc_write_custom_part(const char* dir, const char* spec_json) {
       sid.id_hi = snum;
       sid.id_lo = 1;
       std::vector<int64_t> ts;
-      for (const auto& t : jget(bj, "timestamps").arr) ts.push_back(int64_t(t.num));
+      for (const auto& t : jget(bj, "timestamps").arr) ts.push_back(t.as_i64());
       std::vector<InputColumn> cols;
       for (const auto& cj : jget(bj, "columns").arr) {
         InputColumn c;

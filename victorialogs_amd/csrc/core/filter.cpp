@@ -169,8 +169,8 @@ FilterNode build(const JValue& v) {
     n.children.push_back(build(jget(v, "filter")));
   } else if (type == "time") {
     n.type = FilterNode::Time;
-    n.min_ts = int64_t(jget(v, "min").num);
-    n.max_ts = int64_t(jget(v, "max").num);
+    n.min_ts = jget(v, "min").as_i64();
+    n.max_ts = jget(v, "max").as_i64();
   } else if (type == "range") {
     n.type = FilterNode::Range;
     n.field = jget(v, "field").str;
@@ -304,10 +304,10 @@ FilterNode build(const JValue& v) {
     n.max_u = uint64_t(jget(v, "max").num);
   } else if (type == "day_range" || type == "week_range") {
     n.type = type == "day_range" ? FilterNode::DayRange : FilterNode::WeekRange;
-    n.min_u = uint64_t(jget(v, "start").num);
-    n.max_u = uint64_t(jget(v, "end").num);
+    n.min_u = uint64_t(jget(v, "start").as_i64());
+    n.max_u = uint64_t(jget(v, "end").as_i64());
     auto it = v.obj.find("offset");
-    if (it != v.obj.end()) n.tz_offset = int64_t(it->second.num);
+    if (it != v.obj.end()) n.tz_offset = it->second.as_i64();
   } else if (type == "value_type") {
     n.type = FilterNode::ValueTypeFilter;
     n.field = jget(v, "field").str;

@@ -1,5 +1,8 @@
 #include "json.h"
 
+#include <cerrno>
+#include <cstdlib>
+
 #include <cstdlib>
 #include <cstring>
 
@@ -102,10 +105,24 @@ struct JParser {
         char* endp;
         double d = strtod(p, &endp);
         if (endp == p) err("bad number");
-        p = endp;
         JValue v;
         v.kind = JValue::Num;
         v.num = d;
+        // integral literal (no '.', 'e'): keep the exact int64 too
+        bool integral = true;
+        for (const char* q = p; q < endp; q++) {
+          if (*q == '.' || *q == 'e' || *q == 'E') integral = false;
+        }
+        if (integral) {
+          errno = 0;
+          char* iend;
+          long long iv = strtoll(p, &iend, 10);
+          if (iend == endp && errno != ERANGE) {
+            v.ival = iv;
+            v.is_int = true;
+          }
+        }
+        p = endp;
         return v;
       }
     }
