@@ -224,3 +224,30 @@ def test_multipart_stage(gen_part, typed_part):
     filt.close()
     for p in parts:
         p.close()
+
+
+def test_long_rows_global_fallback(tmp_path):
+    """Rows larger than the 17 KiB wave tile take the direct-global scan
+    path (use_tile=false); results must stay bit-identical."""
+    from victorialogs_amd import write_custom_part
+
+    rows = 64
+    vals = []
+    for i in range(rows):
+        pad = ("x" * 797 + " ") * (30 + (i % 5) * 10)  # ~24-56 KB rows
+        vals.append(f"start {i} {pad} needle_{i % 7} end")
+    spec = {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i for i in range(rows)],
+        "columns": [{"name": "_msg", "values": vals}],
+    }]}
+    d = str(tmp_path / "longrows")
+    write_custom_part(d, spec)
+    for f in [
+        '{"type":"phrase","field":"_msg","phrase":"needle_3"}',
+        '{"type":"phrase","field":"_msg","phrase":"start 7"}',
+        '{"type":"regexp","field":"_msg","re":"needle_(1|2)"}',
+        '{"type":"prefix","field":"_msg","prefix":"start 1"}',
+        '{"type":"regexp","field":"_msg","re":"end$"}',
+    ]:
+        assert_parity(d, f)
