@@ -285,6 +285,72 @@ long orc_regex_match(const char* pat, long pn, const char* s, long sn) {
   }
 }
 
+// tokenizeHashes (hash_tokenizer.go:68-166) over newline-separated values:
+// returns space-separated hex hashes (TestTokenizeHashes port)
+long orc_tokenize_hashes(const char* joined, long jn, char* out, long cap) {
+  std::vector<std::string> values;
+  const char* p = joined;
+  const char* end = joined + jn;
+  while (p < end) {
+    const char* nl = (const char*)memchr(p, '\n', size_t(end - p));
+    if (!nl) nl = end;
+    values.emplace_back(p, nl);
+    p = nl + 1;
+  }
+  if (jn == 0) values.clear();
+  std::vector<strview> vs;
+  for (const auto& v : values) vs.emplace_back(v);
+  auto hashes = tokenize_hashes(vs);
+  std::string outs;
+  char buf[24];
+  for (size_t i = 0; i < hashes.size(); i++) {
+    snprintf(buf, sizeof(buf), "%s%016llX", i ? " " : "",
+             (unsigned long long)hashes[i]);
+    outs += buf;
+  }
+  long n = long(outs.size()) < cap ? long(outs.size()) : cap;
+  memcpy(out, outs.data(), size_t(n));
+  return long(outs.size());
+}
+
+// bloomFilter.containsAll over a marshaled bloom + one query token
+// (bloomfilter_test.go equivalence / false-positive tables)
+long orc_bloom_contains(const unsigned char* bloom, long bn, const char* token,
+                        long tn) {
+  std::vector<uint64_t> words;
+  if (!bloom_unmarshal(words, bloom, size_t(bn))) return -1;
+  std::vector<uint64_t> hashes;
+  append_token_hashes(hashes, strview(token, size_t(tn)));
+  return bloom_contains_all(words.data(), words.size(), hashes.data(),
+                            hashes.size())
+             ? 1
+             : 0;
+}
+
+// tokenizeStrings over multiple values (newline-separated), preserving the
+// cross-value dedup order — for the TestTokenizeStrings table port.
+long orc_tokenize_multi(const char* joined, long jn, char* out, long cap) {
+  std::vector<std::string> values;
+  const char* p = joined;
+  const char* end = joined + jn;
+  while (p < end) {
+    const char* nl = (const char*)memchr(p, '\n', size_t(end - p));
+    if (!nl) nl = end;
+    values.emplace_back(p, nl);
+    p = nl + 1;
+  }
+  if (jn == 0) values.clear();
+  auto tokens = tokenize_strings(values);
+  std::string outs;
+  for (size_t i = 0; i < tokens.size(); i++) {
+    if (i) outs += '\n';
+    outs += tokens[i];
+  }
+  long n = long(outs.size()) < cap ? long(outs.size()) : cap;
+  memcpy(out, outs.data(), size_t(n));
+  return long(outs.size());
+}
+
 // Unicode simple case mapping probes (strings.ToLower/ToUpper restatement)
 long orc_to_lower(const char* s, long sn, char* out, long cap) {
   std::string r = to_lower_str(strview(s, size_t(sn)));

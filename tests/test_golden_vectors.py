@@ -399,3 +399,90 @@ def test_values_encoder_type_selection(lib):
     assert enc(["1.2.3.%d" % i for i in range(n)]) == (IPV4, 16909056, 16909064)
     t, mn, mx = enc(["2011-04-19T03:44:01.%03dZ" % i for i in range(n)])
     assert (t, mn, mx) == (ISO, 1303184641000000000, 1303184641008000000)
+
+
+def test_tokenize_strings_table(lib):
+    """TestTokenizeStrings verbatim (tokenizer_test.go:9-30): cross-value
+    dedup order included."""
+    import ctypes
+
+    def tok(values):
+        joined = "\n".join(values).encode()
+        buf = ctypes.create_string_buffer(1 << 16)
+        n = lib.orc_tokenize_multi(joined, len(joined), buf, 1 << 16)
+        raw = buf.raw[:n].decode()
+        return raw.split("\n") if raw else []
+
+    assert tok([]) == []
+    assert tok([""]) == []
+    assert tok(["foo"]) == ["foo"]
+    assert tok(["foo bar---.!!([baz]!!! %$# TaSte"]) == [
+        "foo", "bar", "baz", "TaSte"]
+    assert tok(["теСТ 1234 f12.34", "34 f12 AS"]) == [
+        "теСТ", "1234", "f12", "34", "AS"]
+    syslog = [
+        "",
+        "Apr 28 13:43:38 localhost whoopsie[2812]: [13:43:38] online",
+        "Apr 28 13:45:01 localhost CRON[12181]: (root) CMD (command -v "
+        "debian-sa1 > /dev/null && debian-sa1 1 1)",
+        "Apr 28 13:48:01 localhost kernel: [36020.497806] CPU0: Core "
+        "temperature above threshold, cpu clock throttled (total events "
+        "= 22034)",
+        "",
+    ]
+    assert tok(syslog) == [
+        "Apr", "28", "13", "43", "38", "localhost", "whoopsie", "2812",
+        "online", "45", "01", "CRON", "12181", "root", "CMD", "command", "v",
+        "debian", "sa1", "dev", "null", "1", "48", "kernel", "36020",
+        "497806", "CPU0", "Core", "temperature", "above", "threshold", "cpu",
+        "clock", "throttled", "total", "events", "22034"]
+
+
+def test_tokenize_hashes_table(lib):
+    """TestTokenizeHashes verbatim (hash_tokenizer_test.go:8-25): exact
+    XXH64 hash stream with hash-level dedup."""
+    import ctypes
+
+    def th(values):
+        joined = "\n".join(values).encode()
+        buf = ctypes.create_string_buffer(1 << 16)
+        n = lib.orc_tokenize_hashes(joined, len(joined), buf, 1 << 16)
+        raw = buf.raw[:n].decode()
+        return [int(x, 16) for x in raw.split()] if raw else []
+
+    assert th([]) == []
+    assert th([""]) == []
+    assert th(["foo"]) == [0x33BF00A859C4BA3F]
+    assert th(["foo foo", "!!foo //"]) == [0x33BF00A859C4BA3F]
+    assert th(["foo bar---.!!([baz]!!! %$# TaSte"]) == [
+        0x33BF00A859C4BA3F, 0x48A37C90AD27A659, 0x42598CF26A247404,
+        0x34709F40A3286E46]
+    assert th(["foo bar---.!!([baz]!!! %$# baz foo TaSte"]) == [
+        0x33BF00A859C4BA3F, 0x48A37C90AD27A659, 0x42598CF26A247404,
+        0x34709F40A3286E46]
+    assert th(["теСТ 1234 f12.34", "34 f12 AS"]) == [
+        0xFE846FA145CEABD1, 0xD8316E61D84F6BA4, 0x6D67BA71C4E03D10,
+        0x5E8D522CA93563ED, 0xED80AED10E029FC8]
+
+
+def test_bloom_equivalence_and_fp_rate(lib):
+    """bloomfilter_test.go:61-103: every added token is found (no false
+    negatives) and the false-positive rate on 20k absent tokens stays under
+    0.11%."""
+    import ctypes
+
+    tokens = ["token_%d" % i for i in range(20000)]
+    joined = "\n".join(tokens).encode()
+    buf = ctypes.create_string_buffer(1 << 20)
+    n = lib.orc_bloom_marshal_tokens(joined, buf, len(buf))
+    assert 0 < n <= len(buf)
+    bloom = buf.raw[:n]
+
+    def contains(tok):
+        t = tok.encode()
+        return lib.orc_bloom_contains(bloom, len(bloom), t, len(t))
+
+    for i in range(0, 20000, 97):
+        assert contains("token_%d" % i) == 1
+    fp = sum(contains("non-existing-token_%d" % i) for i in range(20000))
+    assert fp / 20000 <= 0.0011, f"false positive rate {fp/20000:.4f}"
