@@ -12,6 +12,7 @@
 #include "../victorialogs_amd/csrc/core/match.h"
 #include "../victorialogs_amd/csrc/core/part_writer.h"
 #include "../victorialogs_amd/csrc/core/tokenizer.h"
+#include "../victorialogs_amd/csrc/core/codec.h"
 #include "../victorialogs_amd/csrc/core/regex.h"
 #include "../victorialogs_amd/csrc/core/unicode_case.h"
 #include "../victorialogs_amd/csrc/core/values.h"
@@ -279,6 +280,74 @@ long orc_regex_match(const char* pat, long pn, const char* s, long sn) {
   try {
     RegexProg re = regex_compile(std::string(pat, size_t(pn)));
     return regex_match(re, strview(s, size_t(sn))) ? 1 : 0;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
+// Strings/uint64/int64 block codec round-trip probes
+// (encoding_test.go:17-104,106-148 fixtures; compressed byte lengths are
+// implementation-specific and deliberately not compared — SURVEY.md §8c)
+long orc_strings_block_roundtrip(const char* joined, long jn) {
+  try {
+    std::vector<std::string> values;
+    const char* p = joined;
+    const char* end = joined + jn;
+    while (p < end) {
+      const char* nl = (const char*)memchr(p, '\n', size_t(end - p));
+      if (!nl) nl = end;
+      values.emplace_back(p, nl);
+      p = nl + 1;
+    }
+    if (jn == 0) values.clear();
+    std::vector<strview> vs;
+    for (const auto& v : values) vs.emplace_back(v);
+    bytes data;
+    marshal_strings_block(data, vs);
+    StringsBlockDec dec;
+    unmarshal_strings_block(dec, data.data(), data.size(), values.size());
+    for (size_t i = 0; i < values.size(); i++) {
+      if (!(dec.row(i) == strview(values[i]))) return 0;
+    }
+    return 1;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
+long orc_uint64_block_roundtrip(const unsigned long long* a, long n) {
+  try {
+    bytes data;
+    marshal_uint64_block(data, (const uint64_t*)a, size_t(n));
+    std::vector<uint64_t> back;
+    size_t used = unmarshal_uint64_block(back, data.data(), data.size(),
+                                         uint64_t(n));
+    if (used != data.size()) return 0;
+    for (long i = 0; i < n; i++) {
+      if (back[size_t(i)] != a[i]) return 0;
+    }
+    return 1;
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return -1;
+  }
+}
+
+long orc_int64_array_roundtrip(const long long* a, long n) {
+  try {
+    bytes data;
+    int64_t first;
+    MarshalType mt = marshal_int64_array(data, (const int64_t*)a, size_t(n),
+                                         &first);
+    std::vector<int64_t> back;
+    unmarshal_int64_array(back, data.data(), data.size(), mt, first,
+                          uint64_t(n));
+    for (long i = 0; i < n; i++) {
+      if (back[size_t(i)] != a[i]) return 0;
+    }
+    return 1;
   } catch (const std::exception& e) {
     g_err = e.what();
     return -1;

@@ -486,3 +486,52 @@ def test_bloom_equivalence_and_fp_rate(lib):
         assert contains("token_%d" % i) == 1
     fp = sum(contains("non-existing-token_%d" % i) for i in range(20000))
     assert fp / 20000 <= 0.0011, f"false positive rate {fp/20000:.4f}"
+
+
+def test_block_codec_roundtrips(lib):
+    """Strings / uint64 / int64 block codec round trips over the
+    encoding_test.go fixtures (plus randomized shapes).  Compressed byte
+    lengths are implementation-specific and not compared."""
+    import ctypes
+    import random
+
+    def sb(values):
+        joined = "\n".join(values).encode()
+        r = lib.orc_strings_block_roundtrip(joined, len(joined))
+        assert r == 1, f"strings block round-trip failed ({r})"
+
+    sb([])
+    sb(["foo"])
+    sb(["foo", "bar", "baz"])
+    sb(["x" * 100] * 5)                       # const long strings
+    sb(["payload %d data" % i for i in range(300)])
+    syslog = ("Apr 28 13:39:06 localhost systemd[1]: Started Network Manager "
+              "Script Dispatcher Service.")
+    sb([syslog + str(i) for i in range(64)])
+
+    def ub(values):
+        arr = (ctypes.c_ulonglong * max(len(values), 1))(*values)
+        assert lib.orc_uint64_block_roundtrip(arr, len(values)) == 1, values[:8]
+
+    ub([])
+    ub([1])
+    ub([1, 1, 1])
+    ub([1, 2, 3])
+    ub([1234, 34, 234])
+    ub([123456, 56, 3456])
+    ub([12345678901, 78901, 678901])
+    ub([2**64 - 1, 0, 2**63])
+    rng = random.Random(5)
+    for width in (8, 16, 32, 64):
+        ub([rng.randrange(0, 2**width) for _ in range(257)])
+
+    def ib(values):
+        arr = (ctypes.c_longlong * max(len(values), 1))(*values)
+        assert lib.orc_int64_array_roundtrip(arr, len(values)) == 1, values[:8]
+
+    ib([7])
+    ib([7] * 100)
+    ib([i * 1000 for i in range(100)])          # delta const
+    ib(sorted(rng.randrange(-2**62, 2**62) for _ in range(257)))
+    ib([rng.randrange(-2**62, 2**62) for _ in range(257)])  # unsorted
+    ib([0, 2**62, -2**62, 1, -1] * 50)
