@@ -15,11 +15,21 @@ HIPFLAGS ?= -O3 -std=c++17 -fPIC --offload-arch=$(ARCH)
 CORE_SRCS := $(wildcard victorialogs_amd/csrc/core/*.cpp)
 CORE_OBJS := $(patsubst victorialogs_amd/csrc/core/%.cpp,build/core/%.o,$(CORE_SRCS))
 
-all: oracle hip
+all: oracle hip rowops
 
 oracle: oracle/liboracle.so
 
 hip: victorialogs_amd/libvlogsql.so
+
+# host build of the per-row device code (scan_rowops.h) for CPU-side
+# differential fuzzing against the oracle (tests/test_rowops_fuzz.py)
+rowops: tools/host_rowops/librowops.so
+
+tools/host_rowops/librowops.so: tools/host_rowops/harness.cpp \
+		victorialogs_amd/csrc/hip/scan_rowops.h \
+		victorialogs_amd/csrc/hip/scan_types.h $(CORE_OBJS)
+	$(CXX) $(CXXFLAGS) -shared -Ivictorialogs_amd/csrc/core \
+		tools/host_rowops/harness.cpp $(CORE_OBJS) -o $@ -ldl -lpthread
 
 build/core/%.o: victorialogs_amd/csrc/core/%.cpp $(wildcard victorialogs_amd/csrc/core/*.h) victorialogs_amd/csrc/core/unicode_ranges.inc
 	@mkdir -p build/core

@@ -273,6 +273,31 @@ long orc_encode_values(const char* joined, long jn, char* out, long cap) {
   }
 }
 
+// host matcher probes for the device row-ops differential fuzz
+long orc_match_prefix(const char* s, long sn, const char* pf, long pn) {
+  return match_prefix(strview(s, size_t(sn)), strview(pf, size_t(pn))) ? 1 : 0;
+}
+
+long orc_match_sequence(const char* s, long sn, const char* joined, long jn) {
+  std::vector<std::string> phrases;
+  const char* p = joined;
+  const char* end = joined + jn;
+  while (p < end) {
+    const char* nl = (const char*)memchr(p, '\n', size_t(end - p));
+    if (!nl) nl = end;
+    phrases.emplace_back(p, nl);
+    p = nl + 1;
+  }
+  return match_sequence(strview(s, size_t(sn)), phrases) ? 1 : 0;
+}
+
+long orc_any_case_phrase(const char* s, long sn, const char* lower, long ln) {
+  return match_any_case_phrase(strview(s, size_t(sn)),
+                               strview(lower, size_t(ln)))
+             ? 1
+             : 0;
+}
+
 // Regex probe for differential fuzzing against an independent engine:
 // compiles `pat` (fast paths + Glushkov NFA) and matches `s` unanchored.
 // Returns 1 match, 0 no match, -1 compile-reject.
