@@ -660,7 +660,12 @@ __device__ bool d_parse_ts_secs(const A& a, long s0, long sn, int64_t* secs,
   int64_t yadd = ym >= 0 ? ym / 12 : -((-ym + 11) / 12);
   int64_t mo = ym - yadd * 12 + 1;
   int64_t days = d_days_from_civil(year + yadd, int(mo), day);
-  *secs = days * 86400 + hour * 3600 + minute * 60 + sec;
+  int64_t sv = days * 86400 + hour * 3600 + minute * 60 + sec;
+  // values_encoder.go:545-548: reject timestamps whose nsecs overflow int64
+  if (sv < INT64_MIN / 1000000000 || sv >= INT64_MAX / 1000000000) {
+    return false;
+  }
+  *secs = sv;
   *consumed = i;
   return true;
 }
