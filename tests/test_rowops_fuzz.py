@@ -110,8 +110,11 @@ def test_parse_math_differential(libs):
         lambda: rand_text(rng, 2).decode(),
         lambda: "1_00%d" % rng.randrange(0, 10),
     ]
-    for _ in range(30000):
-        s = rng.choice(shapes)().encode()
+    extremes = [b"2262-06-09T09:53:45Z", b"1677-03-31T14:13:32-11:00",
+                b"1677-09-21T00:12:43Z", b"2262-04-11T23:47:16Z",
+                b"1677-09-21T00:12:44Z", b"2262-04-11T23:47:15Z"]
+    for i in range(30000):
+        s = extremes[i] if i < len(extremes) else rng.choice(shapes)().encode()
         want = orc.orc_parse_math_number(s, len(s))
         got = dev.h_dev_parse_math(s, len(s))
         same = (struct.pack("<d", want) == struct.pack("<d", got) or
