@@ -188,3 +188,68 @@ def test_oracle_multithreaded_equals_single(gen_part):
         assert h1 == h4 and w1 == w4
     finally:
         orc.close()
+
+
+def test_complex_filters_reference_fixture(tmp_path):
+    """TestComplexFilters ported verbatim (filter_test.go:14-156): the
+    reference's own column fixture and expected row sets."""
+    import json
+
+    from victorialogs_amd import OracleScanner, write_custom_part
+
+    values = [
+        "a foo",
+        "a foobar",
+        "aa abc a",
+        "ca afdf a,foobar baz",
+        "a fddf foobarbaz",
+        "a",
+        "a foobar abcdef",
+        "a kjlkjf dfff",
+        "a ТЕСТЙЦУК НГКШ ",
+        "a !!,23.(!1)",
+    ]
+    d = str(tmp_path / "complex")
+    write_custom_part(d, {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i for i in range(len(values))],
+        "columns": [{"name": "foo", "values": values}],
+    }]})
+
+    def ph(p):
+        return {"type": "phrase", "field": "foo", "phrase": p}
+
+    cases = [
+        # (foobar AND NOT baz AND (abcdef OR xyz)) -> [6]
+        ({"type": "and", "filters": [
+            ph("foobar"),
+            {"type": "not", "filter": ph("baz")},
+            {"type": "or", "filters": [ph("abcdef"), ph("xyz")]}]}, [6]),
+        # (foobaz AND NOT baz AND (abcdef OR xyz)) -> []
+        ({"type": "and", "filters": [
+            ph("foobaz"),
+            {"type": "not", "filter": ph("baz")},
+            {"type": "or", "filters": [ph("abcdef"), ph("xyz")]}]}, []),
+        # (foobar AND NOT baz AND (abcdef OR xyz OR a)) -> [1, 6]
+        ({"type": "and", "filters": [
+            ph("foobar"),
+            {"type": "not", "filter": ph("baz")},
+            {"type": "or", "filters": [ph("abcdef"), ph("xyz"), ph("a")]}]},
+         [1, 6]),
+        # (foobar AND NOT qwert AND (abcdef OR xyz OR a)) -> [1, 3, 6]
+        ({"type": "and", "filters": [
+            ph("foobar"),
+            {"type": "not", "filter": ph("qwert")},
+            {"type": "or", "filters": [ph("abcdef"), ph("xyz"), ph("a")]}]},
+         [1, 3, 6]),
+    ]
+    sc = OracleScanner(d)
+    try:
+        for f, want in cases:
+            hits, bits = sc.scan(json.dumps(f), with_bitmaps=True)
+            word = int.from_bytes(bits[:8], "little")
+            rows = [i for i in range(len(values)) if (word >> i) & 1]
+            assert rows == want, f"{f}: got {rows} want {want}"
+            assert hits == len(want)
+    finally:
+        sc.close()

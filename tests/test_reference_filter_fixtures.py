@@ -1,0 +1,115 @@
+"""The reference's own per-filter test fixtures, ported verbatim from
+lib/logstorage/filter_*_test.go by tools/port_filter_tests.py into
+tests/golden/filter_fixtures.json (committed; the reference tree is not
+read at test time).
+
+Every fixture is a column set + a filter + the expected matching row
+indexes.  The oracle must reproduce the reference's expected rows exactly;
+the GPU must match too (gpu-marked variant).  Fixtures whose regex falls
+outside the supported class are tolerated as loud compile rejects and
+counted."""
+
+import json
+import os
+
+import pytest
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+FIXTURES = os.path.join(HERE, "golden", "filter_fixtures.json")
+
+
+def load():
+    with open(FIXTURES) as f:
+        return json.load(f)
+
+
+def run_fixtures(scan_rows, max_reject=25):
+    """scan_rows(columns, filter_json) -> list of matching row indexes or
+    None for a loud compile reject."""
+    data = load()
+    total = checked = rejected = 0
+    failures = []
+    for fname, fixtures in sorted(data.items()):
+        for k, fx in enumerate(fixtures):
+            total += 1
+            rows = scan_rows(fx["columns"], fx["filter"])
+            if rows is None:
+                rejected += 1
+                continue
+            checked += 1
+            if rows != fx["expected"]:
+                failures.append(
+                    f"{fname}[{k}] {json.dumps(fx['filter'])[:120]}: "
+                    f"got {rows} want {fx['expected']}")
+                if len(failures) > 8:
+                    break
+    assert not failures, "\n".join(failures) + f"\n({len(failures)}+ failures)"
+    assert rejected <= max_reject, f"too many compile rejects: {rejected}"
+    assert checked > 1000
+    return checked, rejected
+
+
+def _write_part(tmpdir, idx, columns):
+    from victorialogs_amd import write_custom_part
+
+    n = len(columns[0]["values"])
+    d = os.path.join(tmpdir, f"fx{idx}")
+    write_custom_part(d, {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i for i in range(n)],
+        "columns": [{"name": c["name"], "values": c["values"]}
+                    for c in columns],
+    }]})
+    return d, n
+
+
+def test_reference_fixtures_oracle(tmp_path):
+    from victorialogs_amd import OracleScanner
+
+    counter = [0]
+
+    def scan(columns, filt):
+        d, n = _write_part(str(tmp_path), counter[0], columns)
+        counter[0] += 1
+        sc = OracleScanner(d)
+        try:
+            try:
+                hits, bits = sc.scan(json.dumps(filt), with_bitmaps=True)
+            except RuntimeError:
+                return None
+            word = int.from_bytes(bits[:(n + 63) // 64 * 8], "little")
+            return [i for i in range(n) if (word >> i) & 1]
+        finally:
+            sc.close()
+
+    checked, rejected = run_fixtures(scan)
+    print(f"oracle: {checked} fixtures checked, {rejected} compile-rejected")
+
+
+@pytest.mark.gpu
+def test_reference_fixtures_gpu(tmp_path):
+    from victorialogs_amd import Filter, Part, Stage
+
+    counter = [0]
+
+    def scan(columns, filt):
+        d, n = _write_part(str(tmp_path), counter[0], columns)
+        counter[0] += 1
+        try:
+            f = Filter(json.dumps(filt))
+        except RuntimeError:
+            return None
+        p = Part(d)
+        st = Stage(p, f, device=0)
+        try:
+            st.scan()
+            bits = st.fetch_bitmaps((n + 63) // 64)
+            word = int.from_bytes(bits, "little")
+            return [i for i in range(n) if (word >> i) & 1]
+        finally:
+            st.close()
+            f.close()
+            p.close()
+
+    checked, rejected = run_fixtures(scan)
+    print(f"gpu: {checked} fixtures checked, {rejected} compile-rejected")
