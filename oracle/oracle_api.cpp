@@ -273,6 +273,13 @@ long orc_encode_values(const char* joined, long jn, char* out, long cap) {
   }
 }
 
+long orc_le_values(const char* a, long an, const char* b, long bn, int excl) {
+  return le_values_string(strview(a, size_t(an)), strview(b, size_t(bn)),
+                          excl != 0)
+             ? 1
+             : 0;
+}
+
 // host matcher probes for the device row-ops differential fuzz
 long orc_match_prefix(const char* s, long sn, const char* pf, long pn) {
   return match_prefix(strview(s, size_t(sn)), strview(pf, size_t(pn))) ? 1 : 0;

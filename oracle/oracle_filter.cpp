@@ -1684,6 +1684,147 @@ static void apply_any_case(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
   }
 }
 
+
+// ---- filterEqField / filterLeField (filter_eq_field.go:122-220,
+//      filter_le_field.go:155-282) ----
+
+namespace {
+struct FieldView {
+  int kind = 0;  // 0 column, 1 const, 2 missing
+  std::string cval;
+  ColumnHeader ch;
+  const StringsBlockDec* dec = nullptr;
+};
+}  // namespace
+
+static FieldView make_field_view(BlockCtx& ctx, const std::string& name) {
+  FieldView fv;
+  std::string cv = ctx.const_value(name);
+  if (!cv.empty()) {
+    fv.kind = 1;
+    fv.cval = cv;
+    return fv;
+  }
+  if (!ctx.column_header(name, &fv.ch)) {
+    fv.kind = 2;
+    return fv;
+  }
+  fv.dec = &ctx.values(fv.ch);
+  return fv;
+}
+
+// blockResult.getValues row semantics (block_result.go:306-478): const value
+// for const columns, "" for missing, decoded string form for typed columns
+static std::string fv_row_str(const FieldView& fv, uint64_t idx) {
+  if (fv.kind == 1) return fv.cval;
+  if (fv.kind == 2) return std::string();
+  strview v = fv.dec->row(idx);
+  if (fv.ch.type == ValueType::String) return std::string(v.p, v.n);
+  if (fv.ch.type == ValueType::Dict) return fv.ch.dict[uint8_t(v.p[0])];
+  return format_value(fv.ch.type, v);
+}
+
+static void apply_eq_field(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  const std::string name = canonical_field(f.field);
+  const std::string other = canonical_field(f.min_s);
+  if (name == other) return;
+  FieldView a = make_field_view(ctx, name);
+  FieldView b = make_field_view(ctx, other);
+  if (a.kind == 1 && b.kind == 1) {
+    if (a.cval != b.cval) bm.reset_bits();
+    return;
+  }
+  if (a.kind == 2 && b.kind == 2) return;  // both missing: "" == ""
+  const bool same_type = a.kind == 0 && b.kind == 0 && a.ch.type == b.ch.type;
+  if (!same_type) {
+    // applyFilterString (filter_eq_field.go:182-204)
+    bm.for_each_set_bit([&](uint64_t idx) {
+      return fv_row_str(a, idx) == fv_row_str(b, idx);
+    });
+    return;
+  }
+  if (a.ch.type == ValueType::String) {
+    bm.for_each_set_bit([&](uint64_t idx) {
+      return a.dec->row(idx) == b.dec->row(idx);
+    });
+    return;
+  }
+  if (a.ch.type == ValueType::Dict) {
+    bm.for_each_set_bit([&](uint64_t idx) {
+      return a.ch.dict[uint8_t(a.dec->row(idx).p[0])] ==
+             b.ch.dict[uint8_t(b.dec->row(idx).p[0])];
+    });
+    return;
+  }
+  bm.for_each_set_bit([&](uint64_t idx) {
+    return a.dec->row(idx) == b.dec->row(idx);  // encoded binary equality
+  });
+}
+
+static void apply_le_field(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
+  const std::string name = canonical_field(f.field);
+  const std::string other = canonical_field(f.min_s);
+  const bool excl = f.min_u != 0;
+  if (name == other) {
+    if (excl) bm.reset_bits();
+    return;
+  }
+  FieldView a = make_field_view(ctx, name);
+  FieldView b = make_field_view(ctx, other);
+  if (a.kind == 1 && b.kind == 1) {
+    if (!le_values_string(strview(a.cval), strview(b.cval), excl)) {
+      bm.reset_bits();
+    }
+    return;
+  }
+  if (a.kind == 2 && b.kind == 2) {
+    if (excl) bm.reset_bits();
+    return;
+  }
+  const bool same_type = a.kind == 0 && b.kind == 0 && a.ch.type == b.ch.type;
+  if (!same_type || a.ch.type == ValueType::String) {
+    bm.for_each_set_bit([&](uint64_t idx) {
+      std::string sa = fv_row_str(a, idx);
+      std::string sb = fv_row_str(b, idx);
+      return le_values_string(strview(sa), strview(sb), excl);
+    });
+    return;
+  }
+  switch (a.ch.type) {
+    case ValueType::Dict:
+      bm.for_each_set_bit([&](uint64_t idx) {
+        const std::string& va = a.ch.dict[uint8_t(a.dec->row(idx).p[0])];
+        const std::string& vb = b.ch.dict[uint8_t(b.dec->row(idx).p[0])];
+        return le_values_string(strview(va), strview(vb), excl);
+      });
+      return;
+    case ValueType::Int64:
+      bm.for_each_set_bit([&](uint64_t idx) {
+        int64_t va = get_i64be_zigzag((const uint8_t*)a.dec->row(idx).p);
+        int64_t vb = get_i64be_zigzag((const uint8_t*)b.dec->row(idx).p);
+        return excl ? va < vb : va <= vb;
+      });
+      return;
+    case ValueType::Float64:
+      bm.for_each_set_bit([&](uint64_t idx) {
+        uint64_t ua = get_u64be((const uint8_t*)a.dec->row(idx).p);
+        uint64_t ub = get_u64be((const uint8_t*)b.dec->row(idx).p);
+        double va, vb;
+        memcpy(&va, &ua, 8);
+        memcpy(&vb, &ub, 8);
+        return excl ? va < vb : va <= vb;
+      });
+      return;
+    default:
+      // uint/ipv4/iso8601: leValuesString over the ENCODED bytes
+      // (filter_le_field.go:257-263 quirk, mirrored bit-for-bit)
+      bm.for_each_set_bit([&](uint64_t idx) {
+        return le_values_string(a.dec->row(idx), b.dec->row(idx), excl);
+      });
+      return;
+  }
+}
+
 void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
   switch (f.type) {
     case FilterNode::Phrase:
@@ -1737,6 +1878,12 @@ void apply_filter(const FilterNode& f, BlockCtx& ctx, Bitmap& bm) {
     case FilterNode::AnyCasePhrase:
     case FilterNode::AnyCasePrefix:
       apply_any_case(f, ctx, bm);
+      return;
+    case FilterNode::EqField:
+      apply_eq_field(f, ctx, bm);
+      return;
+    case FilterNode::LeField:
+      apply_le_field(f, ctx, bm);
       return;
     case FilterNode::Time:
       apply_time(f, ctx, bm);

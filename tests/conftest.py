@@ -61,6 +61,17 @@ def typed_part(tmp_path_factory):
                          "3d2h", "100KB", "", "nan?"][i % 10] for i in range(rows)]},
                     {"name": "uni", "values": [
                         ("раз два три" if i % 3 == 0 else "foo bar") for i in range(rows)]},
+                    # second columns for eq_field/le_field pairs
+                    {"name": "lvl2", "values": [dict_vals[(i // 2) % 4] for i in range(rows)]},
+                    {"name": "u8b", "values": [str((i + 1) % 250) for i in range(rows)]},
+                    {"name": "i64b", "values": [str((i - 148) * 37) for i in range(rows)]},
+                    {"name": "f64b", "values": [f"{(i - 149) / 8}" for i in range(rows)]},
+                    {"name": "ipb", "values": [f"10.{(i + 1) % 256}.{(i * 3) % 256}.{(i * 7) % 256}"
+                                               for i in range(rows)]},
+                    {"name": "isob", "values": [
+                        "2024-01-%02dT%02d:%02d:%02d.%03dZ"
+                        % (1 + (i + 1) % 28, i % 24, i % 60, (i * 3) % 60, i % 1000)
+                        for i in range(rows)]},
                     # parseMathNumber legs: scientific/hex/rfc3339/ipv4
                     {"name": "mix2", "values": [
                         ["1e5", "1.5e-3", "-2E2", "0x1F", "0o17", "0b101",
@@ -312,6 +323,27 @@ TYPED_FILTERS = [
     '{"type":"stream_id","ids":[{"account":0,"project":0,"hi":"0","lo":"1"},'
     '{"account":0,"project":0,"hi":"9","lo":"9"}]}',
     '{"type":"stream_id","ids":[]}',
+    # ---- eq_field / le_field (filter_eq_field.go, filter_le_field.go) ----
+    '{"type":"eq_field","field":"u8","other_field":"u8"}',
+    '{"type":"eq_field","field":"lvl","other_field":"lvl2"}',
+    '{"type":"eq_field","field":"u8","other_field":"u8b"}',
+    '{"type":"eq_field","field":"u8","other_field":"u16"}',
+    '{"type":"eq_field","field":"_msg","other_field":"lvl"}',
+    '{"type":"eq_field","field":"constcol","other_field":"_msg"}',
+    '{"type":"eq_field","field":"missing_col","other_field":"lvl"}',
+    '{"type":"eq_field","field":"missing_col","other_field":"missing2"}',
+    '{"type":"le_field","field":"u8","other_field":"u8b"}',
+    '{"type":"le_field","field":"u8","other_field":"u8b","exclude_equal":true}',
+    '{"type":"le_field","field":"u8","other_field":"u16"}',
+    '{"type":"le_field","field":"i64","other_field":"i64b"}',
+    '{"type":"le_field","field":"f64","other_field":"f64b"}',
+    '{"type":"le_field","field":"f64","other_field":"u8"}',
+    '{"type":"le_field","field":"lvl","other_field":"lvl2"}',
+    '{"type":"le_field","field":"ip","other_field":"ipb"}',
+    '{"type":"le_field","field":"iso","other_field":"isob"}',
+    '{"type":"le_field","field":"_msg","other_field":"mix"}',
+    '{"type":"le_field","field":"missing_col","other_field":"u8",'
+    '"exclude_equal":true}',
     # ---- any-case filters (filter_any_case_phrase.go, filter_any_case_prefix.go) ----
     '{"type":"any_case_phrase","field":"lvl","phrase":"ERROR"}',
     '{"type":"any_case_phrase","field":"lvl","phrase":"error"}',

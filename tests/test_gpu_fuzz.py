@@ -50,7 +50,8 @@ def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
                   "prefix", "exact_prefix", "sequence",
                   "in", "contains_any", "contains_all", "string_range",
                   "ipv4_range", "len_range", "day_range", "week_range",
-                  "value_type", "any_case_phrase", "any_case_prefix"]
+                  "value_type", "any_case_phrase", "any_case_prefix",
+                  "eq_field", "le_field"]
     kind = rng.choice(
         leaf_kinds + ["and", "or", "not"] if depth > 0 else leaf_kinds)
     if kind == "phrase":
@@ -61,6 +62,13 @@ def random_tree(rng, phrases, fields_str, fields_num, regexes, depth):
         return {"type": "exact",
                 "field": rng.choice(fields_str + fields_num),
                 "value": rng.choice(phrases)}
+    if kind in ("eq_field", "le_field"):
+        d = {"type": kind,
+             "field": rng.choice(fields_str + fields_num),
+             "other_field": rng.choice(fields_str + fields_num)}
+        if kind == "le_field" and rng.random() < 0.4:
+            d["exclude_equal"] = True
+        return d
     if kind in ("any_case_phrase",):
         return {"type": kind,
                 "field": rng.choice(fields_str + fields_num),

@@ -162,3 +162,25 @@ def test_any_case_differential(libs):
         want = orc.orc_any_case_phrase(s, len(s), ph, len(ph))
         got = dev.h_dev_any_case_phrase(s, len(s), ph, len(ph))
         assert got == want, f"anycase {ph!r} in {s!r}: dev={got} oracle={want}"
+
+
+def test_le_values_differential(libs):
+    """Device leValuesString (two-column le compare core) vs the host
+    restatement, over numeric and byte-noise spans."""
+    dev, orc = libs
+    import random
+    rng = random.Random(616)
+    orc.orc_le_values.restype = ctypes.c_long
+    pool = [b"10", b"9", b"-3", b"1.5", b"1.50", b"100ms", b"2s", b"abc",
+            b"", b"0x10", b"16", b"10.0.0.1", b"167772161", b"inf", b"-inf",
+            b"2024-01-01T00:00:00Z", b"1704067200000000000", b"nan"]
+    for _ in range(20000):
+        if rng.random() < 0.6:
+            a, b = rng.choice(pool), rng.choice(pool)
+        else:
+            a = bytes(rng.randrange(0, 256) for _ in range(rng.randrange(0, 12)))
+            b = bytes(rng.randrange(0, 256) for _ in range(rng.randrange(0, 12)))
+        for excl in (0, 1):
+            want = orc.orc_le_values(a, len(a), b, len(b), excl)
+            got = dev.h_dev_le_values(a, len(a), b, len(b), excl)
+            assert got == want, (a, b, excl, got, want)

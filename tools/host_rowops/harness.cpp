@@ -101,4 +101,11 @@ int h_dev_any_case_phrase(const char* s, long sn, const char* lower, long ln) {
              : 0;
 }
 
+int h_dev_le_values(const char* a, long an, const char* b, long bn, int excl) {
+  return d_le_values_string((const uint8_t*)a, an, (const uint8_t*)b, bn,
+                            excl != 0)
+             ? 1
+             : 0;
+}
+
 }  // extern "C"

@@ -44,6 +44,11 @@ FILTER_MAP = {
     "filterRegexp": ("regexp", {"fieldName": "field", "re": "re"}),
     "filterValueType": ("value_type",
                         {"fieldName": "field", "valueType": "value_type"}),
+    "filterEqField": ("eq_field", {"fieldName": "field",
+                                   "otherFieldName": "other_field"}),
+    "filterLeField": ("le_field",
+                      {"fieldName": "field", "otherFieldName": "other_field",
+                       "excludeEqualValues": "exclude_equal"}),
 }
 
 
@@ -254,6 +259,12 @@ def parse_filter(src, start):
             node[key] = p.parse_go_string()
         elif src.startswith("[]string{", p.i):
             node[key] = p.parse_string_list()
+        elif src.startswith("true", p.i):
+            node[key] = True
+            p.i += 4
+        elif src.startswith("false", p.i):
+            node[key] = False
+            p.i += 5
         else:
             v = p.parse_number()
             if v is None:

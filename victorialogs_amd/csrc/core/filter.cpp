@@ -338,6 +338,16 @@ FilterNode build(const JValue& v) {
     std::vector<std::string> toks_up;
     for (const auto& t : toks) toks_up.push_back(to_upper_str(strview(t)));
     n.all_hashes = probe_hashes(toks_up);
+  } else if (type == "eq_field" || type == "le_field") {
+    // filterEqField / filterLeField (filter_eq_field.go:15-21,
+    // filter_le_field.go:16-24)
+    n.type = type == "eq_field" ? FilterNode::EqField : FilterNode::LeField;
+    n.field = jget(v, "field").str;
+    n.min_s = jget(v, "other_field").str;
+    auto it = v.obj.find("exclude_equal");
+    if (it != v.obj.end() && it->second.kind == JValue::Bool) {
+      n.min_u = it->second.b ? 1 : 0;
+    }
   } else if (type == "noop") {
     n.type = FilterNode::Noop;
   } else {

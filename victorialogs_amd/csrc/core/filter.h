@@ -31,7 +31,7 @@ struct FilterNode {
     Prefix, ExactPrefix, Sequence,
     In, ContainsAny, ContainsAll, StringRange, IPv4Range, LenRange,
     DayRange, WeekRange, ValueTypeFilter, StreamIdFilter,
-    AnyCasePhrase, AnyCasePrefix
+    AnyCasePhrase, AnyCasePrefix, EqField, LeField
   } type;
 
   std::string field;   // phrase/exact/regexp/range (as written in the query)
@@ -44,7 +44,9 @@ struct FilterNode {
   std::vector<std::string> values;   // In/ContainsAny/ContainsAll
   std::string min_s, max_s;          // StringRange; ValueTypeFilter: min_s=type
                                      // AnyCase*: min_s=lowercase, max_s=uppercase
+                                     // EqField/LeField: min_s=otherFieldName
   uint64_t min_u = 0, max_u = 0;     // IPv4Range/LenRange; Day/WeekRange: start/end
+                                     // LeField: min_u=excludeEqualValues
   int64_t tz_offset = 0;             // Day/WeekRange offset (nsecs)
   std::vector<std::array<uint64_t, 3>> stream_ids;  // {acct<<32|proj, hi, lo}
 

@@ -434,6 +434,18 @@ bool try_parse_timestamp_rfc3339(strview s, int64_t* out) {
   return true;
 }
 
+bool le_values_string(strview a, strview b, bool exclude_equal) {
+  // leValuesString (filter_le_field.go:284-299)
+  double fa = parse_math_number(a);
+  if (!std::isnan(fa)) {
+    double fb = parse_math_number(b);
+    if (!std::isnan(fb)) return exclude_equal ? fa < fb : fa <= fb;
+  }
+  int c = memcmp(a.p, b.p, a.n < b.n ? a.n : b.n);
+  if (c == 0) c = a.n < b.n ? -1 : (a.n > b.n ? 1 : 0);
+  return exclude_equal ? c < 0 : c <= 0;
+}
+
 struct SvReader {
   const char* p;
   uint8_t u8(long i) const { return uint8_t(p[i]); }

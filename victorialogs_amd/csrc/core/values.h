@@ -40,11 +40,14 @@ bool try_parse_duration(strview s, int64_t* out);             // :990-1061
 // TryParseTimestampRFC3339Nano (values_encoder.go:340-381); no-timezone
 // inputs use a zero local offset (both runtime boxes are UTC; documented)
 bool try_parse_timestamp_rfc3339(strview s, int64_t* out);
+// leValuesString (filter_le_field.go:284-299): numeric compare when both
+// sides parse via parseMathNumber, else bytewise
+bool le_values_string(strview a, strview b, bool exclude_equal);
 bool try_parse_bytes(strview s, int64_t* out);                // :855-966
 
-// parseMathNumber subset (pipe_math.go:1066-1080 / block_result.go:2710-2735):
-// float64 -> duration -> bytes legs; other legs (hex ints, RFC3339, ipv4)
-// return NaN for now (documented limitation).
+// parseMathNumber (pipe_math.go:1066-1080 / block_result.go:2710-2752), all
+// legs: float64 -> duration -> bytes -> isLikelyNumber(ParseFloat/ParseInt
+// base 0) -> RFC3339Nano -> ipv4; NaN otherwise.
 double parse_math_number(strview s);
 
 // ---- formatters (values_encoder.go:1367-1424) ----

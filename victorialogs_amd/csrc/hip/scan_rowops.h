@@ -1161,6 +1161,135 @@ __device__ __forceinline__ bool d_eval_string_row(const DevLeafBlock& lb,
   return d_eval_string_row_cold(lb, a, s0, sn);
 }
 
+
+// ---- two-column filters (filter_eq_field.go, filter_le_field.go) ----
+
+// leValuesString (filter_le_field.go:284-299) over two global byte spans
+__device__ inline bool d_le_values_string(const uint8_t* ap, long an,
+                                          const uint8_t* bp, long bn,
+                                          bool exclude_equal) {
+  GlobalAcc aa{ap}, ab{bp};
+  double fa = d_parse_math_number(aa, 0, an);
+  if (!(fa != fa)) {  // !isnan
+    double fb = d_parse_math_number(ab, 0, bn);
+    if (!(fb != fb)) return exclude_equal ? fa < fb : fa <= fb;
+  }
+  long lim = an < bn ? an : bn;
+  int c = 0;
+  for (long i = 0; i < lim; i++) {
+    if (ap[i] != bp[i]) {
+      c = ap[i] < bp[i] ? -1 : 1;
+      break;
+    }
+  }
+  if (c == 0) c = an < bn ? -1 : (an > bn ? 1 : 0);
+  return exclude_equal ? c < 0 : c <= 0;
+}
+
+// One side of a two-column string-form compare.  Blob layout per side:
+// u8 mode (0 raw string col, 1 const, 2 missing, 3 formatted fixed-width,
+// 4 dict), u8 fmt<<4|width, u16 aux_len; aux bytes follow both descriptors
+// (mode 1: the const value; mode 4: dict table {u8 n, u16 offs[n+1], bytes}).
+struct DFieldSide {
+  const uint8_t* p;
+  long n;
+  char buf[40];
+};
+
+__device__ inline void d_field_side_resolve(DFieldSide& out, uint8_t mode,
+                                            uint8_t fmtw, const uint8_t* cval,
+                                            uint16_t clen,
+                                            const uint8_t* data,
+                                            const uint32_t* offs,
+                                            uint32_t row) {
+  switch (mode) {
+    case 0:  // raw string column
+      out.p = data + offs[row];
+      out.n = long(offs[row + 1]) - long(offs[row]);
+      return;
+    case 1:  // const value
+      out.p = cval;
+      out.n = clen;
+      return;
+    case 2:  // missing column => ""
+      out.p = (const uint8_t*)out.buf;
+      out.n = 0;
+      return;
+    case 4: {  // dict: 1-byte codes + table in the aux area
+      const uint8_t cnt = cval[0];
+      const uint8_t* offs16 = cval + 1;
+      const uint8_t code = data[row];
+      uint16_t o0 = uint16_t(offs16[code * 2]) |
+                    uint16_t(offs16[code * 2 + 1]) << 8;
+      uint16_t o1 = uint16_t(offs16[code * 2 + 2]) |
+                    uint16_t(offs16[code * 2 + 3]) << 8;
+      out.p = cval + 1 + 2 * (cnt + 1) + o0;
+      out.n = long(o1) - long(o0);
+      return;
+    }
+    default: {  // formatted fixed-width value
+      const uint8_t fmt = fmtw >> 4, w = fmtw & 15;
+      const uint8_t* q = data + size_t(row) * w;
+      int n;
+      switch (fmt) {
+        case kFmtU64: {
+          uint64_t v;
+          switch (w) {
+            case 1: v = q[0]; break;
+            case 2: v = d_get_u16be(q); break;
+            case 4: v = d_get_u32be(q); break;
+            default: v = d_get_u64be(q); break;
+          }
+          n = d_format_u64(out.buf, v);
+          break;
+        }
+        case kFmtI64: {
+          uint64_t u = d_get_u64be(q);
+          n = d_format_i64(out.buf, int64_t(u >> 1) ^ (int64_t(u << 63) >> 63));
+          break;
+        }
+        case kFmtF64:
+          n = vl_ryu::format_f64(out.buf, d_get_u64be(q));
+          break;
+        case kFmtIp:
+          n = d_format_ipv4(out.buf, d_get_u32be(q));
+          break;
+        default:
+          n = d_format_iso8601(out.buf, int64_t(d_get_u64be(q)));
+          break;
+      }
+      out.p = (const uint8_t*)out.buf;
+      out.n = n;
+      return;
+    }
+  }
+}
+
+__device__ inline bool d_eval_field_pair(const DevLeafBlock& lb, uint32_t row,
+                                         bool le) {
+  const uint8_t* blob = lb.operand;
+  const uint8_t amode = blob[0], afmtw = blob[1];
+  const uint8_t bmode = blob[4], bfmtw = blob[5];
+  uint16_t aclen = uint16_t(blob[2]) | uint16_t(blob[3]) << 8;
+  uint16_t bclen = uint16_t(blob[6]) | uint16_t(blob[7]) << 8;
+  const uint8_t* aconst = blob + 8;
+  const uint8_t* bconst = aconst + aclen;
+  DFieldSide sa, sb;
+  d_field_side_resolve(sa, amode, afmtw, aconst, aclen, lb.data, lb.offsets,
+                       row);
+  d_field_side_resolve(sb, bmode, bfmtw, bconst, bclen,
+                       (const uint8_t*)lb.hashes, (const uint32_t*)lb.bloom,
+                       row);
+  if (le) {
+    return d_le_values_string(sa.p, sa.n, sb.p, sb.n, (lb.flags & 1) != 0);
+  }
+  if (sa.n != sb.n) return false;
+  for (long i = 0; i < sa.n; i++) {
+    if (sa.p[i] != sb.p[i]) return false;
+  }
+  return true;
+}
+
 // Cold fixed-width kinds (formatters, regex, parsers) behind a call so the
 // hot scan loop stays small (see d_eval_string_row_cold).
 __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
@@ -1266,6 +1395,43 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
       int64_t off = (lb.ts[row] - off_tz) % (24LL * 3600 * 1000000000);
       return off >= int64_t(lb.vmin) && off <= int64_t(lb.vmax);
     }
+    case kScanEqFieldBin: {
+      const uint8_t* pa = lb.data + size_t(row) * lb.width;
+      const uint8_t* pb = (const uint8_t*)lb.hashes + size_t(row) * lb.width;
+      for (int i = 0; i < lb.width; i++) {
+        if (pa[i] != pb[i]) return false;
+      }
+      return true;
+    }
+    case kScanEqFieldDict:
+    case kScanLeFieldDict: {
+      const uint8_t ca = lb.data[row];
+      const uint8_t cb = ((const uint8_t*)lb.hashes)[row];
+      return (lb.vmin >> (ca * 8 + cb)) & 1;
+    }
+    case kScanLeFieldI64: {
+      uint64_t ua = d_get_u64be(lb.data + size_t(row) * 8);
+      uint64_t ub = d_get_u64be((const uint8_t*)lb.hashes + size_t(row) * 8);
+      int64_t va = int64_t(ua >> 1) ^ (int64_t(ua << 63) >> 63);
+      int64_t vb = int64_t(ub >> 1) ^ (int64_t(ub << 63) >> 63);
+      return (lb.flags & 1) ? va < vb : va <= vb;
+    }
+    case kScanLeFieldF64: {
+      uint64_t ua = d_get_u64be(lb.data + size_t(row) * 8);
+      uint64_t ub = d_get_u64be((const uint8_t*)lb.hashes + size_t(row) * 8);
+      double va = __builtin_bit_cast(double, ua);
+      double vb = __builtin_bit_cast(double, ub);
+      return (lb.flags & 1) ? va < vb : va <= vb;
+    }
+    case kScanLeFieldBinStr: {
+      return d_le_values_string(lb.data + size_t(row) * lb.width, lb.width,
+                                (const uint8_t*)lb.hashes + size_t(row) * lb.width,
+                                lb.width, (lb.flags & 1) != 0);
+    }
+    case kScanEqFieldStr:
+      return d_eval_field_pair(lb, row, false);
+    case kScanLeFieldStr:
+      return d_eval_field_pair(lb, row, true);
     case kScanWeekRange: {
       long long off_tz;
       __builtin_memcpy(&off_tz, lb.operand, 8);

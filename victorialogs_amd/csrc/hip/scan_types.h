@@ -65,6 +65,16 @@ enum ScanKind : uint8_t {
   kScanIPv4RangeBin = 37,   // BE u32 in [vmin,vmax] (filter_ipv4_range.go:166-181)
   kScanAnyCasePhraseStr = 38,  // matchAnyCasePhrase, ASCII rows (filter_any_case_phrase.go:159-181)
   kScanAnyCasePrefixStr = 39,  // matchAnyCasePrefix, ASCII rows (filter_any_case_prefix.go:161-183)
+  // two-column filters (filter_eq_field.go, filter_le_field.go); side B's
+  // data/offsets ride in the unused bloom-gate fields hashes/bloom
+  kScanEqFieldBin = 40,     // fixed-width encoded equality
+  kScanEqFieldDict = 41,    // dict codes vs 8x8 equality matrix in vmin
+  kScanLeFieldDict = 42,    // dict codes vs 8x8 le matrix in vmin
+  kScanLeFieldI64 = 43,     // zig-zag int64 compare (flags bit0 = exclude)
+  kScanLeFieldF64 = 44,     // float64 compare (flags bit0 = exclude)
+  kScanLeFieldBinStr = 45,  // leValuesString over the ENCODED bytes (quirk)
+  kScanEqFieldStr = 46,     // generic string-form equality (sides in operand)
+  kScanLeFieldStr = 47,     // generic string-form le (sides in operand)
 };
 
 // format source for the *Fmt kinds, stored in flags bits 4..7

@@ -2143,6 +2143,163 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
       }
     }
 
+
+    case FilterNode::EqField:
+    case FilterNode::LeField: {
+      // filterEqField / filterLeField (filter_eq_field.go:122-220,
+      // filter_le_field.go:155-282)
+      const bool is_le = f.type == FilterNode::LeField;
+      const bool excl = is_le && f.min_u != 0;
+      const std::string other = canonical_field(f.min_s);
+      if (li.cname == other) {
+        lb.mode = excl ? kModeNone : kModeAll;
+        return;
+      }
+      std::string cva, cvb;
+      bool a_const = pr.get_const_column(ctx.bc, li.cname, &cva) && !cva.empty();
+      bool b_const = pr.get_const_column(ctx.bc, other, &cvb) && !cvb.empty();
+      ColumnHeader cha, chb;
+      bool a_col = !a_const && pr.get_column_header(ctx.bc, li.cname, &cha);
+      bool b_col = !b_const && pr.get_column_header(ctx.bc, other, &chb);
+      if (a_const && b_const) {
+        bool m = is_le ? le_values_string(strview(cva), strview(cvb), excl)
+                       : cva == cvb;
+        lb.mode = m ? kModeAll : kModeNone;
+        return;
+      }
+      if (!a_const && !a_col && !b_const && !b_col) {
+        lb.mode = excl ? kModeNone : kModeAll;  // "" vs ""
+        return;
+      }
+      const bool same_type = a_col && b_col && cha.type == chb.type;
+      lb.mode = kModeScan;
+      lb.flags = excl ? 1 : 0;
+      if (same_type && cha.type != ValueType::String) {
+        const StagedStrCol& sa = ctx.stage_column(cha);
+        const StagedStrCol& sb = ctx.stage_column(chb);
+        // a staged typed column can still be const-encoded; that case falls
+        // through to the generic string-form path below (decoded-string
+        // compares give the same verdicts as the reference's encoded-bytes
+        // compares for same-type columns; the oracle keeps the exact
+        // reference logic)
+        if (!sa.is_const && !sb.is_const) {
+          lb.data = sa.d_data;
+          lb.hashes = (const uint64_t*)sb.d_data;
+          if (cha.type == ValueType::Dict) {
+            // 8x8 verdict matrix over the two dicts
+            uint64_t m = 0;
+            for (size_t i = 0; i < cha.dict.size(); i++) {
+              for (size_t j = 0; j < chb.dict.size(); j++) {
+                bool v = is_le ? le_values_string(strview(cha.dict[i]),
+                                                  strview(chb.dict[j]), excl)
+                               : cha.dict[i] == chb.dict[j];
+                if (v) m |= uint64_t(1) << (i * 8 + j);
+              }
+            }
+            lb.kind = is_le ? kScanLeFieldDict : kScanEqFieldDict;
+            lb.vmin = m;
+            return;
+          }
+          lb.width = width_of_type(cha.type);
+          if (!is_le) {
+            lb.kind = kScanEqFieldBin;
+            return;
+          }
+          if (cha.type == ValueType::Int64) {
+            lb.kind = kScanLeFieldI64;
+            return;
+          }
+          if (cha.type == ValueType::Float64) {
+            lb.kind = kScanLeFieldF64;
+            return;
+          }
+          lb.kind = kScanLeFieldBinStr;  // uint/ipv4/iso quirk path
+          return;
+        }
+      }
+      // generic string-form path: build the two side descriptors
+      bytes blob(8, 0);
+      bytes aux;
+      auto encode_side = [&](bool isc, const std::string& cv, bool iscol,
+                             const ColumnHeader& ch, int base,
+                             const uint8_t*& dptr, const uint32_t*& optr) {
+        dptr = nullptr;
+        optr = nullptr;
+        if (isc) {
+          blob[base] = 1;
+          blob[base + 2] = uint8_t(cv.size());
+          blob[base + 3] = uint8_t(cv.size() >> 8);
+          aux.insert(aux.end(), cv.begin(), cv.end());
+          return;
+        }
+        if (!iscol) {
+          blob[base] = 2;  // missing
+          return;
+        }
+        const StagedStrCol& sc = ctx.stage_column(ch);
+        if (sc.is_const) {
+          // typed column encoded as const: decode its string form once
+          std::string str = ch.type == ValueType::String
+                                ? sc.const_value
+                                : (ch.type == ValueType::Dict
+                                       ? ch.dict[uint8_t(sc.const_value[0])]
+                                       : format_encoded(
+                                             ch.type, strview(sc.const_value)));
+          blob[base] = 1;
+          blob[base + 2] = uint8_t(str.size());
+          blob[base + 3] = uint8_t(str.size() >> 8);
+          aux.insert(aux.end(), str.begin(), str.end());
+          return;
+        }
+        if (ch.type == ValueType::String) {
+          blob[base] = 0;
+          dptr = sc.d_data;
+          optr = sc.d_offsets;
+          return;
+        }
+        if (ch.type == ValueType::Dict) {
+          blob[base] = 4;
+          dptr = sc.d_data;
+          bytes tab;
+          tab.push_back(uint8_t(ch.dict.size()));
+          uint16_t off = 0;
+          for (size_t i = 0; i <= ch.dict.size(); i++) {
+            tab.push_back(uint8_t(off));
+            tab.push_back(uint8_t(off >> 8));
+            if (i < ch.dict.size()) off += uint16_t(ch.dict[i].size());
+          }
+          for (const auto& dv : ch.dict) {
+            tab.insert(tab.end(), dv.begin(), dv.end());
+          }
+          blob[base + 2] = uint8_t(tab.size());
+          blob[base + 3] = uint8_t(tab.size() >> 8);
+          aux.insert(aux.end(), tab.begin(), tab.end());
+          return;
+        }
+        blob[base] = 3;  // formatted fixed-width
+        blob[base + 1] = uint8_t(fmt_of_type(ch.type) << 4 |
+                                 width_of_type(ch.type));
+        dptr = sc.d_data;
+      };
+      const uint8_t* da;
+      const uint32_t* oa;
+      encode_side(a_const, cva, a_col, cha, 0, da, oa);
+      size_t a_aux = aux.size();
+      const uint8_t* db;
+      const uint32_t* ob;
+      encode_side(b_const, cvb, b_col, chb, 4, db, ob);
+      (void)a_aux;
+      blob.insert(blob.end(), aux.begin(), aux.end());
+      lb.kind = is_le ? kScanLeFieldStr : kScanEqFieldStr;
+      lb.operand = (const uint8_t*)st.push(blob.data(), blob.size(), 8);
+      lb.operand_len = uint32_t(blob.size());
+      lb.data = da;
+      lb.offsets = oa;
+      lb.hashes = (const uint64_t*)db;
+      lb.bloom = (const uint64_t*)ob;
+      return;
+    }
+
     default:
       fail("stage_leaf: non-leaf node");
   }
