@@ -15,7 +15,7 @@ HIPFLAGS ?= -O3 -std=c++17 -fPIC --offload-arch=$(ARCH)
 CORE_SRCS := $(wildcard victorialogs_amd/csrc/core/*.cpp)
 CORE_OBJS := $(patsubst victorialogs_amd/csrc/core/%.cpp,build/core/%.o,$(CORE_SRCS))
 
-all: oracle hip rowops
+all: oracle hip rowops emu
 
 oracle: oracle/liboracle.so
 
@@ -24,6 +24,20 @@ hip: victorialogs_amd/libvlogsql.so
 # host build of the per-row device code (scan_rowops.h) for CPU-side
 # differential fuzzing against the oracle (tests/test_rowops_fuzz.py)
 rowops: tools/host_rowops/librowops.so
+
+# CPU emulation build of the WHOLE product pipeline (real staging + real
+# per-row device code, HIP stubbed): tests/test_emu_pipeline.py loads it via
+# VQL_LIB to run the parity batteries and the 1386 reference fixtures end to
+# end without a GPU.  TEST INFRASTRUCTURE — never shipped as the product.
+emu: tools/host_emu/libvlogsql_emu.so
+
+tools/host_emu/libvlogsql_emu.so: victorialogs_amd/csrc/vql_api.cpp \
+		tools/host_emu/emu_kernels.cpp tools/host_emu/hip/hip_runtime.h \
+		victorialogs_amd/csrc/hip/scan_rowops.h \
+		victorialogs_amd/csrc/hip/scan_types.h $(CORE_OBJS)
+	$(CXX) $(CXXFLAGS) -shared -Itools/host_emu -Ivictorialogs_amd/csrc \
+		victorialogs_amd/csrc/vql_api.cpp tools/host_emu/emu_kernels.cpp \
+		$(CORE_OBJS) -o $@ -ldl -lpthread
 
 tools/host_rowops/librowops.so: tools/host_rowops/harness.cpp \
 		victorialogs_amd/csrc/hip/scan_rowops.h \

@@ -332,108 +332,6 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
 
 // ---- gather kernels (blockResult materialization, §8f row 1) ----
 
-__device__ __forceinline__ int d_u64_declen(uint64_t v) {
-  int n = 1;
-  while (v >= 10) {
-    v /= 10;
-    n++;
-  }
-  return n;
-}
-
-// value byte length for one row (pass 1) -- must agree with d_gather_write
-__device__ uint32_t d_gather_len(const DevGatherCol& gc, uint32_t row) {
-  switch (gc.src) {
-    case kGatherStr:
-      return gc.offsets[row + 1] - gc.offsets[row];
-    case kGatherConst:
-      return gc.cval_len;
-    case kGatherDict: {
-      uint8_t code = gc.data[row];
-      return gc.dict_offs[code + 1] - gc.dict_offs[code];
-    }
-    case kGatherFmtU: {
-      const uint8_t* p = gc.data + size_t(row) * gc.width;
-      uint64_t v;
-      switch (gc.width) {
-        case 1: v = p[0]; break;
-        case 2: v = d_get_u16be(p); break;
-        case 4: v = d_get_u32be(p); break;
-        default: v = d_get_u64be(p); break;
-      }
-      return uint32_t(d_u64_declen(v));
-    }
-    case kGatherFmtI: {
-      uint64_t u = d_get_u64be(gc.data + size_t(row) * 8);
-      int64_t v = int64_t(u >> 1) ^ (int64_t(u << 63) >> 63);
-      if (v < 0) return uint32_t(1 + d_u64_declen(~uint64_t(v) + 1));
-      return uint32_t(d_u64_declen(uint64_t(v)));
-    }
-    case kGatherFmtF: {
-      char buf[344];
-      return uint32_t(vl_ryu::format_f64(buf, d_get_u64be(gc.data + size_t(row) * 8)));
-    }
-    case kGatherFmtIp: {
-      uint32_t ip = d_get_u32be(gc.data + size_t(row) * 4);
-      return uint32_t(d_u64_declen((ip >> 24) & 255) + d_u64_declen((ip >> 16) & 255) +
-                      d_u64_declen((ip >> 8) & 255) + d_u64_declen(ip & 255) + 3);
-    }
-    case kGatherFmtIso:
-      return 24;
-    default:  // kGatherMissing
-      return 0;
-  }
-}
-
-__device__ uint32_t d_gather_write(const DevGatherCol& gc, uint32_t row,
-                                   uint8_t* dst) {
-  switch (gc.src) {
-    case kGatherStr: {
-      uint32_t off = gc.offsets[row];
-      uint32_t len = gc.offsets[row + 1] - off;
-      for (uint32_t i = 0; i < len; i++) dst[i] = gc.data[off + i];
-      return len;
-    }
-    case kGatherConst: {
-      for (uint32_t i = 0; i < gc.cval_len; i++) dst[i] = gc.cval[i];
-      return gc.cval_len;
-    }
-    case kGatherDict: {
-      uint8_t code = gc.data[row];
-      uint32_t off = gc.dict_offs[code];
-      uint32_t len = gc.dict_offs[code + 1] - off;
-      for (uint32_t i = 0; i < len; i++) dst[i] = gc.dict_data[off + i];
-      return len;
-    }
-    case kGatherFmtU: {
-      const uint8_t* p = gc.data + size_t(row) * gc.width;
-      uint64_t v;
-      switch (gc.width) {
-        case 1: v = p[0]; break;
-        case 2: v = d_get_u16be(p); break;
-        case 4: v = d_get_u32be(p); break;
-        default: v = d_get_u64be(p); break;
-      }
-      return uint32_t(d_format_u64((char*)dst, v));
-    }
-    case kGatherFmtI: {
-      uint64_t u = d_get_u64be(gc.data + size_t(row) * 8);
-      return uint32_t(
-          d_format_i64((char*)dst, int64_t(u >> 1) ^ (int64_t(u << 63) >> 63)));
-    }
-    case kGatherFmtF:
-      return uint32_t(
-          vl_ryu::format_f64((char*)dst, d_get_u64be(gc.data + size_t(row) * 8)));
-    case kGatherFmtIp:
-      return uint32_t(d_format_ipv4((char*)dst, d_get_u32be(gc.data + size_t(row) * 4)));
-    case kGatherFmtIso:
-      return uint32_t(d_format_iso8601(
-          (char*)dst, int64_t(d_get_u64be(gc.data + size_t(row) * 8))));
-    default:
-      return 0;
-  }
-}
-
 __global__ __launch_bounds__(256) void gather_count_kernel(
     const DevGatherCol* __restrict__ gcols, const DevBlock* __restrict__ blocks,
     const DevChunk* __restrict__ chunks, DevChunkCount* __restrict__ out) {
