@@ -108,3 +108,53 @@ for seed in range(4):
 print("emu fuzz OK")
 """)
     assert "emu fuzz OK" in out
+
+
+def test_emu_gather_and_drivers(tmp_path):
+    """Gather, multi-part staging, vql_scan_query and the bloom build
+    through the emulated pipeline (their gpu-marked twins run the same code
+    against the real kernels)."""
+    out = run_in_emu("""
+import sys, tempfile, pathlib
+sys.path.insert(0, ".")
+import tests.conftest as cf
+import tests.test_gpu_gather as gg
+import tests.test_gpu_parity as gp
+import tests.test_gpu_bloom_build as gb
+
+class TF:
+    def mktemp(self, x):
+        return pathlib.Path(tempfile.mkdtemp())
+
+gen = cf.gen_part.__wrapped__(TF())
+typed = cf.typed_part.__wrapped__(TF())
+
+# gather suite over the typed part
+from victorialogs_amd import Filter, Part, Stage
+part = Part(typed)
+filt = Filter('{"type":"phrase","field":"lvl","phrase":"error"}')
+st = Stage(part, filt, device=0)
+st.scan()
+
+class _S:  # mimic the module fixture
+    pass
+
+gg.test_gather_rowids(st)
+gg.test_gather_string_column(st)
+gg.test_gather_const_column(st)
+gg.test_gather_numeric_columns(st)
+gg.test_gather_missing_column(st)
+st.close(); filt.close(); part.close()
+gg.test_gather_multiblock(typed)
+
+gp.test_multipart_stage(gen, typed)
+gp.test_scan_query_driver(gen, typed)
+gp.test_multichunk_block(pathlib.Path(tempfile.mkdtemp()))
+gp.test_long_rows_global_fallback(pathlib.Path(tempfile.mkdtemp()))
+
+gb.test_bloom_build_simple()
+gb.test_bloom_build_unicode()
+gb.test_bloom_build_edge_cases()
+print("emu gather/drivers OK")
+""")
+    assert "emu gather/drivers OK" in out
