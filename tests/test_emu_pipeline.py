@@ -69,9 +69,11 @@ def test_emu_reference_fixtures():
     out = run_in_emu("""
 import sys
 sys.path.insert(0, ".")
-from tests.test_reference_filter_fixtures import test_reference_fixtures_gpu
+from tests.test_reference_filter_fixtures import (
+    test_reference_fixtures_gpu, test_reference_ts_fixtures_gpu)
 import tempfile, pathlib
 test_reference_fixtures_gpu(pathlib.Path(tempfile.mkdtemp()))
+test_reference_ts_fixtures_gpu(pathlib.Path(tempfile.mkdtemp()))
 print("emu fixtures OK")
 """)
     assert "emu fixtures OK" in out

@@ -113,3 +113,83 @@ def test_reference_fixtures_gpu(tmp_path):
 
     checked, rejected = run_fixtures(scan)
     print(f"gpu: {checked} fixtures checked, {rejected} compile-rejected")
+
+
+TS_FIXTURES = os.path.join(HERE, "golden", "filter_ts_fixtures.json")
+
+
+def run_ts_fixtures(scan_rows):
+    with open(TS_FIXTURES) as f:
+        data = json.load(f)
+    checked = 0
+    failures = []
+    for fname, fixtures in sorted(data.items()):
+        for k, fx in enumerate(fixtures):
+            rows = scan_rows(fx["timestamps"], fx["filter"])
+            checked += 1
+            if rows != fx["expected"]:
+                failures.append(f"{fname}[{k}] {json.dumps(fx['filter'])}: "
+                                f"got {rows} want {fx['expected']}")
+    assert not failures, "\n".join(failures[:8])
+    assert checked >= 25
+    return checked
+
+
+def _write_ts_part(tmpdir, idx, timestamps):
+    from victorialogs_amd import write_custom_part
+
+    n = len(timestamps)
+    d = os.path.join(tmpdir, f"ts{idx}")
+    write_custom_part(d, {"blocks": [{
+        "stream": 0,
+        "timestamps": timestamps,
+        "columns": [{"name": "v", "values": [str(i) for i in range(n)]}],
+    }]})
+    return d, n
+
+
+def test_reference_ts_fixtures_oracle(tmp_path):
+    """filter_time/day_range fixtures (testFilterMatchForTimestamps)."""
+    from victorialogs_amd import OracleScanner
+
+    counter = [0]
+
+    def scan(timestamps, filt):
+        d, n = _write_ts_part(str(tmp_path), counter[0], timestamps)
+        counter[0] += 1
+        sc = OracleScanner(d)
+        try:
+            hits, bits = sc.scan(json.dumps(filt), with_bitmaps=True)
+            word = int.from_bytes(bits[:(n + 63) // 64 * 8], "little")
+            return [i for i in range(n) if (word >> i) & 1]
+        finally:
+            sc.close()
+
+    checked = run_ts_fixtures(scan)
+    print(f"oracle ts fixtures: {checked}")
+
+
+@pytest.mark.gpu
+def test_reference_ts_fixtures_gpu(tmp_path):
+    from victorialogs_amd import Filter, Part, Stage
+
+    counter = [0]
+
+    def scan(timestamps, filt):
+        d, n = _write_ts_part(str(tmp_path), counter[0], timestamps)
+        counter[0] += 1
+        f = Filter(json.dumps(filt))
+        p = Part(d)
+        st = Stage(p, f, device=0)
+        try:
+            st.scan()
+            bits = st.fetch_bitmaps((n + 63) // 64)
+            word = int.from_bytes(bits, "little")
+            return [i for i in range(n) if (word >> i) & 1]
+        finally:
+            st.close()
+            f.close()
+            p.close()
+
+    checked = run_ts_fixtures(scan)
+    print(f"gpu ts fixtures: {checked}")

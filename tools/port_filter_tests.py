@@ -49,6 +49,13 @@ FILTER_MAP = {
     "filterLeField": ("le_field",
                       {"fieldName": "field", "otherFieldName": "other_field",
                        "excludeEqualValues": "exclude_equal"}),
+    # timestamp-harness filters (testFilterMatchForTimestamps)
+    "filterTime": ("time", {"minTimestamp": "min", "maxTimestamp": "max"}),
+    "filterDayRange": ("day_range", {"start": "start", "end": "end",
+                                     "offset": "offset"}),
+    "filterWeekRange": ("week_range", {"startDay": "start", "endDay": "end",
+                                       "start": "start", "end": "end",
+                                       "offset": "offset"}),
 }
 
 
@@ -273,6 +280,70 @@ def parse_filter(src, start):
     return (node, p.i)
 
 
+def parse_int_list(src, start):
+    p = P(src)
+    p.i = src.index("{", start) + 1
+    vals = []
+    while True:
+        p.skip_ws()
+        if src[p.i] == "}":
+            return vals, p.i + 1
+        v = p.parse_number()
+        if v is None:
+            raise ValueError("int list: " + src[p.i:p.i + 30])
+        vals.append(int(v))
+
+
+def port_ts_file(path):
+    """Ports testFilterMatchForTimestamps-based fixtures (time / day_range /
+    week_range): a timestamps list + filter + expected row indexes."""
+    src = open(path).read()
+    fixtures = []
+    skipped = []
+    ts_iter = [(m.start(), m) for m in
+               re.finditer(r"timestamps\s*:?=\s*\[\]int64\{", src)]
+    call_iter = list(re.finditer(
+        r"testFilterMatchForTimestamps\(t,\s*timestamps,\s*(\w+),\s*(nil|\[\]int\{[^}]*\})\)",
+        src))
+    assigns = list(re.finditer(r"(\w+)\s*:?=\s*&(filter\w+)\{", src))
+    for call in call_iter:
+        pos = call.start()
+        ts_def = None
+        for tstart, tm in ts_iter:
+            if tstart < pos:
+                ts_def = tm
+            else:
+                break
+        if ts_def is None:
+            skipped.append("no timestamps")
+            continue
+        try:
+            ts, _ = parse_int_list(src, ts_def.end() - 1)
+        except ValueError:
+            skipped.append("symbolic timestamps")
+            continue
+        var = call.group(1)
+        adef = None
+        for am in assigns:
+            if am.start() < pos and am.group(1) == var:
+                adef = am
+            elif am.start() >= pos:
+                break
+        if adef is None:
+            skipped.append(f"no assign for {var}")
+            continue
+        node = parse_filter(src, src.index("&", adef.start()))
+        if node[0] is None:
+            skipped.append(node[1])
+            continue
+        exp = call.group(2)
+        rows = ([] if exp == "nil" else
+                [int(x) for x in re.findall(r"-?\d+", exp)])
+        fixtures.append({"timestamps": ts, "filter": node[0],
+                         "expected": rows})
+    return fixtures, skipped
+
+
 def port_file(path):
     src = open(path).read()
     fixtures = []
@@ -340,6 +411,7 @@ def port_file(path):
 
 def main():
     out = {}
+    ts_out = {}
     total = 0
     for fn in sorted(os.listdir(REF)):
         if not (fn.startswith("filter_") and fn.endswith("_test.go")):
@@ -352,12 +424,18 @@ def main():
         if fx:
             out[fn] = fx
             total += len(fx)
+        tfx, tsk = port_ts_file(os.path.join(REF, fn))
+        if tfx:
+            ts_out[fn] = tfx
+            total += len(tfx)
         from collections import Counter
-        reasons = Counter(sk)
-        print(f"{fn}: {len(fx)} ported, {len(sk)} skipped"
-              + (f" {dict(reasons)}" if sk else ""))
+        reasons = Counter(sk + tsk)
+        print(f"{fn}: {len(fx)}+{len(tfx)} ported, {len(sk)+len(tsk)} skipped"
+              + (f" {dict(reasons)}" if (sk or tsk) else ""))
     with open("tests/golden/filter_fixtures.json", "w") as f:
         json.dump(out, f, ensure_ascii=False, indent=0)
+    with open("tests/golden/filter_ts_fixtures.json", "w") as f:
+        json.dump(ts_out, f, ensure_ascii=False, indent=0)
     print("TOTAL", total)
 
 
