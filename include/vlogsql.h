@@ -42,9 +42,9 @@ long vql_block_rows(void* part, long block);
  * programmatically).  Node types: phrase, exact, regexp, prefix,
  * exact_prefix, sequence, any_case_phrase, any_case_prefix, in,
  * contains_any, contains_all, string_range, ipv4_range, len_range,
- * day_range, week_range, value_type, stream_id, and, or, not, time, range,
- * noop — every applyToBlockSearch filter except filter_stream (needs
- * indexdb). */
+ * day_range, week_range, value_type, stream_id, eq_field, le_field, and,
+ * or, not, time, range, noop — every applyToBlockSearch filter except
+ * filter_stream (needs indexdb; stream_id is its resolved form). */
 void* vql_compile_filter(const char* json);
 void vql_free_filter(void* filter);
 
