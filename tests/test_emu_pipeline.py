@@ -160,3 +160,26 @@ gb.test_bloom_build_edge_cases()
 print("emu gather/drivers OK")
 """)
     assert "emu gather/drivers OK" in out
+
+
+def test_emu_bench_contract(tmp_path):
+    """bench.py end to end (small rows) under the emulated pipeline: the
+    printed JSON line must satisfy the driver contract."""
+    env = dict(os.environ, VQL_LIB=EMU,
+               VQL_DATA_DIR=str(tmp_path / "bench_data"))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--rows", "200000", "--steps", "2",
+         "--warmup", "1", "--skip-cpu-baseline"],
+        env=env, cwd=ROOT, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-3000:]
+    line = json.loads(r.stdout.strip().splitlines()[-1])
+    for key in ["metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config", "roofline", "cpu_baseline"]:
+        assert key in line, key
+    assert line["metric"] == "matched-rows/sec"
+    assert line["n_gpus"] == 1 and line["scaling"] == "weak"
+    assert line["config"]["workload"].startswith("100M rows") or \
+        "phrase" in line["config"]["filter"]
+    assert line["roofline"]["bound"] == "hbm"
+    assert line["value"] > 0
