@@ -61,6 +61,11 @@ class PartReader {
  private:
   struct FileR {
     int fd = -1;
+    FileR() = default;
+    FileR(const FileR&) = delete;
+    FileR& operator=(const FileR&) = delete;
+    FileR(FileR&& o) noexcept : fd(o.fd) { o.fd = -1; }
+    ~FileR() { close(); }  // a long-lived process opens many parts
     void open(const std::string& path, bool required);
     void pread_full(uint8_t* dst, size_t n, uint64_t off) const;
     bool ok() const { return fd >= 0; }
