@@ -294,10 +294,37 @@ def parse_int_list(src, start):
         vals.append(int(v))
 
 
+GO_WEEKDAYS = {"time.Sunday": 0, "time.Monday": 1, "time.Tuesday": 2,
+               "time.Wednesday": 3, "time.Thursday": 4, "time.Friday": 5,
+               "time.Saturday": 6}
+
+
 def port_ts_file(path):
     """Ports testFilterMatchForTimestamps-based fixtures (time / day_range /
     week_range): a timestamps list + filter + expected row indexes."""
     src = open(path).read()
+    # resolve the week-range file's symbolic timestamps:
+    #   sunday := time.Date(2024, 6, 9, 1, 0, 0, 0, time.UTC).UnixNano()
+    import datetime
+    for m in re.finditer(
+            r"(\w+)\s*:=\s*time\.Date\((\d+),\s*(\d+),\s*(\d+),"
+            r"\s*(\d+),\s*(\d+),\s*(\d+),\s*(\d+),\s*time\.UTC\)"
+            r"\.UnixNano\(\)", src):
+        y, mo, d, h, mi, sec, ns = (int(x) for x in m.groups()[1:])
+        dt = datetime.datetime(y, mo, d, h, mi, sec,
+                               tzinfo=datetime.timezone.utc)
+        v = int(dt.timestamp()) * 10**9 + ns
+        src = src.replace(m.group(0), "")
+        src = re.sub(r"\b" + m.group(1) + r"\s*\+\s*(\d+)\*nsecsPerDay",
+                     lambda g: str(v + int(g.group(1)) * 86400 * 10**9), src)
+        src = re.sub(r"\b" + m.group(1) + r"\b(?!\w)", str(v), src)
+    for name, val in GO_WEEKDAYS.items():
+        src = src.replace(name, str(val))
+    src = src.replace("offset: (12 * nsecsPerHour)", "offset: %d" % (12 * 3600 * 10**9))
+    src = re.sub(r"offset:\s*\(?(-?\d+)\s*\*\s*nsecsPerHour\)?",
+                 lambda g: "offset: %d" % (int(g.group(1)) * 3600 * 10**9), src)
+    src = re.sub(r"offset:\s*\(?(-?\d+)\s*\*\s*nsecsPerDay\)?",
+                 lambda g: "offset: %d" % (int(g.group(1)) * 86400 * 10**9), src)
     fixtures = []
     skipped = []
     ts_iter = [(m.start(), m) for m in
