@@ -37,12 +37,12 @@ AND_REGEX_FILTER = (
 
 WORKLOADS = {
     "phrase": {
-        "name": "100M rows, phrase filter on _msg column, 1xMI355X "
+        "name": "{rows} rows, phrase filter on _msg column "
                 "(bloom + substring kernel)",
         "filter": PHRASE_FILTER,
     },
     "phrase_regex": {
-        "name": "100M rows, AND(phrase, regex) on two string columns, 1xMI355X",
+        "name": "{rows} rows, AND(phrase, regex) on two string columns",
         "filter": AND_REGEX_FILTER,
     },
     "dict_time": {
@@ -258,7 +258,9 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": wl["name"],
+                "workload": wl["name"].format(
+                    rows=f"{args.rows // 10**6}M" if args.rows >= 10**6
+                    else str(args.rows)),
                 "rows_per_gpu": rows,
                 "msg_len": args.msg_len,
                 "filter": filter_json,
