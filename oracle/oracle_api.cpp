@@ -273,6 +273,38 @@ long orc_encode_values(const char* joined, long jn, char* out, long cap) {
   }
 }
 
+// getCommonTokensAndTokenSets probe (in_values_test.go:9-37)
+long orc_common_tokens(const char* joined, long jn, char* out, long cap) {
+  std::vector<std::string> values;
+  const char* p = joined;
+  const char* end = joined + jn;
+  while (p < end) {
+    const char* nl = (const char*)memchr(p, '\n', size_t(end - p));
+    if (!nl) nl = end;
+    values.emplace_back(p, nl);
+    p = nl + 1;
+  }
+  if (jn == 0) values.clear();
+  std::vector<std::string> common;
+  std::vector<std::vector<std::string>> sets;
+  get_common_tokens_and_sets(values, &common, &sets);
+  std::string outs;
+  for (size_t i = 0; i < common.size(); i++) {
+    if (i) outs += ' ';
+    outs += common[i];
+  }
+  for (const auto& ts : sets) {
+    outs += '|';
+    for (size_t i = 0; i < ts.size(); i++) {
+      if (i) outs += ' ';
+      outs += ts[i];
+    }
+  }
+  long n = long(outs.size()) < cap ? long(outs.size()) : cap;
+  memcpy(out, outs.data(), size_t(n));
+  return long(outs.size());
+}
+
 long orc_le_values(const char* a, long an, const char* b, long bn, int excl) {
   return le_values_string(strview(a, size_t(an)), strview(b, size_t(bn)),
                           excl != 0)

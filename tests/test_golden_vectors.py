@@ -535,3 +535,30 @@ def test_block_codec_roundtrips(lib):
     ib(sorted(rng.randrange(-2**62, 2**62) for _ in range(257)))
     ib([rng.randrange(-2**62, 2**62) for _ in range(257)])  # unsorted
     ib([0, 2**62, -2**62, 1, -1] * 50)
+
+
+def test_common_tokens_table(lib):
+    """TestGetCommonTokensAndTokenSets verbatim (in_values_test.go:9-37)."""
+    import ctypes
+
+    def ct(values):
+        joined = "\n".join(values).encode()
+        buf = ctypes.create_string_buffer(4096)
+        n = lib.orc_common_tokens(joined, len(joined), buf, 4096)
+        parts = buf.raw[:n].decode().split("|")
+        common = sorted(parts[0].split()) if parts[0] else []
+        sets = [sorted(p.split()) if p else [] for p in parts[1:]]
+        return common, sets
+
+    assert ct([]) == ([], [])
+    assert ct(["foo"]) == (["foo"], [[]])
+    assert ct(["foo", "foo"]) == (["foo"], [[], []])
+    assert ct(["foo", "bar", "bar", "foo"]) == (
+        [], [["foo"], ["bar"], ["bar"], ["foo"]])
+    assert ct(["foo", "foo bar", "bar foo"]) == (
+        ["foo"], [[], ["bar"], ["bar"]])
+    assert ct(["a foo bar", "bar abc foo", "foo abc a bar"]) == (
+        ["bar", "foo"], [["a"], ["abc"], ["a", "abc"]])
+    assert ct(["a xfoo bar", "xbar abc foo", "foo abc a bar"]) == (
+        [], [["a", "bar", "xfoo"], ["abc", "foo", "xbar"],
+             ["a", "abc", "bar", "foo"]])

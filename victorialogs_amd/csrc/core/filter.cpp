@@ -191,37 +191,10 @@ FilterNode build(const JValue& v) {
         append_hash_hashes(n.all_hashes, th);
       }
     } else {
-      // getTokensHashesAny (in_values.go:104-125): per-value token sets with
-      // the common tokens factored out
-      std::vector<std::vector<std::string>> token_sets;
-      for (const auto& s2 : n.values) {
-        token_sets.push_back(tokenize_strings({s2}));
-      }
+      // getTokensHashesAny (in_values.go:104-125)
       std::vector<std::string> common;
-      if (!token_sets.empty()) {
-        common = token_sets[0];
-        for (size_t i = 1; i < token_sets.size() && !common.empty(); i++) {
-          std::vector<std::string> kept;
-          for (const auto& t : common) {
-            if (std::find(token_sets[i].begin(), token_sets[i].end(), t) !=
-                token_sets[i].end()) {
-              kept.push_back(t);
-            }
-          }
-          common = std::move(kept);
-        }
-      }
-      if (!common.empty()) {
-        for (auto& ts : token_sets) {
-          std::vector<std::string> kept;
-          for (auto& t : ts) {
-            if (std::find(common.begin(), common.end(), t) == common.end()) {
-              kept.push_back(std::move(t));
-            }
-          }
-          ts = std::move(kept);
-        }
-      }
+      std::vector<std::vector<std::string>> token_sets;
+      get_common_tokens_and_sets(n.values, &common, &token_sets);
       n.common_hashes = probe_hashes(common);
       for (const auto& ts : token_sets) {
         n.set_hashes.push_back(probe_hashes(ts));

@@ -1,5 +1,7 @@
 #include "tokenizer.h"
 
+#include <algorithm>
+
 #include <unordered_set>
 
 #include "xxhash64.h"
@@ -162,6 +164,42 @@ std::vector<uint64_t> tokenize_hashes(const std::vector<strview>& a) {
     });
   }
   return dst;
+}
+
+
+void get_common_tokens_and_sets(const std::vector<std::string>& values,
+                                std::vector<std::string>* common,
+                                std::vector<std::vector<std::string>>* sets) {
+  // getCommonTokensAndTokenSets (in_values.go:104-139)
+  sets->clear();
+  common->clear();
+  for (const auto& v : values) {
+    sets->push_back(tokenize_strings({v}));
+  }
+  if (!sets->empty()) {
+    *common = (*sets)[0];
+    for (size_t i = 1; i < sets->size() && !common->empty(); i++) {
+      std::vector<std::string> kept;
+      for (const auto& t : *common) {
+        if (std::find((*sets)[i].begin(), (*sets)[i].end(), t) !=
+            (*sets)[i].end()) {
+          kept.push_back(t);
+        }
+      }
+      *common = std::move(kept);
+    }
+  }
+  if (!common->empty()) {
+    for (auto& ts : *sets) {
+      std::vector<std::string> kept;
+      for (auto& t : ts) {
+        if (std::find(common->begin(), common->end(), t) == common->end()) {
+          kept.push_back(std::move(t));
+        }
+      }
+      ts = std::move(kept);
+    }
+  }
 }
 
 }  // namespace vl

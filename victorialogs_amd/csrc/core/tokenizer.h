@@ -35,4 +35,10 @@ std::vector<std::string> tokenize_strings(const std::vector<std::string>& a);
 // way as the reference since dedup is by hash value.)
 std::vector<uint64_t> tokenize_hashes(const std::vector<strview>& a);
 
+// getCommonTokensAndTokenSets (in_values.go:104-139): per-value token sets
+// with the tokens common to every value factored out.
+void get_common_tokens_and_sets(const std::vector<std::string>& values,
+                                std::vector<std::string>* common,
+                                std::vector<std::vector<std::string>>* sets);
+
 }  // namespace vl
