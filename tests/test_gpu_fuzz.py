@@ -32,8 +32,10 @@ TYPED_PHRASES = [
     "level=error", "took", "row", "warn", "ERROR", "13", "два", "value",
     "1.5", "200ms", "absent", "", "ДВА", "Foo", "LEVEL",
 ]
-TYPED_FIELDS_STR = ["_msg", "lvl", "uni", "mix", "constcol", "nope"]
-TYPED_FIELDS_NUM = ["u8", "u16", "u32", "u64", "i64", "f64", "ip", "iso"]
+TYPED_FIELDS_STR = ["_msg", "lvl", "uni", "mix", "constcol", "nope", "lvl2",
+                    "ipstr", "mix2"]
+TYPED_FIELDS_NUM = ["u8", "u16", "u32", "u64", "i64", "f64", "ip", "iso",
+                    "u8b", "i64b", "f64b", "ipb", "isob"]
 TYPED_REGEXES = [
     "level=(error|warn)", "took 1.*ms", "row [0-9]", "два|foo", "19(2|3)",
     "-3", "2024-01", "fixed.+",
