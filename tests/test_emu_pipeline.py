@@ -183,3 +183,65 @@ def test_emu_bench_contract(tmp_path):
         "phrase" in line["config"]["filter"]
     assert line["roofline"]["bound"] == "hbm"
     assert line["value"] > 0
+
+
+def test_emu_random_parts():
+    """Randomized PARTS (random typed columns incl. contaminated type mixes)
+    x random filters through the emulated pipeline vs the oracle — covers
+    encoder-priority and per-type staging corners beyond the fixed
+    fixtures."""
+    out = run_in_emu("""
+import json, random, sys, tempfile
+sys.path.insert(0, ".")
+from victorialogs_amd import write_custom_part
+from tests.test_gpu_parity import assert_parity
+from tests.test_gpu_fuzz import random_tree
+
+def rand_value(rng, kind):
+    if kind == "u8": return str(rng.randrange(0, 256))
+    if kind == "u16": return str(rng.randrange(0, 1 << 16))
+    if kind == "u32": return str(rng.randrange(0, 1 << 32))
+    if kind == "u64": return str(rng.randrange(0, 1 << 64))
+    if kind == "i64": return str(rng.randrange(-1 << 62, 1 << 62))
+    if kind == "f64": return repr(rng.uniform(-1e6, 1e6))
+    if kind == "ip": return "%d.%d.%d.%d" % tuple(
+        rng.randrange(0, 256) for _ in range(4))
+    if kind == "iso": return "20%02d-%02d-%02dT%02d:%02d:%02d.%03dZ" % (
+        rng.randrange(0, 60), rng.randrange(1, 13), rng.randrange(1, 29),
+        rng.randrange(0, 24), rng.randrange(0, 60), rng.randrange(0, 60),
+        rng.randrange(0, 1000))
+    if kind == "dict": return rng.choice(["alpha", "beta", "gamma", "delta"])
+    return "".join(rng.choice("ab 01_.-!") for _ in range(rng.randrange(0, 20)))
+
+KINDS = ["u8", "u16", "u32", "u64", "i64", "f64", "ip", "iso", "dict", "word"]
+PHRASES = ["alpha", "1", "12", "255", "ab", "0", "-", ".", "", "beta b"]
+REGEXES = ["al.*a", "^1\\\\d", "a|b", "(be|ga)ta", "\\\\d+\\\\.\\\\d+",
+           "1.2", "[0-9]{2,3}$"]
+n = 0
+for seed in range(300, 312):
+    rng = random.Random(seed)
+    rows = rng.randrange(1, 300)
+    cols, names = [], []
+    for c in range(rng.randrange(1, 5)):
+        kind = rng.choice(KINDS)
+        mix = rng.random() < 0.25
+        vals = [rand_value(rng, rng.choice(KINDS))
+                if (mix and rng.random() < 0.1) else rand_value(rng, kind)
+                for _ in range(rows)]
+        names.append("c%d_%s" % (c, kind))
+        cols.append({"name": names[-1], "values": vals})
+    ts0 = rng.randrange(-1 << 60, 1 << 60)
+    d = tempfile.mkdtemp() + "/p"
+    write_custom_part(d, {"blocks": [{
+        "stream": 0,
+        "timestamps": sorted(ts0 + rng.randrange(0, 1 << 40)
+                             for _ in range(rows)),
+        "columns": cols}]})
+    for _ in range(8):
+        t = random_tree(rng, PHRASES, names + ["missing"], names, REGEXES,
+                        depth=3)
+        assert_parity(d, json.dumps(t))
+        n += 1
+print("emu random parts OK", n)
+""")
+    assert "emu random parts OK" in out
