@@ -36,13 +36,15 @@ AND_REGEX_FILTER = (
 )
 
 WORKLOADS = {
+    # names match BASELINE.json configs[1]/[2] verbatim at the default 100M
     "phrase": {
-        "name": "{rows} rows, phrase filter on _msg column "
+        "name": "{rows} rows, phrase filter on _msg column, 1xMI355X "
                 "(bloom + substring kernel)",
         "filter": PHRASE_FILTER,
     },
     "phrase_regex": {
-        "name": "{rows} rows, AND(phrase, regex) on two string columns",
+        "name": "{rows} rows, AND(phrase, re2 regex) on two string columns, "
+                "1xMI355X",
         "filter": AND_REGEX_FILTER,
     },
     "dict_time": {
