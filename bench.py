@@ -47,8 +47,21 @@ WORKLOADS = {
                 "1xMI355X",
         "filter": AND_REGEX_FILTER,
     },
+    # configs[3] shape (the driver shards it over N GPUs weak-scaling)
+    "or8": {
+        "name": "{rows} rows, OR of 8 phrase filters over 4 columns",
+        "filter": '{"type":"or","filters":['
+                  '{"type":"phrase","field":"_msg","phrase":"worker 3"},'
+                  '{"type":"phrase","field":"_msg","phrase":"worker 5"},'
+                  '{"type":"phrase","field":"dict_0","phrase":"error"},'
+                  '{"type":"phrase","field":"dict_0","phrase":"fatal"},'
+                  '{"type":"phrase","field":"dict_1","phrase":"warn"},'
+                  '{"type":"phrase","field":"dict_1","phrase":"debug"},'
+                  '{"type":"phrase","field":"host","phrase":"host_0"},'
+                  '{"type":"phrase","field":"host","phrase":"host_9"}]}',
+    },
     "dict_time": {
-        "name": "dict column + timestamp range",
+        "name": "dict column + timestamp range (configs[4] shape)",
         "filter": '{"type":"and","filters":['
                   '{"type":"phrase","field":"dict_0","phrase":"error"},'
                   '{"type":"time","min":1700000000000000000,'
