@@ -164,6 +164,12 @@ struct Stage {
   uint64_t staged_bytes = 0;   // bytes resident in HBM
   uint64_t algo_bytes = 0;     // algorithmic bytes one scan pass must read
   uint64_t rows = 0;
+  uint64_t live_rows = 0;      // rows of blocks that actually reach the kernel
+
+  // bitmap word ranges of statically-eliminated blocks (program value 0):
+  // zeroed once on the first scan instead of dispatching their chunks
+  std::vector<std::pair<uint64_t, uint64_t>> zero_word_ranges;
+  bool bitmap_zeroed = false;
 
   hipStream_t stream = nullptr;
   hipEvent_t ev0 = nullptr, ev1 = nullptr;
@@ -2305,6 +2311,43 @@ void stage_leaf(const LeafInfo& li, BlockStageCtx& ctx, Stage& st, DevLeafBlock&
   }
 }
 
+// 3-valued static evaluation of the filter program over one block's staged
+// leaf modes: 0 = all rows false, 1 = all rows true, 2 = needs a scan.
+// Blocks whose program value is statically 0 never reach the kernel: their
+// chunks are compacted out of the dispatch and their bitmap words zeroed
+// once (the reference's header prunes have the same effect — e.g.
+// filter_time.go:114-137 zeroes the bitmap without touching rows).  Bloom
+// gates can only turn a kModeScan leaf into 0 at run time, so they keep
+// value 2 here (bloom has no false negatives; the static result is sound).
+static int static_program_value(const std::vector<DevOp>& ops,
+                                const DevLeafBlock* lbs) {
+  std::vector<int> stack;
+  stack.reserve(kMaxStackDepth + 1);
+  for (const DevOp& op : ops) {
+    if (op.kind == kOpLeaf) {
+      const DevLeafBlock& lb = lbs[op.leaf];
+      stack.push_back(lb.mode == kModeNone ? 0 : lb.mode == kModeAll ? 1 : 2);
+    } else if (op.kind == kOpNot) {
+      int& v = stack.back();
+      if (v != 2) v = 1 - v;
+    } else {
+      const int n = op.nargs;
+      int acc = stack[stack.size() - n];
+      for (int k = 1; k < n; k++) {
+        const int v = stack[stack.size() - n + k];
+        if (op.kind == kOpAnd) {
+          acc = (acc == 0 || v == 0) ? 0 : (acc == 1 && v == 1) ? 1 : 2;
+        } else {
+          acc = (acc == 1 || v == 1) ? 1 : (acc == 0 && v == 0) ? 0 : 2;
+        }
+      }
+      stack.resize(stack.size() - n);
+      stack.push_back(acc);
+    }
+  }
+  return stack.back();
+}
+
 Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
                         std::shared_ptr<VqlFilter> filter, int device) {
   if (refs.empty()) fail("build_stage: no blocks");
@@ -2445,6 +2488,12 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
     db.bitmap_out = st->d_bitmap + st->block_word_off[size_t(b)];
     db.hits_out = nullptr;  // filled after d_block_hits is allocated
 
+    // watermark: if the block is statically eliminated below, its staged
+    // column data is reclaimed (bump-allocator rollback)
+    const size_t arena_mark = st->arena_used;
+    const uint64_t staged_mark = st->staged_bytes;
+    uint64_t block_algo = 0;
+
     for (int l = 0; l < nleaves; l++) {
       DevLeafBlock& lb = lbs_h[size_t(b) * size_t(nleaves) + size_t(l)];
       stage_leaf(leaf_infos[size_t(l)], ctx, *st, lb);
@@ -2467,29 +2516,49 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
           case kScanAnyCasePhraseStr:
           case kScanAnyCasePrefixStr: {
             const StagedStrCol& sc = ctx.cols.at(leaf_infos[size_t(l)].cname);
-            st->algo_bytes += sc.data_bytes + (bh.rows_count + 1) * 4;
+            block_algo += sc.data_bytes + (bh.rows_count + 1) * 4;
             break;
           }
           case kScanDict:
-            st->algo_bytes += bh.rows_count;
+            block_algo += bh.rows_count;
             break;
           case kScanTsRange:
-            st->algo_bytes += bh.rows_count * 8;
+            block_algo += bh.rows_count * 8;
             break;
           default:
-            st->algo_bytes += bh.rows_count * (lb.width ? lb.width : 8);
+            block_algo += bh.rows_count * (lb.width ? lb.width : 8);
             break;
         }
       }
     }
+
+    if (static_program_value(filter->ops,
+                             &lbs_h[size_t(b) * size_t(nleaves)]) == 0) {
+      // statically all-zero result: no chunks dispatched, bitmap words zeroed
+      // once at the first scan, staged column data reclaimed
+      st->arena_used = arena_mark;
+      st->staged_bytes = staged_mark;
+      const uint64_t w0 = st->block_word_off[size_t(b)];
+      const uint64_t w1 = st->block_word_off[size_t(b) + 1];
+      if (!st->zero_word_ranges.empty() &&
+          st->zero_word_ranges.back().second == w0) {
+        st->zero_word_ranges.back().second = w1;
+      } else {
+        st->zero_word_ranges.emplace_back(w0, w1);
+      }
+      continue;
+    }
+    st->algo_bytes += block_algo;
+    st->live_rows += bh.rows_count;
+    // bitmap write traffic for this (live) block
+    st->algo_bytes +=
+        (st->block_word_off[size_t(b) + 1] - st->block_word_off[size_t(b)]) * 8;
     // chunks
     uint32_t nch = uint32_t((bh.rows_count + kChunkRows - 1) / kChunkRows);
     for (uint32_t c = 0; c < nch; c++) {
       chunks_h.push_back(DevChunk{uint32_t(b), c});
     }
   }
-  // bitmap write traffic
-  st->algo_bytes += st->bitmap_words * 8;
 
   st->nchunks = uint32_t(chunks_h.size());
   st->chunks_h = chunks_h;
@@ -2510,9 +2579,12 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
   HIP_CHECK(hipMalloc(&st->d_blocks, blocks_h.size() * sizeof(DevBlock)));
   HIP_CHECK(hipMemcpy(st->d_blocks, blocks_h.data(),
                       blocks_h.size() * sizeof(DevBlock), hipMemcpyHostToDevice));
-  HIP_CHECK(hipMalloc(&st->d_chunks, chunks_h.size() * sizeof(DevChunk)));
-  HIP_CHECK(hipMemcpy(st->d_chunks, chunks_h.data(),
-                      chunks_h.size() * sizeof(DevChunk), hipMemcpyHostToDevice));
+  if (!chunks_h.empty()) {
+    HIP_CHECK(hipMalloc(&st->d_chunks, chunks_h.size() * sizeof(DevChunk)));
+    HIP_CHECK(hipMemcpy(st->d_chunks, chunks_h.data(),
+                        chunks_h.size() * sizeof(DevChunk),
+                        hipMemcpyHostToDevice));
+  }
   HIP_CHECK(hipMalloc(&st->d_hits, 8));
 
   return st.release();
@@ -2536,6 +2608,16 @@ long long run_scan(Stage* st) {
   HIP_CHECK(hipMemsetAsync(st->d_hits, 0, 8, st->stream));
   HIP_CHECK(hipMemsetAsync(st->d_block_hits, 0,
                            st->block_refs.size() * 8, st->stream));
+  if (!st->bitmap_zeroed) {
+    // statically-eliminated blocks never reach the kernel; their bitmap
+    // words are zeroed once (they stay zero — live blocks rewrite their own
+    // words every scan)
+    for (const auto& r : st->zero_word_ranges) {
+      HIP_CHECK(hipMemsetAsync(st->d_bitmap + r.first, 0,
+                               (r.second - r.first) * 8, st->stream));
+    }
+    st->bitmap_zeroed = true;
+  }
   HIP_CHECK(hipEventRecord(st->ev0, st->stream));
   HIP_CHECK(vql_launch_scan(st->d_ops, int(st->filter->ops.size()), st->d_lbs,
                             int(st->filter->leaves.size()), st->d_blocks,
