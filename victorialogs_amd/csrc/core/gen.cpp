@@ -19,6 +19,14 @@ uint64_t generate_part(const std::string& dir, const GenConfig& cfg) {
   std::mt19937_64 rng(cfg.seed);
   PartWriter w(dir, 1);
   uint64_t msg_bytes = 0;
+  // Selective-filter invariants recorded at generation time with plain
+  // std::string::find / value equality — independent of every scan path
+  // (oracle, emu, HIP), so a full-size bench assertion against these counts
+  // fails for any kernel that fabricates bitmaps.  "ip=77." can only occur
+  // in the "; ip=a.b.c.d;" fragment (uuid is hex, pad is lowercase), and as
+  // a phrase it needs no trailing token boundary ('.' is a non-token char),
+  // so find() count == phrase-match count.
+  uint64_t sel_msg_ip77 = 0, sel_dict0_error = 0;
 
   uint64_t rows_per_stream = cfg.rows / cfg.streams;
   uint64_t extra = cfg.rows % cfg.streams;
@@ -97,6 +105,7 @@ uint64_t generate_part(const std::string& dir, const GenConfig& cfg) {
           m += "; pad=";
           while (m.size() < cfg.msg_len) m += char('a' + (rng() % 26));
         }
+        if (m.find("ip=77.") != std::string::npos) sel_msg_ip77++;
         msg_bytes += m.size();
         msg.values.push_back(std::move(m));
 
@@ -112,7 +121,9 @@ uint64_t generate_part(const std::string& dir, const GenConfig& cfg) {
                                     std::to_string(rng()));
         }
         for (int j = 0; j < cfg.num_dict_fields; j++) {
-          dictf[j]->values.push_back(kDictValues[rng() % 8]);
+          const char* dv = kDictValues[rng() % 8];
+          if (j == 0 && dv[0] == 'e') sel_dict0_error++;  // exactly "error"
+          dictf[j]->values.push_back(dv);
         }
         if (cfg.extra_typed_fields) {
           u8c->values.push_back(std::to_string(uint8_t(rng())));
@@ -149,6 +160,19 @@ uint64_t generate_part(const std::string& dir, const GenConfig& cfg) {
     }
   }
   w.finish();
+  // manifest last: doubles as a generation-complete marker for cached dirs
+  {
+    std::string path = dir + "/gen_manifest.json";
+    FILE* f = fopen(path.c_str(), "w");
+    if (f) {
+      fprintf(f,
+              "{\"rows\": %" PRIu64 ", \"msg_bytes\": %" PRIu64
+              ", \"sel_msg_ip77\": %" PRIu64 ", \"sel_dict0_error\": %" PRIu64
+              "}\n",
+              cfg.rows, msg_bytes, sel_msg_ip77, sel_dict0_error);
+      fclose(f);
+    }
+  }
   return msg_bytes;
 }
 

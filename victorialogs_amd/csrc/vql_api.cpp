@@ -147,8 +147,18 @@ struct Stage {
   // without caring which part they came from, storage_search.go:1035-1067)
   std::vector<std::pair<VqlPart*, long>> block_refs;
 
-  uint8_t* arena = nullptr;
-  size_t arena_cap = 0, arena_used = 0;
+  // Slab arena: device memory grows with what is actually staged (the whole
+  // part's uncompressed size is NOT a usable bound once only the filter's
+  // columns are staged — a 1B-row part holds ~350 GB of columns of which the
+  // or8 config stages ~265 GB into the 288 GB of HBM3E).  Slabs are never
+  // reallocated, so staged pointers stay valid; every in-slab allocation
+  // keeps >= kSlabTailPad bytes of owned slab memory after it (the string
+  // tile copy rounds groups up to 1 KiB strides and may over-read).
+  static constexpr size_t kSlabBytes = size_t(1) << 30;
+  static constexpr size_t kSlabTailPad = 4096;
+  std::vector<std::pair<uint8_t*, size_t>> slabs;  // (ptr, capacity)
+  uint8_t* cur_slab = nullptr;
+  size_t cur_cap = 0, cur_used = 0;
 
   DevOp* d_ops = nullptr;
   DevLeafBlock* d_lbs = nullptr;
@@ -194,7 +204,7 @@ struct Stage {
     if (ev0) (void)hipEventDestroy(ev0);
     if (ev1) (void)hipEventDestroy(ev1);
     if (stream) (void)hipStreamDestroy(stream);
-    if (arena) (void)hipFree(arena);
+    for (auto& s : slabs) (void)hipFree(s.first);
     if (d_ops) (void)hipFree(d_ops);
     if (d_lbs) (void)hipFree(d_lbs);
     if (d_blocks) (void)hipFree(d_blocks);
@@ -204,22 +214,47 @@ struct Stage {
   }
 
   uint8_t* push(const void* src, size_t n, size_t align = 16) {
-    arena_used = (arena_used + align - 1) & ~(align - 1);
-    if (arena_used + n > arena_cap) {
-      fail("staging arena exhausted; increase the arena estimate");
-    }
-    uint8_t* dst = arena + arena_used;
+    uint8_t* dst = reserve(n, align);
     if (n) HIP_CHECK(hipMemcpy(dst, src, n, hipMemcpyHostToDevice));
-    arena_used += n;
     staged_bytes += n;
     return dst;
   }
   uint8_t* reserve(size_t n, size_t align = 16) {
-    arena_used = (arena_used + align - 1) & ~(align - 1);
-    if (arena_used + n > arena_cap) fail("staging arena exhausted");
-    uint8_t* dst = arena + arena_used;
-    arena_used += n;
+    size_t used = (cur_used + align - 1) & ~(align - 1);
+    if (cur_slab == nullptr || used + n + kSlabTailPad > cur_cap) {
+      const size_t cap = std::max(kSlabBytes, n + kSlabTailPad + align);
+      uint8_t* p = nullptr;
+      HIP_CHECK(hipMalloc(&p, cap));
+      slabs.emplace_back(p, cap);
+      cur_slab = p;
+      cur_cap = cap;
+      cur_used = 0;
+      used = 0;
+    }
+    uint8_t* dst = cur_slab + used;
+    cur_used = used + n;
     return dst;
+  }
+
+  // bump-allocator watermark for statically-eliminated block rollback
+  struct ArenaMark {
+    size_t nslabs;
+    uint8_t* cur_slab;
+    size_t cur_cap, cur_used;
+    uint64_t staged_bytes;
+  };
+  ArenaMark mark() const {
+    return {slabs.size(), cur_slab, cur_cap, cur_used, staged_bytes};
+  }
+  void rollback(const ArenaMark& m) {
+    while (slabs.size() > m.nslabs) {
+      (void)hipFree(slabs.back().first);
+      slabs.pop_back();
+    }
+    cur_slab = m.cur_slab;
+    cur_cap = m.cur_cap;
+    cur_used = m.cur_used;
+    staged_bytes = m.staged_bytes;
   }
 };
 
@@ -2371,22 +2406,12 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
         st->block_refs[size_t(i)].second)];
   };
 
-  // Arena estimate: decoded column data is bounded by the part's uncompressed
-  // size; add offsets (4 B/row/string-col), timestamps (8 B/row), blooms and
-  // the output bitmaps, with slack.
+  // Device memory comes from the slab arena and grows with what is actually
+  // staged (only the filter's columns), so a 1B-row part whose full column
+  // set exceeds HBM still stages when the scanned columns fit.
   uint64_t rows = 0;
-  uint64_t est = 64 << 20;
-  for (long b = 0; b < nblocks; b++) {
-    const BlockHeader& bh = bh_of(b);
-    rows += bh.rows_count;
-    // slack per block covers blooms + offsets + bitmap (measured ~200 KB
-    // for 8192-row blocks; 2 MB/block over-allocated ~24 GB on the 100M-row
-    // multi-part stage)
-    est += bh.uncompressed_size_bytes + bh.rows_count * 24 + (512 << 10);
-  }
+  for (long b = 0; b < nblocks; b++) rows += bh_of(b).rows_count;
   st->rows = rows;
-  st->arena_cap = size_t(est);
-  HIP_CHECK(hipMalloc(&st->arena, st->arena_cap));
 
   // leaf operands + probe hashes (block-independent)
   std::vector<LeafInfo> leaf_infos{filter->leaves.size()};
@@ -2490,8 +2515,7 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
 
     // watermark: if the block is statically eliminated below, its staged
     // column data is reclaimed (bump-allocator rollback)
-    const size_t arena_mark = st->arena_used;
-    const uint64_t staged_mark = st->staged_bytes;
+    const Stage::ArenaMark arena_mark = st->mark();
     uint64_t block_algo = 0;
 
     for (int l = 0; l < nleaves; l++) {
@@ -2536,8 +2560,7 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
                              &lbs_h[size_t(b) * size_t(nleaves)]) == 0) {
       // statically all-zero result: no chunks dispatched, bitmap words zeroed
       // once at the first scan, staged column data reclaimed
-      st->arena_used = arena_mark;
-      st->staged_bytes = staged_mark;
+      st->rollback(arena_mark);
       const uint64_t w0 = st->block_word_off[size_t(b)];
       const uint64_t w1 = st->block_word_off[size_t(b) + 1];
       if (!st->zero_word_ranges.empty() &&
