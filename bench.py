@@ -1,15 +1,20 @@
 #!/usr/bin/env python3
 """Benchmark for the MI355X-native VictoriaLogs block-scan engine.
 
-Workload (BASELINE.json configs[1], the largest single-GPU config the metric
-is quoted on): 100M vlogsgenerator-shaped rows with 256-byte _msg, phrase
-filter on _msg — the per-block filter evaluation of lib/logstorage
-(blockSearch.search, block_search.go:207-226) rebuilt as HIP kernels.
+Headline workload (BASELINE.json configs[2] — the config the metric is
+quoted on): 100M vlogsgenerator-shaped rows with 256-byte _msg,
+AND(phrase on _msg, re2 regex on var_0) — the per-block filter evaluation
+of lib/logstorage (blockSearch.search, block_search.go:207-226) rebuilt as
+HIP kernels.  configs[1] (phrase-only) runs via --workload phrase.
 
 A "step" is one pass of the hot path over the staged dataset: the filter
-program is evaluated over every block's rows (one kernel launch per staged
-part), with inputs already resident in HBM.  value = whole-job rows
-scanned/sec across all ranks (weak scaling: each rank owns its own shard).
+program is evaluated over every block's rows in ONE kernel launch, with
+inputs already resident in HBM.  value = whole-job matched rows/sec across
+all ranks (the headline filters match every generated row by construction,
+so matched == scanned there; both fields are reported).  The bench is
+self-checking at full size: a selective phrase whose expected count was
+recorded by the data generator (gen_manifest.json, computed with plain
+string find — independent of every scan path) must match exactly.
 
 Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--rows R]
 For N>1 the driver launches this under torch.distributed.run, one rank per
@@ -34,6 +39,9 @@ AND_REGEX_FILTER = (
     '{"type":"phrase","field":"_msg","phrase":"message for the stream"},'
     '{"type":"regexp","field":"var_0","re":"some value"}]}'
 )
+# selective phrase: matches rows whose random ip has first octet 77 (~1/256);
+# the generator records the exact count per part in gen_manifest.json
+SELECTIVE_FILTER = '{"type":"phrase","field":"_msg","phrase":"ip=77."}'
 
 WORKLOADS = {
     # names match BASELINE.json configs[1]/[2] verbatim at the default 100M
@@ -41,11 +49,19 @@ WORKLOADS = {
         "name": "{rows} rows, phrase filter on _msg column, 1xMI355X "
                 "(bloom + substring kernel)",
         "filter": PHRASE_FILTER,
+        "expect_all": True,
     },
     "phrase_regex": {
         "name": "{rows} rows, AND(phrase, re2 regex) on two string columns, "
                 "1xMI355X",
         "filter": AND_REGEX_FILTER,
+        "expect_all": True,
+    },
+    "phrase_selective": {
+        "name": "{rows} rows, selective phrase (ip=77., ~1/256 of rows) on "
+                "_msg column",
+        "filter": SELECTIVE_FILTER,
+        "expect_manifest": "sel_msg_ip77",
     },
     # configs[3] shape (the driver shards it over N GPUs weak-scaling)
     "or8": {
@@ -75,7 +91,8 @@ def log(msg):
 
 
 def prepare_parts(data_dir, total_rows, nparts, msg_len, seed_base):
-    """Generates nparts reference-format parts in parallel (cached on disk)."""
+    """Generates nparts reference-format parts in parallel (cached on disk;
+    gen_manifest.json is written last and doubles as a completion marker)."""
     from victorialogs_amd import generate_part
 
     os.makedirs(data_dir, exist_ok=True)
@@ -86,7 +103,7 @@ def prepare_parts(data_dir, total_rows, nparts, msg_len, seed_base):
         rows = rows_per + (total_rows % nparts if i == nparts - 1 else 0)
         d = os.path.join(data_dir, f"part_{rows}_{msg_len}_{seed_base + i}")
         dirs.append(d)
-        if not os.path.exists(os.path.join(d, "metadata.json")):
+        if not os.path.exists(os.path.join(d, "gen_manifest.json")):
             jobs.append((d, rows, seed_base + i))
     if jobs:
         t0 = time.time()
@@ -100,6 +117,39 @@ def prepare_parts(data_dir, total_rows, nparts, msg_len, seed_base):
                 f.result()
         log(f"generation took {time.time() - t0:.1f}s")
     return dirs
+
+
+def read_manifests(part_dirs):
+    """Sums the generator-recorded selective counts over this rank's parts."""
+    totals = {}
+    for d in part_dirs:
+        p = os.path.join(d, "gen_manifest.json")
+        if not os.path.exists(p):
+            return None
+        with open(p) as f:
+            m = json.load(f)
+        for k, v in m.items():
+            totals[k] = totals.get(k, 0) + v
+    return totals
+
+
+def load_pmc_traffic(workload):
+    """PMC-counter HBM traffic per launch for this workload, from the
+    committed rocprofv3 collection (separate --pmc passes; gfx950 FETCH_SIZE
+    correction applied per MI355X_MICROARCH.md).  None if not collected."""
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "profiles", "r02", "pmc_traffic.json")
+    try:
+        with open(path) as f:
+            data = json.load(f)
+        ent = data.get(workload)
+        if not ent:
+            return None, None
+        traffic = ent.get("fetch_bytes_per_launch", 0) + ent.get(
+            "write_bytes_per_launch", 0)
+        return traffic, ent.get("source")
+    except (OSError, ValueError):
+        return None, None
 
 
 def cpu_baseline(part_dirs, filter_json, budget_s=20.0):
@@ -144,10 +194,12 @@ def main():
     ap.add_argument("--rows", type=int,
                     default=int(os.environ.get("VQL_BENCH_ROWS", 100_000_000)))
     ap.add_argument("--msg-len", type=int, default=256)
-    ap.add_argument("--workload", choices=sorted(WORKLOADS), default="phrase")
+    ap.add_argument("--workload", choices=sorted(WORKLOADS),
+                    default="phrase_regex")
     ap.add_argument("--data-dir", default=os.environ.get(
         "VQL_DATA_DIR", "/tmp/vql_bench_data"))
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--skip-selectivity", action="store_true")
     args = ap.parse_args()
 
     import torch
@@ -157,9 +209,11 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     distributed = world > 1
+    ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    device = local_rank % ndev if ndev else 0
     if distributed:
         dist.init_process_group("nccl")
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(device)
 
     from victorialogs_amd import Filter, Part, Stage
 
@@ -168,22 +222,26 @@ def main():
 
     # Each rank generates and stages its own shard (weak scaling; blocks are
     # independent units, storage_search.go:1035-1067).
-    nparts = max(1, min(8, (os.cpu_count() or 8) // max(1, world)))
+    nparts = max(1, min(64, (os.cpu_count() or 8) // max(1, world),
+                        max(8, args.rows // 16_000_000)))
     part_dirs = prepare_parts(
         os.path.join(args.data_dir, f"rank{rank}"), args.rows, nparts,
         args.msg_len, seed_base=1000 * rank + 1)
+    manifest = read_manifests(part_dirs)
 
-    log(f"rank {rank}: staging {len(part_dirs)} parts onto device {local_rank}")
+    log(f"rank {rank}: staging {len(part_dirs)} parts onto device {device}")
     t0 = time.time()
     parts = [Part(d) for d in part_dirs]
     filt = Filter(filter_json)
     # one multi-part stage: the whole pass is ONE kernel launch
-    stages = [Stage(parts, filt, device=local_rank)]
+    stages = [Stage(parts, filt, device=device)]
+    stage_s = time.time() - t0
     staged_bytes = sum(s.staged_bytes for s in stages)
     algo_bytes = sum(s.algo_bytes for s in stages)
     rows = sum(s.rows for s in stages)
+    live_rows = sum(s.live_rows for s in stages)
     log(f"rank {rank}: staged {staged_bytes / 1e9:.2f} GB "
-        f"({rows} rows) in {time.time() - t0:.1f}s")
+        f"({rows} rows, {live_rows} live) in {stage_s:.1f}s")
 
     def one_step():
         hits = 0
@@ -193,16 +251,24 @@ def main():
             kms += s.last_kernel_ms
         return hits, kms
 
-    # warmup + full-size correctness property: the phrase workload's filter
-    # matches every generated row by construction, so the device count must
-    # equal the staged row count (a size-independent invariant at the full
-    # 100M-row config; parity at oracle-checkable sizes lives in tests/)
-    for _ in range(args.warmup):
+    # warmup + full-size correctness property (all-match invariant): the
+    # headline filters match every generated row by construction, so the
+    # device count must equal the staged row count
+    hits = 0
+    for _ in range(max(args.warmup, 1)):
         hits, _ = one_step()
-    if args.workload == "phrase" and hits != rows:
+    if wl.get("expect_all") and hits != rows:
         raise SystemExit(
-            f"correctness check failed: phrase workload matched {hits} of "
-            f"{rows} rows (expected all)")
+            f"correctness check failed: {args.workload} workload matched "
+            f"{hits} of {rows} rows (expected all)")
+    exp_key = wl.get("expect_manifest")
+    if exp_key is not None:
+        if manifest is None:
+            raise SystemExit("no gen_manifest.json for selective workload")
+        if hits != manifest[exp_key]:
+            raise SystemExit(
+                f"correctness check failed: {args.workload} matched {hits}, "
+                f"generator recorded {manifest[exp_key]}")
 
     # timed region: barrier + sync on both sides, MAX over ranks
     if distributed:
@@ -222,11 +288,12 @@ def main():
 
     # whole-job aggregation: MAX(elapsed) over ranks; SUM(rows, hits)
     if distributed:
-        t = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dev = "cuda" if ndev else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
         c = torch.tensor([float(rows), float(hits), float(algo_bytes)],
-                         dtype=torch.float64, device="cuda")
+                         dtype=torch.float64, device=dev)
         dist.all_reduce(c, op=dist.ReduceOp.SUM)
         total_rows, total_hits, total_algo = (int(c[0].item()), int(c[1].item()),
                                               int(c[2].item()))
@@ -236,22 +303,65 @@ def main():
     ms_per_step = elapsed * 1000.0 / args.steps
     rows_per_s = total_rows * args.steps / elapsed
     gb_per_s = total_algo * args.steps / elapsed / 1e9
+    matched_per_s = total_hits * args.steps / elapsed
 
     # roofline for the dominant (only) kernel, HIP-event timed per launch on
     # its own stream; algorithmic bytes per launch / avg launch duration.
     avg_kernel_s = (kernel_ms_total / 1000.0) / (args.steps * max(1, len(stages)))
     algo_per_launch = algo_bytes / max(1, len(stages))
     achieved = algo_per_launch / avg_kernel_s if avg_kernel_s > 0 else 0.0
+    traffic, traffic_src = load_pmc_traffic(args.workload)
     roofline = {
         "bound": "hbm",
         "achieved": achieved,
         "peak": HBM_PEAK_BYTES_PER_S,
         "unit": "B/s",
         "frac": achieved / HBM_PEAK_BYTES_PER_S,
-        "traffic": None,  # PMC counters come from the committed rocprofv3 runs
+        # PMC-counter HBM bytes per launch from the committed rocprofv3
+        # collection on this workload's single-launch kernel (separate --pmc
+        # passes cannot run inside the timed region)
+        "traffic": traffic,
+        "traffic_source": traffic_src,
         "kernel": "scan_program_kernel",
         "avg_kernel_ms": avg_kernel_s * 1000.0,
     }
+
+    # full-size selectivity self-check: a filter with a
+    # generator-known expected count != rows must match it exactly — a
+    # kernel writing all-ones bitmaps fails here
+    sel = None
+    if (not args.skip_selectivity and manifest is not None
+            and args.workload in ("phrase", "phrase_regex")):
+        log("selectivity self-check (phrase ip=77.)...")
+        sfilt = Filter(SELECTIVE_FILTER)
+        sstage = Stage(parts, sfilt, device=device)
+        got = sstage.scan()
+        t0 = time.perf_counter()
+        sel_steps = 3
+        for _ in range(sel_steps):
+            got = sstage.scan()
+        sel_elapsed = time.perf_counter() - t0
+        expected = manifest["sel_msg_ip77"]
+        sstage.close()
+        sfilt.close()
+        if distributed:
+            dev = "cuda" if ndev else "cpu"
+            c = torch.tensor([float(expected), float(got)],
+                             dtype=torch.float64, device=dev)
+            dist.all_reduce(c, op=dist.ReduceOp.SUM)
+            expected, got = int(c[0].item()), int(c[1].item())
+        if got != expected:
+            raise SystemExit(
+                f"selectivity check FAILED: matched {got}, generator "
+                f"recorded {expected}")
+        sel = {
+            "filter": SELECTIVE_FILTER,
+            "expected_matches": expected,
+            "got_matches": got,
+            "ok": True,
+        }
+        if world == 1:
+            sel["scanned_rows_per_sec"] = rows * sel_steps / sel_elapsed
 
     result = None
     if rank == 0:
@@ -261,7 +371,7 @@ def main():
             cb = cpu_baseline(part_dirs, filter_json)
         result = {
             "metric": "matched-rows/sec",
-            "value": rows_per_s,
+            "value": matched_per_s,
             "unit": "rows/s",
             "n_gpus": world,
             "steps": args.steps,
@@ -277,14 +387,28 @@ def main():
                     rows=f"{args.rows // 10**6}M" if args.rows >= 10**6
                     else str(args.rows)),
                 "rows_per_gpu": rows,
+                "live_rows_per_gpu": live_rows,
                 "msg_len": args.msg_len,
                 "filter": filter_json,
                 "matched_rows_per_pass": total_hits,
                 "parallelism": f"dp{world}",
             },
+            # matched == scanned for the all-match headline filters; they
+            # diverge on selective workloads (report both)
+            "scanned_rows_per_sec": rows_per_s,
+            "matched_rows_per_sec": matched_per_s,
             "gb_scanned_per_sec": gb_per_s,
             "hbm_resident_bytes": staged_bytes * world,
             "roofline": roofline,
+            # end-to-end (cold) path: host zstd decode + H2D staging + one
+            # scan (§8d "clearly separated"; the reference re-decompresses
+            # values blocks per scan, block_search.go:444-474)
+            "cold": {
+                "stage_s": stage_s,
+                "staged_gb": staged_bytes / 1e9,
+                "rows_per_s_incl_staging": rows / (stage_s + ms_per_step / 1e3),
+            },
+            "selectivity_check": sel,
             "cpu_baseline": cb,
         }
         print(json.dumps(result), flush=True)

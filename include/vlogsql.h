@@ -61,6 +61,10 @@ void vql_stage_free(void* stage);
 long long vql_stage_bytes(void* stage);      /* bytes resident in HBM */
 long long vql_stage_algo_bytes(void* stage); /* algorithmic bytes per pass */
 long long vql_stage_rows(void* stage);
+/* Rows of blocks that actually reach the kernel (blocks whose filter
+ * program is statically all-zero are pruned from the dispatch, mirroring
+ * the reference's header prunes, e.g. filter_time.go:114-137). */
+long long vql_stage_live_rows(void* stage);
 
 /* One scan pass over the staged blocks (one kernel launch).  Returns the
  * number of matched rows; bitmaps stay device-resident. */

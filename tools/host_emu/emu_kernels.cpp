@@ -22,10 +22,16 @@ using std::trunc;
 #include <hip/hip_runtime.h>
 
 namespace vl {
+int64_t g_vl_local_tz_nsecs = 0;
 #include "../../victorialogs_amd/csrc/hip/scan_rowops.h"
 }  // namespace vl
 
 using namespace vl;
+
+extern "C" int vql_set_local_tz_nsecs(long long v) {
+  vl::g_vl_local_tz_nsecs = v;
+  return 0;
+}
 
 namespace {
 

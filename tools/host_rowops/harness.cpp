@@ -22,8 +22,12 @@ using std::isnan;
 #include "../../victorialogs_amd/csrc/hip/scan_types.h"
 
 namespace vl {
+int64_t g_vl_local_tz_nsecs = 0;
 #include "../../victorialogs_amd/csrc/hip/scan_rowops.h"
 }  // namespace vl
+
+// keep the device-code mirror's local-tz behavior identical to the oracle's
+__attribute__((constructor)) static void rowops_tz_init();
 
 #include "../../victorialogs_amd/csrc/core/match.h"
 #include "../../victorialogs_amd/csrc/core/op_serialize.h"
@@ -31,6 +35,10 @@ namespace vl {
 #include "../../victorialogs_amd/csrc/core/values.h"
 
 using namespace vl;
+
+__attribute__((constructor)) static void rowops_tz_init() {
+  vl::g_vl_local_tz_nsecs = vl::local_tz_offset_nsecs();
+}
 
 extern "C" {
 

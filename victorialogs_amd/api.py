@@ -60,6 +60,8 @@ def load_product() -> ctypes.CDLL:
     lib.vql_stage_algo_bytes.argtypes = [ctypes.c_void_p]
     lib.vql_stage_rows.restype = ctypes.c_longlong
     lib.vql_stage_rows.argtypes = [ctypes.c_void_p]
+    lib.vql_stage_live_rows.restype = ctypes.c_longlong
+    lib.vql_stage_live_rows.argtypes = [ctypes.c_void_p]
     lib.vql_scan_staged.restype = ctypes.c_longlong
     lib.vql_scan_staged.argtypes = [ctypes.c_void_p]
     lib.vql_last_kernel_ms.restype = ctypes.c_double
@@ -294,6 +296,12 @@ class Stage:
     @property
     def rows(self):
         return self.lib.vql_stage_rows(self.h)
+
+    @property
+    def live_rows(self):
+        """Rows of blocks that reach the kernel (statically-pruned blocks
+        are compacted out of the dispatch)."""
+        return self.lib.vql_stage_live_rows(self.h)
 
     def scan(self):
         hits = self.lib.vql_scan_staged(self.h)

@@ -5,6 +5,7 @@
 #include <cmath>
 #include <cstdio>
 #include <cstdlib>
+#include <ctime>
 
 namespace vl {
 
@@ -373,8 +374,23 @@ bool try_parse_bytes(strview s, int64_t* out) {
   return true;
 }
 
-// parseTimezoneOffset + tryParseHHMM (values_encoder.go:383-426); local
-// timezone (no suffix) treated as UTC (both runtime containers are UTC)
+// GetLocalTimezoneOffsetNsecs (vendor/.../lib/timeutil/timezone.go:9-19):
+// the reference samples time.Now().Zone() — the host local offset of the
+// CURRENT time, not of the parsed timestamp — and caches it.  Same here.
+int64_t local_tz_offset_nsecs() {
+  static const int64_t cached = [] {
+    tzset();
+    time_t now = time(nullptr);
+    struct tm tmv;
+    if (localtime_r(&now, &tmv) == nullptr) return int64_t(0);
+    return int64_t(tmv.tm_gmtoff) * int64_t(1000000000);
+  }();
+  return cached;
+}
+
+// parseTimezoneOffset + tryParseHHMM (values_encoder.go:383-426); inputs
+// without a timezone suffix use the host local timezone offset, as the
+// reference does
 static bool parse_tz_offset(strview s, int64_t* off, strview* prefix) {
   if (s.n > 0 && s.p[s.n - 1] == 'Z') {
     *off = 0;
@@ -389,7 +405,7 @@ static bool parse_tz_offset(strview s, int64_t* off, strview* prefix) {
     }
   }
   if (n < 0) {
-    *off = 0;  // GetLocalTimezoneOffsetNsecs; UTC here
+    *off = local_tz_offset_nsecs();
     *prefix = s;
     return true;
   }

@@ -38,8 +38,13 @@ bool try_parse_ipv4(strview s, uint32_t* out);                // :675-730
 bool try_parse_timestamp_iso8601(strview s, int64_t* out);    // :428-466
 bool try_parse_duration(strview s, int64_t* out);             // :990-1061
 // TryParseTimestampRFC3339Nano (values_encoder.go:340-381); no-timezone
-// inputs use a zero local offset (both runtime boxes are UTC; documented)
+// inputs use the host local timezone offset, matching the reference's
+// GetLocalTimezoneOffsetNsecs (sampled from the current time, cached)
 bool try_parse_timestamp_rfc3339(strview s, int64_t* out);
+
+// Host local timezone offset of the current time, in nanoseconds (cached;
+// vendor/.../lib/timeutil/timezone.go:9-19 semantics)
+int64_t local_tz_offset_nsecs();
 // leValuesString (filter_le_field.go:284-299): numeric compare when both
 // sides parse via parseMathNumber, else bytewise
 bool le_values_string(strview a, strview b, bool exclude_equal);

@@ -36,8 +36,18 @@ constexpr uint32_t kWaveTileBytes = 17408;  // 1088 slots: 64 rows x <=272 B,
 // 64-slot strides (no single-lane straggler load per group)
 constexpr uint32_t kNumWaves = 4;
 
+// host local timezone offset for no-suffix RFC3339 parses (see scan_rowops.h)
+__device__ int64_t g_vl_local_tz_nsecs = 0;
 
 #include "scan_rowops.h"
+
+extern "C" int vql_set_local_tz_nsecs(long long v) {
+  int64_t x = v;
+  return hipMemcpyToSymbol(HIP_SYMBOL(g_vl_local_tz_nsecs), &x, sizeof(x)) ==
+                 hipSuccess
+             ? 0
+             : -1;
+}
 
 
 

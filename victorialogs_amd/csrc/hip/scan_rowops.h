@@ -12,6 +12,13 @@
 // Non-ASCII token-rune ranges (tokenizer.go:142-148); see unicode_ranges.inc.
 #include "../core/unicode_ranges.inc"
 
+// Host local timezone offset (nsecs) for RFC3339 inputs without a timezone
+// suffix — the reference uses the process-local zone of the current time
+// (timeutil.GetLocalTimezoneOffsetNsecs).  Defined by scan_kernels.hip
+// (device global, set via hipMemcpyToSymbol per device at stage build) and
+// by the host harness/emu builds (plain global).
+extern __device__ int64_t g_vl_local_tz_nsecs;
+
 __device__ __forceinline__ bool d_is_token_char(uint8_t c) {
   // tokenizer.go:132-140: [a-zA-Z0-9_]
   return (c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z') ||
@@ -702,6 +709,10 @@ __device__ bool d_parse_rfc3339(const A& a, long s0, long sn, int64_t* out) {
       if (a.u8(i + tz) == '-') off = -off;
       nsecs -= off;
       n = tz;
+    } else {
+      // no timezone suffix: host local offset, set once per device at stage
+      // build (GetLocalTimezoneOffsetNsecs semantics, timezone.go:9-19)
+      nsecs -= g_vl_local_tz_nsecs;
     }
   }
   if (n == 0) {

@@ -34,6 +34,7 @@
 #include "hip/scan_types.h"
 
 namespace vl {
+extern "C" int vql_set_local_tz_nsecs(long long);
 extern "C" hipError_t vql_launch_scan(const DevOp*, int, const DevLeafBlock*, int,
                                       const DevBlock*, const DevChunk*, uint32_t,
                                       unsigned long long*, hipStream_t);
@@ -2395,6 +2396,11 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
   st->block_refs = std::move(refs);
 
   HIP_CHECK(hipSetDevice(device));
+  // per-device: no-suffix RFC3339 parses use the host local timezone offset
+  // (the reference's GetLocalTimezoneOffsetNsecs, timezone.go:9-19)
+  if (vql_set_local_tz_nsecs(local_tz_offset_nsecs()) != 0) {
+    fail("vql_set_local_tz_nsecs failed");
+  }
   HIP_CHECK(hipStreamCreate(&st->stream));
   HIP_CHECK(hipEventCreate(&st->ev0));
   HIP_CHECK(hipEventCreate(&st->ev1));
@@ -2722,6 +2728,9 @@ long long vql_stage_algo_bytes(void* s) {
   return (long long)((Stage*)s)->algo_bytes;
 }
 long long vql_stage_rows(void* s) { return (long long)((Stage*)s)->rows; }
+long long vql_stage_live_rows(void* s) {
+  return (long long)((Stage*)s)->live_rows;
+}
 
 long long vql_scan_staged(void* s) {
   try {
