@@ -212,8 +212,11 @@ def main():
     ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
     device = local_rank % ndev if ndev else 0
     if distributed:
-        dist.init_process_group("nccl")
-        torch.cuda.set_device(device)
+        # RCCL over xGMI on GPU boxes; gloo lets the same distributed code
+        # paths run on CPU (world-2 CI test via the emu build)
+        dist.init_process_group("nccl" if ndev else "gloo")
+        if ndev:
+            torch.cuda.set_device(device)
 
     from victorialogs_amd import Filter, Part, Stage
 

@@ -251,3 +251,61 @@ def test_long_rows_global_fallback(tmp_path):
         '{"type":"regexp","field":"_msg","re":"end$"}',
     ]:
         assert_parity(d, f)
+
+
+def test_fetch_block_hits_vs_oracle(gen_part):
+    """vql_fetch_block_hits per-block counts == oracle per-block counts
+    (`| stats count()` fast path, block_result.go:403-413)."""
+    fjson = '{"type":"phrase","field":"_msg","phrase":"the stream 1 and"}'
+    part = Part(gen_part)
+    filt = Filter(fjson)
+    st = Stage(part, filt, device=0)
+    try:
+        hits = st.scan()
+        bh = st.fetch_block_hits(part.blocks)
+        orc = OracleScanner(gen_part)
+        try:
+            exp = [orc.scan(fjson, lo=i, hi=i + 1)[0]
+                   for i in range(part.blocks)]
+        finally:
+            orc.close()
+        assert bh == exp
+        assert sum(bh) == hits and hits > 0
+    finally:
+        st.close()
+        filt.close()
+        part.close()
+
+
+def test_static_elimination_parity(tmp_path):
+    """Blocks statically eliminated (timestamp header prune -> AND all-zero)
+    are compacted out of the dispatch; bitmaps and per-block counts must
+    stay bit-identical, including the zeroed pruned blocks."""
+    from victorialogs_amd import write_custom_part
+
+    blocks = []
+    for b in range(6):
+        base = 1700000000000000000 + b * 10**12
+        rows = 100 + b
+        blocks.append({
+            "stream": 0,
+            "timestamps": [base + i for i in range(rows)],
+            "columns": [{"name": "_msg",
+                         "values": [f"blk {b} row {i} tag_{i % 3}"
+                                    for i in range(rows)]}],
+        })
+    d = str(tmp_path / "elim")
+    write_custom_part(d, {"blocks": blocks})
+    # time range covers only blocks 2..3 -> blocks 0,1,4,5 statically zero
+    tmin = 1700000000000000000 + 2 * 10**12
+    tmax = 1700000000000000000 + 3 * 10**12 + 10**9
+    for f in [
+        '{"type":"and","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"tag_1"},'
+        f'{{"type":"time","min":{tmin},"max":{tmax}}}]}}',
+        f'{{"type":"time","min":{tmin},"max":{tmax}}}',
+        '{"type":"and","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"no_such_token"},'
+        '{"type":"phrase","field":"_msg","phrase":"tag_2"}]}',
+    ]:
+        assert_parity(d, f)
