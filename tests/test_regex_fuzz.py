@@ -87,3 +87,92 @@ def test_regex_differential_fuzz():
         if not rejected:
             checked += 1
     assert checked > 1500  # the generator mostly stays in the supported class
+
+
+# ---- round-2 regex extensions: lazy quantifiers, per-branch anchors, (?i) ----
+
+CASES = [
+    # lazy quantifiers: same accepted language as greedy (existence match);
+    # "ab+?" previously mis-parsed as (ab+)? which wrongly matched "a"
+    ("ab+?", "a", 0), ("ab+?", "ab", 1), ("ab+?", "abb", 1),
+    ("a*?b", "b", 1), ("a*?b", "c", 0), ("a??b", "b", 1),
+    ("x{2,3}?", "xx", 1), ("x{2,3}?", "x", 0),
+    # per-branch anchors (the reference's own fixture patterns)
+    ("^01|04$", "012", 1), ("^01|04$", "104", 1),
+    ("^01|04$", "201", 0), ("^01|04$", "042", 0), ("^01|04$", "04", 1),
+    ("foo|bar|^$", "", 1), ("foo|bar|^$", "x", 0), ("foo|bar|^$", "xbar", 1),
+    ("qwe.+rty|^$", "", 1), ("qwe.+rty|^$", "qweXrty", 1),
+    ("qwe.+rty|^$", "qwerty", 0),
+    ("^a|b$|c", "az", 1), ("^a|b$|c", "za", 0), ("^a|b$|c", "zb", 1),
+    ("^a|b$|c", "bz", 0), ("^a|b$|c", "zcz", 1),
+    # leading (?i): simple case closure + Unicode fold orbits
+    ("(?i)foo", "FOO", 1), ("(?i)foo", "xFoOy", 1), ("(?i)foo", "fo", 0),
+    ("(?i)[ab]c", "AC", 1), ("(?i)[a-c]z", "Bz", 1), ("(?i)[a-c]z", "dz", 0),
+    ("(?i)k", "K", 1), ("(?i)K", "k", 1),
+    ("(?i)kilo", "Kilo", 1),          # KELVIN SIGN folds with k
+    ("(?i)foo|йцу", "ЙЦУ", 1), ("(?i)foo|йцу", "FOO", 1),
+    ("(?i)foo|йцу", "йцу", 1), ("(?i)foo|йцу", "цук", 0),
+    ("(?i)σ", "Σ", 1), ("(?i)Σ", "ς", 1),  # sigma orbit via closure
+]
+
+REJECTS = ["a**", "a{2}{3}", "a\\b", "\\p{L}", "(?m)a", "(?i:a)b", "a*+"]
+
+
+def _engines():
+    lib = oracle_helpers()
+    lib.orc_regex_match.restype = ctypes.c_long
+    import os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    dev = ctypes.CDLL(os.path.join(root, "tools", "host_rowops",
+                                   "librowops.so"))
+    dev.h_dev_regex_match.restype = ctypes.c_int
+    return [("oracle", lambda p, s: lib.orc_regex_match(p, len(p), s, len(s))),
+            ("device", lambda p, s: dev.h_dev_regex_match(p, len(p), s, len(s)))]
+
+
+def test_regex_round2_extensions():
+    for name, match in _engines():
+        for pat, subj, want in CASES:
+            pb, sb = pat.encode(), subj.encode()
+            got = match(pb, sb)
+            assert got == want, f"[{name}] pat={pat!r} s={subj!r} got={got} want={want}"
+        for pat in REJECTS:
+            got = match(pat.encode(), b"abc")
+            assert got < 0, f"[{name}] pattern {pat!r} should reject, got {got}"
+
+
+def test_regex_lazy_fuzz_vs_python():
+    """Random patterns with lazy markers sprinkled in, vs Python re."""
+    lib = oracle_helpers()
+    lib.orc_regex_match.restype = ctypes.c_long
+    rng = random.Random(42)
+    checked = 0
+    for _ in range(800):
+        pat = rand_re(rng, 2)
+        # sprinkle lazy markers after quantifiers
+        out = []
+        i = 0
+        while i < len(pat):
+            out.append(pat[i])
+            if pat[i] in "*+}" and rng.random() < 0.4 and \
+                    (i + 1 == len(pat) or pat[i + 1] not in "*+?{"):
+                out.append("?")
+            i += 1
+        pat = "".join(out)
+        try:
+            cre = pyre.compile(pat)
+        except pyre.error:
+            continue
+        pb = pat.encode()
+        ok = True
+        for s in ["", "ab01", "aabb0011", "a b.a-b", "01ab01ab01"]:
+            sb = s.encode()
+            r = lib.orc_regex_match(pb, len(pb), sb, len(sb))
+            if r < 0:
+                ok = False
+                break
+            want = 1 if cre.search(s) else 0
+            assert r == want, f"pat={pat!r} s={s!r} ours={r} py={want}"
+        if ok:
+            checked += 1
+    assert checked > 400

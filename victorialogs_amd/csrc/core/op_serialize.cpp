@@ -7,6 +7,20 @@ namespace vl {
 
 bytes serialize_regex(const RegexProg& re) {
   bytes b;
+  if (re.is_alt_list) {
+    // alt-list blob: marker 0x00 (a compiled non-alt regex always sets at
+    // least one flag bit), u16 n_alts, then per branch {u32 len, sub-blob}
+    b.push_back(0);
+    b.push_back(uint8_t(re.alts.size()));
+    b.push_back(uint8_t(re.alts.size() >> 8));
+    for (const RegexProg& alt : re.alts) {
+      bytes sb = serialize_regex(alt);
+      uint32_t len = uint32_t(sb.size());
+      for (int i = 0; i < 4; i++) b.push_back(uint8_t(len >> (8 * i)));
+      b.insert(b.end(), sb.begin(), sb.end());
+    }
+    return b;
+  }
   uint8_t flags = 0;
   if (re.is_only_prefix) flags |= kReOnlyPrefix;
   if (re.is_suffix_dot_star) flags |= kReDotStar;

@@ -4,6 +4,7 @@
 
 #include "match.h"
 #include "tokenizer.h"
+#include "unicode_case.h"
 
 namespace vl {
 
@@ -20,10 +21,89 @@ struct RNode {
   bool cls_nonascii = false;
 };
 
+// Case-fold orbits for (?i): closure over the simple lower/upper mappings
+// plus the Unicode CaseFolding special orbits Go's unicode.SimpleFold walks
+// (unicode.CaseOrbit — K/k/KELVIN, S/s/LONG S, the Greek symbol letters,
+// ANGSTROM, MICRO, iota subscript, S-with-dot).  Simple-pair letters and
+// ASCII are exact; exotic multi-member orbits outside this table fall back
+// to the lower/upper closure.
+static const uint32_t kFoldOrbits[][4] = {
+    {0x004B, 0x006B, 0x212A, 0},       // K k KELVIN SIGN
+    {0x0053, 0x0073, 0x017F, 0},       // S s LONG S
+    {0x00B5, 0x039C, 0x03BC, 0},       // MICRO, CAP MU, mu
+    {0x00C5, 0x00E5, 0x212B, 0},       // Å å ANGSTROM
+    {0x0345, 0x0399, 0x03B9, 0x1FBE},  // iota subscript orbit
+    {0x0392, 0x03B2, 0x03D0, 0},       // BETA beta beta-symbol
+    {0x0395, 0x03B5, 0x03F5, 0},       // EPSILON epsilon lunate
+    {0x0398, 0x03B8, 0x03D1, 0x03F4},  // THETA orbit
+    {0x039A, 0x03BA, 0x03F0, 0},       // KAPPA kappa kappa-symbol
+    {0x03A0, 0x03C0, 0x03D6, 0},       // PI pi omega-pi
+    {0x03A1, 0x03C1, 0x03F1, 0},       // RHO rho rho-symbol
+    {0x03A6, 0x03C6, 0x03D5, 0},       // PHI phi phi-symbol
+    {0x03A9, 0x03C9, 0x2126, 0},       // OMEGA omega OHM
+    {0x1E60, 0x1E61, 0x1E9B, 0},       // S-dot orbit
+    {0x03A3, 0x03C2, 0x03C3, 0},       // SIGMA final-sigma sigma
+    {0x01C4, 0x01C5, 0x01C6, 0},       // DŽ Dž dž (title-case digraphs)
+    {0x01C7, 0x01C8, 0x01C9, 0},       // LJ Lj lj
+    {0x01CA, 0x01CB, 0x01CC, 0},       // NJ Nj nj
+    {0x01F1, 0x01F2, 0x01F3, 0},       // DZ Dz dz
+};
+
+static void case_orbit(uint32_t r, std::vector<uint32_t>& out) {
+  out.clear();
+  out.push_back(r);
+  // closure over simple lower/upper (catches sigma/dz-digraph style orbits)
+  for (size_t i = 0; i < out.size() && out.size() < 8; i++) {
+    uint32_t lo = to_lower_rune(out[i]), up = to_upper_rune(out[i]);
+    for (uint32_t x : {lo, up}) {
+      bool seen = false;
+      for (uint32_t y : out) seen |= (y == x);
+      if (!seen) out.push_back(x);
+    }
+  }
+  for (const auto& orbit : kFoldOrbits) {
+    bool hit = false;
+    for (uint32_t m : orbit) hit |= (m != 0 && m == r);
+    if (!hit) {
+      for (uint32_t y : out) {
+        for (uint32_t m : orbit) hit |= (m != 0 && m == y);
+      }
+    }
+    if (hit) {
+      for (uint32_t m : orbit) {
+        if (m == 0) continue;
+        bool seen = false;
+        for (uint32_t y : out) seen |= (y == m);
+        if (!seen) out.push_back(m);
+      }
+    }
+  }
+}
+
+static void utf8_encode(uint32_t r, std::string& out) {
+  if (r < 0x80) {
+    out.push_back(char(r));
+  } else if (r < 0x800) {
+    out.push_back(char(0xC0 | (r >> 6)));
+    out.push_back(char(0x80 | (r & 0x3F)));
+  } else if (r < 0x10000) {
+    out.push_back(char(0xE0 | (r >> 12)));
+    out.push_back(char(0x80 | ((r >> 6) & 0x3F)));
+    out.push_back(char(0x80 | (r & 0x3F)));
+  } else {
+    out.push_back(char(0xF0 | (r >> 18)));
+    out.push_back(char(0x80 | ((r >> 12) & 0x3F)));
+    out.push_back(char(0x80 | ((r >> 6) & 0x3F)));
+    out.push_back(char(0x80 | (r & 0x3F)));
+  }
+}
+
 struct Parser {
   const std::string& s;
   size_t pos = 0;
-  explicit Parser(const std::string& s_) : s(s_) {}
+  bool icase = false;  // leading (?i): fold literals/classes at parse time
+  explicit Parser(const std::string& s_, bool icase_ = false)
+      : s(s_), icase(icase_) {}
 
   [[noreturn]] void err(const std::string& msg) {
     fail("regex: " + msg + " in " + s);
@@ -50,15 +130,22 @@ struct Parser {
     std::vector<RNode> items;
     while (!eof() && peek() != '|' && peek() != ')') {
       RNode atom = parse_atom();
-      // postfix quantifiers bind to the last atom
+      // postfix quantifiers bind to the last atom; at most one quantifier
+      // per atom (Go rejects "a**" as nested repetition), with an optional
+      // trailing '?' lazy marker — lazy and greedy accept the same language,
+      // and MatchString only asks for existence, so the marker is a no-op
+      bool quantified = false;
       while (!eof()) {
         char c = peek();
         if (c == '*' || c == '+' || c == '?') {
+          if (quantified) err("nested repetition operator");
           pos++;
+          if (!eof() && peek() == '?') pos++;  // lazy variant
           RNode q;
           q.kind = c == '*' ? RNode::Star : c == '+' ? RNode::Plus : RNode::Quest;
           q.subs.push_back(std::move(atom));
           atom = std::move(q);
+          quantified = true;
         } else if (c == '{') {
           // {m}, {m,}, {m,n}: expanded by cloning (Go regexp/syntax
           // semantics; a '{' that is not a valid repetition is a literal)
@@ -91,6 +178,8 @@ struct Parser {
             break;
           }
           pos++;
+          if (quantified) err("nested repetition operator");
+          if (!eof() && peek() == '?') pos++;  // lazy {m,n}?: same language
           if (nrep == -2) nrep = m;
           if (m > 1000 || nrep > 1000) err("invalid repeat count");
           if (nrep >= 0 && nrep < m) err("invalid repeat count");
@@ -134,10 +223,7 @@ struct Parser {
           } else {
             atom = std::move(expanded);
           }
-          if (!eof() && (peek() == '*' || peek() == '+' || peek() == '?' ||
-                         peek() == '{')) {
-            err("nested repetition operator");
-          }
+          quantified = true;
         } else {
           break;
         }
@@ -202,9 +288,35 @@ struct Parser {
         int len = c0 < 0x80 ? 1 : (c0 & 0xE0) == 0xC0 ? 2
                   : (c0 & 0xF0) == 0xE0 ? 3 : (c0 & 0xF8) == 0xF0 ? 4 : 1;
         for (int i = 0; i < len && !eof(); i++) n.lit.push_back(s[pos++]);
+        if (icase) return fold_rune_lit(n.lit);
         return n;
       }
     }
+  }
+
+  // (?i): a literal rune becomes the alternation of its case-fold orbit
+  RNode fold_rune_lit(const std::string& lit) {
+    int sz = 0;
+    uint32_t r = uint8_t(lit[0]) < 0x80
+                     ? uint32_t(uint8_t(lit[0]))
+                     : utf8_decode(lit.data(), lit.size(), &sz);
+    std::vector<uint32_t> orbit;
+    case_orbit(r, orbit);
+    if (orbit.size() == 1) {
+      RNode n;
+      n.kind = RNode::Lit;
+      n.lit = lit;
+      return n;
+    }
+    RNode alt;
+    alt.kind = RNode::Alt;
+    for (uint32_t m : orbit) {
+      RNode l;
+      l.kind = RNode::Lit;
+      utf8_encode(m, l.lit);
+      alt.subs.push_back(std::move(l));
+    }
+    return alt;
   }
 
   RNode parse_escape() {
@@ -312,6 +424,31 @@ struct Parser {
     if (eof()) err("missing ]");
     pos++;  // ']'
 
+    // (?i): fold the ASCII set before negation (Go folds class members);
+    // k/K and s/S drag in their non-ASCII orbit members (KELVIN, LONG S)
+    std::vector<uint32_t> extra;
+    if (icase) {
+      uint8_t folded[32];
+      memcpy(folded, bits, 32);
+      for (int b = 0; b < 128; b++) {
+        if (!((bits[b >> 3] >> (b & 7)) & 1)) continue;
+        std::vector<uint32_t> orbit;
+        case_orbit(uint32_t(b), orbit);
+        for (uint32_t m : orbit) {
+          if (m < 0x80) {
+            folded[m >> 3] |= uint8_t(1) << (m & 7);
+          } else {
+            extra.push_back(m);
+          }
+        }
+      }
+      memcpy(bits, folded, 32);
+      if (negated && !extra.empty()) {
+        err("(?i) with a negated class containing k or s is not supported "
+            "(non-ASCII fold-orbit members)");
+      }
+    }
+
     if (negated) {
       RNode cn;
       cn.kind = RNode::Class;
@@ -321,6 +458,7 @@ struct Parser {
     }
     // expand small positive classes as alternation of single chars so the
     // or-values fast path classifies them like Go (regexutil.go:93-107)
+    RNode result;
     int count = 0;
     for (int b = 0; b < 128; b++) count += (bits[b >> 3] >> (b & 7)) & 1;
     if (count == 0) err("empty char class");
@@ -335,13 +473,24 @@ struct Parser {
           alt.subs.push_back(std::move(l));
         }
       }
-      if (alt.subs.size() == 1) return std::move(alt.subs[0]);
+      result = alt.subs.size() == 1 ? std::move(alt.subs[0]) : std::move(alt);
+    } else {
+      result.kind = RNode::Class;
+      memcpy(result.cls, bits, 32);
+    }
+    if (!extra.empty()) {
+      RNode alt;
+      alt.kind = RNode::Alt;
+      alt.subs.push_back(std::move(result));
+      for (uint32_t m : extra) {
+        RNode l;
+        l.kind = RNode::Lit;
+        utf8_encode(m, l.lit);
+        alt.subs.push_back(std::move(l));
+      }
       return alt;
     }
-    RNode cn;
-    cn.kind = RNode::Class;
-    memcpy(cn.cls, bits, 32);
-    return cn;
+    return result;
   }
 };
 
@@ -683,29 +832,86 @@ static bytes g_serialize(const GBuild& b, const GInfo& root,
 
 }  // namespace
 
-RegexProg regex_compile(const std::string& expr) {
-  // Top-level anchors: a leading '^' / trailing unescaped '$' anchor the
-  // whole pattern (Go regexp semantics without multiline).  Anchors inside
-  // the pattern (alternation branches, groups) remain unsupported and error
-  // in the parser.
-  std::string body = expr;
-  bool a_start = false, a_end = false;
+// Splits expr at top-level '|' (outside groups, classes and escapes).
+// Returns a single element when there is no top-level alternation.
+static std::vector<std::string> split_top_level_alt(const std::string& s) {
+  std::vector<std::string> parts;
+  size_t start = 0;
+  int depth = 0;
+  bool in_class = false, class_first = false;
+  for (size_t i = 0; i < s.size(); i++) {
+    char c = s[i];
+    if (c == '\\') {
+      i++;  // skip escaped char
+      continue;
+    }
+    if (in_class) {
+      if (c == ']' && !class_first) in_class = false;
+      class_first = false;
+      continue;
+    }
+    switch (c) {
+      case '[':
+        in_class = true;
+        class_first = true;
+        if (i + 1 < s.size() && s[i + 1] == '^') {
+          i++;  // '^' right after '[' keeps the first-']'-is-literal rule
+        }
+        break;
+      case '(':
+        depth++;
+        break;
+      case ')':
+        depth--;
+        break;
+      case '|':
+        if (depth == 0) {
+          parts.push_back(s.substr(start, i - start));
+          start = i + 1;
+        }
+        break;
+      default:
+        break;
+    }
+  }
+  parts.push_back(s.substr(start));
+  return parts;
+}
+
+static bool strip_edge_anchors(std::string& body, bool* a_start, bool* a_end) {
+  *a_start = *a_end = false;
   if (!body.empty() && body[0] == '^') {
-    a_start = true;
+    *a_start = true;
     body.erase(body.begin());
   }
   if (!body.empty() && body.back() == '$') {
     size_t bs = 0;
     while (bs + 1 < body.size() && body[body.size() - 2 - bs] == '\\') bs++;
     if (bs % 2 == 0) {
-      a_end = true;
+      *a_end = true;
       body.pop_back();
     }
   }
-  Parser p(body);
+  return *a_start || *a_end;
+}
+
+static RegexProg compile_single(const std::string& expr,
+                                const std::string& body_in, bool icase) {
+  // Edge anchors: a leading '^' / trailing unescaped '$' anchor this
+  // pattern (Go regexp semantics without multiline).  Anchors elsewhere
+  // (inside groups, mid-pattern) remain unsupported and error in the
+  // parser; anchored top-level alternatives are handled by the alt-list
+  // split in regex_compile.
+  std::string body = body_in;
+  bool a_start = false, a_end = false;
+  strip_edge_anchors(body, &a_start, &a_end);
+  Parser p(body, icase);
   RNode raw = p.parse_alt();
   if (!p.eof()) p.err("unexpected )");
   if ((a_start || a_end) && raw.kind == RNode::Alt) {
+    // only reachable for anchors around a group-free alternation inside a
+    // single alt-list branch, e.g. "^(a|b)" is fine (Group) but "^a|b"
+    // was already split; keep the loud error as a backstop
     fail("regex: anchors with top-level alternation are not supported in " +
          expr);
   }
@@ -804,6 +1010,42 @@ RegexProg regex_compile(const std::string& expr) {
   return re;
 }
 
+RegexProg regex_compile(const std::string& expr) {
+  std::string body = expr;
+  // a leading (?i) applies to the whole pattern (Go flag semantics); other
+  // (?...) constructs still error in the parser
+  bool icase = false;
+  if (body.size() >= 4 && body.compare(0, 4, "(?i)") == 0) {
+    icase = true;
+    body.erase(0, 4);
+  }
+
+  // Top-level alternation whose branches carry their own anchors compiles
+  // as an any-of list of independently compiled branches ("^01|04$" ==
+  // starts-with-01 OR ends-with-04, Go per-branch anchoring).
+  std::vector<std::string> parts = split_top_level_alt(body);
+  if (parts.size() > 1) {
+    bool any_anchor = false;
+    for (const std::string& p : parts) {
+      std::string tmp = p;
+      bool s = false, e = false;
+      if (strip_edge_anchors(tmp, &s, &e)) any_anchor = true;
+    }
+    if (any_anchor) {
+      RegexProg re;
+      re.expr = expr;
+      re.is_alt_list = true;
+      // no bloom literals: Go GetLiterals returns nil for OpAlternate
+      // (regexutil.go:141-149 via regex.go:101-124)
+      for (const std::string& p : parts) {
+        re.alts.push_back(compile_single(expr, p, icase));
+      }
+      return re;
+    }
+  }
+  return compile_single(expr, body, icase);
+}
+
 // strings.Contains
 static bool contains(strview s, const std::string& sub) {
   if (sub.empty()) return true;
@@ -898,6 +1140,12 @@ bool nfa_match(const uint8_t* blob, strview s) {
 
 bool regex_match(const RegexProg& re, strview s) {
   // Regex.MatchString (regex.go:86-98)
+  if (re.is_alt_list) {
+    for (const RegexProg& alt : re.alts) {
+      if (regex_match(alt, s)) return true;
+    }
+    return false;
+  }
   if (re.has_nfa) {
     if (re.always_true) return true;
     return nfa_match(re.nfa_blob.data(), s);

@@ -41,6 +41,13 @@ struct RegexProg {
   bool has_nfa = false;
   bool always_true = false;  // pattern matches the empty string
   bytes nfa_blob;
+
+  // Top-level alternation whose alternatives carry their own '^'/'$'
+  // anchors (e.g. "^01|04$"): each alternative compiles independently and
+  // the match is any-of.  Mirrors Go's per-branch anchoring semantics; no
+  // bloom literals (Go GetLiterals returns nil for OpAlternate).
+  bool is_alt_list = false;
+  std::vector<RegexProg> alts;
 };
 
 // NFA blob layout: u16 nstates, u16 pad, u32 pad, u64 first_mask,
@@ -48,8 +55,12 @@ struct RegexProg {
 bool nfa_match(const uint8_t* blob, strview s);
 
 // Compiles expr; patterns outside the fast-path classes compile to the NFA
-// fallback; throws vl::Error only for genuinely unsupported syntax
-// ({m,n} repetition, anchors, backreferences, >64 NFA positions).
+// fallback.  Supported beyond the fast paths: {m,n} repetition, lazy
+// quantifiers (same accepted language for existence matching), top-level
+// and per-alternative '^'/'$' anchors, and a leading (?i) (simple case
+// closure + the Unicode CaseFolding special orbits).  Still rejected with
+// a clear error: \b/\B, \p{...}, mid-pattern anchors, (?...) flags other
+// than a leading (?i), and >64 NFA positions.
 RegexProg regex_compile(const std::string& expr);
 
 // Regex.MatchString (regex.go:86-98,131-212).

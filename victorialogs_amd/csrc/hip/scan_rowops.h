@@ -1017,9 +1017,11 @@ __device__ bool d_nfa_match_at(const uint8_t* blob, const A& a, long s0, long sn
   return a_end && (active & last) != 0;
 }
 
-// Regex.MatchString (regex.go:86-212) over accessor bytes [s0, s0+sn).
+// Regex.MatchString (regex.go:86-212) over accessor bytes [s0, s0+sn)
+// for a single compiled branch (no alt-list marker).
 template <typename A>
-__device__ bool d_regex_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
+__device__ bool d_regex_match_single(const uint8_t* blob, const A& a, long s0,
+                                     long sn) {
   if (blob[0] & kReAlways) return true;
   if (blob[0] & kReNfa) {
     // NFA blob sits after the (empty) prefix/substr/or-values header
@@ -1075,6 +1077,25 @@ __device__ bool d_regex_match_at(const uint8_t* blob, const A& a, long s0, long 
     t0 = s0 + n + re.prefix_len;
     tn = sn - n - re.prefix_len;
   }
+}
+
+// Entry point: a 0x00 marker byte means an alt-list blob (top-level
+// alternation with per-branch anchors, e.g. "^01|04$"): u16 n_alts, then
+// per branch {u32 len, sub-blob}; match = any branch matches.
+template <typename A>
+__device__ bool d_regex_match_at(const uint8_t* blob, const A& a, long s0,
+                                 long sn) {
+  if (blob[0] != 0) return d_regex_match_single(blob, a, s0, sn);
+  const int n_alts = int(blob[1]) | int(blob[2]) << 8;
+  const uint8_t* p = blob + 3;
+  for (int i = 0; i < n_alts; i++) {
+    uint32_t len = uint32_t(p[0]) | uint32_t(p[1]) << 8 |
+                   uint32_t(p[2]) << 16 | uint32_t(p[3]) << 24;
+    p += 4;
+    if (d_regex_match_single(p, a, s0, sn)) return true;
+    p += len;
+  }
+  return false;
 }
 
 // ---- per-row predicates ----
