@@ -309,3 +309,37 @@ def test_static_elimination_parity(tmp_path):
         '{"type":"phrase","field":"_msg","phrase":"tag_2"}]}',
     ]:
         assert_parity(d, f)
+
+
+def test_smallrow_supergroup_parity(tmp_path):
+    """Short rows take the super-group loop (lb.sg > 1); a huge outlier row
+    forces the per-super-group tile-overflow fallback; empty rows and the
+    non-multiple-of-64 tail must stay bit-identical."""
+    from victorialogs_amd import write_custom_part
+
+    rows = 3001
+    vals = []
+    for i in range(rows):
+        if i == 1234:
+            vals.append("X" * 30000 + " needle_1 " + "Y" * 2000)
+        elif i % 97 == 0:
+            vals.append("")
+        else:
+            vals.append(f"r{i} needle_{i % 5} tail{i % 11}")
+    spec = {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i for i in range(rows)],
+        "columns": [{"name": "_msg", "values": vals}],
+    }]}
+    d = str(tmp_path / "smallrows")
+    write_custom_part(d, spec)
+    for f in [
+        '{"type":"phrase","field":"_msg","phrase":"needle_3"}',
+        '{"type":"phrase","field":"_msg","phrase":"needle_1"}',
+        '{"type":"regexp","field":"_msg","re":"needle_(1|2)"}',
+        '{"type":"regexp","field":"_msg","re":"tail7$"}',
+        '{"type":"prefix","field":"_msg","prefix":"r10"}',
+        '{"type":"any_case_phrase","field":"_msg","phrase":"NEEDLE_2"}',
+        '{"type":"len_range","field":"_msg","min":0,"max":0}',
+    ]:
+        assert_parity(d, f)

@@ -31,10 +31,7 @@ namespace vl {
 // word per group) with no workgroup barriers: the wave copies its group's
 // bytes with coalesced 16-byte loads (4-deep batches keep ~4 KiB per wave in
 // flight), then matches lane-per-row from the tile.
-constexpr uint32_t kWaveTileBytes = 17408;  // 1088 slots: 64 rows x <=272 B,
-// a multiple of 64*16 B so the copy loop can round every group up to full
-// 64-slot strides (no single-lane straggler load per group)
-constexpr uint32_t kNumWaves = 4;
+// kWaveTileBytes / kNumWaves live in scan_types.h (staging reads them too)
 
 // host local timezone offset for no-suffix RFC3339 parses (see scan_rowops.h)
 __device__ int64_t g_vl_local_tz_nsecs = 0;
@@ -198,6 +195,108 @@ __device__ __forceinline__ void d_string_tile_loop(
   }
 }
 
+// Small-row string loop: the per-word tile loop above serializes one HBM
+// round trip per 64-row group, which is latency-bound when rows are short
+// (a ~30 B-row column moved ~0.8 TB/s vs ~5 TB/s for 256 B rows).  Here a
+// wave fills its tile with SG words' bytes at once (SG = lb.sg, chosen at
+// staging so SG*64*avg_row fits the tile) and evaluates SG ballot words per
+// fill — amortizing the copy latency SG-fold.  Super-groups whose actual
+// byte span overflows the tile (row-length variance) fall back to direct
+// global evaluation for those words, bit-identically.
+// __noinline__: one clone per evaluator, kept out of the hot per-word loop's
+// instruction stream (see the I-cache regression note above).
+template <bool kOvr, typename EvalFn>
+__device__ __noinline__ void d_string_smallrow_loop(
+    const uint8_t* __restrict__ col_data, const uint32_t* __restrict__ col_offs,
+    const uint64_t* __restrict__ ovr_mask, const uint64_t* __restrict__ ovr_val,
+    uint8_t* wtile, uint64_t* out, uint32_t r0, uint32_t r1, uint32_t nwords,
+    uint32_t sgw, int lane, int wave, int nwaves, EvalFn eval) {
+  typedef uint32_t v4u __attribute__((ext_vector_type(4)));
+  v4u* dst = (v4u*)wtile;
+  const uint32_t nsg = (nwords + sgw - 1) / sgw;
+  for (uint32_t sg = wave; sg < nsg; sg += nwaves) {
+    const uint32_t w0 = sg * sgw;
+    const uint32_t g0 = r0 + w0 * 64;
+    uint32_t o[9];
+#pragma unroll
+    for (uint32_t j = 0; j < 8; j++) {
+      o[j] = j < sgw ? col_offs[min(g0 + j * 64 + uint32_t(lane), r1)] : 0;
+    }
+    o[8] = 0;
+    if (lane == 0) o[8] = col_offs[min(g0 + sgw * 64, r1)];
+    const uint32_t byte0 =
+        uint32_t(__builtin_amdgcn_readfirstlane(int(o[0]))) & ~15u;
+    const uint32_t byte1 = uint32_t(__shfl(int(o[8]), 0, 64));
+    const uint32_t nbytes = byte1 - byte0;
+    const bool use_tile = nbytes <= kWaveTileBytes;
+    if (use_tile) {
+      const v4u* src = (const v4u*)(col_data + byte0);
+      const uint32_t n16 = (((nbytes + 15) >> 4) + 63) & ~63u;
+      uint32_t k = lane;
+      for (; k + 448 < n16; k += 512) {
+        v4u a0 = __builtin_nontemporal_load(src + k);
+        v4u a1 = __builtin_nontemporal_load(src + k + 64);
+        v4u a2 = __builtin_nontemporal_load(src + k + 128);
+        v4u a3 = __builtin_nontemporal_load(src + k + 192);
+        v4u a4 = __builtin_nontemporal_load(src + k + 256);
+        v4u a5 = __builtin_nontemporal_load(src + k + 320);
+        v4u a6 = __builtin_nontemporal_load(src + k + 384);
+        v4u a7 = __builtin_nontemporal_load(src + k + 448);
+        dst[k ^ ((k >> 4) & 15)] = a0;
+        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+        dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
+        dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
+        dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
+        dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
+      }
+      for (; k + 192 < n16; k += 256) {
+        v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128], a3 = src[k + 192];
+        dst[k ^ ((k >> 4) & 15)] = a0;
+        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+      }
+      for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
+#pragma unroll
+    for (uint32_t j = 0; j < 8; j++) {
+      if (j >= sgw) break;
+      const uint32_t w = w0 + j;
+      if (w >= nwords) break;
+      const uint32_t gj = g0 + j * 64;
+      const uint32_t ng = min(64u, r1 - gj);
+      const uint32_t wend =
+          uint32_t(__shfl(int(j + 1 < sgw ? o[j + 1] : o[8]), 0, 64));
+      // all 64 lanes execute the shfl; for the word's LAST valid lane the
+      // next lane's o[] may come from an inactive/garbage source, so that
+      // lane takes the broadcast word end instead (the per-word tile loop
+      // does the same with byte1)
+      const uint32_t e0 = uint32_t(__shfl(int(o[j]), lane + 1, 64));
+      bool pred = false;
+      if (uint32_t(lane) < ng) {
+        const long s = o[j];
+        const long e = uint32_t(lane) == ng - 1 ? long(wend) : long(e0);
+        if (use_tile) {
+          TileAcc a{wtile};
+          pred = eval(a, s - byte0, e - s);
+        } else {
+          GlobalAcc a{col_data};
+          pred = eval(a, s, e - s);
+        }
+      }
+      uint64_t word = __ballot(pred);
+      if (kOvr && ovr_mask != nullptr) {
+        const uint64_t mw = ovr_mask[w], vw = ovr_val[w];
+        word = (word & ~mw) | (vw & mw);
+      }
+      if (lane == 0) out[w] = word;
+    }
+  }
+}
+
 // ---- the program kernel ----
 
 __global__ __launch_bounds__(256) void scan_program_kernel(
@@ -260,32 +359,53 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
         const uint8_t* op_ptr = lb.operand;
         const uint32_t op_len = lb.operand_len;
         const uint8_t op_flags = lb.flags;
+        const uint32_t sgw = lb.sg;
         if (lb.kind == kScanPhraseStr) {
           // hot clone: only the phrase matcher in the loop body
-          d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
-                                    wtile, out, r0, r1, nwords, lane, wave,
-                                    nwaves, [=](const auto& a, long s0, long sn) {
-                                      return d_match_phrase_at(a, s0, sn,
-                                                               op_ptr, op_len,
-                                                               op_flags);
-                                    });
+          auto eval = [=](const auto& a, long s0, long sn) {
+            return d_match_phrase_at(a, s0, sn, op_ptr, op_len, op_flags);
+          };
+          if (sgw > 1) {
+            d_string_smallrow_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                          wtile, out, r0, r1, nwords, sgw,
+                                          lane, wave, nwaves, eval);
+          } else {
+            d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                      wtile, out, r0, r1, nwords, lane, wave,
+                                      nwaves, eval);
+          }
         } else if (lb.kind == kScanRegexStr) {
           // second hot clone: regex fast paths + NFA (BASELINE config 3)
-          d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
-                                    wtile, out, r0, r1, nwords, lane, wave,
-                                    nwaves, [=](const auto& a, long s0, long sn) {
-                                      return d_regex_match_at(op_ptr, a, s0, sn);
-                                    });
+          auto eval = [=](const auto& a, long s0, long sn) {
+            return d_regex_match_at(op_ptr, a, s0, sn);
+          };
+          if (sgw > 1) {
+            d_string_smallrow_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                          wtile, out, r0, r1, nwords, sgw,
+                                          lane, wave, nwaves, eval);
+          } else {
+            d_string_tile_loop<false>(col_data, col_offs, nullptr, nullptr,
+                                      wtile, out, r0, r1, nwords, lane, wave,
+                                      nwaves, eval);
+          }
         } else {
           const bool anycase = lb.kind == kScanAnyCasePhraseStr ||
                                lb.kind == kScanAnyCasePrefixStr;
-          d_string_tile_loop<true>(col_data, col_offs,
-                                   anycase ? lb.hashes : nullptr,
-                                   anycase ? lb.bloom : nullptr,
-                                   wtile, out, r0, r1, nwords, lane, wave,
-                                   nwaves, [&](const auto& a, long s0, long sn) {
-                                     return d_eval_string_row(lb, a, s0, sn);
-                                   });
+          auto eval = [&](const auto& a, long s0, long sn) {
+            return d_eval_string_row(lb, a, s0, sn);
+          };
+          if (sgw > 1) {
+            d_string_smallrow_loop<true>(col_data, col_offs,
+                                         anycase ? lb.hashes : nullptr,
+                                         anycase ? lb.bloom : nullptr, wtile,
+                                         out, r0, r1, nwords, sgw, lane, wave,
+                                         nwaves, eval);
+          } else {
+            d_string_tile_loop<true>(col_data, col_offs,
+                                     anycase ? lb.hashes : nullptr,
+                                     anycase ? lb.bloom : nullptr, wtile, out,
+                                     r0, r1, nwords, lane, wave, nwaves, eval);
+          }
         }
       } else if (lb.kind == kScanDict) {
         // word-per-lane: each lane builds one full bitmap word from its 64

@@ -15,6 +15,11 @@ namespace vl {
 
 constexpr uint32_t kChunkRows = 8192;            // rows per workgroup
 constexpr uint32_t kChunkWords = kChunkRows / 64;
+// Per-wave LDS string tile (see scan_kernels.hip): 1088 16-B slots = 64 rows
+// x <=272 B, a multiple of 64*16 B.  Also used by staging to pick the
+// small-row super-group size (DevLeafBlock.sg).
+constexpr uint32_t kWaveTileBytes = 17408;
+constexpr uint32_t kNumWaves = 4;
 constexpr int kMaxProgOps = 64;
 constexpr int kMaxStackDepth = 8;
 
@@ -115,7 +120,8 @@ struct DevLeafBlock {
   uint32_t bloom_words;
   uint32_t dict_mask;      // kScanDict: bit i set if dict value i matches
   uint32_t operand_len;
-  uint32_t pad0;
+  uint32_t sg;             // string kinds: words per tile fill (2/4/8) for
+                           // the small-row loop; 0/1 = per-word tile loop
   const uint64_t* hashes;  // device ptrs
   const uint64_t* bloom;
   const uint8_t* operand;  // phrase bytes / bin value / serialized regex

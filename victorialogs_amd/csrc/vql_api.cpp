@@ -2547,6 +2547,19 @@ Stage* build_stage_refs(std::vector<std::pair<VqlPart*, long>> refs,
           case kScanAnyCasePrefixStr: {
             const StagedStrCol& sc = ctx.cols.at(leaf_infos[size_t(l)].cname);
             block_algo += sc.data_bytes + (bh.rows_count + 1) * 4;
+            // small-row super-group size: how many 64-row bitmap words fit
+            // one wave tile with margin for row-length variance (the kernel
+            // falls back per super-group when the actual span overflows)
+            if (sc.rows > 0) {
+              const uint64_t avg = sc.data_bytes / sc.rows;
+              uint32_t sgw = 1;
+              while (sgw < 8 &&
+                     uint64_t(sgw) * 2 * 64 * (avg + 8) + 2048 <=
+                         kWaveTileBytes) {
+                sgw *= 2;
+              }
+              lb.sg = sgw;
+            }
             break;
           }
           case kScanDict:
