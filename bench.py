@@ -211,10 +211,15 @@ def main():
     distributed = world > 1
     ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
     device = local_rank % ndev if ndev else 0
+    # RCCL over xGMI when each rank owns a GPU (the driver's 8-GPU run);
+    # RCCL refuses two ranks on one device ("Duplicate GPU detected"), so a
+    # world-2 test on a 1-GPU box runs the SAME distributed branches with
+    # gloo carrying the three tiny counter all_reduces while the scans still
+    # run on the GPU; gloo also lets the world-2 CPU CI test run (emu build)
+    backend = "nccl" if ndev and world <= ndev else "gloo"
+    coll_dev = "cuda" if backend == "nccl" else "cpu"
     if distributed:
-        # RCCL over xGMI on GPU boxes; gloo lets the same distributed code
-        # paths run on CPU (world-2 CI test via the emu build)
-        dist.init_process_group("nccl" if ndev else "gloo")
+        dist.init_process_group(backend)
         if ndev:
             torch.cuda.set_device(device)
 
@@ -291,7 +296,7 @@ def main():
 
     # whole-job aggregation: MAX(elapsed) over ranks; SUM(rows, hits)
     if distributed:
-        dev = "cuda" if ndev else "cpu"
+        dev = coll_dev
         t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
@@ -348,7 +353,7 @@ def main():
         sstage.close()
         sfilt.close()
         if distributed:
-            dev = "cuda" if ndev else "cpu"
+            dev = coll_dev
             c = torch.tensor([float(expected), float(got)],
                              dtype=torch.float64, device=dev)
             dist.all_reduce(c, op=dist.ReduceOp.SUM)

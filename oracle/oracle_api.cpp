@@ -134,6 +134,14 @@ long long orc_generate_part(const char* dir, unsigned long long rows,
     cfg.rows_per_block = rows_per_block;
     cfg.msg_len = size_t(msg_len);
     cfg.seed = seed;
+    // lean parts for the 1B-row configs: only the columns the or8/dict_time
+    // workloads scan (_msg, dict_*, host, worker_id, run_id) — halves the
+    // on-disk fixture size and generation time on the GPU box
+    if (getenv("VQL_GEN_LEAN") != nullptr) {
+      cfg.extra_typed_fields = false;
+      cfg.num_const_fields = 0;
+      cfg.num_var_fields = 0;
+    }
     return (long long)generate_part(dir, cfg);
   } catch (const std::exception& e) {
     return set_err(e);
