@@ -75,6 +75,9 @@ WORKLOADS = {
                   '{"type":"phrase","field":"dict_1","phrase":"debug"},'
                   '{"type":"phrase","field":"host","phrase":"host_0"},'
                   '{"type":"phrase","field":"host","phrase":"host_9"}]}',
+        # host_0 is a const column matching every generated row, so the OR
+        # must match all rows — a full-size invariant like the headline's
+        "expect_all": True,
     },
     "dict_time": {
         "name": "dict column + timestamp range (configs[4] shape)",
@@ -117,6 +120,66 @@ def prepare_parts(data_dir, total_rows, nparts, msg_len, seed_base):
                 f.result()
         log(f"generation took {time.time() - t0:.1f}s")
     return dirs
+
+
+def stream_stages(data_dir, total_rows, nparts, msg_len, seed_base, filt,
+                  device):
+    """Generate -> stage -> delete, one part at a time (a window of parts is
+    generated ahead on worker threads).  The GPU box's disk (~80 GB) cannot
+    hold a 1B-row fixture (~65 GB of zstd parts plus headroom), but its HBM
+    can hold the staged columns (~260 GB) — so parts stream through the disk
+    while the staged arena accumulates.  One kernel launch per part-stage.
+    Returns (stages, manifest_totals, rows, gen_stage_s)."""
+    import shutil
+    from concurrent.futures import ThreadPoolExecutor
+
+    from victorialogs_amd import Part, Stage, generate_part
+
+    os.makedirs(data_dir, exist_ok=True)
+    rows_per = total_rows // nparts
+    specs = []
+    for i in range(nparts):
+        rows = rows_per + (total_rows % nparts if i == nparts - 1 else 0)
+        d = os.path.join(data_dir, f"part_{rows}_{msg_len}_{seed_base + i}")
+        specs.append((d, rows, seed_base + i))
+
+    ahead = min(16, nparts)
+    t0 = time.time()
+    stages = []
+    manifests = {}
+    total = 0
+    with ThreadPoolExecutor(max_workers=ahead) as ex:
+        futs = {}
+
+        def ensure(j):
+            if j < len(specs) and j not in futs:
+                d, rows, seed = specs[j]
+                if os.path.exists(os.path.join(d, "gen_manifest.json")):
+                    futs[j] = None
+                else:
+                    futs[j] = ex.submit(generate_part, d, rows, 1, 8192,
+                                        msg_len, seed)
+
+        for j in range(ahead):
+            ensure(j)
+        for i, (d, rows, seed) in enumerate(specs):
+            if futs.get(i) is not None:
+                futs[i].result()
+            ensure(i + ahead)
+            m = read_manifests([d])
+            if m is not None:
+                for k, v in m.items():
+                    manifests[k] = manifests.get(k, 0) + v
+            p = Part(d)
+            st = Stage(p, filt, device=device)
+            p.close()  # staging holds everything in HBM; files can go
+            stages.append(st)
+            total += st.rows
+            shutil.rmtree(d, ignore_errors=True)
+            if i % 8 == 0:
+                log(f"streamed part {i + 1}/{len(specs)} "
+                    f"({total} rows staged)")
+    return stages, manifests, total, time.time() - t0
 
 
 def read_manifests(part_dirs):
@@ -200,6 +263,12 @@ def main():
         "VQL_DATA_DIR", "/tmp/vql_bench_data"))
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-selectivity", action="store_true")
+    ap.add_argument("--stream-parts", action="store_true",
+                    help="generate->stage->delete parts one at a time so "
+                         "configs whose fixture exceeds the box disk (1B "
+                         "rows) still stage fully into HBM; implies "
+                         "--skip-cpu-baseline/--skip-selectivity (the part "
+                         "files are gone after staging)")
     args = ap.parse_args()
 
     import torch
@@ -232,18 +301,28 @@ def main():
     # independent units, storage_search.go:1035-1067).
     nparts = max(1, min(64, (os.cpu_count() or 8) // max(1, world),
                         max(8, args.rows // 16_000_000)))
-    part_dirs = prepare_parts(
-        os.path.join(args.data_dir, f"rank{rank}"), args.rows, nparts,
-        args.msg_len, seed_base=1000 * rank + 1)
-    manifest = read_manifests(part_dirs)
-
-    log(f"rank {rank}: staging {len(part_dirs)} parts onto device {device}")
-    t0 = time.time()
-    parts = [Part(d) for d in part_dirs]
     filt = Filter(filter_json)
-    # one multi-part stage: the whole pass is ONE kernel launch
-    stages = [Stage(parts, filt, device=device)]
-    stage_s = time.time() - t0
+    if args.stream_parts:
+        args.skip_cpu_baseline = True
+        args.skip_selectivity = True
+        part_dirs = []
+        parts = []
+        stages, manifest, _, stage_s = stream_stages(
+            os.path.join(args.data_dir, f"rank{rank}"), args.rows, nparts,
+            args.msg_len, 1000 * rank + 1, filt, device)
+    else:
+        part_dirs = prepare_parts(
+            os.path.join(args.data_dir, f"rank{rank}"), args.rows, nparts,
+            args.msg_len, seed_base=1000 * rank + 1)
+        manifest = read_manifests(part_dirs)
+
+        log(f"rank {rank}: staging {len(part_dirs)} parts onto device "
+            f"{device}")
+        t0 = time.time()
+        parts = [Part(d) for d in part_dirs]
+        # one multi-part stage: the whole pass is ONE kernel launch
+        stages = [Stage(parts, filt, device=device)]
+        stage_s = time.time() - t0
     staged_bytes = sum(s.staged_bytes for s in stages)
     algo_bytes = sum(s.algo_bytes for s in stages)
     rows = sum(s.rows for s in stages)
