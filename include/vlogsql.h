@@ -29,6 +29,11 @@ extern "C" {
 #endif
 
 const char* vql_errstr(void);
+/* 1 if the last error was an unsupported-construct rejection (valid LogsQL
+ * outside the engine's class, e.g. regex \b or \p{...}) rather than
+ * corruption/IO — the cgo shim keeps such filters on the host Go path
+ * (INTEGRATION.md); the engine never silently falls back itself. */
+int vql_error_unsupported(void);
 
 /* Opens a reference-format part directory (FormatVersion 1..3; the file set
  * of lib/logstorage/filenames.go:3-24, read as part.go:105-173 does). */

@@ -78,3 +78,28 @@ def test_part_metadata_via_product(gen_part):
     assert nb == 8
     assert sum(lib.vql_block_rows(part, i) for i in range(nb)) == 30000
     lib.vql_close_part(part)
+
+
+def test_error_unsupported_classification():
+    """vql_error_unsupported distinguishes valid-LogsQL-outside-the-class
+    rejections (shim falls back to the Go path) from
+    corruption/IO errors."""
+    lib = load_product()
+    lib.vql_error_unsupported.restype = ctypes.c_int
+    # residual regex class: loud reject, classified unsupported
+    f = lib.vql_compile_filter(
+        b'{"type":"regexp","field":"x","re":"a\\\\b"}')
+    assert not f
+    assert b"regex" in lib.vql_errstr()
+    assert lib.vql_error_unsupported() == 1
+    # malformed input: error, NOT classified unsupported
+    f = lib.vql_compile_filter(b'{"type":"nope"}')
+    assert not f
+    assert lib.vql_error_unsupported() == 0
+    # supported since round 2: these must compile
+    for ok in [b'{"type":"regexp","field":"x","re":"^01|04$"}',
+               b'{"type":"regexp","field":"x","re":"(?i)foo"}',
+               b'{"type":"regexp","field":"x","re":"ab+?"}']:
+        h = lib.vql_compile_filter(ok)
+        assert h, lib.vql_errstr()
+        lib.vql_free_filter(h)
