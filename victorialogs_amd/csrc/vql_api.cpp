@@ -155,7 +155,10 @@ struct Stage {
   // reallocated, so staged pointers stay valid; every in-slab allocation
   // keeps >= kSlabTailPad bytes of owned slab memory after it (the string
   // tile copy rounds groups up to 1 KiB strides and may over-read).
-  static constexpr size_t kSlabBytes = size_t(1) << 30;
+  // 256 MiB: small enough that the per-stage tail waste stays negligible
+  // even when a 1B-row config streams through 64 separate part-stages
+  // (1 GiB slabs wasted ~32 GB there and OOM'd a 288 GiB device)
+  static constexpr size_t kSlabBytes = size_t(256) << 20;
   static constexpr size_t kSlabTailPad = 4096;
   std::vector<std::pair<uint8_t*, size_t>> slabs;  // (ptr, capacity)
   uint8_t* cur_slab = nullptr;
