@@ -56,7 +56,13 @@ build/oracle/%.o: oracle/%.cpp oracle/oracle_filter.h $(wildcard victorialogs_am
 oracle/liboracle.so: $(CORE_OBJS) build/oracle/oracle_filter.o build/oracle/oracle_api.o
 	$(CXX) -shared $^ -o $@ -ldl -lpthread
 
-build/hip/scan_kernels.o: victorialogs_amd/csrc/hip/scan_kernels.hip victorialogs_amd/csrc/hip/scan_types.h victorialogs_amd/csrc/core/unicode_ranges.inc
+build/hip/scan_kernels.o: victorialogs_amd/csrc/hip/scan_kernels.hip \
+		victorialogs_amd/csrc/hip/scan_types.h \
+		victorialogs_amd/csrc/hip/scan_rowops.h \
+		victorialogs_amd/csrc/core/parse_float.h \
+		victorialogs_amd/csrc/core/ryu.h \
+		victorialogs_amd/csrc/core/xxhash64.h \
+		victorialogs_amd/csrc/core/unicode_ranges.inc
 	@mkdir -p build/hip
 	$(HIPCC) $(HIPFLAGS) -x hip -c $< -o $@
 
