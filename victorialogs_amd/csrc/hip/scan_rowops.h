@@ -50,6 +50,13 @@ struct GlobalAcc {
   __device__ __forceinline__ uint64_t u64a(long off) const {  // off % 8 == 0
     return *(const uint64_t*)(base + off);
   }
+  // off % 16 == 0: both halves of one 16-byte window (single dwordx4 load)
+  __device__ __forceinline__ void u64a2(long off, uint64_t* x0,
+                                        uint64_t* x1) const {
+    const uint64_t* p = (const uint64_t*)(base + off);
+    *x0 = p[0];
+    *x1 = p[1];
+  }
   __device__ __forceinline__ uint8_t u8(long off) const { return base[off]; }
 };
 
@@ -64,6 +71,14 @@ struct TileAcc {
   __device__ __forceinline__ uint64_t u64a(long off) const {
     return *(const uint64_t*)(tile + swz(off));
   }
+  // off % 16 == 0: the whole window lives in ONE swizzled slot, so this is
+  // a single ds_read_b128
+  __device__ __forceinline__ void u64a2(long off, uint64_t* x0,
+                                        uint64_t* x1) const {
+    const uint64_t* p = (const uint64_t*)(tile + swz(off));
+    *x0 = p[0];
+    *x1 = p[1];
+  }
   __device__ __forceinline__ uint8_t u8(long off) const { return tile[swz(off)]; }
 };
 
@@ -74,13 +89,21 @@ struct TileAcc {
 template <typename A>
 struct LowerAcc {
   A a;
-  __device__ __forceinline__ uint64_t u64a(long off) const {
-    uint64_t x = a.u64a(off);
+  __device__ __forceinline__ static uint64_t lower64(uint64_t x) {
     uint64_t low7 = x & 0x7F7F7F7F7F7F7F7FULL;
     uint64_t ge_a = low7 + 0x3F3F3F3F3F3F3F3FULL;   // high bit: byte >= 0x41
     uint64_t ge_z1 = low7 + 0x2525252525252525ULL;  // high bit: byte >= 0x5B
     uint64_t is_az = ge_a & ~ge_z1 & ~x & 0x8080808080808080ULL;
     return x | (is_az >> 2);
+  }
+  __device__ __forceinline__ uint64_t u64a(long off) const {
+    return lower64(a.u64a(off));
+  }
+  __device__ __forceinline__ void u64a2(long off, uint64_t* x0,
+                                        uint64_t* x1) const {
+    a.u64a2(off, x0, x1);
+    *x0 = lower64(*x0);
+    *x1 = lower64(*x1);
   }
   __device__ __forceinline__ uint8_t u8(long off) const {
     uint8_t c = a.u8(off);
@@ -141,6 +164,10 @@ __device__ uint32_t d_utf8_decode_last(const A& a, long off, long n, int* size) 
 // verifies per 256 B of random text, and the divergent verify loop was the
 // dominant cost of non-matching scans (18.4 ms vs 5.1 ms per 100M-row pass
 // measured on the phrase kernel).  Head/tail bytes are masked out.
+__device__ __forceinline__ uint64_t d_swar_zero(uint64_t t) {
+  return (t - 0x0101010101010101ULL) & ~t & 0x8080808080808080ULL;
+}
+
 template <typename A>
 __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
                            long subn) {
@@ -148,26 +175,45 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
   if (subn > sn) return -1;
   const uint8_t c0 = sub[0];
   const uint64_t pat = 0x0101010101010101ULL * c0;
-  const uint64_t pat1 =
-      subn > 1 ? 0x0101010101010101ULL * sub[1] : 0;
+  const uint64_t pat1 = subn > 1 ? 0x0101010101010101ULL * sub[1] : 0;
   const long last = s0 + sn - subn;  // last valid start (absolute)
-  long w = s0 & ~7L;
-  for (; w <= last; w += 8) {
-    uint64_t x = a.u64a(w);
-    uint64_t t = x ^ pat;
-    uint64_t hit = (t - 0x0101010101010101ULL) & ~t & 0x8080808080808080ULL;
-    if (subn > 1 && hit) {
-      const uint64_t t1 = x ^ pat1;
-      const uint64_t hit1 =
-          (t1 - 0x0101010101010101ULL) & ~t1 & 0x8080808080808080ULL;
-      hit &= (hit1 >> 8) | 0x8000000000000000ULL;
+  // 16-byte windows: one ds_read_b128 per iteration (a swizzled tile slot
+  // holds the whole window), halving the serial per-window chain of the
+  // 8-byte version
+  for (long w = s0 & ~15L; w <= last; w += 16) {
+    uint64_t x0, x1;
+    a.u64a2(w, &x0, &x1);
+    uint64_t hitA = d_swar_zero(x0 ^ pat);
+    uint64_t hitB = d_swar_zero(x1 ^ pat);
+    if (subn > 1 && (hitA | hitB)) {
+      // two-byte anchor: require the second pattern byte at k+1 (the
+      // zero-scan has no false negatives, so pruning is sound; the last
+      // byte's successor lives in the next window and keeps its bit)
+      const uint64_t h1A = d_swar_zero(x0 ^ pat1);
+      const uint64_t h1B = d_swar_zero(x1 ^ pat1);
+      hitA &= (h1A >> 8) | ((h1B & 0x80) << 56);
+      hitB &= (h1B >> 8) | 0x8000000000000000ULL;
     }
     // mask hits before s0 (head window) -- hit bit for byte k is bit 8k+7
-    if (w < s0) hit &= ~((uint64_t(1) << ((s0 - w) * 8)) - 1);
-    while (hit) {
-      long k = long(__builtin_ctzll(hit) >> 3);
-      hit &= hit - 1;
-      long pos = w + k;
+    if (w < s0) {
+      const long off = s0 - w;
+      if (off >= 8) {
+        hitA = 0;
+        hitB &= ~((uint64_t(1) << ((off - 8) * 8)) - 1);
+      } else {
+        hitA &= ~((uint64_t(1) << (off * 8)) - 1);
+      }
+    }
+    while (hitA | hitB) {
+      long k;
+      if (hitA) {
+        k = long(__builtin_ctzll(hitA) >> 3);
+        hitA &= hitA - 1;
+      } else {
+        k = 8 + long(__builtin_ctzll(hitB) >> 3);
+        hitB &= hitB - 1;
+      }
+      const long pos = w + k;
       if (pos > last) return -1;
       // verify from byte 0: the SWAR zero-scan's borrow cascade can flag a
       // byte equal to c0^1 right after a true candidate ("101" vs "11")
@@ -385,6 +431,11 @@ struct BufAcc {
     uint64_t v;
     __builtin_memcpy(&v, base + off, 8);
     return v;
+  }
+  __device__ __forceinline__ void u64a2(long off, uint64_t* x0,
+                                        uint64_t* x1) const {
+    __builtin_memcpy(x0, base + off, 8);
+    __builtin_memcpy(x1, base + off + 8, 8);
   }
   __device__ __forceinline__ uint8_t u8(long off) const { return base[off]; }
 };
@@ -1238,7 +1289,7 @@ __device__ inline bool d_le_values_string(const uint8_t* ap, long an,
 struct DFieldSide {
   const uint8_t* p;
   long n;
-  char buf[40];
+  char buf[48];
 };
 
 __device__ inline void d_field_side_resolve(DFieldSide& out, uint8_t mode,
@@ -1386,7 +1437,7 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
         case 4: v = d_get_u32be(p); break;
         default: v = d_get_u64be(p); break;
       }
-      char buf[24];
+      char buf[32];
       int n = d_format_u64(buf, v);
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
@@ -1394,7 +1445,7 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
     case kScanRegexI: {
       uint64_t u = d_get_u64be(lb.data + size_t(row) * 8);
       int64_t v = int64_t(u >> 1) ^ (int64_t(u << 63) >> 63);
-      char buf[24];
+      char buf[32];
       int n = d_format_i64(buf, v);
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
@@ -1414,13 +1465,13 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
     case kScanPhraseF64: {
       // matchFloat64ByPhrase slow path (filter_phrase.go:175-186): format the
       // stored float with Ryu (== Go strconv 'f' -1) and substring-match
-      char buf[344];
+      char buf[352];
       int n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
       BufAcc a{(const uint8_t*)buf};
       return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
     }
     case kScanRegexF64: {
-      char buf[344];
+      char buf[352];
       int n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
@@ -1495,7 +1546,7 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
     case kScanAllPhrasesFmt:
     case kScanStrRangeFmt:
     case kScanLenRangeFmt: {
-      char buf[344];
+      char buf[352];
       int n;
       switch (lb.flags >> 4) {
         case kFmtU64: {
@@ -1637,7 +1688,7 @@ __device__ uint32_t d_gather_len(const DevGatherCol& gc, uint32_t row) {
       return uint32_t(d_u64_declen(uint64_t(v)));
     }
     case kGatherFmtF: {
-      char buf[344];
+      char buf[352];
       return uint32_t(vl_ryu::format_f64(buf, d_get_u64be(gc.data + size_t(row) * 8)));
     }
     case kGatherFmtIp: {
