@@ -168,6 +168,34 @@ __device__ __forceinline__ uint64_t d_swar_zero(uint64_t t) {
   return (t - 0x0101010101010101ULL) & ~t & 0x8080808080808080ULL;
 }
 
+// bytes [pos, pos+subn) == sub, compared 8 bytes per step through aligned
+// u64 windows (a byte-wise verify cost one swizzled LDS byte-load per
+// pattern byte — ~22 loads per row for the headline phrase)
+template <typename A>
+__device__ __forceinline__ bool d_bytes_eq_at(const A& a, long pos,
+                                              const uint8_t* sub, long subn) {
+  const long base = pos & ~7L;
+  const int sh = int(pos - base) * 8;
+  uint64_t cur = a.u64a(base);
+  long i = 0;
+  while (i < subn) {
+    const uint64_t nxt = a.u64a(base + i + 8);
+    const uint64_t v = sh ? (cur >> sh) | (nxt << (64 - sh)) : cur;
+    const long rem = subn - i;
+    uint64_t want = 0;
+    __builtin_memcpy(&want, sub + i, rem >= 8 ? 8 : size_t(rem));
+    if (rem >= 8) {
+      if (v != want) return false;
+    } else {
+      const uint64_t m = (uint64_t(1) << (rem * 8)) - 1;
+      if ((v ^ want) & m) return false;
+    }
+    cur = nxt;
+    i += 8;
+  }
+  return true;
+}
+
 template <typename A>
 __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
                            long subn) {
@@ -217,14 +245,7 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
       if (pos > last) return -1;
       // verify from byte 0: the SWAR zero-scan's borrow cascade can flag a
       // byte equal to c0^1 right after a true candidate ("101" vs "11")
-      bool eq = true;
-      for (long i = 0; i < subn; i++) {
-        if (a.u8(pos + i) != sub[i]) {
-          eq = false;
-          break;
-        }
-      }
-      if (eq) return pos - s0;
+      if (d_bytes_eq_at(a, pos, sub, subn)) return pos - s0;
     }
   }
   return -1;
@@ -1289,7 +1310,7 @@ __device__ inline bool d_le_values_string(const uint8_t* ap, long an,
 struct DFieldSide {
   const uint8_t* p;
   long n;
-  char buf[48];
+  char buf[64];
 };
 
 __device__ inline void d_field_side_resolve(DFieldSide& out, uint8_t mode,
@@ -1417,13 +1438,13 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
       return v >= mn && v <= mx;
     }
     case kScanPhraseIp: {
-      char buf[16];
+      char buf[32];
       int n = d_format_ipv4(buf, d_get_u32be(lb.data + size_t(row) * 4));
       BufAcc a{(const uint8_t*)buf};
       return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
     }
     case kScanPhraseIso: {
-      char buf[32];
+      char buf[48];
       int n = d_format_iso8601(buf, int64_t(d_get_u64be(lb.data + size_t(row) * 8)));
       BufAcc a{(const uint8_t*)buf};
       return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
@@ -1437,7 +1458,7 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
         case 4: v = d_get_u32be(p); break;
         default: v = d_get_u64be(p); break;
       }
-      char buf[32];
+      char buf[48];
       int n = d_format_u64(buf, v);
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
@@ -1445,19 +1466,19 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
     case kScanRegexI: {
       uint64_t u = d_get_u64be(lb.data + size_t(row) * 8);
       int64_t v = int64_t(u >> 1) ^ (int64_t(u << 63) >> 63);
-      char buf[32];
+      char buf[48];
       int n = d_format_i64(buf, v);
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
     }
     case kScanRegexIp: {
-      char buf[16];
+      char buf[32];
       int n = d_format_ipv4(buf, d_get_u32be(lb.data + size_t(row) * 4));
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
     }
     case kScanRegexIso: {
-      char buf[32];
+      char buf[48];
       int n = d_format_iso8601(buf, int64_t(d_get_u64be(lb.data + size_t(row) * 8)));
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
@@ -1465,13 +1486,13 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
     case kScanPhraseF64: {
       // matchFloat64ByPhrase slow path (filter_phrase.go:175-186): format the
       // stored float with Ryu (== Go strconv 'f' -1) and substring-match
-      char buf[352];
+      char buf[368];
       int n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
       BufAcc a{(const uint8_t*)buf};
       return d_match_phrase_at(a, 0, n, lb.operand, lb.operand_len, lb.flags);
     }
     case kScanRegexF64: {
-      char buf[352];
+      char buf[368];
       int n = vl_ryu::format_f64(buf, d_get_u64be(lb.data + size_t(row) * 8));
       BufAcc a{(const uint8_t*)buf};
       return d_regex_match_at(lb.operand, a, 0, n);
@@ -1546,7 +1567,7 @@ __device__ __noinline__ bool d_eval_fixed_row_cold(const DevLeafBlock& lb,
     case kScanAllPhrasesFmt:
     case kScanStrRangeFmt:
     case kScanLenRangeFmt: {
-      char buf[352];
+      char buf[368];
       int n;
       switch (lb.flags >> 4) {
         case kFmtU64: {
@@ -1688,7 +1709,7 @@ __device__ uint32_t d_gather_len(const DevGatherCol& gc, uint32_t row) {
       return uint32_t(d_u64_declen(uint64_t(v)));
     }
     case kGatherFmtF: {
-      char buf[352];
+      char buf[368];
       return uint32_t(vl_ryu::format_f64(buf, d_get_u64be(gc.data + size_t(row) * 8)));
     }
     case kGatherFmtIp: {
