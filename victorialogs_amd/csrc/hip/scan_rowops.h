@@ -167,34 +167,11 @@ __device__ uint32_t d_utf8_decode_last(const A& a, long off, long n, int* size) 
 __device__ __forceinline__ uint64_t d_swar_zero(uint64_t t) {
   return (t - 0x0101010101010101ULL) & ~t & 0x8080808080808080ULL;
 }
-
-// bytes [pos, pos+subn) == sub, compared 8 bytes per step through aligned
-// u64 windows (a byte-wise verify cost one swizzled LDS byte-load per
-// pattern byte — ~22 loads per row for the headline phrase)
-template <typename A>
-__device__ __forceinline__ bool d_bytes_eq_at(const A& a, long pos,
-                                              const uint8_t* sub, long subn) {
-  const long base = pos & ~7L;
-  const int sh = int(pos - base) * 8;
-  uint64_t cur = a.u64a(base);
-  long i = 0;
-  while (i < subn) {
-    const uint64_t nxt = a.u64a(base + i + 8);
-    const uint64_t v = sh ? (cur >> sh) | (nxt << (64 - sh)) : cur;
-    const long rem = subn - i;
-    uint64_t want = 0;
-    __builtin_memcpy(&want, sub + i, rem >= 8 ? 8 : size_t(rem));
-    if (rem >= 8) {
-      if (v != want) return false;
-    } else {
-      const uint64_t m = (uint64_t(1) << (rem * 8)) - 1;
-      if ((v ^ want) & m) return false;
-    }
-    cur = nxt;
-    i += 8;
-  }
-  return true;
-}
+// NOTE (measured dead end): replacing the byte-wise candidate verify with
+// aligned u64-window compares regressed EVERY config (headline 11.7 -> 9.2
+// G rows/s, non-matching 10.5 -> 18.4 ms) — the extra live values collapse
+// the hot clone's codegen (same class as the round-1 descriptor-reload
+// regression).  The byte loop below stays.
 
 template <typename A>
 __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
@@ -245,7 +222,14 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
       if (pos > last) return -1;
       // verify from byte 0: the SWAR zero-scan's borrow cascade can flag a
       // byte equal to c0^1 right after a true candidate ("101" vs "11")
-      if (d_bytes_eq_at(a, pos, sub, subn)) return pos - s0;
+      bool eq = true;
+      for (long i = 0; i < subn; i++) {
+        if (a.u8(pos + i) != sub[i]) {
+          eq = false;
+          break;
+        }
+      }
+      if (eq) return pos - s0;
     }
   }
   return -1;
