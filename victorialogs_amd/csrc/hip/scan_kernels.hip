@@ -297,6 +297,120 @@ __device__ __noinline__ void d_string_smallrow_loop(
   }
 }
 
+// Dual-phrase tile loop: two phrase leaves over the SAME string column
+// evaluate from one tile fill, halving the column's HBM reads (an OR/AND of
+// two phrases on one column — configs[3]'s or8 — otherwise streams the
+// column once per leaf).  Same structure as d_string_tile_loop; the copy is
+// shared, the two matchers run back-to-back from LDS.
+__device__ __noinline__ void d_string_phrase2_loop(
+    const uint8_t* __restrict__ col_data, const uint32_t* __restrict__ col_offs,
+    uint8_t* wtile, uint64_t* out1, uint64_t* out2, uint32_t r0, uint32_t r1,
+    uint32_t nwords, int lane, int wave, int nwaves, const uint8_t* p1,
+    uint32_t n1, uint8_t f1, const uint8_t* p2, uint32_t n2, uint8_t f2) {
+  typedef uint32_t v4u __attribute__((ext_vector_type(4)));
+  v4u* dst = (v4u*)wtile;
+  uint32_t wd = wave;
+  uint32_t o_lane = 0, o_end = 0;
+  if (wd < nwords) {
+    o_lane = col_offs[min(r0 + wd * 64 + uint32_t(lane), r1)];
+    if (lane == 0) o_end = col_offs[min(r0 + wd * 64 + 64, r1)];
+  }
+  while (wd < nwords) {
+    const uint32_t g0 = r0 + wd * 64;
+    const uint32_t g1 = min(g0 + 64, r1);
+    const uint32_t ng = g1 - g0;
+    const uint32_t byte0 =
+        uint32_t(__builtin_amdgcn_readfirstlane(int(o_lane))) & ~15u;
+    const uint32_t byte1 = uint32_t(__shfl(int(o_end), 0, 64));
+    const uint32_t nbytes = byte1 - byte0;
+    const bool use_tile = nbytes <= kWaveTileBytes;
+    if (use_tile) {
+      const v4u* src = (const v4u*)(col_data + byte0);
+      const uint32_t n16 = (((nbytes + 15) >> 4) + 63) & ~63u;
+      uint32_t k = lane;
+      for (; k + 448 < n16; k += 512) {
+        v4u a0 = __builtin_nontemporal_load(src + k);
+        v4u a1 = __builtin_nontemporal_load(src + k + 64);
+        v4u a2 = __builtin_nontemporal_load(src + k + 128);
+        v4u a3 = __builtin_nontemporal_load(src + k + 192);
+        v4u a4 = __builtin_nontemporal_load(src + k + 256);
+        v4u a5 = __builtin_nontemporal_load(src + k + 320);
+        v4u a6 = __builtin_nontemporal_load(src + k + 384);
+        v4u a7 = __builtin_nontemporal_load(src + k + 448);
+        dst[k ^ ((k >> 4) & 15)] = a0;
+        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+        dst[(k + 256) ^ (((k + 256) >> 4) & 15)] = a4;
+        dst[(k + 320) ^ (((k + 320) >> 4) & 15)] = a5;
+        dst[(k + 384) ^ (((k + 384) >> 4) & 15)] = a6;
+        dst[(k + 448) ^ (((k + 448) >> 4) & 15)] = a7;
+      }
+      for (; k + 192 < n16; k += 256) {
+        v4u a0 = src[k], a1 = src[k + 64], a2 = src[k + 128], a3 = src[k + 192];
+        dst[k ^ ((k >> 4) & 15)] = a0;
+        dst[(k + 64) ^ (((k + 64) >> 4) & 15)] = a1;
+        dst[(k + 128) ^ (((k + 128) >> 4) & 15)] = a2;
+        dst[(k + 192) ^ (((k + 192) >> 4) & 15)] = a3;
+      }
+      for (; k < n16; k += 64) dst[k ^ ((k >> 4) & 15)] = src[k];
+    }
+    const uint32_t next_wd = wd + nwaves;
+    uint32_t o_next = 0, o_end_next = 0;
+    if (next_wd < nwords) {
+      o_next = col_offs[min(r0 + next_wd * 64 + uint32_t(lane), r1)];
+      if (lane == 0) o_end_next = col_offs[min(r0 + next_wd * 64 + 64, r1)];
+    }
+    if (use_tile) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
+    bool pred1 = false, pred2 = false;
+    if (uint32_t(lane) < ng) {
+      const long s = o_lane;
+      const long e = uint32_t(__shfl(int(o_lane), lane + 1, 64));
+      const long e_fix = uint32_t(lane) == ng - 1 ? long(byte1) : e;
+      if (use_tile) {
+        TileAcc a{wtile};
+        pred1 = d_match_phrase_at(a, s - byte0, e_fix - s, p1, n1, f1);
+        pred2 = d_match_phrase_at(a, s - byte0, e_fix - s, p2, n2, f2);
+      } else {
+        GlobalAcc a{col_data};
+        pred1 = d_match_phrase_at(a, s, e_fix - s, p1, n1, f1);
+        pred2 = d_match_phrase_at(a, s, e_fix - s, p2, n2, f2);
+      }
+    }
+    const uint64_t word1 = __ballot(pred1);
+    const uint64_t word2 = __ballot(pred2);
+    if (lane == 0) {
+      out1[wd] = word1;
+      out2[wd] = word2;
+    }
+    o_lane = o_next;
+    o_end = o_end_next;
+    wd = next_wd;
+  }
+}
+
+// Cooperative bloom gate (bloomfilter.go:173-191); uniform control flow.
+__device__ __forceinline__ bool d_bloom_gate_ok(const DevLeafBlock& lb,
+                                                int tid, int bdim,
+                                                int* shared_flag) {
+  if (!lb.nhashes) return true;
+  if (tid == 0) *shared_flag = 1;
+  __syncthreads();
+  if (lb.bloom_words > 0) {
+    const uint64_t max_bits = uint64_t(lb.bloom_words) * 64;
+    bool miss = false;
+    for (uint32_t k = tid; k < lb.nhashes; k += bdim) {
+      uint64_t idx = lb.hashes[k] % max_bits;
+      if (((lb.bloom[idx >> 6] >> (idx & 63)) & 1) == 0) miss = true;
+    }
+    if (miss) *shared_flag = 0;
+  }
+  __syncthreads();
+  return *shared_flag != 0;
+}
+
 // ---- the program kernel ----
 
 __global__ __launch_bounds__(256) void scan_program_kernel(
@@ -325,6 +439,31 @@ __global__ __launch_bounds__(256) void scan_program_kernel(
     if (op.kind == kOpLeaf) {
       const DevLeafBlock& lb = lbs[size_t(ck.block) * nleaves + op.leaf];
       uint64_t* out = stack[sp];
+      // phrase-pair fusion: the next op is another phrase leaf over the
+      // SAME column -> evaluate both from one tile fill (halves this
+      // column's HBM reads).  Falls through to the single path when a
+      // bloom gate misses (rare; the single path re-evaluates it).
+      if (i + 1 < nops && ops[i + 1].kind == kOpLeaf &&
+          lb.kind == kScanPhraseStr && lb.mode == kModeScan && lb.sg <= 1 &&
+          sp + 1 < kMaxStackDepth) {
+        const DevLeafBlock& lb2 =
+            lbs[size_t(ck.block) * nleaves + ops[i + 1].leaf];
+        if (lb2.kind == kScanPhraseStr && lb2.mode == kModeScan &&
+            lb2.data == lb.data && lb2.offsets == lb.offsets) {
+          const bool ok1 = d_bloom_gate_ok(lb, tid, blockDim.x, &bloom_ok);
+          const bool ok2 = d_bloom_gate_ok(lb2, tid, blockDim.x, &bloom_ok);
+          if (ok1 && ok2) {
+            d_string_phrase2_loop(lb.data, lb.offsets,
+                                  tile + wave * kWaveTileBytes, stack[sp],
+                                  stack[sp + 1], r0, r1, nwords, lane, wave,
+                                  nwaves, lb.operand, lb.operand_len, lb.flags,
+                                  lb2.operand, lb2.operand_len, lb2.flags);
+            sp += 2;
+            i++;
+            continue;
+          }
+        }
+      }
       if (lb.mode == kModeNone || lb.mode == kModeAll) {
         uint64_t fill = lb.mode == kModeAll ? ~uint64_t(0) : 0;
         for (uint32_t w = tid; w < nwords; w += blockDim.x) out[w] = fill;
