@@ -343,3 +343,35 @@ def test_smallrow_supergroup_parity(tmp_path):
         '{"type":"len_range","field":"_msg","min":0,"max":0}',
     ]:
         assert_parity(d, f)
+
+
+def test_phrase_pair_fusion_parity(gen_part, tmp_path):
+    """Adjacent same-column phrase leaves fuse into one tile pass in the
+    kernel; results must stay bit-identical, including when one phrase's
+    bloom gate misses (fallback to the single-leaf path)."""
+    for f in [
+        # both present
+        '{"type":"or","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"the stream 1 and"},'
+        '{"type":"phrase","field":"_msg","phrase":"ip="}]}',
+        '{"type":"and","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"message"},'
+        '{"type":"phrase","field":"_msg","phrase":"uuid="}]}',
+        # one side bloom-gated off (token absent from every block)
+        '{"type":"or","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"zz_no_such_token"},'
+        '{"type":"phrase","field":"_msg","phrase":"message"}]}',
+        '{"type":"and","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"message"},'
+        '{"type":"phrase","field":"_msg","phrase":"zz_no_such_token"}]}',
+        # three in a row (fuses the first pair, single third)
+        '{"type":"or","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"worker"},'
+        '{"type":"phrase","field":"_msg","phrase":"stream"},'
+        '{"type":"phrase","field":"_msg","phrase":"ip="}]}',
+        # different columns adjacent: must NOT fuse
+        '{"type":"and","filters":['
+        '{"type":"phrase","field":"_msg","phrase":"message"},'
+        '{"type":"phrase","field":"dict_0","phrase":"error"}]}',
+    ]:
+        assert_parity(gen_part, f)
