@@ -180,13 +180,7 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
   if (subn > sn) return -1;
   const uint8_t c0 = sub[0];
   const uint64_t pat = 0x0101010101010101ULL * c0;
-  // second anchor byte: the pattern's LAST byte when it is within 7 bytes
-  // of the start (text sharing a prefix with the pattern — "worker 0" vs
-  // the phrase "worker 3" — defeats a second-byte anchor and made every
-  // row pay a divergent verify), else byte 1
-  const long d = subn <= 8 ? subn - 1 : 1;
-  const uint64_t pat1 = subn > 1 ? 0x0101010101010101ULL * sub[d] : 0;
-  const int dsh = int(d) * 8;
+  const uint64_t pat1 = subn > 1 ? 0x0101010101010101ULL * sub[1] : 0;
   const long last = s0 + sn - subn;  // last valid start (absolute)
   // 16-byte windows: one ds_read_b128 per iteration (a swizzled tile slot
   // holds the whole window), halving the serial per-window chain of the
@@ -197,13 +191,13 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
     uint64_t hitA = d_swar_zero(x0 ^ pat);
     uint64_t hitB = d_swar_zero(x1 ^ pat);
     if (subn > 1 && (hitA | hitB)) {
-      // two-byte anchor: require the anchor byte at k+d (the zero-scan has
-      // no false negatives, so pruning is sound; candidates whose anchor
-      // position lies beyond this 16-byte window keep their bit)
+      // two-byte anchor: require the second pattern byte at k+1 (the
+      // zero-scan has no false negatives, so pruning is sound; the last
+      // byte's successor lives in the next window and keeps its bit)
       const uint64_t h1A = d_swar_zero(x0 ^ pat1);
       const uint64_t h1B = d_swar_zero(x1 ^ pat1);
-      hitA &= (h1A >> dsh) | (h1B << (64 - dsh));
-      hitB &= (h1B >> dsh) | (~uint64_t(0) << (64 - dsh));
+      hitA &= (h1A >> 8) | ((h1B & 0x80) << 56);
+      hitB &= (h1B >> 8) | 0x8000000000000000ULL;
     }
     // mask hits before s0 (head window) -- hit bit for byte k is bit 8k+7
     if (w < s0) {
