@@ -447,8 +447,17 @@ def main():
     # generator-known expected count != rows must match it exactly — a
     # kernel writing all-ones bitmaps fails here
     sel = None
-    if (not args.skip_selectivity and manifest is not None
-            and args.workload in ("phrase", "phrase_regex")):
+    run_sel = (not args.skip_selectivity and manifest is not None
+               and args.workload in ("phrase", "phrase_regex"))
+    if distributed:
+        # ranks must agree (a rank that fell back to stream-parts has no
+        # files left to restage and skips; mismatched schedules would hang
+        # the all_reduce below)
+        t = torch.tensor([0.0 if run_sel else 1.0], dtype=torch.float64,
+                         device=coll_dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        run_sel = t.item() == 0.0
+    if run_sel:
         log("selectivity self-check (phrase ip=77.)...")
         sfilt = Filter(SELECTIVE_FILTER)
         sstage = Stage(parts, sfilt, device=device)
