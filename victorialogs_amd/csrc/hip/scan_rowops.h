@@ -182,24 +182,34 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
   const uint64_t pat = 0x0101010101010101ULL * c0;
   const uint64_t pat1 = subn > 1 ? 0x0101010101010101ULL * sub[1] : 0;
   const long last = s0 + sn - subn;  // last valid start (absolute)
-  // 16-byte windows: one ds_read_b128 per iteration (a swizzled tile slot
-  // holds the whole window), halving the serial per-window chain of the
-  // 8-byte version
-  for (long w = s0 & ~15L; w <= last; w += 16) {
-    uint64_t x0, x1;
+  // 32-byte iterations: two ds_read_b128 windows issued together (a
+  // swizzled tile slot holds one 16-byte window), overlapping the LDS
+  // latency that serialized the original one-8-byte-window-per-iteration
+  // loop.  Scratch-buffer accessors are sized for reads up to 32 bytes
+  // past the aligned start of the last valid position.
+  for (long w = s0 & ~15L; w <= last; w += 32) {
+    uint64_t x0, x1, x2, x3;
     a.u64a2(w, &x0, &x1);
+    a.u64a2(w + 16, &x2, &x3);
     uint64_t hitA = d_swar_zero(x0 ^ pat);
     uint64_t hitB = d_swar_zero(x1 ^ pat);
-    if (subn > 1 && (hitA | hitB)) {
+    uint64_t hitC = d_swar_zero(x2 ^ pat);
+    uint64_t hitD = d_swar_zero(x3 ^ pat);
+    if (subn > 1 && (hitA | hitB | hitC | hitD)) {
       // two-byte anchor: require the second pattern byte at k+1 (the
       // zero-scan has no false negatives, so pruning is sound; the last
       // byte's successor lives in the next window and keeps its bit)
       const uint64_t h1A = d_swar_zero(x0 ^ pat1);
       const uint64_t h1B = d_swar_zero(x1 ^ pat1);
+      const uint64_t h1C = d_swar_zero(x2 ^ pat1);
+      const uint64_t h1D = d_swar_zero(x3 ^ pat1);
       hitA &= (h1A >> 8) | ((h1B & 0x80) << 56);
-      hitB &= (h1B >> 8) | 0x8000000000000000ULL;
+      hitB &= (h1B >> 8) | ((h1C & 0x80) << 56);
+      hitC &= (h1C >> 8) | ((h1D & 0x80) << 56);
+      hitD &= (h1D >> 8) | 0x8000000000000000ULL;
     }
-    // mask hits before s0 (head window) -- hit bit for byte k is bit 8k+7
+    // mask hits before s0 (first iteration only) -- hit bit for byte k is
+    // bit 8k+7; s0 - w is in [0, 15]
     if (w < s0) {
       const long off = s0 - w;
       if (off >= 8) {
@@ -209,14 +219,20 @@ __device__ long d_index_at(const A& a, long s0, long sn, const uint8_t* sub,
         hitA &= ~((uint64_t(1) << (off * 8)) - 1);
       }
     }
-    while (hitA | hitB) {
+    while (hitA | hitB | hitC | hitD) {
       long k;
       if (hitA) {
         k = long(__builtin_ctzll(hitA) >> 3);
         hitA &= hitA - 1;
-      } else {
+      } else if (hitB) {
         k = 8 + long(__builtin_ctzll(hitB) >> 3);
         hitB &= hitB - 1;
+      } else if (hitC) {
+        k = 16 + long(__builtin_ctzll(hitC) >> 3);
+        hitC &= hitC - 1;
+      } else {
+        k = 24 + long(__builtin_ctzll(hitD) >> 3);
+        hitD &= hitD - 1;
       }
       const long pos = w + k;
       if (pos > last) return -1;
