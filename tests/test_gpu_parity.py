@@ -375,3 +375,41 @@ def test_phrase_pair_fusion_parity(gen_part, tmp_path):
         '{"type":"phrase","field":"dict_0","phrase":"error"}]}',
     ]:
         assert_parity(gen_part, f)
+
+
+def test_smallrow_anycase_override_parity(tmp_path):
+    """Short rows (super-group loop) + any-case filters + non-ASCII rows:
+    the host-resolved override bitmaps must merge per word at the right
+    indexes in d_string_smallrow_loop (kOvr path)."""
+    from victorialogs_amd import write_custom_part
+
+    rows = 2500
+    vals = []
+    for i in range(rows):
+        if i % 7 == 0:
+            vals.append(f"Straße {i} GRÜN")       # non-ASCII: host override
+        elif i % 7 == 3:
+            vals.append(f"MiXeD needle_{i % 4}")
+        elif i % 97 == 0:
+            vals.append("")
+        else:
+            vals.append(f"plain needle_{i % 4} t{i % 9}")
+    spec = {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i for i in range(rows)],
+        "columns": [{"name": "_msg", "values": vals}],
+    }]}
+    d = str(tmp_path / "anycase_small")
+    write_custom_part(d, spec)
+    for f in [
+        '{"type":"any_case_phrase","field":"_msg","phrase":"NEEDLE_2"}',
+        '{"type":"any_case_phrase","field":"_msg","phrase":"mixed"}',
+        '{"type":"any_case_phrase","field":"_msg","phrase":"strasse"}',
+        '{"type":"any_case_phrase","field":"_msg","phrase":"grün"}',
+        '{"type":"any_case_prefix","field":"_msg","prefix":"sTrAß"}',
+        '{"type":"any_case_prefix","field":"_msg","prefix":"PLAIN"}',
+        '{"type":"and","filters":['
+        '{"type":"any_case_phrase","field":"_msg","phrase":"needle_1"},'
+        '{"type":"phrase","field":"_msg","phrase":"t5"}]}',
+    ]:
+        assert_parity(d, f)
