@@ -176,3 +176,56 @@ def test_regex_lazy_fuzz_vs_python():
         if ok:
             checked += 1
     assert checked > 400
+
+
+def test_wide_nfa_65_to_128_positions():
+    """Patterns needing 65..128 Glushkov positions use the wide (two-word)
+    NFA blob on oracle and device; >128 still rejects loudly."""
+    import re as _re
+
+    cases = [
+        # [ab]{65}: 65 class positions -> wide
+        ("[ab]{65}", ["a" * 64, "a" * 65, "ab" * 40, "c" + "b" * 70]),
+        # 12 x 6-char alternative = 72 positions
+        ("(foo|bar){12}", ["foobar" * 6, "foobar" * 5, "barbar" * 6 + "x"]),
+        ("x[0-9]{79}y", ["x" + "5" * 79 + "y", "x" + "5" * 78 + "y"]),
+        ("^[ab]{65}$", ["a" * 65, "a" * 66, "b" * 65]),
+    ]
+    for name, match in _engines():
+        for pat, subjects in cases:
+            cre = _re.compile(pat)
+            for subj in subjects:
+                got = match(pat.encode(), subj.encode())
+                want = 1 if cre.search(subj) else 0
+                assert got == want, (
+                    f"[{name}] pat={pat!r} s={subj!r} got={got} want={want}")
+        # beyond 128 positions: loud reject
+        assert match(b"[ab]{129}", b"a" * 130) < 0, name
+        assert match(b"(foo|bar){24}", b"foobar" * 12) < 0, name
+
+
+def test_wide_nfa_fuzz_vs_python():
+    lib = oracle_helpers()
+    lib.orc_regex_match.restype = ctypes.c_long
+    rng = random.Random(11)
+    checked = 0
+    for _ in range(200):
+        n = rng.randrange(65, 120)
+        cls = rng.choice(["[ab]", "[a-c]", "(a|b0)"])
+        pat = "%s{%d}" % (cls, n // (2 if cls == "(a|b0)" else 1))
+        try:
+            cre = pyre.compile(pat)
+        except pyre.error:
+            continue
+        pb = pat.encode()
+        for _ in range(4):
+            s = "".join(rng.choice("abc0") for _ in range(rng.randrange(50, 150)))
+            sb = s.encode()
+            r = lib.orc_regex_match(pb, len(pb), sb, len(sb))
+            if r < 0:
+                break
+            want = 1 if cre.search(s) else 0
+            assert r == want, f"pat={pat!r} s={s!r} ours={r} py={want}"
+        else:
+            checked += 1
+    assert checked > 100

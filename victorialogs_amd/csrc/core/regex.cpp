@@ -184,8 +184,8 @@ struct Parser {
           if (m > 1000 || nrep > 1000) err("invalid repeat count");
           if (nrep >= 0 && nrep < m) err("invalid repeat count");
           const long max_copies = nrep < 0 ? m + 1 : nrep;
-          if (max_copies > 64) {
-            err("repetition too large for the 64-position NFA");
+          if (max_copies > 128) {
+            err("repetition too large for the 128-position NFA");
           }
           // X{m,n} = X^m (X (X ... )?)? ; X{m,} = X^m X*
           RNode expanded;
@@ -623,7 +623,7 @@ void collect_literals(const RNode& raw, std::vector<std::string>& out) {
 
 // ---- Glushkov position automaton for the general class ----
 // Positions (atoms) are byte-consuming; first/last/follow sets are u64 masks
-// (<= 64 positions).  Multi-byte runes (Dot, negated classes) use a relaxed
+// (<= 128 positions; two-u64 masks, narrow blob when <= 64).  Multi-byte runes (Dot, negated classes) use a relaxed
 // UTF-8 shape: [00-7F] | [C0-DF][80-BF] | [E0-EF][80-BF]{2} | [F0-F4][80-BF]{3}
 // -- identical to Go on valid UTF-8 input (DESIGN.md notes the invalid-UTF-8
 // divergence).
@@ -632,26 +632,51 @@ struct GAtom {
   uint8_t set[32];
 };
 
+// position mask: up to 128 byte-level positions (two u64 words); patterns
+// with <= 64 positions serialize to the narrow single-word blob
+struct PMask {
+  uint64_t w0 = 0, w1 = 0;
+  void set(int i) { (i < 64 ? w0 : w1) |= uint64_t(1) << (i & 63); }
+  void operator|=(const PMask& o) {
+    w0 |= o.w0;
+    w1 |= o.w1;
+  }
+};
+
+template <typename F>
+static void pm_foreach(const PMask& m, F f) {
+  uint64_t v = m.w0;
+  while (v) {
+    f(__builtin_ctzll(v));
+    v &= v - 1;
+  }
+  v = m.w1;
+  while (v) {
+    f(64 + __builtin_ctzll(v));
+    v &= v - 1;
+  }
+}
+
 struct GBuild {
   std::vector<GAtom> atoms;
-  std::vector<uint64_t> follow;
+  std::vector<PMask> follow;
   [[noreturn]] void overflow(const std::string& expr) {
     fail("regex: NFA fallback for \"" + expr +
-         "\" needs more than 64 positions; simplify the pattern");
+         "\" needs more than 128 positions; simplify the pattern");
   }
 };
 
 struct GInfo {
   bool nullable;
-  uint64_t first, last;
+  PMask first, last;
 };
 
 static int g_add_atom(GBuild& b, const uint8_t* set, const std::string& expr) {
-  if (b.atoms.size() >= 64) b.overflow(expr);
+  if (b.atoms.size() >= 128) b.overflow(expr);
   GAtom a;
   memcpy(a.set, set, 32);
   b.atoms.push_back(a);
-  b.follow.push_back(0);
+  b.follow.push_back(PMask{});
   return int(b.atoms.size()) - 1;
 }
 
@@ -661,30 +686,43 @@ static void g_range_set(uint8_t* set, int lo, int hi) {
 
 static GInfo g_cat(GBuild& b, GInfo x, GInfo y) {
   // follow: last(x) -> first(y)
-  uint64_t m = x.last;
-  while (m) {
-    int i = __builtin_ctzll(m);
-    m &= m - 1;
-    b.follow[i] |= y.first;
-  }
+  pm_foreach(x.last, [&](int i) { b.follow[i] |= y.first; });
   GInfo r;
   r.nullable = x.nullable && y.nullable;
-  r.first = x.first | (x.nullable ? y.first : 0);
-  r.last = y.last | (y.nullable ? x.last : 0);
+  r.first = x.first;
+  if (x.nullable) r.first |= y.first;
+  r.last = y.last;
+  if (y.nullable) r.last |= x.last;
   return r;
 }
 
 static GInfo g_alt(GInfo x, GInfo y) {
-  return GInfo{x.nullable || y.nullable, x.first | y.first, x.last | y.last};
+  GInfo r;
+  r.nullable = x.nullable || y.nullable;
+  r.first = x.first;
+  r.first |= y.first;
+  r.last = x.last;
+  r.last |= y.last;
+  return r;
 }
 
 static void g_loop(GBuild& b, const GInfo& x) {
-  uint64_t m = x.last;
-  while (m) {
-    int i = __builtin_ctzll(m);
-    m &= m - 1;
-    b.follow[i] |= x.first;
-  }
+  pm_foreach(x.last, [&](int i) { b.follow[i] |= x.first; });
+}
+
+
+static GInfo g_pos(int a) {
+  GInfo r;
+  r.nullable = false;
+  r.first.set(a);
+  r.last.set(a);
+  return r;
+}
+
+static GInfo g_empty(bool nullable) {
+  GInfo r;
+  r.nullable = nullable;
+  return r;
 }
 
 // relaxed UTF-8 multi-byte rune
@@ -695,11 +733,9 @@ static GInfo g_multibyte(GBuild& b, const std::string& expr) {
   g_range_set(l3, 0xE0, 0xEF);
   g_range_set(l4, 0xF0, 0xF4);
   auto seq = [&](const uint8_t* lead, int ncont) {
-    int a = g_add_atom(b, lead, expr);
-    GInfo r{false, uint64_t(1) << a, uint64_t(1) << a};
+    GInfo r = g_pos(g_add_atom(b, lead, expr));
     for (int i = 0; i < ncont; i++) {
-      int c = g_add_atom(b, cont, expr);
-      r = g_cat(b, r, GInfo{false, uint64_t(1) << c, uint64_t(1) << c});
+      r = g_cat(b, r, g_pos(g_add_atom(b, cont, expr)));
     }
     return r;
   };
@@ -745,34 +781,31 @@ static RNode compress_classes(const RNode& n) {
 static GInfo g_build(GBuild& b, const RNode& n, const std::string& expr) {
   switch (n.kind) {
     case RNode::Empty:
-      return GInfo{true, 0, 0};
+      return g_empty(true);
     case RNode::Group:
       return g_build(b, n.subs[0], expr);
     case RNode::Lit: {
-      GInfo r{true, 0, 0};
+      GInfo r = g_empty(true);
       for (unsigned char c : n.lit) {
         uint8_t set[32] = {0};
         set[c >> 3] = uint8_t(1) << (c & 7);
-        int a = g_add_atom(b, set, expr);
-        r = g_cat(b, r, GInfo{false, uint64_t(1) << a, uint64_t(1) << a});
+        r = g_cat(b, r, g_pos(g_add_atom(b, set, expr)));
       }
       return r;
     }
     case RNode::Dot: {
       uint8_t ascii[32] = {0};
       g_range_set(ascii, 0x00, 0x7F);  // (?s) DotNL: '.' matches any rune
-      int a = g_add_atom(b, ascii, expr);
-      GInfo r{false, uint64_t(1) << a, uint64_t(1) << a};
+      GInfo r = g_pos(g_add_atom(b, ascii, expr));
       return g_alt(r, g_multibyte(b, expr));
     }
     case RNode::Class: {
-      int a = g_add_atom(b, n.cls, expr);
-      GInfo r{false, uint64_t(1) << a, uint64_t(1) << a};
+      GInfo r = g_pos(g_add_atom(b, n.cls, expr));
       if (n.cls_nonascii) r = g_alt(r, g_multibyte(b, expr));
       return r;
     }
     case RNode::Concat: {
-      GInfo r{true, 0, 0};
+      GInfo r = g_empty(true);
       for (const auto& sub : n.subs) r = g_cat(b, r, g_build(b, sub, expr));
       return r;
     }
@@ -807,25 +840,31 @@ static bytes g_serialize(const GBuild& b, const GInfo& root,
                          bool a_start = false, bool a_end = false,
                          bool nullable = false) {
   bytes out;
-  uint16_t n = uint16_t(b.atoms.size());
+  const uint16_t n = uint16_t(b.atoms.size());
+  const bool wide = n > 64;  // two-word position masks (65..128 positions)
   out.push_back(uint8_t(n));
   out.push_back(uint8_t(n >> 8));
-  // byte 2: anchor flags (1 = ^ anchored, 2 = $ anchored, 4 = nullable root)
+  // byte 2: flags (1 = ^ anchored, 2 = $ anchored, 4 = nullable root,
+  // 8 = wide/128-position layout: all masks are 16-byte pairs)
   out.push_back(uint8_t((a_start ? 1 : 0) | (a_end ? 2 : 0) |
-                        (nullable ? 4 : 0)));
+                        (nullable ? 4 : 0) | (wide ? 8 : 0)));
   out.resize(8, 0);
   auto put64 = [&](uint64_t v) {
     for (int i = 0; i < 8; i++) out.push_back(uint8_t(v >> (8 * i)));
   };
-  put64(root.first);
-  put64(root.last);
-  for (uint16_t i = 0; i < n; i++) put64(b.follow[i]);
+  auto put_mask = [&](const PMask& m) {
+    put64(m.w0);
+    if (wide) put64(m.w1);
+  };
+  put_mask(root.first);
+  put_mask(root.last);
+  for (uint16_t i = 0; i < n; i++) put_mask(b.follow[i]);
   for (int c = 0; c < 256; c++) {
-    uint64_t m = 0;
+    PMask m;
     for (uint16_t i = 0; i < n; i++) {
-      if ((b.atoms[i].set[c >> 3] >> (c & 7)) & 1) m |= uint64_t(1) << i;
+      if ((b.atoms[i].set[c >> 3] >> (c & 7)) & 1) m.set(i);
     }
-    put64(m);
+    put_mask(m);
   }
   return out;
 }
@@ -1106,36 +1145,48 @@ bool nfa_match(const uint8_t* blob, strview s) {
   uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
   const uint8_t anchors = blob[2];
   const bool a_start = anchors & 1, a_end = anchors & 2;
-  auto rd64 = [&](size_t off) {
-    uint64_t v;
-    memcpy(&v, blob + off, 8);
-    return v;
+  const bool wide = (anchors & 8) != 0;  // 65..128 positions: 16-byte masks
+  const size_t msz = wide ? 16 : 8;
+  auto rd = [&](const uint8_t* p, uint64_t* w0, uint64_t* w1) {
+    memcpy(w0, p, 8);
+    if (wide) {
+      memcpy(w1, p + 8, 8);
+    } else {
+      *w1 = 0;
+    }
   };
-  const uint64_t first = rd64(8);
-  const uint64_t last = rd64(16);
-  const uint8_t* follow = blob + 24;
-  const uint8_t* table = blob + 24 + size_t(n) * 8;
+  uint64_t first0, first1, last0, last1;
+  rd(blob + 8, &first0, &first1);
+  rd(blob + 8 + msz, &last0, &last1);
+  const uint8_t* follow = blob + 8 + 2 * msz;
+  const uint8_t* table = follow + size_t(n) * msz;
   if (s.n == 0) return (anchors & 4) != 0;  // nullable root matches ""
-  uint64_t active = 0;
+  uint64_t active0 = 0, active1 = 0;
   for (size_t i = 0; i < s.n; i++) {
     // '^' anchored: new matches may start only at offset 0
-    uint64_t targets = (a_start && i > 0) ? 0 : first;
-    uint64_t m = active;
-    while (m) {
-      int x = __builtin_ctzll(m);
-      m &= m - 1;
-      uint64_t f;
-      memcpy(&f, follow + size_t(x) * 8, 8);
-      targets |= f;
-    }
-    uint64_t tb;
-    memcpy(&tb, table + size_t(uint8_t(s.p[i])) * 8, 8);
-    const uint64_t entered = targets & tb;
-    if (!a_end && (entered & last)) return true;
-    active = entered;
+    uint64_t t0 = (a_start && i > 0) ? 0 : first0;
+    uint64_t t1 = (a_start && i > 0) ? 0 : first1;
+    auto accum = [&](uint64_t m, int base) {
+      while (m) {
+        int x = base + __builtin_ctzll(m);
+        m &= m - 1;
+        uint64_t f0, f1;
+        rd(follow + size_t(x) * msz, &f0, &f1);
+        t0 |= f0;
+        t1 |= f1;
+      }
+    };
+    accum(active0, 0);
+    accum(active1, 64);
+    uint64_t tb0, tb1;
+    rd(table + size_t(uint8_t(s.p[i])) * msz, &tb0, &tb1);
+    const uint64_t e0 = t0 & tb0, e1 = t1 & tb1;
+    if (!a_end && ((e0 & last0) | (e1 & last1))) return true;
+    active0 = e0;
+    active1 = e1;
   }
   // '$' anchored: accept only with a final position active at string end
-  return a_end && (active & last) != 0;
+  return a_end && ((active0 & last0) | (active1 & last1)) != 0;
 }
 
 bool regex_match(const RegexProg& re, strview s) {

@@ -60,7 +60,7 @@ bool nfa_match(const uint8_t* blob, strview s);
 // and per-alternative '^'/'$' anchors, and a leading (?i) (simple case
 // closure + the Unicode CaseFolding special orbits).  Still rejected with
 // a clear error: \b/\B, \p{...}, mid-pattern anchors, (?...) flags other
-// than a leading (?i), and >64 NFA positions.
+// than a leading (?i), and >128 NFA positions.
 RegexProg regex_compile(const std::string& expr);
 
 // Regex.MatchString (regex.go:86-98,131-212).

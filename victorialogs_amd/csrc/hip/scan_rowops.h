@@ -1070,12 +1070,59 @@ __device__ bool d_regex_or_hasprefix(const DRegex& re, const A& a, long s0, long
 }
 
 // Glushkov NFA executor (general regex class; see core/regex.cpp g_build).
-// blob: u16 nstates, pad to 8, u64 first, u64 last, u64 follow[n], u64 table[256]
+// blob: u16 nstates, u8 flags (1 '^', 2 '$', 4 nullable, 8 wide), pad to 8,
+// then first/last/follow[n]/table[256] masks — 8-byte masks, or 16-byte
+// pairs when the wide flag is set (65..128 positions).
 template <typename A>
 __device__ bool d_nfa_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
   const uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
-  const uint8_t anchors = blob[2];  // 1 = '^', 2 = '$', 4 = nullable root
+  const uint8_t anchors = blob[2];
   const bool a_start = anchors & 1, a_end = anchors & 2;
+  if (anchors & 8) {
+    // wide: two-word position masks
+    const uint8_t* p = blob + 8;
+    uint64_t first0, first1, last0, last1;
+    __builtin_memcpy(&first0, p, 8);
+    __builtin_memcpy(&first1, p + 8, 8);
+    __builtin_memcpy(&last0, p + 16, 8);
+    __builtin_memcpy(&last1, p + 24, 8);
+    const uint8_t* follow = p + 32;
+    const uint8_t* table = follow + size_t(n) * 16;
+    if (sn == 0) return (anchors & 4) != 0;
+    uint64_t active0 = 0, active1 = 0;
+    for (long i = 0; i < sn; i++) {
+      uint64_t t0 = (a_start && i > 0) ? 0 : first0;
+      uint64_t t1 = (a_start && i > 0) ? 0 : first1;
+      uint64_t m = active0;
+      while (m) {
+        int x = __builtin_ctzll(m);
+        m &= m - 1;
+        uint64_t f0, f1;
+        __builtin_memcpy(&f0, follow + size_t(x) * 16, 8);
+        __builtin_memcpy(&f1, follow + size_t(x) * 16 + 8, 8);
+        t0 |= f0;
+        t1 |= f1;
+      }
+      m = active1;
+      while (m) {
+        int x = 64 + __builtin_ctzll(m);
+        m &= m - 1;
+        uint64_t f0, f1;
+        __builtin_memcpy(&f0, follow + size_t(x) * 16, 8);
+        __builtin_memcpy(&f1, follow + size_t(x) * 16 + 8, 8);
+        t0 |= f0;
+        t1 |= f1;
+      }
+      uint64_t tb0, tb1;
+      __builtin_memcpy(&tb0, table + size_t(a.u8(s0 + i)) * 16, 8);
+      __builtin_memcpy(&tb1, table + size_t(a.u8(s0 + i)) * 16 + 8, 8);
+      const uint64_t e0 = t0 & tb0, e1 = t1 & tb1;
+      if (!a_end && ((e0 & last0) | (e1 & last1))) return true;
+      active0 = e0;
+      active1 = e1;
+    }
+    return a_end && ((active0 & last0) | (active1 & last1)) != 0;
+  }
   uint64_t first, last;
   __builtin_memcpy(&first, blob + 8, 8);
   __builtin_memcpy(&last, blob + 16, 8);
