@@ -413,3 +413,35 @@ def test_smallrow_anycase_override_parity(tmp_path):
         '{"type":"phrase","field":"_msg","phrase":"t5"}]}',
     ]:
         assert_parity(d, f)
+
+
+def test_wide_nfa_parity(tmp_path):
+    """65..128-position regexes execute the wide (two-word) NFA blob on
+    device, bit-identical to the oracle."""
+    from victorialogs_amd import write_custom_part
+
+    rows = 600
+    vals = []
+    for i in range(rows):
+        if i % 5 == 0:
+            vals.append("a" * (60 + i % 20))
+        elif i % 5 == 1:
+            vals.append("x" + "7" * (70 + i % 15) + "y")
+        elif i % 5 == 2:
+            vals.append("foobar" * (4 + i % 10))
+        else:
+            vals.append(f"plain text {i} ab{'c' * (i % 90)}")
+    spec = {"blocks": [{
+        "stream": 0,
+        "timestamps": [1700000000000000000 + i for i in range(rows)],
+        "columns": [{"name": "_msg", "values": vals}],
+    }]}
+    d = str(tmp_path / "widenfa")
+    write_custom_part(d, spec)
+    for f in [
+        '{"type":"regexp","field":"_msg","re":"[ab]{65}"}',
+        '{"type":"regexp","field":"_msg","re":"x[0-9]{79}y"}',
+        '{"type":"regexp","field":"_msg","re":"(foo|bar){12}"}',
+        '{"type":"regexp","field":"_msg","re":"^[ac-z ]{70}"}',
+    ]:
+        assert_parity(d, f)
