@@ -30,7 +30,7 @@ extern "C" {
 
 const char* vql_errstr(void);
 /* 1 if the last error was an unsupported-construct rejection (valid LogsQL
- * outside the engine's class, e.g. regex \b or \p{...}) rather than
+ * outside the engine class, e.g. regex \p{...}) rather than
  * corruption/IO — the cgo shim keeps such filters on the host Go path
  * (INTEGRATION.md); the engine never silently falls back itself. */
 int vql_error_unsupported(void);

@@ -41,7 +41,7 @@ def test_filter_compile_errors():
         '{"type":"phrase"}',                       # missing fields
         '{"type":"wat","field":"x","phrase":"y"}', # unknown type
         'not json at all',
-        '{"type":"regexp","field":"x","re":"a\\\\b"}',  # word-boundary assertion
+        '{"type":"regexp","field":"x","re":"a\\\\p{L}"}',  # unicode class
         '{"type":"regexp","field":"x","re":"a{2000}"}', # repeat count too big
         '{"type":"regexp","field":"x","re":"(a|^b)c"}', # mid-pattern anchor
     ]
@@ -88,7 +88,7 @@ def test_error_unsupported_classification():
     lib.vql_error_unsupported.restype = ctypes.c_int
     # residual regex class: loud reject, classified unsupported
     f = lib.vql_compile_filter(
-        b'{"type":"regexp","field":"x","re":"a\\\\b"}')
+        b'{"type":"regexp","field":"x","re":"\\\\p{L}x"}')
     assert not f
     assert b"regex" in lib.vql_errstr()
     assert lib.vql_error_unsupported() == 1
@@ -99,7 +99,8 @@ def test_error_unsupported_classification():
     # supported since round 2: these must compile
     for ok in [b'{"type":"regexp","field":"x","re":"^01|04$"}',
                b'{"type":"regexp","field":"x","re":"(?i)foo"}',
-               b'{"type":"regexp","field":"x","re":"ab+?"}']:
+               b'{"type":"regexp","field":"x","re":"ab+?"}',
+               b'{"type":"regexp","field":"x","re":"\\\\bfoo\\\\b"}']:
         h = lib.vql_compile_filter(ok)
         assert h, lib.vql_errstr()
         lib.vql_free_filter(h)

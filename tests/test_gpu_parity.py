@@ -116,7 +116,7 @@ def test_unsupported_regex_fails_at_compile(typed_part):
     """Unsupported constructs must raise a clear error at compile time,
     never fall back to CPU silently."""
     with pytest.raises(RuntimeError, match="not supported|fast-path"):
-        Filter('{"type":"regexp","field":"_msg","re":"a\\\\bb"}')
+        Filter('{"type":"regexp","field":"_msg","re":"a\\\\p{L}b"}')
 
 
 def test_multichunk_block(tmp_path):
@@ -445,3 +445,15 @@ def test_wide_nfa_parity(tmp_path):
         '{"type":"regexp","field":"_msg","re":"^[ac-z ]{70}"}',
     ]:
         assert_parity(d, f)
+
+
+def test_word_boundary_regex_parity(gen_part):
+    """\\b/\\B regexes (assert-layout NFA blobs) on device vs oracle."""
+    for f in [
+        '{"type":"regexp","field":"_msg","re":"\\\\bstream\\\\b"}',
+        '{"type":"regexp","field":"_msg","re":"\\\\bip=\\\\d+\\\\b"}',
+        '{"type":"regexp","field":"_msg","re":"uuid\\\\B"}',
+        '{"type":"regexp","field":"_msg","re":"\\\\bfor\\\\b.*\\\\bstream"}',
+        '{"type":"regexp","field":"var_0","re":"value\\\\b \\\\b\\\\d"}',
+    ]:
+        assert_parity(gen_part, f)

@@ -115,7 +115,7 @@ CASES = [
     ("(?i)σ", "Σ", 1), ("(?i)Σ", "ς", 1),  # sigma orbit via closure
 ]
 
-REJECTS = ["a**", "a{2}{3}", "a\\b", "\\p{L}", "(?m)a", "(?i:a)b", "a*+"]
+REJECTS = ["a**", "a{2}{3}", "\\p{L}", "(?m)a", "(?i:a)b", "a*+"]
 
 
 def _engines():
@@ -229,3 +229,70 @@ def test_wide_nfa_fuzz_vs_python():
         else:
             checked += 1
     assert checked > 100
+
+
+# ---- word-boundary assertions (\b/\B), supported since round 2 ----
+
+BOUNDARY_CASES = [
+    ('\\bfoo', [('foo bar', 1), ('xfoo', 0), (' foo', 1), ('9foo', 0), ('-foo', 1)]),
+    ('foo\\b', [('foo bar', 1), ('foox', 0), ('foo9', 0), ('foo', 1)]),
+    ('\\bfoo\\b', [('a foo b', 1), ('afoob', 0), ('foo.bar', 1), ('xfoo ', 0)]),
+    ('\\b', [('', 0), ('a', 1), (' ', 0), ('.a', 1), ('_', 1)]),
+    ('\\B', [('', 0), ('a', 0), ('ab', 1), (' ', 1), ('a b', 0), ('.', 1)]),
+    ('a\\Bb', [('ab', 1), ('a b', 0), ('xaby', 1), ('a-b', 0)]),
+    ('\\bw\\w+\\b', [('the word here', 1), ('w', 0), ('ww', 1)]),
+    ('^\\bfoo', [('foo x', 1), (' foo', 0), ('xfoo', 0)]),
+    ('foo\\b$', [('x foo', 1), ('foo ', 0), ('afoo', 1), ('foo', 1)]),
+    ('\\b(cat|dog)s?\\b', [('cats!', 1), ('dogcat', 0), ('a dog', 1), ('catsx', 0)]),
+    ('\\b\\B', [('a', 0), (' ', 0), ('ab', 0)]),
+]
+
+
+def test_regex_word_boundaries():
+    for name, match in _engines():
+        for pat, cases in BOUNDARY_CASES:
+            for subj, want in cases:
+                got = match(pat.encode(), subj.encode())
+                assert got == want, (
+                    f"[{name}] pat={pat!r} s={subj!r} got={got} want={want}")
+        # \b with >64 positions stays a loud reject
+        assert match(rb"\b[ab]{70}", b"a" * 80) < 0, name
+
+
+def test_regex_boundary_fuzz_vs_python():
+    """Random patterns with \b/\B sprinkled in, vs Python re (bytes
+    patterns: ASCII \w, matching Go/RE2 semantics)."""
+    rng = random.Random(77)
+    atoms = ["a", "b", "1", "_", ".", "[ab1]", "a?", "b*", "(ab|1_)",
+             "a{1,2}", r"\b", r"\B", r"\w", r"\W", "x+"]
+    lib = oracle_helpers()
+    lib.orc_regex_match.restype = ctypes.c_long
+    checked = 0
+    for _ in range(4000):
+        pat = "".join(rng.choice(atoms) for _ in range(rng.randrange(1, 6)))
+        if rng.random() < 0.2:
+            pat = r"\b" + pat
+        if rng.random() < 0.2:
+            pat = pat + r"\b"
+        if rng.random() < 0.15:
+            pat = "^" + pat
+        if rng.random() < 0.15:
+            pat = pat + "$"
+        try:
+            cre = pyre.compile(pat.encode())
+        except pyre.error:
+            continue
+        pb = pat.encode()
+        ok = True
+        for _ in range(5):
+            s = "".join(rng.choice("ab1_ .x-")
+                        for _ in range(rng.randrange(0, 12))).encode()
+            r = lib.orc_regex_match(pb, len(pb), s, len(s))
+            if r < 0:
+                ok = False
+                break
+            want = 1 if cre.search(s) else 0
+            assert r == want, f"pat={pat!r} s={s!r} ours={r} py={want}"
+        if ok:
+            checked += 1
+    assert checked > 2500

@@ -13,12 +13,17 @@ namespace {
 // Minimal RE2-subset AST.  Group nodes are preserved so GetLiterals sees the
 // same literal boundaries as Go's parse tree (regex.go:101-124).
 struct RNode {
-  enum Kind { Lit, Dot, Concat, Alt, Star, Plus, Quest, Group, Empty, Class } kind;
+  enum Kind {
+    Lit, Dot, Concat, Alt, Star, Plus, Quest, Group, Empty, Class, Assert
+  } kind;
   std::string lit;
   std::vector<RNode> subs;
   // Class: 256-bit byte set (ASCII bits 0..127 used) + "all non-ASCII runes"
   uint8_t cls[32] = {0};
   bool cls_nonascii = false;
+  // Assert: 1 = \b (word boundary), 2 = \B (non-boundary); zero-width,
+  // checked between the previous and next byte (RE2 ASCII \w semantics)
+  uint8_t assert_kind = 0;
 };
 
 // Case-fold orbits for (?i): closure over the simple lower/upper mappings
@@ -354,7 +359,13 @@ struct Parser {
         }
         return cn;
       }
-      case 'b': case 'B': case 'p': case 'P':
+      case 'b': case 'B': {
+        RNode an;
+        an.kind = RNode::Assert;
+        an.assert_kind = c == 'b' ? 1 : 2;
+        return an;
+      }
+      case 'p': case 'P':
         err(std::string("escape class \\") + c + " is not supported");
       default:
         if ((c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z') ||
@@ -657,9 +668,34 @@ static void pm_foreach(const PMask& m, F f) {
   }
 }
 
+// Edge/entry assertion classes: transitions (and fragment entries/exits)
+// are classified by the zero-width assertion that must hold between the
+// previous and next byte at that point.  The matcher evaluates the word
+// boundary once per step (it has both bytes), so assertions cost one mask
+// select.  plain composes as identity; b∘nb is infeasible (dead).
+constexpr int kAcPlain = 0, kAcB = 1, kAcNB = 2, kAcDead = 3;
+
+static int ac_compose(int a, int b) {
+  if (a == kAcDead || b == kAcDead) return kAcDead;
+  if (a == kAcPlain) return b;
+  if (b == kAcPlain) return a;
+  return a == b ? a : kAcDead;
+}
+
+struct CMask {
+  PMask c[3];  // per assertion class
+  void operator|=(const CMask& o) {
+    for (int k = 0; k < 3; k++) c[k] |= o.c[k];
+  }
+  bool any_assert() const {
+    return (c[1].w0 | c[1].w1 | c[2].w0 | c[2].w1) != 0;
+  }
+};
+
 struct GBuild {
   std::vector<GAtom> atoms;
-  std::vector<PMask> follow;
+  std::vector<CMask> follow;
+  bool has_assert = false;
   [[noreturn]] void overflow(const std::string& expr) {
     fail("regex: NFA fallback for \"" + expr +
          "\" needs more than 128 positions; simplify the pattern");
@@ -667,8 +703,8 @@ struct GBuild {
 };
 
 struct GInfo {
-  bool nullable;
-  PMask first, last;
+  uint8_t null_mask = 0;  // bit c: fragment matches "" under class c
+  CMask first, last;
 };
 
 static int g_add_atom(GBuild& b, const uint8_t* set, const std::string& expr) {
@@ -676,7 +712,7 @@ static int g_add_atom(GBuild& b, const uint8_t* set, const std::string& expr) {
   GAtom a;
   memcpy(a.set, set, 32);
   b.atoms.push_back(a);
-  b.follow.push_back(PMask{});
+  b.follow.push_back(CMask{});
   return int(b.atoms.size()) - 1;
 }
 
@@ -685,20 +721,47 @@ static void g_range_set(uint8_t* set, int lo, int hi) {
 }
 
 static GInfo g_cat(GBuild& b, GInfo x, GInfo y) {
-  // follow: last(x) -> first(y)
-  pm_foreach(x.last, [&](int i) { b.follow[i] |= y.first; });
+  // follow: last(x) -> first(y), classes composed across the junction
+  for (int cx = 0; cx < 3; cx++) {
+    for (int cy = 0; cy < 3; cy++) {
+      const int c = ac_compose(cx, cy);
+      if (c == kAcDead) continue;
+      pm_foreach(x.last.c[cx],
+                 [&](int i) { b.follow[i].c[c] |= y.first.c[cy]; });
+    }
+  }
   GInfo r;
-  r.nullable = x.nullable && y.nullable;
   r.first = x.first;
-  if (x.nullable) r.first |= y.first;
   r.last = y.last;
-  if (y.nullable) r.last |= x.last;
+  for (int nx = 0; nx < 3; nx++) {
+    if (!((x.null_mask >> nx) & 1)) continue;
+    for (int cy = 0; cy < 3; cy++) {
+      const int c = ac_compose(nx, cy);
+      if (c != kAcDead) r.first.c[c] |= y.first.c[cy];
+    }
+  }
+  for (int ny = 0; ny < 3; ny++) {
+    if (!((y.null_mask >> ny) & 1)) continue;
+    for (int cx = 0; cx < 3; cx++) {
+      const int c = ac_compose(cx, ny);
+      if (c != kAcDead) r.last.c[c] |= x.last.c[cx];
+    }
+  }
+  r.null_mask = 0;
+  for (int nx = 0; nx < 3; nx++) {
+    if (!((x.null_mask >> nx) & 1)) continue;
+    for (int ny = 0; ny < 3; ny++) {
+      if (!((y.null_mask >> ny) & 1)) continue;
+      const int c = ac_compose(nx, ny);
+      if (c != kAcDead) r.null_mask |= uint8_t(1) << c;
+    }
+  }
   return r;
 }
 
 static GInfo g_alt(GInfo x, GInfo y) {
   GInfo r;
-  r.nullable = x.nullable || y.nullable;
+  r.null_mask = x.null_mask | y.null_mask;
   r.first = x.first;
   r.first |= y.first;
   r.last = x.last;
@@ -707,21 +770,35 @@ static GInfo g_alt(GInfo x, GInfo y) {
 }
 
 static void g_loop(GBuild& b, const GInfo& x) {
-  pm_foreach(x.last, [&](int i) { b.follow[i] |= x.first; });
+  for (int cx = 0; cx < 3; cx++) {
+    for (int cy = 0; cy < 3; cy++) {
+      const int c = ac_compose(cx, cy);
+      if (c == kAcDead) continue;
+      pm_foreach(x.last.c[cx],
+                 [&](int i) { b.follow[i].c[c] |= x.first.c[cy]; });
+    }
+  }
 }
 
 
 static GInfo g_pos(int a) {
   GInfo r;
-  r.nullable = false;
-  r.first.set(a);
-  r.last.set(a);
+  r.first.c[kAcPlain].set(a);
+  r.last.c[kAcPlain].set(a);
   return r;
 }
 
 static GInfo g_empty(bool nullable) {
   GInfo r;
-  r.nullable = nullable;
+  if (nullable) r.null_mask = uint8_t(1) << kAcPlain;
+  return r;
+}
+
+static GInfo g_assert(GBuild& b, uint8_t kind) {
+  // \b / \B: zero-width; matches "" only where the class condition holds
+  b.has_assert = true;
+  GInfo r;
+  r.null_mask = uint8_t(1) << (kind == 1 ? kAcB : kAcNB);
   return r;
 }
 
@@ -756,6 +833,7 @@ static RNode compress_classes(const RNode& n) {
   out.lit = n.lit;
   memcpy(out.cls, n.cls, sizeof(out.cls));
   out.cls_nonascii = n.cls_nonascii;
+  out.assert_kind = n.assert_kind;
   out.subs.reserve(n.subs.size());
   for (const auto& sub : n.subs) out.subs.push_back(compress_classes(sub));
   if (out.kind != RNode::Alt) return out;
@@ -819,7 +897,8 @@ static GInfo g_build(GBuild& b, const RNode& n, const std::string& expr) {
     case RNode::Star: {
       GInfo x = g_build(b, n.subs[0], expr);
       g_loop(b, x);
-      return GInfo{true, x.first, x.last};
+      x.null_mask |= uint8_t(1) << kAcPlain;
+      return x;
     }
     case RNode::Plus: {
       GInfo x = g_build(b, n.subs[0], expr);
@@ -828,26 +907,41 @@ static GInfo g_build(GBuild& b, const RNode& n, const std::string& expr) {
     }
     case RNode::Quest: {
       GInfo x = g_build(b, n.subs[0], expr);
-      return GInfo{true, x.first, x.last};
+      x.null_mask |= uint8_t(1) << kAcPlain;
+      return x;
     }
+    case RNode::Assert:
+      return g_assert(b, n.assert_kind);
   }
   fail("regex: unreachable node kind");
 }
 
-// blob: u16 nstates, u16 pad, u32 pad, u64 first, u64 last,
-//       u64 follow[nstates], u64 byte_table[256]
+// blob layouts (selected by flags byte 2):
+//   plain/wide:  u16 n, u8 flags, pad to 8, first, last, follow[n],
+//                table[256] — 8-byte masks, or 16-byte pairs when flag 8
+//                (wide, 65..128 positions) is set.
+//   assertions (flag 16, <= 64 positions): u16 n, u8 flags, u8 null_mask,
+//                pad to 8, u64 first[3], u64 last[3], u64 follow[n][3],
+//                u64 table[256] — masks per assertion class
+//                {plain, \b, \B}; null_mask bits say which classes admit
+//                an empty match.
 static bytes g_serialize(const GBuild& b, const GInfo& root,
-                         bool a_start = false, bool a_end = false,
-                         bool nullable = false) {
+                         const std::string& expr, bool a_start = false,
+                         bool a_end = false, uint8_t null_mask = 0) {
   bytes out;
   const uint16_t n = uint16_t(b.atoms.size());
   const bool wide = n > 64;  // two-word position masks (65..128 positions)
+  const bool has_assert = b.has_assert;
+  if (has_assert && wide) {
+    fail("regex: \\b/\\B with more than 64 NFA positions is not supported"
+         " in " + expr);
+  }
   out.push_back(uint8_t(n));
   out.push_back(uint8_t(n >> 8));
-  // byte 2: flags (1 = ^ anchored, 2 = $ anchored, 4 = nullable root,
-  // 8 = wide/128-position layout: all masks are 16-byte pairs)
   out.push_back(uint8_t((a_start ? 1 : 0) | (a_end ? 2 : 0) |
-                        (nullable ? 4 : 0) | (wide ? 8 : 0)));
+                        (((null_mask >> kAcPlain) & 1) ? 4 : 0) |
+                        (wide ? 8 : 0) | (has_assert ? 16 : 0)));
+  out.push_back(has_assert ? null_mask : 0);
   out.resize(8, 0);
   auto put64 = [&](uint64_t v) {
     for (int i = 0; i < 8; i++) out.push_back(uint8_t(v >> (8 * i)));
@@ -856,9 +950,18 @@ static bytes g_serialize(const GBuild& b, const GInfo& root,
     put64(m.w0);
     if (wide) put64(m.w1);
   };
-  put_mask(root.first);
-  put_mask(root.last);
-  for (uint16_t i = 0; i < n; i++) put_mask(b.follow[i]);
+  if (has_assert) {
+    for (int c = 0; c < 3; c++) put64(root.first.c[c].w0);
+    for (int c = 0; c < 3; c++) put64(root.last.c[c].w0);
+    for (uint16_t i = 0; i < n; i++) {
+      for (int c = 0; c < 3; c++) put64(b.follow[i].c[c].w0);
+    }
+  } else {
+    // no assertions anywhere: every mask lives in class 0
+    put_mask(root.first.c[0]);
+    put_mask(root.last.c[0]);
+    for (uint16_t i = 0; i < n; i++) put_mask(b.follow[i].c[0]);
+  }
   for (int c = 0; c < 256; c++) {
     PMask m;
     for (uint16_t i = 0; i < n; i++) {
@@ -1035,7 +1138,8 @@ static RegexProg compile_single(const std::string& expr,
     re.has_nfa = true;
     // nullable root matches "": unanchored or half-anchored => every string
     // has an empty prefix/suffix match; both-anchored only matches ""
-    re.always_true = root.nullable && !(a_start && a_end);
+    re.always_true =
+        ((root.null_mask >> kAcPlain) & 1) != 0 && !(a_start && a_end);
     re.is_only_prefix = false;
     re.is_suffix_dot_star = false;
     re.is_suffix_dot_plus = false;
@@ -1043,7 +1147,8 @@ static RegexProg compile_single(const std::string& expr,
     re.substr_dot_plus.clear();
     re.has_or_values = false;
     re.or_values.clear();
-    re.nfa_blob = g_serialize(b, root, a_start, a_end, root.nullable);
+    re.nfa_blob =
+        g_serialize(b, root, expr, a_start, a_end, root.null_mask);
     re.prefix.clear();  // NFA matches the whole pattern; ignore the prefix
   }
   return re;
@@ -1141,9 +1246,76 @@ static bool match_with_prefix(const RegexProg& re, strview s) {
   }
 }
 
+static bool ascii_word(uint8_t c) {
+  // RE2 \b is over ASCII \w = [0-9A-Za-z_]
+  return c == '_' || (c >= '0' && c <= '9') || (c >= 'a' && c <= 'z') ||
+         (c >= 'A' && c <= 'Z');
+}
+
+// assert-layout matcher: per-class first/last/follow masks; the word
+// boundary between the previous and current byte selects which class's
+// transitions are live at each step (see g_serialize layout comment)
+static bool nfa_match_assert(const uint8_t* blob, strview s) {
+  const uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t anchors = blob[2];
+  const uint8_t null_mask = blob[3];
+  const bool a_start = anchors & 1, a_end = anchors & 2;
+  auto rd64 = [](const uint8_t* p) {
+    uint64_t v;
+    memcpy(&v, p, 8);
+    return v;
+  };
+  uint64_t first[3], last[3];
+  for (int c = 0; c < 3; c++) {
+    first[c] = rd64(blob + 8 + 8 * c);
+    last[c] = rd64(blob + 32 + 8 * c);
+  }
+  const uint8_t* follow = blob + 56;
+  const uint8_t* table = follow + size_t(n) * 24;
+  if (s.n == 0) return (null_mask >> kAcPlain) & 1;  // "" has no boundary
+  if (((null_mask >> kAcPlain) & 1) && !(a_start && a_end)) return true;
+  uint64_t active = 0;
+  bool prev_w = false;  // BOF behaves as a non-word char
+  for (size_t i = 0; i < s.n; i++) {
+    const bool cur_w = ascii_word(uint8_t(s.p[i]));
+    const bool bnd = prev_w != cur_w;
+    // empty match under an assertion class at offset i
+    if (!a_end && (!a_start || i == 0)) {
+      if ((null_mask >> (bnd ? kAcB : kAcNB)) & 1) return true;
+    }
+    uint64_t targets = 0;
+    if (!a_start || i == 0) {
+      targets = first[0] | (bnd ? first[1] : first[2]);
+    }
+    uint64_t m = active;
+    while (m) {
+      const int x = __builtin_ctzll(m);
+      m &= m - 1;
+      const uint8_t* f = follow + size_t(x) * 24;
+      targets |= rd64(f) | (bnd ? rd64(f + 8) : rd64(f + 16));
+    }
+    const uint64_t entered = targets & rd64(table + size_t(uint8_t(s.p[i])) * 8);
+    if (entered && !a_end) {
+      const bool next_w = i + 1 < s.n ? ascii_word(uint8_t(s.p[i + 1])) : false;
+      const bool bnd2 = cur_w != next_w;
+      if (entered & (last[0] | (bnd2 ? last[1] : last[2]))) return true;
+    }
+    active = entered;
+    prev_w = cur_w;
+  }
+  const bool bnd_eof = ascii_word(uint8_t(s.p[s.n - 1]));  // vs EOF non-word
+  if (active & (last[0] | (bnd_eof ? last[1] : last[2]))) {
+    if (a_end) return true;  // !a_end acceptances were taken in the loop
+  }
+  // empty match at offset s.n (allowed unless '^' pins the match to 0)
+  if (!a_start && ((null_mask >> (bnd_eof ? kAcB : kAcNB)) & 1)) return true;
+  return false;
+}
+
 bool nfa_match(const uint8_t* blob, strview s) {
   uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
   const uint8_t anchors = blob[2];
+  if (anchors & 16) return nfa_match_assert(blob, s);
   const bool a_start = anchors & 1, a_end = anchors & 2;
   const bool wide = (anchors & 8) != 0;  // 65..128 positions: 16-byte masks
   const size_t msz = wide ? 16 : 8;

@@ -58,9 +58,10 @@ bool nfa_match(const uint8_t* blob, strview s);
 // fallback.  Supported beyond the fast paths: {m,n} repetition, lazy
 // quantifiers (same accepted language for existence matching), top-level
 // and per-alternative '^'/'$' anchors, and a leading (?i) (simple case
-// closure + the Unicode CaseFolding special orbits).  Still rejected with
-// a clear error: \b/\B, \p{...}, mid-pattern anchors, (?...) flags other
-// than a leading (?i), and >128 NFA positions.
+// closure + the Unicode CaseFolding special orbits).  Word-boundary assertions
+// \b/\B are supported (assertion-classed NFA edges).  Still rejected
+// with a clear error: \p{...}, mid-pattern anchors, (?...) flags other
+// than a leading (?i), >128 NFA positions, and \b with >64 positions.
 RegexProg regex_compile(const std::string& expr);
 
 // Regex.MatchString (regex.go:86-98,131-212).

@@ -1073,10 +1073,73 @@ __device__ bool d_regex_or_hasprefix(const DRegex& re, const A& a, long s0, long
 // blob: u16 nstates, u8 flags (1 '^', 2 '$', 4 nullable, 8 wide), pad to 8,
 // then first/last/follow[n]/table[256] masks — 8-byte masks, or 16-byte
 // pairs when the wide flag is set (65..128 positions).
+// assert-layout executor (flag 16): per-class {plain, \b, \B} masks; the
+// ASCII word boundary between the previous and current byte selects which
+// class's transitions are live (core/regex.cpp nfa_match_assert mirror;
+// d_is_token_char IS ASCII \w)
+template <typename A>
+__device__ bool d_nfa_match_assert_at(const uint8_t* blob, const A& a, long s0,
+                                      long sn) {
+  const uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
+  const uint8_t anchors = blob[2];
+  const uint8_t null_mask = blob[3];
+  const bool a_start = anchors & 1, a_end = anchors & 2;
+  uint64_t first[3], last[3];
+  for (int c = 0; c < 3; c++) {
+    __builtin_memcpy(&first[c], blob + 8 + 8 * c, 8);
+    __builtin_memcpy(&last[c], blob + 32 + 8 * c, 8);
+  }
+  const uint8_t* follow = blob + 56;
+  const uint8_t* table = follow + size_t(n) * 24;
+  if (sn == 0) return (null_mask >> 0) & 1;  // "" has no boundary
+  if (((null_mask >> 0) & 1) && !(a_start && a_end)) return true;
+  uint64_t active = 0;
+  bool prev_w = false;  // BOF behaves as a non-word char
+  for (long i = 0; i < sn; i++) {
+    const bool cur_w = d_is_token_char(a.u8(s0 + i));
+    const bool bnd = prev_w != cur_w;
+    if (!a_end && (!a_start || i == 0)) {
+      if ((null_mask >> (bnd ? 1 : 2)) & 1) return true;
+    }
+    uint64_t targets = 0;
+    if (!a_start || i == 0) {
+      targets = first[0] | (bnd ? first[1] : first[2]);
+    }
+    uint64_t m = active;
+    while (m) {
+      const int x = __builtin_ctzll(m);
+      m &= m - 1;
+      const uint8_t* f = follow + size_t(x) * 24;
+      uint64_t f0, fb;
+      __builtin_memcpy(&f0, f, 8);
+      __builtin_memcpy(&fb, f + (bnd ? 8 : 16), 8);
+      targets |= f0 | fb;
+    }
+    uint64_t tb;
+    __builtin_memcpy(&tb, table + size_t(a.u8(s0 + i)) * 8, 8);
+    const uint64_t entered = targets & tb;
+    if (entered && !a_end) {
+      const bool next_w =
+          i + 1 < sn ? d_is_token_char(a.u8(s0 + i + 1)) : false;
+      const bool bnd2 = cur_w != next_w;
+      if (entered & (last[0] | (bnd2 ? last[1] : last[2]))) return true;
+    }
+    active = entered;
+    prev_w = cur_w;
+  }
+  const bool bnd_eof = d_is_token_char(a.u8(s0 + sn - 1));
+  if ((active & (last[0] | (bnd_eof ? last[1] : last[2]))) && a_end) {
+    return true;
+  }
+  if (!a_start && ((null_mask >> (bnd_eof ? 1 : 2)) & 1)) return true;
+  return false;
+}
+
 template <typename A>
 __device__ bool d_nfa_match_at(const uint8_t* blob, const A& a, long s0, long sn) {
   const uint16_t n = uint16_t(blob[0]) | uint16_t(blob[1]) << 8;
   const uint8_t anchors = blob[2];
+  if (anchors & 16) return d_nfa_match_assert_at(blob, a, s0, sn);
   const bool a_start = anchors & 1, a_end = anchors & 2;
   if (anchors & 8) {
     // wide: two-word position masks

@@ -2685,7 +2685,7 @@ const char* vql_errstr() { return g_err.c_str(); }
 
 /* 1 if the last error on this thread was an unsupported-construct
  * rejection (a valid LogsQL filter outside the engine's class — e.g. a
- * regex using \b or \p{...}) rather than corruption/IO.  A drop-in shim
+ * regex using \p{...}) rather than corruption/IO.  A drop-in shim
  * routes such filters back to the host's own Go implementation
  * (INTEGRATION.md "Unsupported-filter fallback"); corruption errors keep
  * the reference's panic semantics. */
