@@ -296,3 +296,39 @@ def test_regex_boundary_fuzz_vs_python():
         if ok:
             checked += 1
     assert checked > 2500
+
+
+def test_regex_icase_fuzz_vs_python():
+    """Random (?i) patterns vs Python re.IGNORECASE on BYTES (ASCII
+    folding — matching our fold orbits for ASCII subjects)."""
+    rng = random.Random(31)
+    atoms = ["a", "B", "k", "S", "1", "_", ".", "[aK]", "[b-f]", "(ab|Kx)",
+             "a?", "B*", "s{1,2}", "x+"]
+    for name, match in _engines():
+        checked = 0
+        for _ in range(1500):
+            pat = "(?i)" + "".join(rng.choice(atoms)
+                                   for _ in range(rng.randrange(1, 6)))
+            if rng.random() < 0.15:
+                pat = pat[:4] + "^" + pat[4:]
+            if rng.random() < 0.15:
+                pat = pat + "$"
+            try:
+                cre = pyre.compile(pat.encode(), pyre.IGNORECASE)
+            except pyre.error:
+                continue
+            pb = pat.encode()
+            ok = True
+            for _ in range(4):
+                s = "".join(rng.choice("aAbBkKsS1_ .xXfF")
+                            for _ in range(rng.randrange(0, 12))).encode()
+                r = match(pb, s)
+                if r < 0:
+                    ok = False
+                    break
+                want = 1 if cre.search(s) else 0
+                assert r == want, (
+                    f"[{name}] pat={pat!r} s={s!r} ours={r} py={want}")
+            if ok:
+                checked += 1
+        assert checked > 1000
