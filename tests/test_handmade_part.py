@@ -54,6 +54,23 @@ CASES = [
     ('{"type":"not","filter":'
      '{"type":"phrase","field":"_msg","phrase":"alpha"}}',
      [[2, 3, 4, 5, 6, 7, 8], [1]]),
+    # typed columns (block 1 only; missing in block 2)
+    ('{"type":"range","field":"u16c","min":302,"max":305}',
+     [[2, 3, 4, 5], []]),
+    ('{"type":"exact","field":"u32c","value":"70007"}', [[7], []]),
+    ('{"type":"range","field":"u64c","min":5000000001,"max":5000000002}',
+     [[1, 2], []]),
+    ('{"type":"exact","field":"i64c","value":"-3"}', [[1], []]),
+    ('{"type":"range","field":"i64c","min":-2,"max":1}',
+     [[2, 3, 4, 5], []]),
+    ('{"type":"range","field":"f64c","min":1.0,"max":3.0}', [[1, 2], []]),
+    ('{"type":"phrase","field":"ipc","phrase":"10.0.0.4"}', [[4], []]),
+    ('{"type":"ipv4_range","field":"ipc","min":167772162,"max":167772165}',
+     [[2, 3, 4, 5], []]),
+    ('{"type":"exact","field":"isoc","value":"2024-01-01T00:00:03.000Z"}',
+     [[3], []]),
+    ('{"type":"phrase","field":"isoc","phrase":"2024"}',
+     [[0, 1, 2, 3, 4, 5, 6, 7, 8], []]),
 ]
 
 BLOCK_ROWS = [9, 3]

@@ -267,6 +267,7 @@ def build_part(out_dir):
             # 9 uniques, numeric, max 9 -> uint8
             "code": [b"1", b"2", b"3", b"4", b"5", b"6", b"7", b"8", b"9"],
             "host": b"h1",
+            "typed": True,  # adds one column per remaining valueType
         },
         {
             "rows": 3,
@@ -292,7 +293,8 @@ def build_part(out_dir):
         ts_bin += payload
 
     # ---- column name table (ID = array index; column_names.go:101-134) ----
-    names = ["", "level", "code", "host"]
+    names = ["", "level", "code", "host", "u16c", "u32c", "u64c", "i64c",
+             "f64c", "ipc", "isoc"]
     name_id = {n: i for i, n in enumerate(names)}
     cn_plain = varuint(len(names)) + b"".join(marshal_bytes(n.encode())
                                               for n in names)
@@ -300,7 +302,8 @@ def build_part(out_dir):
 
     # column -> (values/bloom) shard: _msg goes to message_*; others shard 0
     # (marshalColumnIdxs: varuint count + {varuint nameID, varuint shard})
-    idxs = [(name_id["level"], 0), (name_id["code"], 0)]
+    idxs = [(name_id[n], 0) for n in ("level", "code", "u16c", "u32c",
+                                      "u64c", "i64c", "f64c", "ipc", "isoc")]
     column_idxs_bin = varuint(len(idxs)) + b"".join(
         varuint(a) + varuint(b) for a, b in idxs)
 
@@ -326,6 +329,52 @@ def build_part(out_dir):
             if max(nums) < 256:
                 return 3, [bytes([n]) for n in nums], None, min(nums), max(nums)
         return 1, list(values), None, None, None
+
+
+    def zigzag(v):
+        return ((v << 1) ^ (v >> 63)) & MASK64
+
+    def iso_ns(i):
+        # 2024-01-01T00:00:0i.000Z
+        import calendar
+        return (calendar.timegm((2024, 1, 1, 0, 0, i)) * 10**9)
+
+    def typed_columns():
+        """One column per remaining valueType (block_header.go:634-712
+        per-type min/max layouts; values_encoder per-row BE encodings).
+        9 distinct values defeat the dict encoding (<= 8 entries)."""
+        import struct as st
+        u16v = list(range(300, 309))
+        u32v = list(range(70000, 70009))
+        u64v = [5000000000 + i for i in range(9)]
+        i64v = list(range(-4, 5))
+        f64v = [0.5 + i for i in range(9)]
+        ipv = [(10 << 24) | i for i in range(9)]
+        isov = [iso_ns(i) for i in range(9)]
+        fbits = lambda f: st.unpack(">Q", st.pack(">d", f))[0]
+        cols = []
+        cols.append(("u16c", 4, [str(v).encode() for v in u16v],
+                     [be16(v) for v in u16v], be16(u16v[0]) + be16(u16v[-1])))
+        cols.append(("u32c", 5, [str(v).encode() for v in u32v],
+                     [be32(v) for v in u32v], be32(u32v[0]) + be32(u32v[-1])))
+        cols.append(("u64c", 6, [str(v).encode() for v in u64v],
+                     [be64(v) for v in u64v], be64(u64v[0]) + be64(u64v[-1])))
+        # int64: encoding.MarshalInt64 is itself zig-zag + BE (int.go:69-74)
+        # — used for BOTH the rows and the columnHeader min/max
+        cols.append(("i64c", 10, [str(v).encode() for v in i64v],
+                     [be64(zigzag(v)) for v in i64v],
+                     be64(zigzag(i64v[0])) + be64(zigzag(i64v[-1]))))
+        cols.append(("f64c", 7, [repr(v).encode() for v in f64v],
+                     [be64(fbits(v)) for v in f64v],
+                     be64(fbits(f64v[0])) + be64(fbits(f64v[-1]))))
+        cols.append(("ipc", 8,
+                     [("10.0.0.%d" % i).encode() for i in range(9)],
+                     [be32(v) for v in ipv], be32(ipv[0]) + be32(ipv[-1])))
+        cols.append(("isoc", 9,
+                     [("2024-01-01T00:00:0%d.000Z" % i).encode()
+                      for i in range(9)],
+                     [be64(v) for v in isov], be64(isov[0]) + be64(isov[-1])))
+        return cols
 
     # ---- per-block columns_header{,_index} + values/bloom files ----
     ch_bin = bytearray()      # columns_header.bin
@@ -358,6 +407,20 @@ def build_part(out_dir):
                 h += varuint(voff) + varuint(len(vblob))
                 h += varuint(boff) + varuint(len(bloom))
             cols.append((cname, h))
+
+        if b.get("typed"):
+            for cname, vt, raws, rows_enc, minmax in typed_columns():
+                vblob = strings_block(rows_enc)
+                voff = len(shard_values)
+                shard_values.extend(vblob)
+                bloom = bloom_marshal(
+                    [t for t in tokenize_unique([r.decode() for r in raws])])
+                boff = len(shard_bloom)
+                shard_bloom.extend(bloom)
+                h = bytes([vt]) + minmax
+                h += varuint(voff) + varuint(len(vblob))
+                h += varuint(boff) + varuint(len(bloom))
+                cols.append((cname, h))
 
         # columnsHeader region (block_header.go:454-484): varuint count,
         # headers (offsets relative to region start), varuint const count,
